@@ -1,0 +1,395 @@
+"""The per-rank DPPO training engine.
+
+Implements the reference's intended Chief/Worker round protocol
+ in the MI355X-native shape: one process per GPU,
+each rank = one reference "Worker" with a batch of E synthetic envs;
+rank 0 carries the Chief's bookkeeping; cross-rank coordination is
+RCCL collectives instead of Events/deques.
+
+One training round (reference Chief.check + Worker.work):
+  1. oldpi <- pi            (sync_pis at round start; PPO.py:47, Worker.py:42)
+  2. rollout MAX_EPOCH_STEPS env steps per env, with the epsilon-greedy
+     exploration overlay and its linear anneal (Worker.py:140-153);
+     repeat (discarding, like the partial-buffer discard rule
+     Worker.py:43-47) until >=1 episode completed (push guard
+     Worker.py:135), up to MAX_ROLLOUT_RETRIES.
+  3. host GAE -> device GAE scan: advantages, returns, whitening
+     (Worker.py:82-92; HIP segmented scan on GPU).
+  4. evaluate the 4 losses pre-update for the logs vector (Worker.py:117-118)
+     and build logs = [score, epr_min, epr_max, epr_mean, policyLoss,
+     valueLoss, entropyLoss, totalLoss, CUR_EP] (Worker.py:123-133).
+  5. all_gather [logs(9), cur_lr, valid] across ranks; the best valid
+     rank by logs[2] (max episode reward — the Chief's sort key,
+     Chief.py:51) supplies l_mul for everyone (Chief.py:58-63) and the
+     stop decision (CUR_EP >= STOP_EPOCH, Chief.py:85-87).
+  6. UPDATE_STEPS repeated full-batch updates on the SAME data
+     (Chief.py:64): fwd/bwd -> flat-bucket all_reduce(AVG)
+     (PPO.py:55-65 analog) -> Adam with lr = LEARNING_RATE * l_mul and
+     clip = CLIP_PARAM * l_mul (PPO.py:19-20).
+  7. periodic rank-0 param broadcast as drift guard (replaces the
+     per-round assign broadcast Chief.py:67-70 — replicas are already
+     bit-identical under all-reduced grads + identical Adam state).
+
+A rank with no completed episode still participates in every collective
+(validity flag; SURVEY.md §7 "hard parts" last bullet) so the ring never
+deadlocks.
+"""
+
+from __future__ import annotations
+
+import math
+from dataclasses import dataclass, field
+from typing import Dict, List, Optional, Tuple
+
+import torch
+
+from .config import DPPOConfig, game_spaces
+from .distributions import make_pdtype
+from .envs.synthetic import BatchedSyntheticEnv, make_env
+from .models.mlp import PolicyValueMLP
+from .ops import gae_advantages, ppo_losses, PPOLossCoeffs
+from .parallel.comm import Comm, FlatBuffers
+from .utils.logging import ScalarLogger
+from .utils.timers import PhaseTimers
+from . import spaces
+
+STATS_DIM = 11  # logs[9] + cur_lr + valid
+
+
+@dataclass
+class RolloutBatch:
+    states: torch.Tensor      # [T*E, obs]
+    actions: torch.Tensor     # [T*E] long or [T*E, A] float
+    adv: torch.Tensor         # [T*E] whitened
+    etr: torch.Tensor         # [T*E]
+    cur_lr: float
+    episode_rewards: torch.Tensor  # [n_completed] (device)
+    valid: bool
+
+
+class DPPOEngine:
+    def __init__(
+        self,
+        cfg: DPPOConfig,
+        comm: Optional[Comm] = None,
+        scope: str = "Worker",
+        seed_offset: int = 0,
+    ):
+        self.cfg = cfg
+        self.comm = comm if comm is not None else Comm()
+        self.scope = scope
+        self.device = self.comm.device
+        self.dtype = cfg.torch_dtype()
+
+        obs_space, act_space = game_spaces(cfg.GAME)
+        self.obs_space, self.act_space = obs_space, act_space
+        self._discrete = isinstance(act_space, spaces.Discrete)
+
+        seed = cfg.SEED + 1000 * (self.comm.rank + seed_offset)
+        torch.manual_seed(seed)
+
+        def build_net() -> PolicyValueMLP:
+            return PolicyValueMLP(
+                obs_dim=obs_space.shape[0],
+                action_space=act_space,
+                hidden_sizes=cfg.HIDDEN_SIZES,
+                activation=cfg.ACTIVATION,
+                init_std=cfg.INIT_STD,
+            ).to(self.device)
+
+        # pi / oldpi two-network scheme (PPO.py:21-22)
+        self.pi = build_net()
+        self.oldpi = build_net()
+        self.flat_pi = FlatBuffers(self.pi)
+        self.flat_old = FlatBuffers(self.oldpi)
+        # initial broadcast: chief(rank0) pi AND oldpi -> everyone (main.py:48-50)
+        self.comm.broadcast_(self.flat_pi.flat_param, src=0)
+        self.sync_oldpi()
+
+        self.optimizer = torch.optim.Adam([self.flat_pi.flat_param], lr=cfg.LEARNING_RATE)
+        # Adam moments live on every rank but stay identical (SURVEY.md §2.3
+        # "parameter broadcast" row) — the reference keeps them only on the
+        # Chief (PPO.py:20,53); replicated-but-identical is the DP-native form.
+
+        self.env: BatchedSyntheticEnv = make_env(cfg, str(self.device), seed + 17)
+        self.obs = self.env.reset()
+        self.epr = torch.zeros(cfg.NUM_ENVS, device=self.device, dtype=torch.float32)
+
+        self.CUR_EP = 0
+        self.timers = PhaseTimers(cuda_sync=False)
+        self.logger = ScalarLogger(
+            cfg.LOG_FILE_PATH, enabled=(self.comm.rank == 0), name=f"{scope}_rank0"
+        )
+        self._round = 0
+
+    # ------------------------------------------------------------------
+    @torch.no_grad()
+    def sync_oldpi(self) -> None:
+        """oldpi <- pi, one flat copy (sync_pis, PPO.py:47)."""
+        self.flat_old.flat_param.copy_(self.flat_pi.flat_param)
+
+    def exploration_rate(self) -> float:
+        """Linear epsilon anneal MAX->MIN over AC_EXP_PERCENTAGE*EPOCH_MAX
+        epochs (Worker.py:140-144)."""
+        c = self.cfg
+        ac_exp_epoch = c.AC_EXP_PERCENTAGE * c.EPOCH_MAX
+        if self.CUR_EP >= ac_exp_epoch:
+            return c.MIN_AC_EXP_RATE
+        return c.MAX_AC_EXP_RATE + self.CUR_EP * (
+            c.MIN_AC_EXP_RATE - c.MAX_AC_EXP_RATE
+        ) / ac_exp_epoch
+
+    def current_lr_mul(self) -> float:
+        """LR/clip anneal multiplier (Worker.py:77-80)."""
+        c = self.cfg
+        if c.SCHEDULE == "constant":
+            return 1.0
+        return max(1.0 - float(self.CUR_EP) / c.EPOCH_MAX, 0.0)
+
+    # ------------------------------------------------------------------
+    def _random_actions(self, n: int) -> torch.Tensor:
+        """Uniform random actions for the epsilon-greedy overlay
+        (Worker.py:149-152; extended to Box spaces: uniform in [low, high])."""
+        if self._discrete:
+            return torch.randint(
+                self.act_space.n, (n,), device=self.device, dtype=torch.long
+            )
+        low = float(self.act_space.low.flat[0])
+        high = float(self.act_space.high.flat[0])
+        a_dim = self.act_space.shape[0]
+        return low + (high - low) * torch.rand(
+            n, a_dim, device=self.device, dtype=self.dtype
+        )
+
+    @torch.no_grad()
+    def act_batch(self, obs: torch.Tensor, eps: float) -> Tuple[torch.Tensor, torch.Tensor]:
+        """Sample actions + values for a batch of states, with the
+        epsilon-greedy overlay.  The old-policy logp of explored actions is
+        NOT recorded — the ratio re-evaluates oldpi on the batch later,
+        exactly the reference's formulation (Worker.py:149-152 + PPO.py:31;
+        SURVEY.md §7 'hard parts')."""
+        v, pdflat = self.pi(obs)
+        pd = self.pi.pdtype.pdfromflat(pdflat)
+        a = pd.sample()
+        if eps > 0.0:
+            E = obs.shape[0]
+            explore = torch.rand(E, device=self.device) < eps
+            rand_a = self._random_actions(E)
+            if self._discrete:
+                a = torch.where(explore, rand_a, a)
+            else:
+                a = torch.where(explore.unsqueeze(-1), rand_a, a)
+        return a, v
+
+    @torch.no_grad()
+    def rollout_once(self) -> Tuple[RolloutBatch, Dict[str, float]]:
+        """Collect one iteration of T = MAX_EPOCH_STEPS batched env steps
+        (Worker.py:39-65), then GAE (Worker.py:82-92)."""
+        c, E = self.cfg, self.cfg.NUM_ENVS
+        T = c.MAX_EPOCH_STEPS
+        obs_dim = self.obs_space.shape[0]
+        eps = self.exploration_rate()
+
+        states = torch.empty(T, E, obs_dim, device=self.device, dtype=self.dtype)
+        if self._discrete:
+            actions = torch.empty(T, E, device=self.device, dtype=torch.long)
+        else:
+            actions = torch.empty(
+                T, E, self.act_space.shape[0], device=self.device, dtype=self.dtype
+            )
+        rewards = torch.empty(T, E, device=self.device, dtype=torch.float32)
+        dones = torch.empty(T, E, device=self.device, dtype=torch.float32)
+        values = torch.empty(T, E, device=self.device, dtype=torch.float32)
+
+        ep_rewards: List[torch.Tensor] = []
+        obs = self.obs
+        for t in range(T):
+            a, v = self.act_batch(obs, eps)
+            states[t] = obs
+            actions[t] = a
+            values[t] = v.float()
+            obs, r, done, _ = self.env.step(a)
+            rewards[t] = r.float()
+            dones[t] = done.float()
+            # episode-reward bookkeeping (Worker.py:57-65)
+            self.epr += r.float()
+            if done.any():
+                ep_rewards.append(self.epr[done])
+                self.epr = torch.where(done, torch.zeros_like(self.epr), self.epr)
+        self.obs = obs
+
+        with torch.no_grad():
+            boot_v, _ = self.pi(obs)
+        adv, etr = gae_advantages(
+            rewards, values, dones, boot_v.float(),
+            c.GAMMA, c.LAM, whiten=True, eps=c.ADV_EPS,
+            policy=c.USE_HIP_KERNELS,
+        )
+
+        epr_cat = (
+            torch.cat(ep_rewards)
+            if ep_rewards
+            else torch.empty(0, device=self.device)
+        )
+        batch = RolloutBatch(
+            states=states.reshape(T * E, obs_dim),
+            actions=actions.reshape(T * E, *actions.shape[2:]),
+            adv=adv.reshape(T * E).to(self.dtype),
+            etr=etr.reshape(T * E).to(self.dtype),
+            cur_lr=self.current_lr_mul(),
+            episode_rewards=epr_cat,
+            valid=epr_cat.numel() > 0,
+        )
+        return batch, {"exploration_rate": eps}
+
+    def collect(self) -> RolloutBatch:
+        """Rollout with the push guard: retry (discarding) until at least
+        one episode completed (Worker.py:135 + while-loop Worker.py:30)."""
+        for _ in range(self.cfg.MAX_ROLLOUT_RETRIES):
+            batch, _ = self.rollout_once()
+            if batch.valid:
+                return batch
+        return batch  # invalid; the validity flag keeps collectives alive
+
+    # ------------------------------------------------------------------
+    def eval_losses(self, batch: RolloutBatch, l_mul: float) -> Dict[str, float]:
+        """Pre-update loss evaluation for the logs vector (Worker.py:117-118)."""
+        with torch.no_grad():
+            losses = self._losses(batch, l_mul)
+        return {k: float(v) for k, v in losses.items()}
+
+    def _losses(self, batch: RolloutBatch, l_mul: float) -> Dict[str, torch.Tensor]:
+        v, pdflat = self.pi(batch.states)
+        with torch.no_grad():
+            oldv, oldflat = self.oldpi(batch.states)
+        pd = self.pi.pdtype.pdfromflat(pdflat)
+        oldpd = self.pi.pdtype.pdfromflat(oldflat)
+        coeffs = PPOLossCoeffs(
+            clip_param=self.cfg.CLIP_PARAM * l_mul,
+            entcoeff=self.cfg.ENTCOEFF,
+            vcoeff=self.cfg.VCOEFF,
+        )
+        return ppo_losses(
+            pd, oldpd, v, oldv, batch.actions, batch.adv, batch.etr, coeffs,
+            policy=self.cfg.USE_HIP_KERNELS,
+        )
+
+    def stats_row(self, batch: RolloutBatch, losses: Dict[str, float]) -> torch.Tensor:
+        """The 9-float logs vector + cur_lr + valid (Worker.py:123-133).
+
+        score = epr.mean()/epr.std() is NaN-prone for a single episode
+        (Worker.py:121) — computed but guarded to 0."""
+        epr = batch.episode_rewards
+        if batch.valid and epr.numel() > 1 and float(epr.std(unbiased=False)) > 0:
+            score = float(epr.mean() / epr.std(unbiased=False))
+        else:
+            score = 0.0
+        if batch.valid:
+            mn, mx, mean = float(epr.min()), float(epr.max()), float(epr.mean())
+        else:
+            mn = mx = mean = -math.inf  # never wins the best-rank sort
+        row = torch.tensor(
+            [
+                score, mn, mx, mean,
+                losses["policyLoss"], losses["valueLoss"],
+                losses["entropyLoss"], losses["total_loss"],
+                float(self.CUR_EP),
+                batch.cur_lr,
+                1.0 if batch.valid else 0.0,
+            ],
+            device=self.device, dtype=torch.float32,
+        )
+        assert row.numel() == STATS_DIM
+        return row
+
+    # ------------------------------------------------------------------
+    def update(self, batch: RolloutBatch, l_mul: float) -> Dict[str, float]:
+        """UPDATE_STEPS repeated full-batch steps on the same data
+        (Chief.py:64), DP gradient mean across ranks each step."""
+        for g in self.optimizer.param_groups:
+            g["lr"] = self.cfg.LEARNING_RATE * l_mul
+        last: Dict[str, torch.Tensor] = {}
+        for _ in range(self.cfg.UPDATE_STEPS):
+            self.flat_pi.zero_grad()
+            losses = self._losses(batch, l_mul)
+            losses["total_loss"].backward()
+            self.comm.allreduce_mean_(self.flat_pi.flat_grad)
+            self.optimizer.step()
+            last = losses
+        return {k: float(v.detach()) for k, v in last.items()}
+
+    # ------------------------------------------------------------------
+    def train_round(self) -> Tuple[Dict[str, float], bool]:
+        """One full synchronous round. Returns (rank0-view stats, stop)."""
+        c = self.cfg
+        with self.timers.phase("sync_oldpi"):
+            self.sync_oldpi()
+        with self.timers.phase("rollout"):
+            batch = self.collect()
+        with self.timers.phase("eval_losses"):
+            losses = self.eval_losses(batch, batch.cur_lr)
+        with self.timers.phase("stats_allgather"):
+            row = self.stats_row(batch, losses)
+            gathered = self.comm.all_gather_rows(row)  # [W, STATS_DIM]
+        # Chief semantics: best valid rank by logs[2] (max episode reward)
+        valid = gathered[:, 10] > 0.5
+        any_valid = bool(valid.any())
+        if any_valid:
+            key = gathered[:, 2].clone()
+            key[~valid] = -math.inf
+            best = int(torch.argmax(key))
+            l_mul = float(gathered[best, 9])
+            best_cur_ep = float(gathered[best, 8])
+        else:
+            best = -1
+            l_mul = batch.cur_lr
+            best_cur_ep = float(self.CUR_EP)
+
+        if any_valid:
+            with self.timers.phase("update"):
+                self.update(batch, l_mul)
+
+        self._round += 1
+        if c.BROADCAST_INTERVAL > 0 and self._round % c.BROADCAST_INTERVAL == 0:
+            with self.timers.phase("broadcast"):
+                self.comm.broadcast_(self.flat_pi.flat_param, src=0)
+
+        # stop rule: best rank's epoch count >= STOP_EPOCH (Chief.py:85-87)
+        stop = best_cur_ep >= c.STOP_EPOCH
+        self.CUR_EP += 1
+
+        stats = {
+            "l_mul": l_mul,
+            "best_rank": float(best),
+            "score": float(gathered[max(best, 0), 0]),
+            "epr_max": float(gathered[max(best, 0), 2]),
+            "epr_mean": float(gathered[max(best, 0), 3]),
+            **losses,
+        }
+        if self.comm.rank == 0:
+            self.logger.log(self.CUR_EP, stats)
+        return stats, stop
+
+    def train(self, max_rounds: Optional[int] = None) -> Dict[str, float]:
+        """Run rounds until the stop rule fires (or max_rounds)."""
+        n = 0
+        stats: Dict[str, float] = {}
+        while True:
+            stats, stop = self.train_round()
+            n += 1
+            if stop or (max_rounds is not None and n >= max_rounds):
+                break
+        return stats
+
+    # ------------------------------------------------------------------
+    @torch.no_grad()
+    def act(self, s) -> torch.Tensor:
+        """Greedy-eval action for a single state: sample from pi with no
+        epsilon overlay (Chief.act, Chief.py:89-92 — samples, not mode)."""
+        s_t = torch.as_tensor(s, device=self.device, dtype=self.dtype)
+        squeeze = s_t.dim() == 1
+        if squeeze:
+            s_t = s_t.unsqueeze(0)
+        _, pdflat = self.pi(s_t)
+        a = self.pi.pdtype.pdfromflat(pdflat).sample()
+        return a[0] if squeeze else a

@@ -1,0 +1,117 @@
+"""Multi-process DP tests on a CPU gloo process group (BASELINE.json
+config 1: Pendulum-shaped, 2 CPU workers — the distributed-logic harness
+that runs without a GPU)."""
+
+import json
+import os
+import pickle
+import sys
+
+import pytest
+import torch
+import torch.multiprocessing as mp
+
+from dppo_amd.config import DPPOConfig
+
+WORLD = 2
+
+
+def _worker(rank, world, port, out_dir):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    os.environ["RANK"] = str(rank)
+    os.environ["WORLD_SIZE"] = str(world)
+    os.environ["LOCAL_RANK"] = str(rank)
+    import torch.distributed as dist
+
+    from dppo_amd.parallel.comm import Comm
+    from dppo_amd.trainer import DPPOEngine
+
+    cfg = DPPOConfig(
+        GAME="Pendulum-v1", NUM_ENVS=4, MAX_EPOCH_STEPS=12, EPOCH_MAX=8,
+        STOP_EPOCH=8, LEARNING_RATE=1e-3, NUM_WORKERS=world,
+        LOG_FILE_PATH=os.path.join(out_dir, "logs"), DEVICE="cpu",
+        BROADCAST_INTERVAL=0,  # rely on determinism; test checks bit-identity
+    )
+    comm = Comm(backend="gloo", device="cpu")
+    eng = DPPOEngine(cfg, comm=comm)
+
+    # ranks start from different seeds BUT the initial broadcast must have
+    # aligned parameters (main.py:48-50 analog)
+    p0 = eng.flat_pi.flat_param.clone()
+    gathered = comm.all_gather_rows(p0)
+    init_identical = bool(torch.allclose(gathered[0], gathered[1]))
+
+    stats_list = []
+    for _ in range(3):
+        stats, stop = eng.train_round()
+        stats_list.append(stats)
+
+    pf = eng.flat_pi.flat_param.clone()
+    gathered_after = comm.all_gather_rows(pf)
+    # replicas must remain BIT-identical under all-reduced grads +
+    # identical Adam state (SURVEY.md §2.3 broadcast-elimination claim)
+    final_identical = bool(torch.equal(gathered_after[0], gathered_after[1]))
+
+    # l_mul consensus: both ranks used the same best-rank multiplier
+    with open(os.path.join(out_dir, f"rank{rank}.pkl"), "wb") as f:
+        pickle.dump(
+            {
+                "init_identical": init_identical,
+                "final_identical": final_identical,
+                "l_muls": [s["l_mul"] for s in stats_list],
+                "best_ranks": [s["best_rank"] for s in stats_list],
+                "param_sum": float(pf.sum()),
+            },
+            f,
+        )
+    comm.shutdown()
+
+
+@pytest.mark.timeout(300)
+def test_two_rank_gloo_training(tmp_path):
+    port = 29741
+    mp.spawn(_worker, args=(WORLD, port, str(tmp_path)), nprocs=WORLD, join=True)
+    results = []
+    for r in range(WORLD):
+        with open(tmp_path / f"rank{r}.pkl", "rb") as f:
+            results.append(pickle.load(f))
+    for r in results:
+        assert r["init_identical"], "initial broadcast failed"
+        assert r["final_identical"], "replicas diverged"
+    assert results[0]["l_muls"] == results[1]["l_muls"], "l_mul consensus broken"
+    assert results[0]["best_ranks"] == results[1]["best_ranks"]
+    assert results[0]["param_sum"] == results[1]["param_sum"]
+
+
+def _allreduce_worker(rank, world, port, out_dir):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    os.environ["RANK"] = str(rank)
+    os.environ["WORLD_SIZE"] = str(world)
+    from dppo_amd.parallel.comm import Comm, FlatBuffers
+
+    comm = Comm(backend="gloo", device="cpu")
+    t = torch.full((10,), float(rank + 1))
+    comm.allreduce_mean_(t)
+    ok = torch.allclose(t, torch.full((10,), 1.5))
+
+    lin = torch.nn.Linear(4, 3)
+    fb = FlatBuffers(lin)
+    assert fb.numel == 4 * 3 + 3
+    # backward accumulates into the flat grad buffer
+    loss = lin(torch.randn(5, 4)).pow(2).sum()
+    loss.backward()
+    grad_ok = bool(fb.flat_grad.abs().sum() > 0)
+    with open(os.path.join(out_dir, f"ar{rank}.json"), "w") as f:
+        json.dump({"ok": bool(ok), "grad_ok": grad_ok}, f)
+    comm.shutdown()
+
+
+@pytest.mark.timeout(120)
+def test_flat_bucket_allreduce(tmp_path):
+    mp.spawn(_allreduce_worker, args=(WORLD, 29753, str(tmp_path)), nprocs=WORLD, join=True)
+    for r in range(WORLD):
+        with open(tmp_path / f"ar{r}.json") as f:
+            d = json.load(f)
+        assert d["ok"] and d["grad_ok"]
