@@ -1,0 +1,141 @@
+"""Flagship benchmark: Humanoid-shaped MLP DPPO training throughput.
+
+Measures the BASELINE.json headline metric — env-steps/sec (whole node)
+for Humanoid-shaped (obs=376, act=17) MLP DPPO on synthetic data with
+random-init weights — on N GPUs of one node (weak scaling: per-GPU work
+fixed as N grows).
+
+One bench "step" = one full DPPO training round: a rollout of T env
+steps on each of E device-resident envs, the GAE scan + whitening, the
+cross-rank stats all-gather (Chief semantics), and UPDATE_STEPS repeated
+full-batch PPO updates with flat-bucket gradient all-reduce.  env-steps
+counted = E * T per rank per step.
+
+Usage:
+    python bench.py --gpus N --steps K --warmup W
+For N > 1 launch via:
+    python -m torch.distributed.run --nnodes=1 --nproc-per-node N \
+        --master-addr 127.0.0.1 bench.py --gpus N --steps K --warmup W
+"""
+
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import time
+
+import torch
+
+from dppo_amd.config import DPPOConfig
+from dppo_amd.parallel.comm import Comm
+from dppo_amd.trainer import DPPOEngine
+
+PRESETS = {
+    # BASELINE.json config 3 (the headline): Humanoid-shaped MLP DPPO
+    "humanoid": dict(
+        GAME="Humanoid-v4", HIDDEN_SIZES=(64, 64), ACTIVATION="tanh",
+        NUM_ENVS=4096, MAX_EPOCH_STEPS=64, DTYPE="float32",
+    ),
+    # BASELINE.json config 2: HalfCheetah-shaped, 64 envs, 1 GPU
+    "halfcheetah": dict(
+        GAME="HalfCheetah-v4", HIDDEN_SIZES=(64, 64), ACTIVATION="tanh",
+        NUM_ENVS=64, MAX_EPOCH_STEPS=100, DTYPE="float32",
+    ),
+    # BASELINE.json config 5: wide MFMA-bound MLP
+    "wide4096": dict(
+        GAME="Wide-4096", HIDDEN_SIZES=(4096, 4096, 4096, 4096),
+        ACTIVATION="tanh", NUM_ENVS=1024, MAX_EPOCH_STEPS=16, DTYPE="float32",
+    ),
+}
+
+
+def main() -> None:
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=20)
+    p.add_argument("--warmup", type=int, default=5)
+    p.add_argument("--preset", type=str, default="humanoid", choices=sorted(PRESETS))
+    p.add_argument("--num-envs", type=int, default=None)
+    p.add_argument("--rollout", type=int, default=None)
+    p.add_argument("--device", type=str, default=None)
+    args = p.parse_args()
+
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    n_gpus = max(args.gpus, world)
+
+    preset = dict(PRESETS[args.preset])
+    if args.num_envs:
+        preset["NUM_ENVS"] = args.num_envs
+    if args.rollout:
+        preset["MAX_EPOCH_STEPS"] = args.rollout
+    cfg = DPPOConfig(
+        **preset,
+        EPOCH_MAX=1_000_000, STOP_EPOCH=1_000_000,
+        LEARNING_RATE=3e-4,
+        NUM_WORKERS=n_gpus,
+        LOG_FILE_PATH="/tmp/dppo_bench_logs",
+        SEED=1234,
+        DEVICE=args.device or "auto",
+    )
+
+    comm = Comm(device=args.device)
+    engine = DPPOEngine(cfg, comm=comm, scope=f"Worker_N{comm.rank}")
+    on_gpu = comm.device.type == "cuda"
+
+    def sync():
+        comm.barrier()
+        if on_gpu:
+            torch.cuda.synchronize()
+
+    for _ in range(args.warmup):
+        engine.train_round()
+
+    sync()
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        engine.train_round()
+    sync()
+    elapsed = time.perf_counter() - t0
+
+    # MAX over ranks
+    e_t = torch.tensor([elapsed], device=comm.device, dtype=torch.float64)
+    if comm.distributed:
+        import torch.distributed as dist
+
+        dist.all_reduce(e_t, op=dist.ReduceOp.MAX)
+    elapsed = float(e_t)
+
+    E, T = cfg.NUM_ENVS, cfg.MAX_EPOCH_STEPS
+    env_steps = E * T * args.steps * world
+    value = env_steps / elapsed
+    if comm.rank == 0:
+        result = {
+            "metric": "env-steps/sec (whole node), Humanoid-shaped MLP DPPO at 1/2/4/8 MI355X",
+            "value": value,
+            "unit": "env-steps/s",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": elapsed / args.steps * 1000.0,
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": cfg.DTYPE.replace("float32", "fp32").replace("bfloat16", "bf16"),
+            "data": "synthetic",
+            "config": {
+                "model": f"{cfg.GAME} MLP {'x'.join(str(h) for h in cfg.HIDDEN_SIZES)} {cfg.ACTIVATION}",
+                "preset": args.preset,
+                "global_batch": E * T * world,
+                "num_envs_per_gpu": E,
+                "rollout_T": T,
+                "update_steps": cfg.UPDATE_STEPS,
+                "parallelism": f"dp{world}",
+            },
+        }
+        print(json.dumps(result))
+    comm.shutdown()
+
+
+if __name__ == "__main__":
+    main()

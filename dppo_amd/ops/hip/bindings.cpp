@@ -1,0 +1,35 @@
+// Python bindings for the dppo_amd gfx950 kernel library.
+#include <torch/extension.h>
+
+#include <vector>
+
+std::vector<torch::Tensor> gae_scan(torch::Tensor rewards, torch::Tensor values,
+                                    torch::Tensor dones, torch::Tensor boot,
+                                    double gamma, double lam, bool whiten,
+                                    double eps);
+
+torch::Tensor ppo_loss_gauss_fwd(torch::Tensor pdpi, torch::Tensor pdold,
+                                 torch::Tensor vpred, torch::Tensor oldv,
+                                 torch::Tensor act, torch::Tensor adv,
+                                 torch::Tensor etr, double clip,
+                                 double entcoeff, double vcoeff);
+
+std::vector<torch::Tensor> ppo_loss_gauss_bwd(
+    torch::Tensor pdpi, torch::Tensor pdold, torch::Tensor vpred,
+    torch::Tensor oldv, torch::Tensor act, torch::Tensor adv,
+    torch::Tensor etr, double clip, double entcoeff, double vcoeff,
+    torch::Tensor gtotal);
+
+void adam_step(torch::Tensor p, torch::Tensor g, torch::Tensor m,
+               torch::Tensor v, int64_t step, double lr, double beta1,
+               double beta2, double eps);
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
+  mod.def("gae_scan", &gae_scan,
+          "segmented GAE reverse scan + whitening (gfx950)");
+  mod.def("ppo_loss_gauss_fwd", &ppo_loss_gauss_fwd,
+          "fused DiagGaussian PPO loss forward (gfx950)");
+  mod.def("ppo_loss_gauss_bwd", &ppo_loss_gauss_bwd,
+          "fused DiagGaussian PPO loss backward (gfx950)");
+  mod.def("adam_step", &adam_step, "fused flat Adam step (gfx950)");
+}

@@ -1,0 +1,41 @@
+// Common device helpers for dppo_amd CDNA4 (gfx950) kernels.
+//
+// Written for MI355X: 64-wide wavefronts (every warp idiom uses width 64),
+// wave-level __shfl reductions, one atomic per wave (guideline 12), and
+// double-precision global accumulators (global_atomic_add_f64 is native on
+// CDNA) so large-batch loss reductions keep fp32-class accuracy.
+#pragma once
+
+#include <hip/hip_runtime.h>
+
+#define DEV_INLINE __device__ __forceinline__
+
+constexpr int WAVE = 64;  // CDNA wavefront width (not 32)
+
+DEV_INLINE float wave_reduce_sum(float x) {
+  #pragma unroll
+  for (int off = WAVE / 2; off > 0; off >>= 1) x += __shfl_down(x, off, WAVE);
+  return x;
+}
+
+// Per-wave reduce then ONE double atomic per wave.
+DEV_INLINE void wave_atomic_add(double* dst, float val) {
+  float w = wave_reduce_sum(val);
+  if ((threadIdx.x & (WAVE - 1)) == 0) atomicAdd(dst, static_cast<double>(w));
+}
+
+DEV_INLINE int64_t gidx() {
+  return static_cast<int64_t>(blockIdx.x) * blockDim.x + threadIdx.x;
+}
+
+DEV_INLINE int64_t gstride() {
+  return static_cast<int64_t>(gridDim.x) * blockDim.x;
+}
+
+// Grid sizing for memory-bound elementwise/reduction kernels: cap the
+// grid and grid-stride the rest (guideline 11).
+inline int elementwise_grid(int64_t n, int block) {
+  int64_t g = (n + block - 1) / block;
+  const int64_t cap = 2048;
+  return static_cast<int>(g < cap ? (g > 0 ? g : 1) : cap);
+}

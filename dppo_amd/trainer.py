@@ -106,10 +106,23 @@ class DPPOEngine:
         self.comm.broadcast_(self.flat_pi.flat_param, src=0)
         self.sync_oldpi()
 
-        self.optimizer = torch.optim.Adam([self.flat_pi.flat_param], lr=cfg.LEARNING_RATE)
         # Adam moments live on every rank but stay identical (SURVEY.md §2.3
         # "parameter broadcast" row) — the reference keeps them only on the
         # Chief (PPO.py:20,53); replicated-but-identical is the DP-native form.
+        # GPU: fused flat Adam HIP kernel (K10); CPU/'never': torch Adam.
+        from .ops import use_hip
+
+        if use_hip(self.device, cfg.USE_HIP_KERNELS):
+            from .ops.adam import FusedFlatAdam
+
+            self.optimizer = FusedFlatAdam(
+                self.flat_pi.flat_param, self.flat_pi.flat_grad,
+                lr=cfg.LEARNING_RATE,
+            )
+        else:
+            self.optimizer = torch.optim.Adam(
+                [self.flat_pi.flat_param], lr=cfg.LEARNING_RATE
+            )
 
         self.env: BatchedSyntheticEnv = make_env(cfg, str(self.device), seed + 17)
         self.obs = self.env.reset()

@@ -1,0 +1,152 @@
+"""HIP kernel numerics vs plain-PyTorch fp32 references (gfx950).
+
+Every kernel is compared against the eager reference path of the same op
+(kernel-vs-torch tolerance tests).
+"""
+
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+from dppo_amd.distributions import DiagGaussianPdType
+from dppo_amd.ops import hip_ext, require_hip_ext
+from dppo_amd.ops.gae import gae_advantages_ref
+from dppo_amd.ops.ppo_loss import PPOLossCoeffs, ppo_losses_ref
+
+
+@pytest.fixture(scope="module")
+def ext():
+    return require_hip_ext()
+
+
+def test_gae_scan_matches_ref(ext):
+    T, E = 128, 512
+    g = torch.Generator(device="cuda").manual_seed(0)
+    r = torch.randn(T, E, device="cuda", generator=g)
+    v = torch.randn(T, E, device="cuda", generator=g)
+    d = (torch.rand(T, E, device="cuda", generator=g) < 0.05).float()
+    boot = torch.randn(E, device="cuda", generator=g)
+    adv, etr = ext.gae_scan(r, v, d, boot, 0.99, 0.95, True, 1e-8)
+    adv_ref, etr_ref = gae_advantages_ref(r, v, d, boot, 0.99, 0.95, True, 1e-8)
+    torch.testing.assert_close(adv, adv_ref, atol=2e-4, rtol=2e-4)
+    torch.testing.assert_close(etr, etr_ref, atol=1e-4, rtol=1e-4)
+
+
+def test_gae_scan_long_rollout(ext):
+    """BASELINE config 4 shape: 65k-step scans."""
+    T, E = 65536, 64
+    r = torch.randn(T, E, device="cuda")
+    v = torch.randn(T, E, device="cuda")
+    d = (torch.rand(T, E, device="cuda") < 0.01).float()
+    boot = torch.randn(E, device="cuda")
+    adv, etr = ext.gae_scan(r, v, d, boot, 0.99, 0.95, False, 1e-8)
+    adv_ref, etr_ref = gae_advantages_ref(r, v, d, boot, 0.99, 0.95, False)
+    torch.testing.assert_close(adv, adv_ref, atol=5e-4, rtol=5e-4)
+
+
+def test_gae_whiten_constant_guard(ext):
+    T, E = 8, 64
+    r = torch.ones(T, E, device="cuda")
+    v = torch.zeros(T, E, device="cuda")
+    d = torch.ones(T, E, device="cuda")
+    boot = torch.zeros(E, device="cuda")
+    adv, _ = ext.gae_scan(r, v, d, boot, 0.99, 0.95, True, 1e-8)
+    assert torch.isfinite(adv).all()
+
+
+def _loss_case(B=8192, A=17, seed=0, clip=0.2):
+    g = torch.Generator(device="cuda").manual_seed(seed)
+    pdflat = torch.randn(B, 2 * A, device="cuda", generator=g) * 0.5
+    oldflat = pdflat + 0.1 * torch.randn(B, 2 * A, device="cuda", generator=g)
+    v = torch.randn(B, device="cuda", generator=g)
+    oldv = v + 0.3 * torch.randn(B, device="cuda", generator=g)
+    pdt = DiagGaussianPdType(A)
+    with torch.no_grad():
+        actions = pdt.pdfromflat(oldflat).sample()
+    adv = torch.randn(B, device="cuda", generator=g)
+    etr = torch.randn(B, device="cuda", generator=g)
+    return pdt, pdflat, oldflat, v, oldv, actions, adv, etr
+
+
+def test_ppo_loss_fwd_matches_ref(ext):
+    pdt, pdflat, oldflat, v, oldv, a, adv, etr = _loss_case()
+    clip, entc, vc = 0.2, 0.01, 0.5
+    losses = ext.ppo_loss_gauss_fwd(pdflat, oldflat, v, oldv, a, adv, etr,
+                                    clip, entc, vc)
+    ref = ppo_losses_ref(pdt.pdfromflat(pdflat), pdt.pdfromflat(oldflat),
+                         v, oldv, a, adv, etr, PPOLossCoeffs(clip, entc, vc))
+    torch.testing.assert_close(losses[0], ref["policyLoss"], atol=1e-5, rtol=1e-4)
+    torch.testing.assert_close(losses[1], ref["entropyLoss"], atol=1e-5, rtol=1e-4)
+    torch.testing.assert_close(losses[2], ref["valueLoss"], atol=1e-5, rtol=1e-4)
+    torch.testing.assert_close(losses[3], ref["total_loss"], atol=2e-5, rtol=1e-4)
+
+
+def test_ppo_loss_bwd_matches_autograd(ext):
+    pdt, pdflat, oldflat, v, oldv, a, adv, etr = _loss_case(B=4096, seed=3)
+    clip, entc, vc = 0.2, 0.01, 0.5
+    pdflat_r = pdflat.clone().requires_grad_(True)
+    v_r = v.clone().requires_grad_(True)
+    ref = ppo_losses_ref(pdt.pdfromflat(pdflat_r), pdt.pdfromflat(oldflat),
+                         v_r, oldv, a, adv, etr, PPOLossCoeffs(clip, entc, vc))
+    ref["total_loss"].backward()
+    gt = torch.ones((), device="cuda")
+    g_pdflat, g_v = ext.ppo_loss_gauss_bwd(pdflat, oldflat, v, oldv, a, adv,
+                                           etr, clip, entc, vc, gt)
+    torch.testing.assert_close(g_pdflat, pdflat_r.grad, atol=1e-6, rtol=1e-4)
+    torch.testing.assert_close(g_v, v_r.grad, atol=1e-6, rtol=1e-4)
+
+
+def test_ppo_loss_autograd_function_end_to_end(ext):
+    """The autograd.Function wrapper routes gradients like the eager path."""
+    from dppo_amd.ops.ppo_loss import ppo_losses
+
+    pdt, pdflat, oldflat, v, oldv, a, adv, etr = _loss_case(B=2048, seed=5)
+    coeffs = PPOLossCoeffs(0.2, 0.01, 0.5)
+
+    p1 = pdflat.clone().requires_grad_(True)
+    v1 = v.clone().requires_grad_(True)
+    out = ppo_losses(pdt.pdfromflat(p1), pdt.pdfromflat(oldflat), v1, oldv,
+                     a, adv, etr, coeffs, policy="always")
+    out["total_loss"].backward()
+
+    p2 = pdflat.clone().requires_grad_(True)
+    v2 = v.clone().requires_grad_(True)
+    ref = ppo_losses_ref(pdt.pdfromflat(p2), pdt.pdfromflat(oldflat), v2, oldv,
+                         a, adv, etr, coeffs)
+    ref["total_loss"].backward()
+
+    torch.testing.assert_close(out["total_loss"], ref["total_loss"],
+                               atol=2e-5, rtol=1e-4)
+    torch.testing.assert_close(p1.grad, p2.grad, atol=1e-6, rtol=1e-4)
+    torch.testing.assert_close(v1.grad, v2.grad, atol=1e-6, rtol=1e-4)
+
+
+def test_adam_matches_torch(ext):
+    n = 100_003  # odd size exercises the float4 tail
+    p = torch.randn(n, device="cuda")
+    gref = torch.randn(n, device="cuda")
+
+    p1 = p.clone().requires_grad_(True)
+    p1.grad = gref.clone()
+    opt = torch.optim.Adam([p1], lr=1e-3)
+
+    p2 = p.clone()
+    m = torch.zeros(n, device="cuda")
+    vv = torch.zeros(n, device="cuda")
+    for step in range(1, 4):
+        opt.step()
+        ext.adam_step(p2, gref, m, vv, step, 1e-3, 0.9, 0.999, 1e-8)
+    torch.testing.assert_close(p2, p1.detach(), atol=1e-6, rtol=1e-5)
+    st = opt.state_dict()["state"][0]
+    torch.testing.assert_close(m, st["exp_avg"], atol=1e-7, rtol=1e-6)
+    torch.testing.assert_close(vv, st["exp_avg_sq"], atol=1e-7, rtol=1e-6)
+
+
+def test_native_extension_is_loaded():
+    """The ops layer must be running the native path on GPU, not a silent
+    eager fallback."""
+    import _dppo_hip  # built in-tree; import must succeed on a GPU box
+
+    assert hip_ext() is not None
+    assert hasattr(_dppo_hip, "gae_scan")

@@ -1,0 +1,88 @@
+"""End-to-end GPU training tests (single MI355X)."""
+
+import math
+
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+from dppo_amd.config import DPPOConfig
+from dppo_amd.parallel.comm import Comm
+from dppo_amd.trainer import DPPOEngine
+
+
+def _cfg(**kw):
+    base = dict(
+        GAME="Humanoid-v4", HIDDEN_SIZES=(64, 64), ACTIVATION="tanh",
+        NUM_ENVS=256, MAX_EPOCH_STEPS=16, EPOCH_MAX=1000, STOP_EPOCH=1000,
+        LEARNING_RATE=3e-4, NUM_WORKERS=1,
+        LOG_FILE_PATH="/tmp/dppo_gpu_test_logs", DEVICE="cuda",
+    )
+    base.update(kw)
+    return DPPOConfig(**base)
+
+
+def test_gpu_round_runs():
+    eng = DPPOEngine(_cfg(), comm=Comm(device="cuda:0"))
+    p0 = eng.flat_pi.flat_param.detach().clone()
+    stats, stop = eng.train_round()
+    torch.cuda.synchronize()
+    assert not stop
+    assert all(math.isfinite(v) for v in stats.values())
+    assert not torch.equal(p0, eng.flat_pi.flat_param.detach())
+
+
+def test_gpu_uses_fused_adam():
+    from dppo_amd.ops.adam import FusedFlatAdam
+
+    eng = DPPOEngine(_cfg(), comm=Comm(device="cuda:0"))
+    assert isinstance(eng.optimizer, FusedFlatAdam)
+
+
+def test_hip_vs_eager_training_close():
+    """Several rounds with HIP kernels vs the eager path from identical
+    init stay numerically close (same seeds => same rollouts up to RNG
+    stream differences are avoided by comparing loss magnitudes)."""
+    torch.manual_seed(0)
+    e1 = DPPOEngine(_cfg(USE_HIP_KERNELS="always", SEED=7), comm=Comm(device="cuda:0"))
+    torch.manual_seed(0)
+    e2 = DPPOEngine(_cfg(USE_HIP_KERNELS="never", SEED=7), comm=Comm(device="cuda:0"))
+    torch.testing.assert_close(e1.flat_pi.flat_param, e2.flat_pi.flat_param)
+
+    for _ in range(3):
+        torch.manual_seed(123)
+        e1.env._noise_gen.manual_seed(99)
+        s1, _ = e1.train_round()
+        torch.manual_seed(123)
+        e2.env._noise_gen.manual_seed(99)
+        s2, _ = e2.train_round()
+    # identical seeds -> identical rollouts -> near-identical updates
+    torch.testing.assert_close(
+        e1.flat_pi.flat_param, e2.flat_pi.flat_param, atol=5e-4, rtol=1e-3
+    )
+    assert abs(s1["total_loss"] - s2["total_loss"]) < 1e-2
+
+
+def test_discrete_game_on_gpu():
+    eng = DPPOEngine(
+        _cfg(GAME="CartPole-v0", HIDDEN_SIZES=(16,), ACTIVATION="relu",
+             NUM_ENVS=128),
+        comm=Comm(device="cuda:0"),
+    )
+    stats, _ = eng.train_round()
+    assert math.isfinite(stats["total_loss"])
+
+
+def test_checkpoint_roundtrip_gpu(tmp_path):
+    from dppo_amd.checkpoint import load_state, save_state
+
+    eng = DPPOEngine(_cfg(), comm=Comm(device="cuda:0"))
+    eng.train_round()
+    path = str(tmp_path / "gpu_ckpt.pt")
+    save_state(path, eng)
+    eng2 = DPPOEngine(_cfg(SEED=55), comm=Comm(device="cuda:0"))
+    load_state(path, eng2)
+    torch.testing.assert_close(eng2.flat_pi.flat_param, eng.flat_pi.flat_param)
+    # fused Adam moments restored
+    torch.testing.assert_close(eng2.optimizer.exp_avg, eng.optimizer.exp_avg)
