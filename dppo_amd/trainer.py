@@ -62,8 +62,20 @@ class RolloutBatch:
     actions: torch.Tensor     # [T*E] long or [T*E, A] float
     adv: torch.Tensor         # [T*E] whitened
     etr: torch.Tensor         # [T*E]
+    oldflat: torch.Tensor     # [T*E, P] pi's pd-params recorded at rollout
+                              # time == oldpi outputs (oldpi==pi during the
+                              # rollout, PPO.py:47 sync at round start), so
+                              # the update path never re-runs oldpi.
+    oldv: torch.Tensor        # [T*E] values recorded at rollout time
     cur_lr: float
-    episode_rewards: torch.Tensor  # [n_completed] (device)
+    # Episode-reward moments, accumulated on-device during the rollout so
+    # the hot loop never syncs to host (the reference's per-step python
+    # bookkeeping, Worker.py:57-65, becomes masked tensor ops):
+    ep_count: torch.Tensor    # scalar
+    ep_sum: torch.Tensor
+    ep_sumsq: torch.Tensor
+    ep_min: torch.Tensor
+    ep_max: torch.Tensor
     valid: bool
 
 
@@ -192,7 +204,7 @@ class DPPOEngine:
                 a = torch.where(explore, rand_a, a)
             else:
                 a = torch.where(explore.unsqueeze(-1), rand_a, a)
-        return a, v
+        return a, v, pdflat
 
     @torch.no_grad()
     def rollout_once(self) -> Tuple[RolloutBatch, Dict[str, float]]:
@@ -203,32 +215,50 @@ class DPPOEngine:
         obs_dim = self.obs_space.shape[0]
         eps = self.exploration_rate()
 
-        states = torch.empty(T, E, obs_dim, device=self.device, dtype=self.dtype)
+        P = self.pi.pdtype.param_shape()[0]
+        dev = self.device
+        states = torch.empty(T, E, obs_dim, device=dev, dtype=self.dtype)
+        pdflats = torch.empty(T, E, P, device=dev, dtype=self.dtype)
         if self._discrete:
-            actions = torch.empty(T, E, device=self.device, dtype=torch.long)
+            actions = torch.empty(T, E, device=dev, dtype=torch.long)
         else:
             actions = torch.empty(
-                T, E, self.act_space.shape[0], device=self.device, dtype=self.dtype
+                T, E, self.act_space.shape[0], device=dev, dtype=self.dtype
             )
-        rewards = torch.empty(T, E, device=self.device, dtype=torch.float32)
-        dones = torch.empty(T, E, device=self.device, dtype=torch.float32)
-        values = torch.empty(T, E, device=self.device, dtype=torch.float32)
+        rewards = torch.empty(T, E, device=dev, dtype=torch.float32)
+        dones = torch.empty(T, E, device=dev, dtype=torch.float32)
+        values = torch.empty(T, E, device=dev, dtype=torch.float32)
 
-        ep_rewards: List[torch.Tensor] = []
+        # device-side episode-reward moments: the reference appends python
+        # floats per done (Worker.py:62-65); here masked tensor ops keep
+        # the whole loop sync-free (no host round trip per step).
+        ep_count = torch.zeros((), device=dev)
+        ep_sum = torch.zeros((), device=dev)
+        ep_sumsq = torch.zeros((), device=dev)
+        ep_min = torch.full((), math.inf, device=dev)
+        ep_max = torch.full((), -math.inf, device=dev)
+        ninf = torch.full((), -math.inf, device=dev)
+        pinf = torch.full((), math.inf, device=dev)
+
         obs = self.obs
         for t in range(T):
-            a, v = self.act_batch(obs, eps)
+            a, v, pdflat = self.act_batch(obs, eps)
             states[t] = obs
             actions[t] = a
             values[t] = v.float()
+            pdflats[t] = pdflat
             obs, r, done, _ = self.env.step(a)
             rewards[t] = r.float()
-            dones[t] = done.float()
-            # episode-reward bookkeeping (Worker.py:57-65)
+            donef = done.float()
+            dones[t] = donef
+            # episode-reward bookkeeping (Worker.py:57-65), device-side
             self.epr += r.float()
-            if done.any():
-                ep_rewards.append(self.epr[done])
-                self.epr = torch.where(done, torch.zeros_like(self.epr), self.epr)
+            ep_count += donef.sum()
+            ep_sum += (self.epr * donef).sum()
+            ep_sumsq += (self.epr.square() * donef).sum()
+            ep_min = torch.minimum(ep_min, torch.where(done, self.epr, pinf).min())
+            ep_max = torch.maximum(ep_max, torch.where(done, self.epr, ninf).max())
+            self.epr *= 1.0 - donef
         self.obs = obs
 
         with torch.no_grad():
@@ -239,19 +269,17 @@ class DPPOEngine:
             policy=c.USE_HIP_KERNELS,
         )
 
-        epr_cat = (
-            torch.cat(ep_rewards)
-            if ep_rewards
-            else torch.empty(0, device=self.device)
-        )
         batch = RolloutBatch(
             states=states.reshape(T * E, obs_dim),
             actions=actions.reshape(T * E, *actions.shape[2:]),
             adv=adv.reshape(T * E).to(self.dtype),
             etr=etr.reshape(T * E).to(self.dtype),
+            oldflat=pdflats.reshape(T * E, P),
+            oldv=values.reshape(T * E).to(self.dtype),
             cur_lr=self.current_lr_mul(),
-            episode_rewards=epr_cat,
-            valid=epr_cat.numel() > 0,
+            ep_count=ep_count, ep_sum=ep_sum, ep_sumsq=ep_sumsq,
+            ep_min=ep_min, ep_max=ep_max,
+            valid=float(ep_count) > 0,  # the rollout's single host sync
         )
         return batch, {"exploration_rate": eps}
 
@@ -266,25 +294,37 @@ class DPPOEngine:
 
     # ------------------------------------------------------------------
     def eval_losses(self, batch: RolloutBatch, l_mul: float) -> Dict[str, float]:
-        """Pre-update loss evaluation for the logs vector (Worker.py:117-118)."""
+        """Pre-update loss evaluation for the logs vector (Worker.py:117-118).
+
+        The reference evaluates its losses right after the rollout, when
+        pi still equals oldpi (sync_pis ran at round start), so this needs
+        NO forward pass: both networks' outputs on the batch are the
+        recorded rollout outputs."""
         with torch.no_grad():
-            losses = self._losses(batch, l_mul)
+            losses = self._losses(batch, l_mul, recorded_pi=True)
         return {k: float(v) for k, v in losses.items()}
 
-    def _losses(self, batch: RolloutBatch, l_mul: float) -> Dict[str, torch.Tensor]:
-        v, pdflat = self.pi(batch.states)
-        with torch.no_grad():
-            oldv, oldflat = self.oldpi(batch.states)
+    def _losses(
+        self, batch: RolloutBatch, l_mul: float, recorded_pi: bool = False
+    ) -> Dict[str, torch.Tensor]:
+        """PPO losses on a batch.  oldpi's outputs are the recorded rollout
+        outputs (batch.oldflat/oldv) — bit-identical to re-running oldpi,
+        which equals pi at rollout time (PPO.py:47 + Worker.py:42), without
+        the 3 extra GEMMs per update step."""
+        if recorded_pi:
+            v, pdflat = batch.oldv, batch.oldflat
+        else:
+            v, pdflat = self.pi(batch.states)
         pd = self.pi.pdtype.pdfromflat(pdflat)
-        oldpd = self.pi.pdtype.pdfromflat(oldflat)
+        oldpd = self.pi.pdtype.pdfromflat(batch.oldflat)
         coeffs = PPOLossCoeffs(
             clip_param=self.cfg.CLIP_PARAM * l_mul,
             entcoeff=self.cfg.ENTCOEFF,
             vcoeff=self.cfg.VCOEFF,
         )
         return ppo_losses(
-            pd, oldpd, v, oldv, batch.actions, batch.adv, batch.etr, coeffs,
-            policy=self.cfg.USE_HIP_KERNELS,
+            pd, oldpd, v, batch.oldv, batch.actions, batch.adv, batch.etr,
+            coeffs, policy=self.cfg.USE_HIP_KERNELS,
         )
 
     def stats_row(self, batch: RolloutBatch, losses: Dict[str, float]) -> torch.Tensor:
@@ -292,14 +332,15 @@ class DPPOEngine:
 
         score = epr.mean()/epr.std() is NaN-prone for a single episode
         (Worker.py:121) — computed but guarded to 0."""
-        epr = batch.episode_rewards
-        if batch.valid and epr.numel() > 1 and float(epr.std(unbiased=False)) > 0:
-            score = float(epr.mean() / epr.std(unbiased=False))
+        if batch.valid:
+            n = float(batch.ep_count)
+            mean = float(batch.ep_sum) / n
+            var = max(float(batch.ep_sumsq) / n - mean * mean, 0.0)
+            std = math.sqrt(var)  # population std, matching numpy's epr.std()
+            score = mean / std if (n > 1 and std > 0) else 0.0
+            mn, mx = float(batch.ep_min), float(batch.ep_max)
         else:
             score = 0.0
-        if batch.valid:
-            mn, mx, mean = float(epr.min()), float(epr.max()), float(epr.mean())
-        else:
             mn = mx = mean = -math.inf  # never wins the best-rank sort
         row = torch.tensor(
             [

@@ -70,7 +70,7 @@ class Worker:
         scalar for Discrete, numpy array for Box)."""
         eng = self.engine
         s_t = torch.as_tensor(s, device=eng.device, dtype=eng.dtype).unsqueeze(0)
-        a, v = eng.act_batch(s_t, eng.exploration_rate())
+        a, v, _ = eng.act_batch(s_t, eng.exploration_rate())
         a0 = a[0]
         if eng._discrete:
             return int(a0), float(v[0])
