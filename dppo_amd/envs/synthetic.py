@@ -88,6 +88,13 @@ class BatchedSyntheticEnv:
         span = max(horizon, 2)
         self.horizons = (horizon // 2 + (e * 2654435761 % span)).to(self.device)
         self.horizons = torch.clamp(self.horizons, min=2)
+        self.horizons_i32 = self.horizons.to(torch.int32).contiguous()
+
+        # Transposed parameter layouts for the fused HIP rollout kernel
+        # (per-lane contiguous streams in the env phase; rollout.hip).
+        self.Vt = self.V.t().contiguous()   # [r, D]
+        self.Ut = self.U.t().contiguous()   # [D, r]
+        self.Bt = self.B.t().contiguous()   # [D, A] (or [D, n] discrete)
 
         self._noise_gen = torch.Generator(device=self.device.type).manual_seed(seed + 1)
         self.x = torch.zeros(num_envs, obs_dim, device=self.device, dtype=dtype)

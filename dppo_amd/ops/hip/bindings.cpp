@@ -24,6 +24,16 @@ void adam_step(torch::Tensor p, torch::Tensor g, torch::Tensor m,
                torch::Tensor v, int64_t step, double lr, double beta1,
                double beta2, double eps);
 
+std::vector<torch::Tensor> rollout_run(
+    std::vector<torch::Tensor> Ws, std::vector<torch::Tensor> bs,
+    torch::Tensor Wv, torch::Tensor bv, torch::Tensor Wp, torch::Tensor bp,
+    int64_t activation,
+    torch::Tensor env_d, torch::Tensor env_Vt, torch::Tensor env_Ut,
+    torch::Tensor env_Bt, torch::Tensor horizons, double noise,
+    double act_low, double act_high, double eps_explore,
+    torch::Tensor x, torch::Tensor t, torch::Tensor epr,
+    int64_t T, int64_t seed);
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("gae_scan", &gae_scan,
           "segmented GAE reverse scan + whitening (gfx950)");
@@ -32,4 +42,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("ppo_loss_gauss_bwd", &ppo_loss_gauss_bwd,
           "fused DiagGaussian PPO loss backward (gfx950)");
   mod.def("adam_step", &adam_step, "fused flat Adam step (gfx950)");
+  mod.def("rollout_run", &rollout_run,
+          "fused T-step rollout: MLP fwd + sample + synthetic env (gfx950)");
 }

@@ -206,8 +206,86 @@ class DPPOEngine:
                 a = torch.where(explore.unsqueeze(-1), rand_a, a)
         return a, v, pdflat
 
+    def _can_fuse_rollout(self) -> bool:
+        """Eligibility for the fused HIP rollout kernel (rollout.hip):
+        Box/DiagGaussian policy, dims within kernel limits, fp32, GPU."""
+        from .ops import use_hip
+
+        c = self.cfg
+        if self._discrete or c.DTYPE != "float32":
+            return False
+        if not use_hip(self.device, c.USE_HIP_KERNELS):
+            return False
+        if not (1 <= len(c.HIDDEN_SIZES) <= 3):
+            return False
+        if max(c.HIDDEN_SIZES) > 128 or self.obs_space.shape[0] > 512:
+            return False
+        if self.act_space.shape[0] > 32 or self.env.Vt.size(0) > 32:
+            return False
+        return True
+
     @torch.no_grad()
     def rollout_once(self) -> Tuple[RolloutBatch, Dict[str, float]]:
+        if self._can_fuse_rollout():
+            return self._rollout_once_hip()
+        return self._rollout_once_eager()
+
+    @torch.no_grad()
+    def _rollout_once_hip(self) -> Tuple[RolloutBatch, Dict[str, float]]:
+        """One rollout iteration in a single fused kernel launch
+        (ops/hip/rollout.hip): MLP forward + sampling + epsilon-greedy +
+        env dynamics + episode bookkeeping for all T steps."""
+        from .ops import hip_ext
+
+        ext = hip_ext()
+        c, E = self.cfg, self.cfg.NUM_ENVS
+        T = c.MAX_EPOCH_STEPS
+        eps = self.exploration_rate()
+        env = self.env
+        self._rollout_counter = getattr(self, "_rollout_counter", 0) + 1
+        seed = (
+            c.SEED * 1_000_003
+            + self.comm.rank * 7_919
+            + self._rollout_counter * 104_729
+        ) & 0x7FFFFFFFFFFFFFFF
+        low = float(self.act_space.low.flat[0])
+        high = float(self.act_space.high.flat[0])
+        (states, pdflats, actions, values, rewards, dones, boot_v,
+         moments) = ext.rollout_run(
+            [l.weight for l in self.pi.hidden],
+            [l.bias for l in self.pi.hidden],
+            self.pi.vf.weight.contiguous(), self.pi.vf.bias,
+            self.pi.pi.weight.contiguous(), self.pi.pi.bias,
+            1 if c.ACTIVATION == "tanh" else 0,
+            env.d, env.Vt, env.Ut, env.Bt, env.horizons_i32,
+            float(env.NOISE), low, high, float(eps),
+            env.x, env.t, self.epr, T, seed,
+        )
+        self.obs = env.x  # updated in place by the kernel
+
+        adv, etr = gae_advantages(
+            rewards, values, dones, boot_v,
+            c.GAMMA, c.LAM, whiten=True, eps=c.ADV_EPS,
+            policy=c.USE_HIP_KERNELS,
+        )
+        obs_dim = self.obs_space.shape[0]
+        P = self.pi.pdtype.param_shape()[0]
+        batch = RolloutBatch(
+            states=states.reshape(T * E, obs_dim),
+            actions=actions.reshape(T * E, self.act_space.shape[0]),
+            adv=adv.reshape(T * E),
+            etr=etr.reshape(T * E),
+            oldflat=pdflats.reshape(T * E, P),
+            oldv=values.reshape(T * E),
+            cur_lr=self.current_lr_mul(),
+            ep_count=moments[0], ep_sum=moments[1], ep_sumsq=moments[2],
+            ep_min=moments[3], ep_max=moments[4],
+            valid=float(moments[0]) > 0,  # the rollout's single host sync
+        )
+        return batch, {"exploration_rate": eps}
+
+    @torch.no_grad()
+    def _rollout_once_eager(self) -> Tuple[RolloutBatch, Dict[str, float]]:
         """Collect one iteration of T = MAX_EPOCH_STEPS batched env steps
         (Worker.py:39-65), then GAE (Worker.py:82-92)."""
         c, E = self.cfg, self.cfg.NUM_ENVS
