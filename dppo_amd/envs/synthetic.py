@@ -95,6 +95,12 @@ class BatchedSyntheticEnv:
         self.Vt = self.V.t().contiguous()   # [r, D]
         self.Ut = self.U.t().contiguous()   # [D, r]
         self.Bt = self.B.t().contiguous()   # [D, A] (or [D, n] discrete)
+        # single packed blob (d | Vt | Ut | Bt) so the kernel takes one pointer
+        self.blob = torch.cat(
+            [self.d.reshape(-1), self.Vt.reshape(-1), self.Ut.reshape(-1),
+             self.Bt.reshape(-1)]
+        ).contiguous()
+        self.rank_eff = self.Vt.shape[0]
 
         self._noise_gen = torch.Generator(device=self.device.type).manual_seed(seed + 1)
         self.x = torch.zeros(num_envs, obs_dim, device=self.device, dtype=dtype)

@@ -25,14 +25,12 @@ void adam_step(torch::Tensor p, torch::Tensor g, torch::Tensor m,
                double beta2, double eps);
 
 std::vector<torch::Tensor> rollout_run(
-    std::vector<torch::Tensor> Ws, std::vector<torch::Tensor> bs,
-    torch::Tensor Wv, torch::Tensor bv, torch::Tensor Wp, torch::Tensor bp,
-    int64_t activation,
-    torch::Tensor env_d, torch::Tensor env_Vt, torch::Tensor env_Ut,
-    torch::Tensor env_Bt, torch::Tensor horizons, double noise,
-    double act_low, double act_high, double eps_explore,
+    torch::Tensor params, std::vector<int64_t> offsets,
+    std::vector<int64_t> dims, int64_t activation,
+    torch::Tensor envblob, int64_t rank, torch::Tensor horizons,
+    double noise, double act_low, double act_high, double eps_explore,
     torch::Tensor x, torch::Tensor t, torch::Tensor epr,
-    int64_t T, int64_t seed);
+    int64_t T, int64_t act_dim, int64_t seed);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("gae_scan", &gae_scan,

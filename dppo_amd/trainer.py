@@ -250,16 +250,17 @@ class DPPOEngine:
         ) & 0x7FFFFFFFFFFFFFFF
         low = float(self.act_space.low.flat[0])
         high = float(self.act_space.high.flat[0])
+        # weight offsets into the flat parameter buffer, in parameters()
+        # registration order: hidden (W, b)*, vf.W, vf.b, pi.W, pi.b
+        offsets = [sl.start for sl in self.flat_pi.slices]
+        dims = [self.obs_space.shape[0], *c.HIDDEN_SIZES]
         (states, pdflats, actions, values, rewards, dones, boot_v,
          moments) = ext.rollout_run(
-            [l.weight for l in self.pi.hidden],
-            [l.bias for l in self.pi.hidden],
-            self.pi.vf.weight.contiguous(), self.pi.vf.bias,
-            self.pi.pi.weight.contiguous(), self.pi.pi.bias,
+            self.flat_pi.flat_param.detach(), offsets, dims,
             1 if c.ACTIVATION == "tanh" else 0,
-            env.d, env.Vt, env.Ut, env.Bt, env.horizons_i32,
+            env.blob, env.rank_eff, env.horizons_i32,
             float(env.NOISE), low, high, float(eps),
-            env.x, env.t, self.epr, T, seed,
+            env.x, env.t, self.epr, T, self.act_space.shape[0], seed,
         )
         self.obs = env.x  # updated in place by the kernel
 

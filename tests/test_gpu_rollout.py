@@ -38,15 +38,14 @@ def run_kernel(eng, T=32, eps=0.0, noise=None, seed=1234):
         env.NOISE = noise
     low = float(eng.act_space.low.flat[0])
     high = float(eng.act_space.high.flat[0])
+    offsets = [sl.start for sl in eng.flat_pi.slices]
+    dims = [eng.obs_space.shape[0], *eng.cfg.HIDDEN_SIZES]
     return ext.rollout_run(
-        [l.weight for l in eng.pi.hidden],
-        [l.bias for l in eng.pi.hidden],
-        eng.pi.vf.weight.contiguous(), eng.pi.vf.bias,
-        eng.pi.pi.weight.contiguous(), eng.pi.pi.bias,
+        eng.flat_pi.flat_param.detach(), offsets, dims,
         1 if eng.cfg.ACTIVATION == "tanh" else 0,
-        env.d, env.Vt, env.Ut, env.Bt, env.horizons_i32,
+        env.blob, env.rank_eff, env.horizons_i32,
         float(env.NOISE), low, high, float(eps),
-        env.x, env.t, eng.epr, T, seed,
+        env.x, env.t, eng.epr, T, eng.act_space.shape[0], seed,
     )
 
 
