@@ -24,6 +24,22 @@ void adam_step(torch::Tensor p, torch::Tensor g, torch::Tensor m,
                torch::Tensor v, int64_t step, double lr, double beta1,
                double beta2, double eps);
 
+std::vector<torch::Tensor> mlp_fwd(torch::Tensor params,
+                                   std::vector<int64_t> offsets,
+                                   std::vector<int64_t> dims,
+                                   int64_t activation, torch::Tensor states,
+                                   int64_t act_dim);
+
+std::vector<torch::Tensor> mlp_bwd_rows(
+    torch::Tensor params, std::vector<int64_t> offsets,
+    std::vector<int64_t> dims, int64_t activation, torch::Tensor acts,
+    torch::Tensor pdflat, torch::Tensor oldflat, torch::Tensor v,
+    torch::Tensor oldv, torch::Tensor actions, torch::Tensor adv,
+    torch::Tensor etr, double clip, double entcoeff, double vcoeff);
+
+void dw_accum(torch::Tensor delta, torch::Tensor acts, torch::Tensor grad_buf,
+              int64_t w_off, int64_t b_off);
+
 std::vector<torch::Tensor> rollout_run(
     torch::Tensor params, std::vector<int64_t> offsets,
     std::vector<int64_t> dims, int64_t activation,
@@ -42,4 +58,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("adam_step", &adam_step, "fused flat Adam step (gfx950)");
   mod.def("rollout_run", &rollout_run,
           "fused T-step rollout: MLP fwd + sample + synthetic env (gfx950)");
+  mod.def("mlp_fwd", &mlp_fwd, "fused MLP forward, saves activations (gfx950)");
+  mod.def("mlp_bwd_rows", &mlp_bwd_rows,
+          "fused PPO-loss + MLP backward rows -> dz/g_pd/g_v (gfx950)");
+  mod.def("dw_accum", &dw_accum,
+          "dW += delta^T @ acts, db += sum(delta) into flat grad (gfx950)");
 }

@@ -436,20 +436,88 @@ class DPPOEngine:
         return row
 
     # ------------------------------------------------------------------
-    def update(self, batch: RolloutBatch, l_mul: float) -> Dict[str, float]:
+    def _can_fuse_update(self) -> bool:
+        """Eligibility for the fused HIP update path (mlp_train.hip)."""
+        from .ops import use_hip
+
+        c = self.cfg
+        if self._discrete or c.DTYPE != "float32":
+            return False
+        if not use_hip(self.device, c.USE_HIP_KERNELS):
+            return False
+        if not (1 <= len(c.HIDDEN_SIZES) <= 3) or max(c.HIDDEN_SIZES) > 128:
+            return False
+        if self.act_space.shape[0] > 32:
+            return False
+        dims = [self.obs_space.shape[0], *c.HIDDEN_SIZES]
+        P, HL = 2 * self.act_space.shape[0], dims[-1]
+        hsz = sum(dims[l + 1] * dims[l] for l in range(1, len(dims) - 1))
+        return hsz + P * HL + HL <= 24576  # bwd staged-weight LDS budget
+
+    def update(self, batch: RolloutBatch, l_mul: float) -> None:
         """UPDATE_STEPS repeated full-batch steps on the same data
-        (Chief.py:64), DP gradient mean across ranks each step."""
+        (Chief.py:64), DP gradient mean across ranks each step.  The
+        reference's Chief fetches only the train op (Chief.py:64), so no
+        loss values are materialized here."""
         for g in self.optimizer.param_groups:
             g["lr"] = self.cfg.LEARNING_RATE * l_mul
-        last: Dict[str, torch.Tensor] = {}
+        if self._can_fuse_update():
+            self._update_fused(batch, l_mul)
+            return
         for _ in range(self.cfg.UPDATE_STEPS):
             self.flat_pi.zero_grad()
             losses = self._losses(batch, l_mul)
             losses["total_loss"].backward()
             self.comm.allreduce_mean_(self.flat_pi.flat_grad)
             self.optimizer.step()
-            last = losses
-        return {k: float(v.detach()) for k, v in last.items()}
+
+    def _update_fused(self, batch: RolloutBatch, l_mul: float) -> None:
+        """Fused HIP update steps: mlp_fwd -> mlp_bwd_rows -> dw_accum x L
+        -> all-reduce -> fused Adam (~7 launches per step, no autograd)."""
+        from .ops import hip_ext
+
+        ext = hip_ext()
+        c = self.cfg
+        A = self.act_space.shape[0]
+        offsets = [sl.start for sl in self.flat_pi.slices]
+        dims = [self.obs_space.shape[0], *c.HIDDEN_SIZES]
+        act_code = 1 if c.ACTIVATION == "tanh" else 0
+        clip = c.CLIP_PARAM * l_mul
+        B = batch.states.shape[0]
+        n_hidden = len(c.HIDDEN_SIZES)
+        g_v_2d = None
+        for _ in range(c.UPDATE_STEPS):
+            acts, v, pdflat = ext.mlp_fwd(
+                self.flat_pi.flat_param.detach(), offsets, dims, act_code,
+                batch.states, A,
+            )
+            self.flat_pi.zero_grad()
+            dz, g_pd, g_v = ext.mlp_bwd_rows(
+                self.flat_pi.flat_param.detach(), offsets, dims, act_code,
+                acts, pdflat, batch.oldflat, v, batch.oldv,
+                batch.actions, batch.adv, batch.etr,
+                clip, c.ENTCOEFF, c.VCOEFF,
+            )
+            # carve per-layer activation/dz views out of the blobs
+            a_views, dz_views, o = [], [], 0
+            for l in range(n_hidden):
+                n = B * dims[l + 1]
+                a_views.append(acts.narrow(0, o, n).view(B, dims[l + 1]))
+                dz_views.append(dz.narrow(0, o, n).view(B, dims[l + 1]))
+                o += n
+            grad = self.flat_pi.flat_grad
+            ext.dw_accum(dz_views[0], batch.states, grad, offsets[0], offsets[1])
+            for l in range(1, n_hidden):
+                ext.dw_accum(dz_views[l], a_views[l - 1], grad,
+                             offsets[2 * l], offsets[2 * l + 1])
+            h_last = a_views[-1]
+            off_wv, off_bv = offsets[2 * n_hidden], offsets[2 * n_hidden + 1]
+            off_wp, off_bp = offsets[2 * n_hidden + 2], offsets[2 * n_hidden + 3]
+            ext.dw_accum(g_pd, h_last, grad, off_wp, off_bp)
+            g_v_2d = g_v.view(B, 1)
+            ext.dw_accum(g_v_2d, h_last, grad, off_wv, off_bv)
+            self.comm.allreduce_mean_(grad)
+            self.optimizer.step()
 
     # ------------------------------------------------------------------
     def train_round(self) -> Tuple[Dict[str, float], bool]:
