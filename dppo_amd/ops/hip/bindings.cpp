@@ -40,6 +40,16 @@ std::vector<torch::Tensor> mlp_bwd_rows(
 void dw_accum(torch::Tensor delta, torch::Tensor acts, torch::Tensor grad_buf,
               int64_t w_off, int64_t b_off);
 
+void gemm_fwd(torch::Tensor X, torch::Tensor Wt, torch::Tensor bias,
+              int64_t activation, int64_t heads, torch::Tensor C,
+              torch::Tensor v);
+
+void dw_mfma(torch::Tensor delta, torch::Tensor acts, torch::Tensor grad_buf,
+             int64_t w_off, int64_t b_off);
+
+void dwv(torch::Tensor gv, torch::Tensor acts, torch::Tensor grad_buf,
+         int64_t w_off, int64_t b_off);
+
 std::vector<torch::Tensor> rollout_run(
     torch::Tensor params, std::vector<int64_t> offsets,
     std::vector<int64_t> dims, int64_t activation,
@@ -61,6 +71,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("mlp_fwd", &mlp_fwd, "fused MLP forward, saves activations (gfx950)");
   mod.def("mlp_bwd_rows", &mlp_bwd_rows,
           "fused PPO-loss + MLP backward rows -> dz/g_pd/g_v (gfx950)");
+  mod.def("gemm_fwd", &gemm_fwd,
+          "MFMA f32 layer forward C=act(X@Wt+b), fused tanh (gfx950)");
+  mod.def("dw_mfma", &dw_mfma,
+          "MFMA f32 split-K dW += delta^T@acts into flat grad (gfx950)");
+  mod.def("dwv", &dwv, "value-head weight grad reduction (gfx950)");
   mod.def("dw_accum", &dw_accum,
           "dW += delta^T @ acts, db += sum(delta) into flat grad (gfx950)");
 }

@@ -1,0 +1,345 @@
+// MFMA f32 GEMM kernels for the MLP update path (gfx950).
+//
+// gfx950 has exact fp32-input MFMA (v_mfma_f32_32x32x2_f32: bitwise a
+// k-ordered fmaf chain, 157 TF chip peak = the f32 vector peak — cdna
+// guide §3 "FP32-input MFMA") — ~2.4x a VALU f32 GEMM at identical
+// numerics, and an order of magnitude over rocBLAS/Tensile's fp32
+// tall-skinny picks on these shapes (<1 TB/s effective measured).
+//
+//   gemm_fwd:   C[B,N] = act(X[B,K] @ Wt[K,N] + bias) — one hidden layer,
+//               fused activation (tanh/relu), activations ARE the saved
+//               forward outputs.  X tiles are LDS-staged (the MFMA A
+//               fragment is a column read, lane = row); Wt streams from
+//               L2 coalesced (reused by every row block).
+//   gemm_heads: same, N = 2A+1, no activation; epilogue splits columns
+//               into pdflat[B,2A] and v[B].
+//   dw_mfma:    dW[out,in] += delta^T @ acts (split-K over row blocks,
+//               fp32 atomics into the flat grad; fused db += sum(delta)).
+//   dwv:        the out==1 case (value-head weight grad) as a plain
+//               wave-per-column reduction (MFMA would waste 31/32 lanes).
+//
+// Fragment layout (cdna guide §3, v_mfma_f32_32x32x2_f32):
+//   lane l: A[i = l&31][k = l>>5], B[k = l>>5][j = l&31]
+//   C/D reg r (of 16): row = (r&3) + 8*(r>>2) + 4*(l>>5), col = l&31.
+
+#include <hip/hip_runtime.h>
+#include <torch/extension.h>
+#include <c10/hip/HIPStream.h>
+
+#include "common.h"
+
+namespace {
+
+using f32x16 = __attribute__((ext_vector_type(16))) float;
+
+constexpr int BK = 32;        // K-step per stage
+constexpr int M_WAVE = 32;    // rows per wave tile
+constexpr int FWD_WAVES = 4;  // waves per block (each owns 32 rows)
+constexpr int FWD_M = FWD_WAVES * M_WAVE;  // 128 rows per block
+constexpr int MAX_NT = 4;     // 32-col accumulator tiles per wave
+
+DEV_INLINE int cd_row(int reg, int lane) {
+  return (reg & 3) + 8 * (reg >> 2) + 4 * (lane >> 5);
+}
+
+// ---------------------------------------------------------------------------
+// Forward layer: C = act(X @ Wt + b).  heads_mode splits the epilogue.
+// ---------------------------------------------------------------------------
+
+struct FwdArgs {
+  const float* X;     // [B][K]
+  const float* Wt;    // [K][N]
+  const float* bias;  // [N]
+  float* C;           // [B][N]      (heads: pdflat [B][N-1])
+  float* v;           // [B] heads only
+  int64_t B;
+  int K, N;
+  int activation;     // 0 relu, 1 tanh, 2 none
+  int heads;          // if 1: last column -> v, rest -> C (pdflat)
+};
+
+__launch_bounds__(FWD_WAVES * 64)
+__global__ void gemm_fwd_kernel(FwdArgs a) {
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wave = threadIdx.x / WAVE;
+  const int NT = (a.N + M_WAVE - 1) / M_WAVE;  // <= MAX_NT
+
+  // X tile [FWD_M][BK] with +1 padding (A-fragment read is a column read)
+  __shared__ float xs[FWD_M][BK + 1];
+
+  const int i_l = lane & 31;   // A row within wave tile
+  const int k_l = lane >> 5;   // A k within pair
+
+  for (int64_t tile = blockIdx.x; tile * FWD_M < a.B; tile += gridDim.x) {
+    const int64_t b0 = tile * FWD_M;
+    f32x16 acc[MAX_NT];
+    #pragma unroll
+    for (int t = 0; t < MAX_NT; ++t)
+      #pragma unroll
+      for (int r = 0; r < 16; ++r) acc[t][r] = 0.f;
+
+    for (int kb = 0; kb < a.K; kb += BK) {
+      // ---- stage X[b0:b0+128][kb:kb+BK] (coalesced; zero-pad tails) ----
+      __syncthreads();
+      for (int idx = threadIdx.x; idx < FWD_M * BK; idx += FWD_WAVES * 64) {
+        const int r = idx / BK, c = idx % BK;
+        const int64_t row = b0 + r;
+        const int col = kb + c;
+        xs[r][c] = (row < a.B && col < a.K) ? a.X[row * a.K + col] : 0.f;
+      }
+      __syncthreads();
+
+      const int ksteps = min(BK, a.K - kb);
+      for (int k2 = 0; k2 < ksteps; k2 += 2) {
+        const float av = xs[wave * M_WAVE + i_l][k2 + k_l];
+        #pragma unroll
+        for (int t = 0; t < MAX_NT; ++t) {
+          if (t < NT) {
+            const int col = t * M_WAVE + i_l;
+            const int krow = kb + k2 + k_l;
+            const float bv = (col < a.N && krow < a.K)
+                                 ? a.Wt[(int64_t)krow * a.N + col]
+                                 : 0.f;
+            acc[t] = __builtin_amdgcn_mfma_f32_32x32x2f32(av, bv, acc[t], 0, 0, 0);
+          }
+        }
+      }
+    }
+
+    // ---- epilogue: bias + activation + store ----
+    #pragma unroll
+    for (int t = 0; t < MAX_NT; ++t) {
+      if (t < NT) {
+        const int col = t * M_WAVE + i_l;
+        if (col < a.N) {
+          const float bv = a.bias[col];
+          #pragma unroll
+          for (int r = 0; r < 16; ++r) {
+            const int64_t row = b0 + wave * M_WAVE + cd_row(r, lane);
+            if (row < a.B) {
+              float x = acc[t][r] + bv;
+              if (a.activation == 0) x = fmaxf(x, 0.f);
+              else if (a.activation == 1) x = tanhf(x);
+              if (a.heads) {
+                if (col == a.N - 1) a.v[row] = x;
+                else a.C[row * (a.N - 1) + col] = x;
+              } else {
+                a.C[row * a.N + col] = x;
+              }
+            }
+          }
+        }
+      }
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// dW accumulation: dW[out][in] += delta^T @ acts (+ db += sum delta)
+// grid = (m_tiles, n_tiles, splits); one wave per block.
+// ---------------------------------------------------------------------------
+
+struct DwArgs {
+  const float* delta;  // [B][out]
+  const float* acts;   // [B][in]
+  float* dW;           // grad + w_off  ([out][in])
+  float* db;           // grad + b_off or nullptr
+  int64_t B;
+  int out_dim, in_dim;
+  int nt;      // 32-col tiles per wave (<= MAX_NT)
+  int splits;  // K splits
+};
+
+__launch_bounds__(64)
+__global__ void dw_mfma_kernel(DwArgs a) {
+  const int lane = threadIdx.x;
+  const int m0 = blockIdx.x * M_WAVE;
+  const int n_base = blockIdx.y * a.nt * M_WAVE;
+  const int split = blockIdx.z;
+
+  const int64_t rows_per = (a.B + a.splits - 1) / a.splits;
+  const int64_t k0 = split * rows_per;
+  const int64_t k1 = min(a.B, k0 + rows_per);
+
+  const int i_l = lane & 31;
+  const int k_l = lane >> 5;
+  const int mcol = m0 + i_l;          // delta column this lane reads
+  const bool m_ok = mcol < a.out_dim;
+
+  f32x16 acc[MAX_NT];
+  #pragma unroll
+  for (int t = 0; t < MAX_NT; ++t)
+    #pragma unroll
+    for (int r = 0; r < 16; ++r) acc[t][r] = 0.f;
+  float dbacc = 0.f;
+
+  for (int64_t k = k0; k + 1 < k1; k += 2) {
+    const float av = m_ok ? a.delta[(k + k_l) * a.out_dim + mcol] : 0.f;
+    dbacc += av;
+    #pragma unroll
+    for (int t = 0; t < MAX_NT; ++t) {
+      if (t < a.nt) {
+        const int col = n_base + t * M_WAVE + i_l;
+        const float bv = (col < a.in_dim)
+                             ? a.acts[(k + k_l) * a.in_dim + col]
+                             : 0.f;
+        acc[t] = __builtin_amdgcn_mfma_f32_32x32x2f32(av, bv, acc[t], 0, 0, 0);
+      }
+    }
+  }
+  // odd tail row
+  if (((k1 - k0) & 1) && k1 > k0) {
+    const int64_t k = k1 - 1;
+    const float av = (m_ok && k_l == 0) ? a.delta[k * a.out_dim + mcol] : 0.f;
+    if (k_l == 0) dbacc += av;
+    #pragma unroll
+    for (int t = 0; t < MAX_NT; ++t) {
+      if (t < a.nt) {
+        const int col = n_base + t * M_WAVE + i_l;
+        const float bv = (col < a.in_dim && k_l == 0)
+                             ? a.acts[k * a.in_dim + col]
+                             : 0.f;
+        acc[t] = __builtin_amdgcn_mfma_f32_32x32x2f32(av, bv, acc[t], 0, 0, 0);
+      }
+    }
+  }
+
+  // ---- scatter accumulators ----
+  #pragma unroll
+  for (int t = 0; t < MAX_NT; ++t) {
+    if (t < a.nt) {
+      const int col = n_base + t * M_WAVE + i_l;
+      if (col < a.in_dim) {
+        #pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          const int mrow = m0 + cd_row(r, lane);
+          if (mrow < a.out_dim && acc[t][r] != 0.f) {
+            atomicAdd(&a.dW[(int64_t)mrow * a.in_dim + col], acc[t][r]);
+          }
+        }
+      }
+    }
+  }
+  // db: lane's dbacc covers rows k with k_l parity; combine lane and
+  // lane+32 (same m-column), then one atomic per m from the low half.
+  if (a.db != nullptr && blockIdx.y == 0) {
+    const float other = __shfl(dbacc, lane ^ 32, WAVE);
+    if (lane < 32 && m_ok) atomicAdd(&a.db[mcol], dbacc + other);
+  }
+}
+
+// ---------------------------------------------------------------------------
+// out == 1 weight grad: dWv[n] += sum_k gv[k] * acts[k][n]; dbv += sum gv
+// ---------------------------------------------------------------------------
+
+__launch_bounds__(64)
+__global__ void dwv_kernel(const float* __restrict__ gv,    // [B]
+                           const float* __restrict__ acts,  // [B][in]
+                           float* __restrict__ dWv,         // [in]
+                           float* __restrict__ dbv,         // [1] or null
+                           int64_t B, int in_dim, int splits) {
+  const int lane = threadIdx.x;
+  const int n0 = blockIdx.x * WAVE;
+  const int split = blockIdx.y;
+  const int64_t rows_per = (B + splits - 1) / splits;
+  const int64_t k0 = split * rows_per;
+  const int64_t k1 = min(B, k0 + rows_per);
+  const int col = n0 + lane;
+  float acc = 0.f, bacc = 0.f;
+  for (int64_t k = k0; k < k1; ++k) {
+    const float g = gv[k];
+    bacc += g;
+    if (col < in_dim) acc += g * acts[k * in_dim + col];
+  }
+  if (col < in_dim) atomicAdd(&dWv[col], acc);
+  if (dbv != nullptr && blockIdx.x == 0 && lane == 0) atomicAdd(dbv, bacc);
+}
+
+}  // namespace
+
+// ---------------------------------------------------------------------------
+// Bindings
+// ---------------------------------------------------------------------------
+
+void gemm_fwd(torch::Tensor X, torch::Tensor Wt, torch::Tensor bias,
+              int64_t activation, int64_t heads, torch::Tensor C,
+              torch::Tensor v) {
+  // C and (for heads) v are caller-allocated so activations can land
+  // directly in the backward's blob layout.
+  const int64_t B = X.size(0);
+  const int K = static_cast<int>(X.size(1));
+  const int N = static_cast<int>(Wt.size(1));
+  TORCH_CHECK(X.is_cuda() && X.is_contiguous() && Wt.is_contiguous());
+  TORCH_CHECK(C.is_contiguous());
+  TORCH_CHECK(Wt.size(0) == K && bias.numel() == N);
+  TORCH_CHECK(N <= MAX_NT * M_WAVE, "N exceeds MFMA fwd tile budget");
+  if (heads) {
+    TORCH_CHECK(C.numel() == B * (N - 1) && v.numel() == B);
+  } else {
+    TORCH_CHECK(C.numel() == B * N);
+  }
+
+  FwdArgs a{};
+  a.X = X.data_ptr<float>();
+  a.Wt = Wt.data_ptr<float>();
+  a.bias = bias.data_ptr<float>();
+  a.B = B;
+  a.K = K;
+  a.N = N;
+  a.activation = static_cast<int>(activation);
+  a.heads = static_cast<int>(heads);
+  a.C = C.data_ptr<float>();
+  a.v = heads ? v.data_ptr<float>() : nullptr;
+
+  hipStream_t stream = c10::hip::getCurrentHIPStream().stream();
+  const int64_t tiles = (B + FWD_M - 1) / FWD_M;
+  const int grid = static_cast<int>(std::min<int64_t>(tiles, 4096));
+  hipLaunchKernelGGL(gemm_fwd_kernel, dim3(grid), dim3(FWD_WAVES * 64), 0,
+                     stream, a);
+}
+
+void dw_mfma(torch::Tensor delta, torch::Tensor acts, torch::Tensor grad_buf,
+             int64_t w_off, int64_t b_off) {
+  const int64_t B = delta.size(0);
+  const int out_dim = static_cast<int>(delta.size(1));
+  const int in_dim = static_cast<int>(acts.size(1));
+  TORCH_CHECK(delta.is_contiguous() && acts.is_contiguous());
+
+  DwArgs a{};
+  a.delta = delta.data_ptr<float>();
+  a.acts = acts.data_ptr<float>();
+  a.dW = grad_buf.data_ptr<float>() + w_off;
+  a.db = (b_off >= 0) ? grad_buf.data_ptr<float>() + b_off : nullptr;
+  a.B = B;
+  a.out_dim = out_dim;
+  a.in_dim = in_dim;
+  // tile shape: maximize N per wave to avoid re-reading delta
+  a.nt = std::min(MAX_NT, (in_dim + M_WAVE - 1) / M_WAVE);
+  const int m_tiles = (out_dim + M_WAVE - 1) / M_WAVE;
+  const int n_tiles = (in_dim + a.nt * M_WAVE - 1) / (a.nt * M_WAVE);
+  // enough blocks to fill the chip
+  const int target_blocks = 1024;
+  a.splits = std::max(1, target_blocks / std::max(1, m_tiles * n_tiles));
+  a.splits = static_cast<int>(
+      std::min<int64_t>(a.splits, std::max<int64_t>(1, B / 256)));
+
+  hipStream_t stream = c10::hip::getCurrentHIPStream().stream();
+  hipLaunchKernelGGL(dw_mfma_kernel,
+                     dim3(m_tiles, n_tiles, a.splits), dim3(WAVE), 0, stream,
+                     a);
+}
+
+void dwv(torch::Tensor gv, torch::Tensor acts, torch::Tensor grad_buf,
+         int64_t w_off, int64_t b_off) {
+  const int64_t B = gv.numel();
+  const int in_dim = static_cast<int>(acts.size(1));
+  const int n_blocks = (in_dim + WAVE - 1) / WAVE;
+  const int splits = static_cast<int>(
+      std::max<int64_t>(1, std::min<int64_t>(512, B / 1024)));
+  hipStream_t stream = c10::hip::getCurrentHIPStream().stream();
+  hipLaunchKernelGGL(dwv_kernel, dim3(n_blocks, splits), dim3(WAVE), 0, stream,
+                     gv.data_ptr<float>(), acts.data_ptr<float>(),
+                     grad_buf.data_ptr<float>() + w_off,
+                     (b_off >= 0) ? grad_buf.data_ptr<float>() + b_off : nullptr,
+                     B, in_dim, splits);
+}

@@ -193,8 +193,11 @@ __global__ void mlp_bwd_rows_kernel(BwdArgs a) {
   const int P = 2 * A;
   const int HL = a.dims[a.n_hidden];
 
-  __shared__ float wbuf[WBUF_FLOATS];
-  __shared__ float scratch[WAVES_PER_BLOCK][2 * MAX_A + 2 * MAX_H];
+  // dynamic LDS sized to the ACTUAL staged weights (a fixed 96 KiB
+  // static buffer capped occupancy at 1 block/CU and left every L2 load
+  // latency exposed)
+  extern __shared__ __attribute__((aligned(16))) float dynlds[];
+  float* wbuf = dynlds;
 
   // stage W_2..W_n (hidden-to-hidden), then Wp [P][HL], then Wv [HL]
   int hsz = 0;
@@ -222,10 +225,11 @@ __global__ void mlp_bwd_rows_kernel(BwdArgs a) {
   }
   const float* Wp_l = &wbuf[hsz];
   const float* Wv_l = &wbuf[hsz + P * HL];
-
-  float* gpd_s = &scratch[wave][0];         // [2A]
-  float* dz_s = &scratch[wave][2 * MAX_A];  // [MAX_H]
-  float* red_s = dz_s + MAX_H;              // [MAX_H]
+  const int wtotal = ((hsz + P * HL + HL) + 3) & ~3;
+  float* scratch = &wbuf[wtotal];  // [WAVES_PER_BLOCK][2*MAX_A + 2*MAX_H]
+  float* gpd_s = scratch + wave * (2 * MAX_A + 2 * MAX_H);
+  float* dz_s = gpd_s + 2 * MAX_A;  // [MAX_H]
+  float* red_s = dz_s + MAX_H;      // [MAX_H]
 
   int64_t act_base_last = 0;
   for (int l = 0; l < a.n_hidden - 1; ++l) act_base_last += a.B * a.dims[l + 1];
@@ -277,6 +281,7 @@ __global__ void mlp_bwd_rows_kernel(BwdArgs a) {
     // ---- dz_last = (g_pd @ Wp + g_v * Wv) * act'(a_last) ----
     for (int u = lane; u < HL; u += WAVE) {
       float acc = g.g_v * Wv_l[u];
+      #pragma unroll 4
       for (int j = 0; j < P; ++j) acc += gpd_s[j] * Wp_l[j * HL + u];
       const float h = a.acts[act_base_last + b * HL + u];
       const float dact = a.activation ? (1.f - h * h) : (h > 0.f ? 1.f : 0.f);
@@ -297,6 +302,7 @@ __global__ void mlp_bwd_rows_kernel(BwdArgs a) {
       act_base -= a.B * in_dim;
       for (int u = lane; u < in_dim; u += WAVE) {
         float acc = 0.f;
+        #pragma unroll 4
         for (int k = 0; k < out_dim; ++k) acc += dz_s[k] * Wl[k * in_dim + u];
         const float h = a.acts[act_base + b * in_dim + u];
         const float dact = a.activation ? (1.f - h * h) : (h > 0.f ? 1.f : 0.f);
@@ -481,8 +487,11 @@ std::vector<torch::Tensor> mlp_bwd_rows(
 
   hipStream_t stream = c10::hip::getCurrentHIPStream().stream();
   const int grid = 2048;
+  const int wtotal = ((hsz + P * HL + HL) + 3) & ~3;
+  const size_t lds_bytes =
+      (wtotal + WAVES_PER_BLOCK * (2 * MAX_A + 2 * MAX_H)) * sizeof(float);
   hipLaunchKernelGGL(mlp_bwd_rows_kernel, dim3(grid),
-                     dim3(WAVES_PER_BLOCK * 64), 0, stream, a);
+                     dim3(WAVES_PER_BLOCK * 64), lds_bytes, stream, a);
   return {dz, g_pd, g_v};
 }
 

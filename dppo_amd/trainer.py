@@ -471,9 +471,48 @@ class DPPOEngine:
             self.comm.allreduce_mean_(self.flat_pi.flat_grad)
             self.optimizer.step()
 
+    def _fused_forward(self, states: torch.Tensor):
+        """MFMA forward through the MLP (gemm_fwd per layer + heads).
+
+        Returns (acts_blob, a_views, v, pdflat): activations land in one
+        contiguous blob laid out [B*H1 | B*H2 | ...] for mlp_bwd_rows.
+        Weight transposes are cheap (<100 KB) and re-done per call since
+        params change every update step."""
+        from .ops import hip_ext
+
+        ext = hip_ext()
+        c = self.cfg
+        B = states.shape[0]
+        dims = [self.obs_space.shape[0], *c.HIDDEN_SIZES]
+        n_hidden = len(c.HIDDEN_SIZES)
+        act_code = 1 if c.ACTIVATION == "tanh" else 0
+        with torch.no_grad():
+            Wts = [lay.weight.t().contiguous() for lay in self.pi.hidden]
+            bs = [lay.bias.detach() for lay in self.pi.hidden]
+            Wth = torch.cat(
+                [self.pi.pi.weight.t(), self.pi.vf.weight.t()], dim=1
+            ).contiguous()
+            bh = torch.cat([self.pi.pi.bias, self.pi.vf.bias]).contiguous()
+        total = sum(B * dims[l + 1] for l in range(n_hidden))
+        acts = torch.empty(total, device=states.device, dtype=states.dtype)
+        a_views, o = [], 0
+        x = states
+        for l in range(n_hidden):
+            n = B * dims[l + 1]
+            cview = acts.narrow(0, o, n).view(B, dims[l + 1])
+            ext.gemm_fwd(x, Wts[l], bs[l], act_code, 0, cview, cview)
+            x = cview
+            a_views.append(cview)
+            o += n
+        P = 2 * self.act_space.shape[0]
+        pdflat = torch.empty(B, P, device=states.device, dtype=states.dtype)
+        v = torch.empty(B, device=states.device, dtype=states.dtype)
+        ext.gemm_fwd(x, Wth, bh, 2, 1, pdflat, v)
+        return acts, a_views, v, pdflat
+
     def _update_fused(self, batch: RolloutBatch, l_mul: float) -> None:
-        """Fused HIP update steps: mlp_fwd -> mlp_bwd_rows -> dw_accum x L
-        -> all-reduce -> fused Adam (~7 launches per step, no autograd)."""
+        """Fused MFMA update steps: gemm_fwd per layer -> mlp_bwd_rows ->
+        dw_mfma/dwv into the flat grad -> all-reduce -> fused Adam."""
         from .ops import hip_ext
 
         ext = hip_ext()
@@ -485,12 +524,8 @@ class DPPOEngine:
         clip = c.CLIP_PARAM * l_mul
         B = batch.states.shape[0]
         n_hidden = len(c.HIDDEN_SIZES)
-        g_v_2d = None
         for _ in range(c.UPDATE_STEPS):
-            acts, v, pdflat = ext.mlp_fwd(
-                self.flat_pi.flat_param.detach(), offsets, dims, act_code,
-                batch.states, A,
-            )
+            acts, a_views, v, pdflat = self._fused_forward(batch.states)
             self.flat_pi.zero_grad()
             dz, g_pd, g_v = ext.mlp_bwd_rows(
                 self.flat_pi.flat_param.detach(), offsets, dims, act_code,
@@ -498,27 +533,23 @@ class DPPOEngine:
                 batch.actions, batch.adv, batch.etr,
                 clip, c.ENTCOEFF, c.VCOEFF,
             )
-            # carve per-layer activation/dz views out of the blobs
-            a_views, dz_views, o = [], [], 0
+            dz_views, o = [], 0
             for l in range(n_hidden):
                 n = B * dims[l + 1]
-                a_views.append(acts.narrow(0, o, n).view(B, dims[l + 1]))
                 dz_views.append(dz.narrow(0, o, n).view(B, dims[l + 1]))
                 o += n
             grad = self.flat_pi.flat_grad
-            ext.dw_accum(dz_views[0], batch.states, grad, offsets[0], offsets[1])
+            ext.dw_mfma(dz_views[0], batch.states, grad, offsets[0], offsets[1])
             for l in range(1, n_hidden):
-                ext.dw_accum(dz_views[l], a_views[l - 1], grad,
-                             offsets[2 * l], offsets[2 * l + 1])
+                ext.dw_mfma(dz_views[l], a_views[l - 1], grad,
+                            offsets[2 * l], offsets[2 * l + 1])
             h_last = a_views[-1]
             off_wv, off_bv = offsets[2 * n_hidden], offsets[2 * n_hidden + 1]
             off_wp, off_bp = offsets[2 * n_hidden + 2], offsets[2 * n_hidden + 3]
-            ext.dw_accum(g_pd, h_last, grad, off_wp, off_bp)
-            g_v_2d = g_v.view(B, 1)
-            ext.dw_accum(g_v_2d, h_last, grad, off_wv, off_bv)
+            ext.dw_mfma(g_pd, h_last, grad, off_wp, off_bp)
+            ext.dwv(g_v, h_last, grad, off_wv, off_bv)
             self.comm.allreduce_mean_(grad)
             self.optimizer.step()
-
     # ------------------------------------------------------------------
     def train_round(self) -> Tuple[Dict[str, float], bool]:
         """One full synchronous round. Returns (rank0-view stats, stop)."""

@@ -1,4 +1,8 @@
-"""Fused MLP update-path kernels (mlp_train.hip) vs autograd references."""
+"""Fused MFMA MLP update-path kernels vs autograd references.
+
+The forward test doubles as a transpose detector (cdna guide §5.4 rule
+16): weights and states are random, so a swapped C-write or operand
+layout cannot pass."""
 
 import math
 
@@ -25,43 +29,31 @@ def make_engine(**kw):
     return DPPOEngine(DPPOConfig(**base), comm=Comm(device="cuda:0"))
 
 
-def fused_pieces(eng, B=4096):
-    ext = require_hip_ext()
-    D = eng.obs_space.shape[0]
-    A = eng.act_space.shape[0]
-    states = torch.randn(B, D, device="cuda") * 0.5
-    offsets = [sl.start for sl in eng.flat_pi.slices]
-    dims = [D, *eng.cfg.HIDDEN_SIZES]
-    acts, v, pdflat = ext.mlp_fwd(
-        eng.flat_pi.flat_param.detach(), offsets, dims, 1, states, A
-    )
-    return ext, states, offsets, dims, acts, v, pdflat, A
-
-
-def test_mlp_fwd_matches_eager():
+def test_mfma_fwd_matches_eager():
     eng = make_engine()
-    ext, states, offsets, dims, acts, v, pdflat, A = fused_pieces(eng)
+    B = 4099  # non-multiple of the 128-row tile: exercises tail masking
+    states = torch.randn(B, eng.obs_space.shape[0], device="cuda") * 0.5
+    acts, a_views, v, pdflat = eng._fused_forward(states)
     with torch.no_grad():
         v_ref, flat_ref = eng.pi(states)
     torch.testing.assert_close(v, v_ref, atol=3e-5, rtol=3e-5)
     torch.testing.assert_close(pdflat, flat_ref, atol=3e-5, rtol=3e-5)
-    # saved activations match the eager hidden outputs
-    B = states.shape[0]
     h = states
-    o = 0
-    for layer in eng.pi.hidden:
+    for layer, av in zip(eng.pi.hidden, a_views):
         h = torch.tanh(layer(h))
-        n = B * h.shape[1]
-        torch.testing.assert_close(
-            acts.narrow(0, o, n).view_as(h), h, atol=3e-5, rtol=3e-5
-        )
-        o += n
+        torch.testing.assert_close(av, h, atol=3e-5, rtol=3e-5)
 
 
 def test_fused_backward_matches_autograd():
     eng = make_engine()
-    ext, states, offsets, dims, acts, v, pdflat, A = fused_pieces(eng)
-    B = states.shape[0]
+    B = 4096
+    A = eng.act_space.shape[0]
+    states = torch.randn(B, eng.obs_space.shape[0], device="cuda") * 0.5
+    ext = require_hip_ext()
+    offsets = [sl.start for sl in eng.flat_pi.slices]
+    dims = [eng.obs_space.shape[0], *eng.cfg.HIDDEN_SIZES]
+    acts, a_views, v, pdflat = eng._fused_forward(states)
+
     pdt = DiagGaussianPdType(A)
     with torch.no_grad():
         oldflat = pdflat + 0.05 * torch.randn_like(pdflat)
@@ -71,7 +63,7 @@ def test_fused_backward_matches_autograd():
         etr = torch.randn(B, device="cuda")
     clip, entc, vc = 0.2, 0.01, 0.5
 
-    # autograd reference gradient into flat_grad
+    # autograd reference
     eng.flat_pi.zero_grad()
     v2, flat2 = eng.pi(states)
     out = ppo_losses_ref(pdt.pdfromflat(flat2), pdt.pdfromflat(oldflat),
@@ -88,23 +80,22 @@ def test_fused_backward_matches_autograd():
     )
     grad = eng.flat_pi.flat_grad
     n_hidden = len(eng.cfg.HIDDEN_SIZES)
-    a_views, dz_views, o = [], [], 0
+    dz_views, o = [], 0
     for l in range(n_hidden):
         n = B * dims[l + 1]
-        a_views.append(acts.narrow(0, o, n).view(B, dims[l + 1]))
         dz_views.append(dz.narrow(0, o, n).view(B, dims[l + 1]))
         o += n
-    ext.dw_accum(dz_views[0], states, grad, offsets[0], offsets[1])
+    ext.dw_mfma(dz_views[0], states, grad, offsets[0], offsets[1])
     for l in range(1, n_hidden):
-        ext.dw_accum(dz_views[l], a_views[l - 1], grad,
-                     offsets[2 * l], offsets[2 * l + 1])
-    ext.dw_accum(g_pd, a_views[-1], grad,
-                 offsets[2 * n_hidden + 2], offsets[2 * n_hidden + 3])
-    ext.dw_accum(g_v.view(B, 1), a_views[-1], grad,
-                 offsets[2 * n_hidden], offsets[2 * n_hidden + 1])
+        ext.dw_mfma(dz_views[l], a_views[l - 1], grad,
+                    offsets[2 * l], offsets[2 * l + 1])
+    ext.dw_mfma(g_pd, a_views[-1], grad,
+                offsets[2 * n_hidden + 2], offsets[2 * n_hidden + 3])
+    ext.dwv(g_v, a_views[-1], grad,
+            offsets[2 * n_hidden], offsets[2 * n_hidden + 1])
 
-    scale = ref_grad.abs().max()
-    torch.testing.assert_close(grad, ref_grad, atol=float(scale) * 2e-4 + 1e-8,
+    scale = float(ref_grad.abs().max())
+    torch.testing.assert_close(grad, ref_grad, atol=scale * 2e-4 + 1e-8,
                                rtol=2e-3)
 
 
@@ -127,11 +118,10 @@ def test_fused_update_matches_autograd_update():
     eng1.sync_oldpi()
     eng2.sync_oldpi()
     batch = eng1.collect()
-    eng2_batch = batch  # same tensors; updates don't mutate the batch
     eng1._can_fuse_update = lambda: True
     eng2._can_fuse_update = lambda: False
     eng1.update(batch, 0.9)
-    eng2.update(eng2_batch, 0.9)
+    eng2.update(batch, 0.9)
     torch.testing.assert_close(
         eng1.flat_pi.flat_param, eng2.flat_pi.flat_param, atol=2e-5, rtol=1e-3
     )
@@ -142,3 +132,14 @@ def test_single_hidden_layer_path():
     assert eng._can_fuse_update()
     stats, _ = eng.train_round()
     assert math.isfinite(stats["total_loss"])
+
+
+def test_relu_and_128_wide_path():
+    eng = make_engine(HIDDEN_SIZES=(128, 128), ACTIVATION="relu", NUM_ENVS=64)
+    B = 512
+    states = torch.randn(B, eng.obs_space.shape[0], device="cuda") * 0.5
+    acts, a_views, v, pdflat = eng._fused_forward(states)
+    with torch.no_grad():
+        v_ref, flat_ref = eng.pi(states)
+    torch.testing.assert_close(v, v_ref, atol=3e-5, rtol=3e-5)
+    torch.testing.assert_close(pdflat, flat_ref, atol=3e-5, rtol=3e-5)
