@@ -58,28 +58,31 @@ struct FwdArgs {
   int heads;          // if 1: last column -> v, rest -> C (pdflat)
 };
 
+template <int NT>
 __launch_bounds__(FWD_WAVES * 64)
 __global__ void gemm_fwd_kernel(FwdArgs a) {
   const int lane = threadIdx.x & (WAVE - 1);
   const int wave = threadIdx.x / WAVE;
-  const int NT = (a.N + M_WAVE - 1) / M_WAVE;  // <= MAX_NT
 
-  // X tile [FWD_M][BK] with +1 padding (A-fragment read is a column read)
+  // X tile [FWD_M][BK] (+1 pad: the A-fragment read is a column read);
+  // Wt tile [BK][NT*32] staged coalesced so the MFMA B-operand comes
+  // from LDS instead of a fresh L2 round trip per k-step.
   __shared__ float xs[FWD_M][BK + 1];
+  __shared__ float ws[BK][MAX_NT * M_WAVE];
 
   const int i_l = lane & 31;   // A row within wave tile
   const int k_l = lane >> 5;   // A k within pair
+  const int NW = NT * M_WAVE;
 
   for (int64_t tile = blockIdx.x; tile * FWD_M < a.B; tile += gridDim.x) {
     const int64_t b0 = tile * FWD_M;
-    f32x16 acc[MAX_NT];
+    f32x16 acc[NT];
     #pragma unroll
-    for (int t = 0; t < MAX_NT; ++t)
+    for (int t = 0; t < NT; ++t)
       #pragma unroll
       for (int r = 0; r < 16; ++r) acc[t][r] = 0.f;
 
     for (int kb = 0; kb < a.K; kb += BK) {
-      // ---- stage X[b0:b0+128][kb:kb+BK] (coalesced; zero-pad tails) ----
       __syncthreads();
       for (int idx = threadIdx.x; idx < FWD_M * BK; idx += FWD_WAVES * 64) {
         const int r = idx / BK, c = idx % BK;
@@ -87,45 +90,43 @@ __global__ void gemm_fwd_kernel(FwdArgs a) {
         const int col = kb + c;
         xs[r][c] = (row < a.B && col < a.K) ? a.X[row * a.K + col] : 0.f;
       }
+      for (int idx = threadIdx.x; idx < BK * NW; idx += FWD_WAVES * 64) {
+        const int r = idx / NW, c = idx % NW;
+        const int krow = kb + r;
+        ws[r][c] = (krow < a.K && c < a.N) ? a.Wt[(int64_t)krow * a.N + c] : 0.f;
+      }
       __syncthreads();
 
       const int ksteps = min(BK, a.K - kb);
+      #pragma unroll 4
       for (int k2 = 0; k2 < ksteps; k2 += 2) {
         const float av = xs[wave * M_WAVE + i_l][k2 + k_l];
         #pragma unroll
-        for (int t = 0; t < MAX_NT; ++t) {
-          if (t < NT) {
-            const int col = t * M_WAVE + i_l;
-            const int krow = kb + k2 + k_l;
-            const float bv = (col < a.N && krow < a.K)
-                                 ? a.Wt[(int64_t)krow * a.N + col]
-                                 : 0.f;
-            acc[t] = __builtin_amdgcn_mfma_f32_32x32x2f32(av, bv, acc[t], 0, 0, 0);
-          }
+        for (int t = 0; t < NT; ++t) {
+          const float bv = ws[k2 + k_l][t * M_WAVE + i_l];
+          acc[t] = __builtin_amdgcn_mfma_f32_32x32x2f32(av, bv, acc[t], 0, 0, 0);
         }
       }
     }
 
     // ---- epilogue: bias + activation + store ----
     #pragma unroll
-    for (int t = 0; t < MAX_NT; ++t) {
-      if (t < NT) {
-        const int col = t * M_WAVE + i_l;
-        if (col < a.N) {
-          const float bv = a.bias[col];
-          #pragma unroll
-          for (int r = 0; r < 16; ++r) {
-            const int64_t row = b0 + wave * M_WAVE + cd_row(r, lane);
-            if (row < a.B) {
-              float x = acc[t][r] + bv;
-              if (a.activation == 0) x = fmaxf(x, 0.f);
-              else if (a.activation == 1) x = tanhf(x);
-              if (a.heads) {
-                if (col == a.N - 1) a.v[row] = x;
-                else a.C[row * (a.N - 1) + col] = x;
-              } else {
-                a.C[row * a.N + col] = x;
-              }
+    for (int t = 0; t < NT; ++t) {
+      const int col = t * M_WAVE + i_l;
+      if (col < a.N) {
+        const float bv = a.bias[col];
+        #pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          const int64_t row = b0 + wave * M_WAVE + cd_row(r, lane);
+          if (row < a.B) {
+            float x = acc[t][r] + bv;
+            if (a.activation == 0) x = fmaxf(x, 0.f);
+            else if (a.activation == 1) x = tanhf(x);
+            if (a.heads) {
+              if (col == a.N - 1) a.v[row] = x;
+              else a.C[row * (a.N - 1) + col] = x;
+            } else {
+              a.C[row * a.N + col] = x;
             }
           }
         }
@@ -150,6 +151,7 @@ struct DwArgs {
   int splits;  // K splits
 };
 
+template <int NT>
 __launch_bounds__(64)
 __global__ void dw_mfma_kernel(DwArgs a) {
   const int lane = threadIdx.x;
@@ -166,56 +168,63 @@ __global__ void dw_mfma_kernel(DwArgs a) {
   const int mcol = m0 + i_l;          // delta column this lane reads
   const bool m_ok = mcol < a.out_dim;
 
-  f32x16 acc[MAX_NT];
+  f32x16 acc[NT];
   #pragma unroll
-  for (int t = 0; t < MAX_NT; ++t)
+  for (int t = 0; t < NT; ++t)
     #pragma unroll
     for (int r = 0; r < 16; ++r) acc[t][r] = 0.f;
   float dbacc = 0.f;
 
-  for (int64_t k = k0; k + 1 < k1; k += 2) {
-    const float av = m_ok ? a.delta[(k + k_l) * a.out_dim + mcol] : 0.f;
-    dbacc += av;
-    #pragma unroll
-    for (int t = 0; t < MAX_NT; ++t) {
-      if (t < a.nt) {
-        const int col = n_base + t * M_WAVE + i_l;
-        const float bv = (col < a.in_dim)
-                             ? a.acts[(k + k_l) * a.in_dim + col]
-                             : 0.f;
-        acc[t] = __builtin_amdgcn_mfma_f32_32x32x2f32(av, bv, acc[t], 0, 0, 0);
-      }
-    }
+  // column guards hoisted out of the K loop (tail tiles load col 0 and
+  // discard at scatter time — a harmless re-read keeps the loop branchless)
+  int bcol[NT];
+  bool col_ok[NT];
+  #pragma unroll
+  for (int t = 0; t < NT; ++t) {
+    const int col = n_base + t * M_WAVE + i_l;
+    col_ok[t] = col < a.in_dim;
+    bcol[t] = col_ok[t] ? col : 0;
   }
-  // odd tail row
-  if (((k1 - k0) & 1) && k1 > k0) {
-    const int64_t k = k1 - 1;
+
+  #pragma unroll 1
+  for (int64_t k = k0; k + 3 < k1; k += 4) {
+    // two k-pairs per iteration: loads of the second pair issue while the
+    // first pair's MFMAs run
+    const float av0 = m_ok ? a.delta[(k + k_l) * a.out_dim + mcol] : 0.f;
+    const float av1 = m_ok ? a.delta[(k + 2 + k_l) * a.out_dim + mcol] : 0.f;
+    float bv0[NT], bv1[NT];
+    #pragma unroll
+    for (int t = 0; t < NT; ++t) {
+      bv0[t] = a.acts[(k + k_l) * a.in_dim + bcol[t]];
+      bv1[t] = a.acts[(k + 2 + k_l) * a.in_dim + bcol[t]];
+    }
+    dbacc += av0 + av1;
+    #pragma unroll
+    for (int t = 0; t < NT; ++t)
+      acc[t] = __builtin_amdgcn_mfma_f32_32x32x2f32(av0, bv0[t], acc[t], 0, 0, 0);
+    #pragma unroll
+    for (int t = 0; t < NT; ++t)
+      acc[t] = __builtin_amdgcn_mfma_f32_32x32x2f32(av1, bv1[t], acc[t], 0, 0, 0);
+  }
+  for (int64_t k = k0 + ((k1 - k0) & ~3); k < k1; ++k) {
     const float av = (m_ok && k_l == 0) ? a.delta[k * a.out_dim + mcol] : 0.f;
     if (k_l == 0) dbacc += av;
     #pragma unroll
-    for (int t = 0; t < MAX_NT; ++t) {
-      if (t < a.nt) {
-        const int col = n_base + t * M_WAVE + i_l;
-        const float bv = (col < a.in_dim && k_l == 0)
-                             ? a.acts[k * a.in_dim + col]
-                             : 0.f;
-        acc[t] = __builtin_amdgcn_mfma_f32_32x32x2f32(av, bv, acc[t], 0, 0, 0);
-      }
+    for (int t = 0; t < NT; ++t) {
+      const float bv = (k_l == 0) ? a.acts[k * a.in_dim + bcol[t]] : 0.f;
+      acc[t] = __builtin_amdgcn_mfma_f32_32x32x2f32(av, bv, acc[t], 0, 0, 0);
     }
   }
 
   // ---- scatter accumulators ----
   #pragma unroll
-  for (int t = 0; t < MAX_NT; ++t) {
-    if (t < a.nt) {
-      const int col = n_base + t * M_WAVE + i_l;
-      if (col < a.in_dim) {
-        #pragma unroll
-        for (int r = 0; r < 16; ++r) {
-          const int mrow = m0 + cd_row(r, lane);
-          if (mrow < a.out_dim && acc[t][r] != 0.f) {
-            atomicAdd(&a.dW[(int64_t)mrow * a.in_dim + col], acc[t][r]);
-          }
+  for (int t = 0; t < NT; ++t) {
+    if (col_ok[t]) {
+      #pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const int mrow = m0 + cd_row(r, lane);
+        if (mrow < a.out_dim && acc[t][r] != 0.f) {
+          atomicAdd(&a.dW[(int64_t)mrow * a.in_dim + bcol[t]], acc[t][r]);
         }
       }
     }
@@ -245,13 +254,24 @@ __global__ void dwv_kernel(const float* __restrict__ gv,    // [B]
   const int64_t k0 = split * rows_per;
   const int64_t k1 = min(B, k0 + rows_per);
   const int col = n0 + lane;
-  float acc = 0.f, bacc = 0.f;
-  for (int64_t k = k0; k < k1; ++k) {
+  const int ccol = (col < in_dim) ? col : 0;
+  float acc[4] = {0.f, 0.f, 0.f, 0.f};
+  float bacc = 0.f;
+  int64_t k = k0;
+  for (; k + 3 < k1; k += 4) {
+    #pragma unroll
+    for (int q = 0; q < 4; ++q) {
+      const float g = gv[k + q];
+      bacc += g;
+      acc[q] += g * acts[(k + q) * in_dim + ccol];
+    }
+  }
+  for (; k < k1; ++k) {
     const float g = gv[k];
     bacc += g;
-    if (col < in_dim) acc += g * acts[k * in_dim + col];
+    acc[0] += g * acts[k * in_dim + ccol];
   }
-  if (col < in_dim) atomicAdd(&dWv[col], acc);
+  if (col < in_dim) atomicAdd(&dWv[col], acc[0] + acc[1] + acc[2] + acc[3]);
   if (dbv != nullptr && blockIdx.x == 0 && lane == 0) atomicAdd(dbv, bacc);
 }
 
@@ -294,8 +314,25 @@ void gemm_fwd(torch::Tensor X, torch::Tensor Wt, torch::Tensor bias,
   hipStream_t stream = c10::hip::getCurrentHIPStream().stream();
   const int64_t tiles = (B + FWD_M - 1) / FWD_M;
   const int grid = static_cast<int>(std::min<int64_t>(tiles, 4096));
-  hipLaunchKernelGGL(gemm_fwd_kernel, dim3(grid), dim3(FWD_WAVES * 64), 0,
-                     stream, a);
+  const int NT = (N + M_WAVE - 1) / M_WAVE;
+  switch (NT) {
+    case 1:
+      hipLaunchKernelGGL(gemm_fwd_kernel<1>, dim3(grid), dim3(FWD_WAVES * 64),
+                         0, stream, a);
+      break;
+    case 2:
+      hipLaunchKernelGGL(gemm_fwd_kernel<2>, dim3(grid), dim3(FWD_WAVES * 64),
+                         0, stream, a);
+      break;
+    case 3:
+      hipLaunchKernelGGL(gemm_fwd_kernel<3>, dim3(grid), dim3(FWD_WAVES * 64),
+                         0, stream, a);
+      break;
+    default:
+      hipLaunchKernelGGL(gemm_fwd_kernel<4>, dim3(grid), dim3(FWD_WAVES * 64),
+                         0, stream, a);
+      break;
+  }
 }
 
 void dw_mfma(torch::Tensor delta, torch::Tensor acts, torch::Tensor grad_buf,
@@ -317,16 +354,28 @@ void dw_mfma(torch::Tensor delta, torch::Tensor acts, torch::Tensor grad_buf,
   a.nt = std::min(MAX_NT, (in_dim + M_WAVE - 1) / M_WAVE);
   const int m_tiles = (out_dim + M_WAVE - 1) / M_WAVE;
   const int n_tiles = (in_dim + a.nt * M_WAVE - 1) / (a.nt * M_WAVE);
-  // enough blocks to fill the chip
-  const int target_blocks = 1024;
+  // enough waves to hide the streamed-operand latency (~2k waves)
+  const int target_blocks = 2048;
   a.splits = std::max(1, target_blocks / std::max(1, m_tiles * n_tiles));
   a.splits = static_cast<int>(
       std::min<int64_t>(a.splits, std::max<int64_t>(1, B / 256)));
 
   hipStream_t stream = c10::hip::getCurrentHIPStream().stream();
-  hipLaunchKernelGGL(dw_mfma_kernel,
-                     dim3(m_tiles, n_tiles, a.splits), dim3(WAVE), 0, stream,
-                     a);
+  const dim3 grid(m_tiles, n_tiles, a.splits);
+  switch (a.nt) {
+    case 1:
+      hipLaunchKernelGGL(dw_mfma_kernel<1>, grid, dim3(WAVE), 0, stream, a);
+      break;
+    case 2:
+      hipLaunchKernelGGL(dw_mfma_kernel<2>, grid, dim3(WAVE), 0, stream, a);
+      break;
+    case 3:
+      hipLaunchKernelGGL(dw_mfma_kernel<3>, grid, dim3(WAVE), 0, stream, a);
+      break;
+    default:
+      hipLaunchKernelGGL(dw_mfma_kernel<4>, grid, dim3(WAVE), 0, stream, a);
+      break;
+  }
 }
 
 void dwv(torch::Tensor gv, torch::Tensor acts, torch::Tensor grad_buf,
@@ -334,8 +383,9 @@ void dwv(torch::Tensor gv, torch::Tensor acts, torch::Tensor grad_buf,
   const int64_t B = gv.numel();
   const int in_dim = static_cast<int>(acts.size(1));
   const int n_blocks = (in_dim + WAVE - 1) / WAVE;
-  const int splits = static_cast<int>(
-      std::max<int64_t>(1, std::min<int64_t>(512, B / 1024)));
+  // scale splits so the launch has ~2k waves regardless of in_dim
+  const int splits = static_cast<int>(std::max<int64_t>(
+      1, std::min<int64_t>(2048 / std::max(1, n_blocks), B / 256)));
   hipStream_t stream = c10::hip::getCurrentHIPStream().stream();
   hipLaunchKernelGGL(dwv_kernel, dim3(n_blocks, splits), dim3(WAVE), 0, stream,
                      gv.data_ptr<float>(), acts.data_ptr<float>(),

@@ -11,9 +11,13 @@
 // T times, with env state resident in LDS, so a whole rollout costs ONE
 // launch and all policy/env weights stream from L1/L2.
 //
-// Decomposition: one wave (64 lanes) per block owns ENV_TILE=4 envs for
-// the full T-step loop (E/4 blocks ≈ 1024 at the bench config → 4 blocks
-// per CU, 1 wave/SIMD).  No inter-block communication exists (each
+// Decomposition: one block of FOUR waves (256 threads) owns ENV_TILE=4
+// envs for the full T-step loop (E/4 blocks ≈ 1024 at the bench config).
+// The four waves K-SPLIT each layer's dot products (wave w accumulates
+// the k-quarter of every unit, partials combined through LDS), so the
+// chip runs 4096 waves (≈4 waves/SIMD) at UNCHANGED weight traffic —
+// the single-wave variant measured latency-bound at 1 wave/SIMD, every
+// W-row L2 access exposed.  No inter-block communication exists (each
 // block's envs are private, output rows disjoint), so the in-kernel
 // T-loop needs no grid sync by construction.
 //   - Layer phase: lane u computes output unit u for all 4 envs
@@ -56,8 +60,10 @@ constexpr int H1_OFF = H0_OFF + ENV_TILE * MAX_H;       // [ENV_TILE][MAX_H]
 constexpr int PD_OFF = H1_OFF + ENV_TILE * MAX_H;       // [ENV_TILE][2*MAX_A]
 constexpr int ACT_OFF = PD_OFF + ENV_TILE * 2 * MAX_A;  // [ENV_TILE][MAX_A]
 constexpr int XV_OFF = ACT_OFF + ENV_TILE * MAX_A;      // [ENV_TILE][MAX_R]
-constexpr int MISC_OFF = XV_OFF + ENV_TILE * MAX_R;     // val, rsum, epr
-constexpr int LDS_FLOATS = MISC_OFF + 3 * ENV_TILE;
+constexpr int NWAVES = 4;                               // waves per block
+constexpr int PART_OFF = XV_OFF + ENV_TILE * MAX_R;     // [NWAVES][ENV_TILE][MAX_H]
+constexpr int MISC_OFF = PART_OFF + NWAVES * ENV_TILE * MAX_H;  // val, rsum, epr
+constexpr int LDS_FLOATS = MISC_OFF + 3 * ENV_TILE + ENV_TILE;  // + racc
 
 struct RolloutArgs {
   const float* params;   // flat parameter buffer (pi)
@@ -118,9 +124,11 @@ DEV_INLINE float u2f_mono(unsigned u) {
 
 // ---- the rollout kernel ---------------------------------------------------
 
-__launch_bounds__(64)
+__launch_bounds__(NWAVES * 64)
 __global__ void rollout_kernel(RolloutArgs a) {
-  const int lane = threadIdx.x;  // block = one wave of 64
+  const int tid = threadIdx.x;      // block = 4 waves of 64
+  const int lane = tid & (WAVE - 1);
+  const int wv = tid / WAVE;        // k-split wave index
   const int e0 = blockIdx.x * ENV_TILE;
   const int nE = min(ENV_TILE, a.E - e0);
   const int D = a.D;
@@ -132,6 +140,7 @@ __global__ void rollout_kernel(RolloutArgs a) {
   float* val_lds = &lds[MISC_OFF];
   float* rsum_lds = &lds[MISC_OFF + ENV_TILE];
   float* epr_lds = &lds[MISC_OFF + 2 * ENV_TILE];
+  float* racc_lds = &lds[MISC_OFF + 3 * ENV_TILE];
   __shared__ int tc_lds[ENV_TILE];
   __shared__ int done_lds[ENV_TILE];
 
@@ -151,75 +160,34 @@ __global__ void rollout_kernel(RolloutArgs a) {
 
   // ---- load persistent state; zero tail envs ----
   for (int e = 0; e < ENV_TILE; ++e) {
-    for (int d = lane; d < D; d += WAVE)
+    for (int d = tid; d < D; d += NWAVES * WAVE)
       lds[X_OFF + e * MAX_D + d] =
           (e < nE) ? a.x[(int64_t)(e0 + e) * D + d] : 0.f;
   }
-  if (lane < nE) {
-    epr_lds[lane] = a.epr[e0 + lane];
-    tc_lds[lane] = a.t[e0 + lane];
+  if (tid < nE) {
+    epr_lds[tid] = a.epr[e0 + tid];
+    tc_lds[tid] = a.t[e0 + tid];
   }
   __syncthreads();
 
-  for (int step = 0; step < T; ++step) {
-    // ---- write current obs ----
-    for (int e = 0; e < nE; ++e) {
-      const int64_t base = ((int64_t)step * E + e0 + e) * D;
-      for (int d = lane; d < D; d += WAVE)
-        out_states[base + d] = lds[X_OFF + e * MAX_D + d];
-    }
-
-    // ---- policy MLP forward ----
-    int in_off = X_OFF, in_stride = MAX_D, in_dim = D;
-    for (int l = 0; l < a.n_hidden; ++l) {
-      const int out_dim = a.dims[l + 1];
-      const int out_off = (l & 1) ? H1_OFF : H0_OFF;
-      const float* W = a.params + a.off_W[l];
-      const float* bias = a.params + a.off_b[l];
-      for (int u = lane; u < out_dim; u += WAVE) {
-        float acc[ENV_TILE];
-        const float bu = bias[u];
-        #pragma unroll
-        for (int e = 0; e < ENV_TILE; ++e) acc[e] = bu;
-        const float* Wrow = W + (int64_t)u * in_dim;
-        #pragma unroll 2
-        for (int k = 0; k + 4 <= in_dim; k += 4) {
-          const float4 w4 = *reinterpret_cast<const float4*>(Wrow + k);
-          #pragma unroll
-          for (int e = 0; e < ENV_TILE; ++e) {
-            const float4 i4 = *reinterpret_cast<const float4*>(
-                &lds[in_off + e * in_stride + k]);
-            acc[e] += w4.x * i4.x + w4.y * i4.y + w4.z * i4.z + w4.w * i4.w;
-          }
-        }
-        for (int k = in_dim & ~3; k < in_dim; ++k) {
-          const float w = Wrow[k];
-          #pragma unroll
-          for (int e = 0; e < ENV_TILE; ++e)
-            acc[e] += w * lds[in_off + e * in_stride + k];
-        }
-        #pragma unroll
-        for (int e = 0; e < ENV_TILE; ++e) {
-          const float v = acc[e];
-          lds[out_off + e * MAX_H + u] = a.activation ? tanhf(v) : fmaxf(v, 0.f);
-        }
-      }
-      __syncthreads();
-      in_off = out_off;
-      in_stride = MAX_H;
-      in_dim = out_dim;
-    }
-
-    // ---- heads: pd params (u < P) and value (u == P) ----
-    for (int u = lane; u < P + 1; u += WAVE) {
-      const bool is_v = (u == P);
-      const float* Wrow =
-          a.params + (is_v ? (int64_t)a.off_Wv : a.off_Wp + (int64_t)u * in_dim);
+  // K-split layer forward: wave wv accumulates its k-quarter of every
+  // unit into the partial slab; a combine pass sums the 4 partials,
+  // adds bias and applies the activation.
+  auto layer_kpart = [&](const float* W, int off_Wv_u, int in_off,
+                         int in_stride, int in_dim, int out_dim,
+                         bool heads) {
+    const int kq = (((in_dim + NWAVES * 4 - 1) / (NWAVES * 4)) * 4);
+    const int k0 = wv * kq;
+    const int k1 = min(in_dim, k0 + kq);
+    for (int u = lane; u < out_dim; u += WAVE) {
       float acc[ENV_TILE];
-      const float bu = a.params[is_v ? a.off_bv : a.off_bp + u];
       #pragma unroll
-      for (int e = 0; e < ENV_TILE; ++e) acc[e] = bu;
-      for (int k = 0; k + 4 <= in_dim; k += 4) {
+      for (int e = 0; e < ENV_TILE; ++e) acc[e] = 0.f;
+      const bool is_v = heads && (u == out_dim - 1);
+      const float* Wrow =
+          is_v ? (a.params + off_Wv_u) : (W + (int64_t)u * in_dim);
+      #pragma unroll 2
+      for (int k = k0; k + 4 <= k1; k += 4) {
         const float4 w4 = *reinterpret_cast<const float4*>(Wrow + k);
         #pragma unroll
         for (int e = 0; e < ENV_TILE; ++e) {
@@ -228,38 +196,81 @@ __global__ void rollout_kernel(RolloutArgs a) {
           acc[e] += w4.x * i4.x + w4.y * i4.y + w4.z * i4.z + w4.w * i4.w;
         }
       }
-      for (int k = in_dim & ~3; k < in_dim; ++k) {
+      for (int k = max(k0, k1 & ~3); k < k1; ++k) {
         const float w = Wrow[k];
         #pragma unroll
         for (int e = 0; e < ENV_TILE; ++e)
           acc[e] += w * lds[in_off + e * in_stride + k];
       }
       #pragma unroll
-      for (int e = 0; e < ENV_TILE; ++e) {
-        if (is_v) val_lds[e] = acc[e];
-        else lds[PD_OFF + e * 2 * MAX_A + u] = acc[e];
+      for (int e = 0; e < ENV_TILE; ++e)
+        lds[PART_OFF + (wv * ENV_TILE + e) * MAX_H + u] = acc[e];
+    }
+  };
+
+  for (int step = 0; step < T; ++step) {
+    // ---- write current obs ----
+    for (int e = 0; e < nE; ++e) {
+      const int64_t base = ((int64_t)step * E + e0 + e) * D;
+      for (int d = tid; d < D; d += NWAVES * WAVE)
+        out_states[base + d] = lds[X_OFF + e * MAX_D + d];
+    }
+
+    // ---- policy MLP forward (K-split + combine per layer) ----
+    int in_off = X_OFF, in_stride = MAX_D, in_dim = D;
+    for (int l = 0; l < a.n_hidden; ++l) {
+      const int out_dim = a.dims[l + 1];
+      const int out_off = (l & 1) ? H1_OFF : H0_OFF;
+      layer_kpart(a.params + a.off_W[l], 0, in_off, in_stride, in_dim,
+                  out_dim, false);
+      __syncthreads();
+      const float* bias = a.params + a.off_b[l];
+      for (int idx = tid; idx < ENV_TILE * out_dim; idx += NWAVES * WAVE) {
+        const int e = idx / out_dim, u = idx % out_dim;
+        float sum = bias[u];
+        #pragma unroll
+        for (int w = 0; w < NWAVES; ++w)
+          sum += lds[PART_OFF + (w * ENV_TILE + e) * MAX_H + u];
+        lds[out_off + e * MAX_H + u] =
+            a.activation ? tanhf(sum) : fmaxf(sum, 0.f);
       }
+      __syncthreads();
+      in_off = out_off;
+      in_stride = MAX_H;
+      in_dim = out_dim;
+    }
+
+    // ---- heads (u < P: pd params; u == P: value), no activation ----
+    layer_kpart(a.params + a.off_Wp, a.off_Wv, in_off, in_stride, in_dim,
+                P + 1, true);
+    __syncthreads();
+    for (int idx = tid; idx < ENV_TILE * (P + 1); idx += NWAVES * WAVE) {
+      const int e = idx / (P + 1), u = idx % (P + 1);
+      float sum = a.params[(u == P) ? a.off_bv : a.off_bp + u];
+      #pragma unroll
+      for (int w = 0; w < NWAVES; ++w)
+        sum += lds[PART_OFF + (w * ENV_TILE + e) * MAX_H + u];
+      if (u == P) val_lds[e] = sum;
+      else lds[PD_OFF + e * 2 * MAX_A + u] = sum;
     }
     __syncthreads();
 
-    // ---- sample actions (lanes = (env, dim) pairs: 2 envs x 32 dims) ----
+    // ---- sample actions (threads = (env, dim) pairs) ----
     {
-      const int e = lane / MAX_A;
-      const int j = lane % MAX_A;
-      for (int ee = e; ee < nE; ee += WAVE / MAX_A) {
-        if (j < A) {
-          const int ge = e0 + ee;
-          const float mean = lds[PD_OFF + ee * 2 * MAX_A + j];
-          const float logstd = lds[PD_OFF + ee * 2 * MAX_A + A + j];
-          float act = mean + __expf(logstd) * rng_normal(a.seed, ge, step, j);
-          // epsilon-greedy overlay (Worker.py:149-152)
-          const float u_dec = rng_uniform(a.seed, ge, step, 90001);
-          if (u_dec < a.eps_explore) {
-            const float u = rng_uniform(a.seed, ge, step, 90010 + j);
-            act = a.act_low + (a.act_high - a.act_low) * u;
-          }
-          lds[ACT_OFF + ee * MAX_A + j] = act;
+      const int e = tid / MAX_A;   // 8 groups x 32 dims
+      const int j = tid % MAX_A;
+      if (e < nE && j < A) {
+        const int ge = e0 + e;
+        const float mean = lds[PD_OFF + e * 2 * MAX_A + j];
+        const float logstd = lds[PD_OFF + e * 2 * MAX_A + A + j];
+        float act = mean + __expf(logstd) * rng_normal(a.seed, ge, step, j);
+        // epsilon-greedy overlay (Worker.py:149-152)
+        const float u_dec = rng_uniform(a.seed, ge, step, 90001);
+        if (u_dec < a.eps_explore) {
+          const float u = rng_uniform(a.seed, ge, step, 90010 + j);
+          act = a.act_low + (a.act_high - a.act_low) * u;
         }
+        lds[ACT_OFF + e * MAX_A + j] = act;
       }
     }
     __syncthreads();
@@ -267,35 +278,34 @@ __global__ void rollout_kernel(RolloutArgs a) {
     // ---- write pdflat / action / value rows ----
     for (int e = 0; e < nE; ++e) {
       const int64_t row = (int64_t)step * E + e0 + e;
-      if (lane < P) out_pdflats[row * P + lane] = lds[PD_OFF + e * 2 * MAX_A + lane];
-      if (lane < A) out_actions[row * A + lane] = lds[ACT_OFF + e * MAX_A + lane];
+      if (tid < P) out_pdflats[row * P + tid] = lds[PD_OFF + e * 2 * MAX_A + tid];
+      if (tid < A) out_actions[row * A + tid] = lds[ACT_OFF + e * MAX_A + tid];
     }
-    if (lane < nE) out_values[(int64_t)step * E + e0 + lane] = val_lds[lane];
+    if (tid < nE) out_values[(int64_t)step * E + e0 + tid] = val_lds[tid];
+    if (tid < ENV_TILE) racc_lds[tid] = 0.f;
 
-    // ---- env low-rank projection: xv = x @ V ----
-    {
-      const int per = WAVE / ENV_TILE;  // 16 lanes per env
-      const int e = lane / per;
-      for (int rr = lane % per; rr < a.rank; rr += per) {
-        const float* Vrow = env_Vt + (int64_t)rr * D;
-        float accv = 0.f;
-        for (int k = 0; k + 4 <= D; k += 4) {
-          const float4 v4 = *reinterpret_cast<const float4*>(Vrow + k);
-          const float4 x4 =
-              *reinterpret_cast<const float4*>(&lds[X_OFF + e * MAX_D + k]);
-          accv += v4.x * x4.x + v4.y * x4.y + v4.z * x4.z + v4.w * x4.w;
-        }
-        for (int k = D & ~3; k < D; ++k) accv += Vrow[k] * lds[X_OFF + e * MAX_D + k];
-        if (e < nE) lds[XV_OFF + e * MAX_R + rr] = accv;
+    // ---- env low-rank projection: wave wv handles env wv ----
+    if (wv < nE && lane < a.rank) {
+      const int e = wv;
+      const int rr = lane;
+      const float* Vrow = env_Vt + (int64_t)rr * D;
+      float accv = 0.f;
+      for (int k = 0; k + 4 <= D; k += 4) {
+        const float4 v4 = *reinterpret_cast<const float4*>(Vrow + k);
+        const float4 x4 =
+            *reinterpret_cast<const float4*>(&lds[X_OFF + e * MAX_D + k]);
+        accv += v4.x * x4.x + v4.y * x4.y + v4.z * x4.z + v4.w * x4.w;
       }
+      for (int k = D & ~3; k < D; ++k) accv += Vrow[k] * lds[X_OFF + e * MAX_D + k];
+      lds[XV_OFF + e * MAX_R + rr] = accv;
     }
     __syncthreads();
 
-    // ---- env state update + reward partials (lanes split d) ----
+    // ---- env state update + reward partials (threads split d) ----
     float racc[ENV_TILE];
     #pragma unroll
     for (int e = 0; e < ENV_TILE; ++e) racc[e] = 0.f;
-    for (int d = lane; d < D; d += WAVE) {
+    for (int d = tid; d < D; d += NWAVES * WAVE) {
       const float dd = env_d[d];
       const float* Ut_row = env_Ut + (int64_t)d * a.rank;
       const float* Bt_row = env_Bt + (int64_t)d * A;
@@ -312,17 +322,16 @@ __global__ void rollout_kernel(RolloutArgs a) {
       }
     }
     #pragma unroll
-    for (int e = 0; e < ENV_TILE; ++e) racc[e] = wave_reduce_sum(racc[e]);
-    if (lane == 0) {
-      #pragma unroll
-      for (int e = 0; e < ENV_TILE; ++e) rsum_lds[e] = 1.0f - racc[e] / D;
+    for (int e = 0; e < ENV_TILE; ++e) {
+      const float w = wave_reduce_sum(racc[e]);
+      if (lane == 0 && e < nE) atomicAdd(&racc_lds[e], w);
     }
     __syncthreads();
 
-    // ---- reward, done, episode bookkeeping (lane e handles env e) ----
-    if (lane < nE) {
-      const int e = lane;
-      const float r = rsum_lds[e];
+    // ---- reward, done, episode bookkeeping (thread e handles env e) ----
+    if (tid < nE) {
+      const int e = tid;
+      const float r = 1.0f - racc_lds[e] / D;
       out_rewards[(int64_t)step * E + e0 + e] = r;
       epr_lds[e] += r;
       int tc = tc_lds[e] + 1;
@@ -340,7 +349,7 @@ __global__ void rollout_kernel(RolloutArgs a) {
     // ---- reset finished envs ----
     for (int e = 0; e < nE; ++e) {
       if (done_lds[e]) {
-        for (int d = lane; d < D; d += WAVE)
+        for (int d = tid; d < D; d += NWAVES * WAVE)
           lds[X_OFF + e * MAX_D + d] =
               0.1f * rng_normal(a.seed, e0 + e, step, 5000 + d);
       }
@@ -354,48 +363,43 @@ __global__ void rollout_kernel(RolloutArgs a) {
     for (int l = 0; l < a.n_hidden; ++l) {
       const int out_dim = a.dims[l + 1];
       const int out_off = (l & 1) ? H1_OFF : H0_OFF;
-      const float* W = a.params + a.off_W[l];
+      layer_kpart(a.params + a.off_W[l], 0, in_off, in_stride, in_dim,
+                  out_dim, false);
+      __syncthreads();
       const float* bias = a.params + a.off_b[l];
-      for (int u = lane; u < out_dim; u += WAVE) {
-        float acc[ENV_TILE];
-        const float bu = bias[u];
+      for (int idx = tid; idx < ENV_TILE * out_dim; idx += NWAVES * WAVE) {
+        const int e = idx / out_dim, u = idx % out_dim;
+        float sum = bias[u];
         #pragma unroll
-        for (int e = 0; e < ENV_TILE; ++e) acc[e] = bu;
-        const float* Wrow = W + (int64_t)u * in_dim;
-        for (int k = 0; k < in_dim; ++k) {
-          const float w = Wrow[k];
-          #pragma unroll
-          for (int e = 0; e < ENV_TILE; ++e)
-            acc[e] += w * lds[in_off + e * in_stride + k];
-        }
-        #pragma unroll
-        for (int e = 0; e < ENV_TILE; ++e) {
-          const float v = acc[e];
-          lds[out_off + e * MAX_H + u] = a.activation ? tanhf(v) : fmaxf(v, 0.f);
-        }
+        for (int w = 0; w < NWAVES; ++w)
+          sum += lds[PART_OFF + (w * ENV_TILE + e) * MAX_H + u];
+        lds[out_off + e * MAX_H + u] =
+            a.activation ? tanhf(sum) : fmaxf(sum, 0.f);
       }
       __syncthreads();
       in_off = out_off;
       in_stride = MAX_H;
       in_dim = out_dim;
     }
-    if (lane < nE) {
-      float acc = a.params[a.off_bv];
+    // value head: wave wv reduces env wv over lanes
+    if (wv < nE) {
       const float* Wv = a.params + a.off_Wv;
-      for (int k = 0; k < in_dim; ++k)
-        acc += Wv[k] * lds[in_off + lane * in_stride + k];
-      out_boot[e0 + lane] = acc;
+      float acc = 0.f;
+      for (int k = lane; k < in_dim; k += WAVE)
+        acc += Wv[k] * lds[in_off + wv * in_stride + k];
+      const float total = wave_reduce_sum(acc);
+      if (lane == 0) out_boot[e0 + wv] = total + a.params[a.off_bv];
     }
   }
 
   // ---- persist env state ----
   for (int e = 0; e < nE; ++e) {
-    for (int d = lane; d < D; d += WAVE)
+    for (int d = tid; d < D; d += NWAVES * WAVE)
       a.x[(int64_t)(e0 + e) * D + d] = lds[X_OFF + e * MAX_D + d];
   }
-  if (lane < nE) {
-    a.epr[e0 + lane] = epr_lds[lane];
-    a.t[e0 + lane] = tc_lds[lane];
+  if (tid < nE) {
+    a.epr[e0 + tid] = epr_lds[tid];
+    a.t[e0 + tid] = tc_lds[tid];
   }
 }
 
@@ -504,7 +508,7 @@ std::vector<torch::Tensor> rollout_run(
 
   hipStream_t stream = c10::hip::getCurrentHIPStream().stream();
   const int grid = static_cast<int>((E + ENV_TILE - 1) / ENV_TILE);
-  hipLaunchKernelGGL(rollout_kernel, dim3(grid), dim3(WAVE), 0, stream, a);
+  hipLaunchKernelGGL(rollout_kernel, dim3(grid), dim3(NWAVES * WAVE), 0, stream, a);
 
   // carve views out of the blob
   int64_t o = 0;
