@@ -101,10 +101,25 @@ DEV_INLINE float rng_uniform(unsigned seed, int env, int step, int slot) {
   return ((h >> 8) + 1) * (1.0f / 16777217.0f);
 }
 
+// One hash -> two 16-bit uniforms -> one Box-Muller pair.  The rollout
+// draws ~1500 normals per block-step (env noise dominates); the naive
+// 2-hash + log + cos per normal made the whole kernel transcendental-
+// bound.  16-bit resolution truncates the tails at ~4.7 sigma, which is
+// statistically invisible for synthetic-env noise and recorded actions.
+DEV_INLINE float2 rng_normal2(unsigned seed, int env, int step, int slot) {
+  const unsigned h = lowbias32(seed ^ (unsigned)env * 0x9E3779B9U ^
+                               (unsigned)step * 0x85EBCA6BU ^
+                               (unsigned)slot * 0xC2B2AE35U);
+  const float u1 = ((h >> 16) + 1) * (1.0f / 65537.0f);
+  const float u2 = (h & 0xFFFFu) * (1.0f / 65536.0f);
+  const float r = sqrtf(-2.0f * __logf(u1));
+  float sn, cs;
+  __sincosf(6.2831853071795865f * u2, &sn, &cs);
+  return make_float2(r * cs, r * sn);
+}
+
 DEV_INLINE float rng_normal(unsigned seed, int env, int step, int slot) {
-  const float u1 = rng_uniform(seed, env, step, 2 * slot + 100000);
-  const float u2 = rng_uniform(seed, env, step, 2 * slot + 100001);
-  return sqrtf(-2.0f * __logf(u1)) * __cosf(6.2831853071795865f * u2);
+  return rng_normal2(seed, env, step, slot).x;
 }
 
 // monotone float<->uint encoding for atomic max (min via -x); raw 0 is the
@@ -309,14 +324,22 @@ __global__ void rollout_kernel(RolloutArgs a) {
       const float dd = env_d[d];
       const float* Ut_row = env_Ut + (int64_t)d * a.rank;
       const float* Bt_row = env_Bt + (int64_t)d * A;
+      // one Box-Muller pair per (even env, d) feeds two envs
+      float nz[ENV_TILE];
+      #pragma unroll
+      for (int q = 0; q < ENV_TILE / 2; ++q) {
+        const float2 p = rng_normal2(a.seed, e0 + 2 * q, step, 1000 + d);
+        nz[2 * q] = p.x;
+        nz[2 * q + 1] = p.y;
+      }
       for (int e = 0; e < nE; ++e) {
         float low = 0.f;
         for (int rr = 0; rr < a.rank; ++rr)
           low += lds[XV_OFF + e * MAX_R + rr] * Ut_row[rr];
         float ain = 0.f;
         for (int j = 0; j < A; ++j) ain += lds[ACT_OFF + e * MAX_A + j] * Bt_row[j];
-        const float n = a.noise * rng_normal(a.seed, e0 + e, step, 1000 + d);
-        const float xn = tanhf(lds[X_OFF + e * MAX_D + d] * dd + low + ain + n);
+        const float xn = tanhf(lds[X_OFF + e * MAX_D + d] * dd + low + ain +
+                               a.noise * nz[e]);
         lds[X_OFF + e * MAX_D + d] = xn;
         racc[e] += xn * xn;
       }
