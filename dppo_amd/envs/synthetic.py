@@ -95,10 +95,12 @@ class BatchedSyntheticEnv:
         self.Vt = self.V.t().contiguous()   # [r, D]
         self.Ut = self.U.t().contiguous()   # [D, r]
         self.Bt = self.B.t().contiguous()   # [D, A] (or [D, n] discrete)
-        # single packed blob (d | Vt | Ut | Bt) so the kernel takes one pointer
+        # single packed blob (d | Vt | U | B) so the kernel takes one
+        # pointer.  Vt rows feed the per-lane xv dot products; U[r][D] and
+        # B[A][D] are read row-wise with lane==d, i.e. fully coalesced.
         self.blob = torch.cat(
-            [self.d.reshape(-1), self.Vt.reshape(-1), self.Ut.reshape(-1),
-             self.Bt.reshape(-1)]
+            [self.d.reshape(-1), self.Vt.reshape(-1), self.U.reshape(-1),
+             self.B.reshape(-1)]
         ).contiguous()
         self.rank_eff = self.Vt.shape[0]
 

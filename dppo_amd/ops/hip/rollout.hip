@@ -159,11 +159,11 @@ __global__ void rollout_kernel(RolloutArgs a) {
   __shared__ int tc_lds[ENV_TILE];
   __shared__ int done_lds[ENV_TILE];
 
-  // env blob offsets
+  // env blob offsets (d | Vt[r][D] | U[r][D] | B[A][D])
   const float* env_d = a.envblob;
   const float* env_Vt = env_d + D;
-  const float* env_Ut = env_Vt + (int64_t)a.rank * D;
-  const float* env_Bt = env_Ut + (int64_t)D * a.rank;
+  const float* env_U = env_Vt + (int64_t)a.rank * D;
+  const float* env_B = env_U + (int64_t)a.rank * D;
   // output blob offsets
   float* out_states = a.out;
   float* out_pdflats = out_states + (int64_t)T * E * D;
@@ -322,8 +322,6 @@ __global__ void rollout_kernel(RolloutArgs a) {
     for (int e = 0; e < ENV_TILE; ++e) racc[e] = 0.f;
     for (int d = tid; d < D; d += NWAVES * WAVE) {
       const float dd = env_d[d];
-      const float* Ut_row = env_Ut + (int64_t)d * a.rank;
-      const float* Bt_row = env_Bt + (int64_t)d * A;
       // one Box-Muller pair per (even env, d) feeds two envs
       float nz[ENV_TILE];
       #pragma unroll
@@ -332,14 +330,30 @@ __global__ void rollout_kernel(RolloutArgs a) {
         nz[2 * q] = p.x;
         nz[2 * q + 1] = p.y;
       }
+      // U[rr][d] / B[j][d] row loads: lane == d, fully coalesced, ONE
+      // load per rr/j shared by all four envs (the previous transposed
+      // per-d rows cost 33 uncoalesced loads per env per d — the kernel
+      // measured VMEM-wait bound, SQ_WAIT_ANY 27x SQ_BUSY).
+      float low[ENV_TILE], ain[ENV_TILE];
+      #pragma unroll
+      for (int e = 0; e < ENV_TILE; ++e) low[e] = ain[e] = 0.f;
+      #pragma unroll 2
+      for (int rr = 0; rr < a.rank; ++rr) {
+        const float uv = env_U[(int64_t)rr * D + d];
+        #pragma unroll
+        for (int e = 0; e < ENV_TILE; ++e)
+          low[e] += lds[XV_OFF + e * MAX_R + rr] * uv;
+      }
+      #pragma unroll 2
+      for (int j = 0; j < A; ++j) {
+        const float bvv = env_B[(int64_t)j * D + d];
+        #pragma unroll
+        for (int e = 0; e < ENV_TILE; ++e)
+          ain[e] += lds[ACT_OFF + e * MAX_A + j] * bvv;
+      }
       for (int e = 0; e < nE; ++e) {
-        float low = 0.f;
-        for (int rr = 0; rr < a.rank; ++rr)
-          low += lds[XV_OFF + e * MAX_R + rr] * Ut_row[rr];
-        float ain = 0.f;
-        for (int j = 0; j < A; ++j) ain += lds[ACT_OFF + e * MAX_A + j] * Bt_row[j];
-        const float xn = tanhf(lds[X_OFF + e * MAX_D + d] * dd + low + ain +
-                               a.noise * nz[e]);
+        const float xn = tanhf(lds[X_OFF + e * MAX_D + d] * dd + low[e] +
+                               ain[e] + a.noise * nz[e]);
         lds[X_OFF + e * MAX_D + d] = xn;
         racc[e] += xn * xn;
       }
