@@ -14,6 +14,12 @@ torch::Tensor ppo_loss_gauss_fwd(torch::Tensor pdpi, torch::Tensor pdold,
                                  torch::Tensor etr, double clip,
                                  double entcoeff, double vcoeff);
 
+torch::Tensor ppo_loss_gauss_gh(torch::Tensor pdflat, torch::Tensor oldflat,
+                                torch::Tensor vpred, torch::Tensor oldv,
+                                torch::Tensor act, torch::Tensor adv,
+                                torch::Tensor etr, double clip,
+                                double entcoeff, double vcoeff);
+
 std::vector<torch::Tensor> ppo_loss_gauss_bwd(
     torch::Tensor pdpi, torch::Tensor pdold, torch::Tensor vpred,
     torch::Tensor oldv, torch::Tensor act, torch::Tensor adv,
@@ -24,28 +30,13 @@ void adam_step(torch::Tensor p, torch::Tensor g, torch::Tensor m,
                torch::Tensor v, int64_t step, double lr, double beta1,
                double beta2, double eps);
 
-std::vector<torch::Tensor> mlp_fwd(torch::Tensor params,
-                                   std::vector<int64_t> offsets,
-                                   std::vector<int64_t> dims,
-                                   int64_t activation, torch::Tensor states,
-                                   int64_t act_dim);
-
-std::vector<torch::Tensor> mlp_bwd_rows(
-    torch::Tensor params, std::vector<int64_t> offsets,
-    std::vector<int64_t> dims, int64_t activation, torch::Tensor acts,
-    torch::Tensor pdflat, torch::Tensor oldflat, torch::Tensor v,
-    torch::Tensor oldv, torch::Tensor actions, torch::Tensor adv,
-    torch::Tensor etr, double clip, double entcoeff, double vcoeff);
-
-void dw_accum(torch::Tensor delta, torch::Tensor acts, torch::Tensor grad_buf,
-              int64_t w_off, int64_t b_off);
-
 void gemm_fwd(torch::Tensor X, torch::Tensor Wt, torch::Tensor bias,
               int64_t activation, int64_t heads, torch::Tensor C,
-              torch::Tensor v);
+              torch::Tensor v, torch::Tensor aux);
 
 void dw_mfma(torch::Tensor delta, torch::Tensor acts, torch::Tensor grad_buf,
-             int64_t w_off, int64_t b_off);
+             int64_t w_off, int64_t b_off, int64_t split_row, int64_t w_off2,
+             int64_t b_off2);
 
 void dwv(torch::Tensor gv, torch::Tensor acts, torch::Tensor grad_buf,
          int64_t w_off, int64_t b_off);
@@ -68,14 +59,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("adam_step", &adam_step, "fused flat Adam step (gfx950)");
   mod.def("rollout_run", &rollout_run,
           "fused T-step rollout: MLP fwd + sample + synthetic env (gfx950)");
-  mod.def("mlp_fwd", &mlp_fwd, "fused MLP forward, saves activations (gfx950)");
-  mod.def("mlp_bwd_rows", &mlp_bwd_rows,
-          "fused PPO-loss + MLP backward rows -> dz/g_pd/g_v (gfx950)");
+  mod.def("ppo_loss_gauss_gh", &ppo_loss_gauss_gh,
+          "wave-per-row PPO loss gradient -> [g_pd | g_v] (gfx950)");
   mod.def("gemm_fwd", &gemm_fwd,
           "MFMA f32 layer forward C=act(X@Wt+b), fused tanh (gfx950)");
   mod.def("dw_mfma", &dw_mfma,
           "MFMA f32 split-K dW += delta^T@acts into flat grad (gfx950)");
   mod.def("dwv", &dwv, "value-head weight grad reduction (gfx950)");
-  mod.def("dw_accum", &dw_accum,
-          "dW += delta^T @ acts, db += sum(delta) into flat grad (gfx950)");
 }

@@ -32,7 +32,7 @@ namespace {
 
 using f32x16 = __attribute__((ext_vector_type(16))) float;
 
-constexpr int BK = 32;        // K-step per stage
+constexpr int BK = 64;        // K-step per stage
 constexpr int M_WAVE = 32;    // rows per wave tile
 constexpr int FWD_WAVES = 4;  // waves per block (each owns 32 rows)
 constexpr int FWD_M = FWD_WAVES * M_WAVE;  // 128 rows per block
@@ -50,11 +50,14 @@ struct FwdArgs {
   const float* X;     // [B][K]
   const float* Wt;    // [K][N]
   const float* bias;  // [N]
+  const float* aux;   // [B][N] forward activations (dgrad modes)
   float* C;           // [B][N]      (heads: pdflat [B][N-1])
   float* v;           // [B] heads only
   int64_t B;
   int K, N;
-  int activation;     // 0 relu, 1 tanh, 2 none
+  int activation;     // 0 relu, 1 tanh, 2 none;
+                      // 3 tanh-grad: C = (1-aux^2)*acc (no bias)
+                      // 4 relu-grad: C = (aux>0)*acc   (no bias)
   int heads;          // if 1: last column -> v, rest -> C (pdflat)
 };
 
@@ -114,14 +117,21 @@ __global__ void gemm_fwd_kernel(FwdArgs a) {
     for (int t = 0; t < NT; ++t) {
       const int col = t * M_WAVE + i_l;
       if (col < a.N) {
-        const float bv = a.bias[col];
+        const float bv = (a.activation >= 3) ? 0.f : a.bias[col];
         #pragma unroll
         for (int r = 0; r < 16; ++r) {
           const int64_t row = b0 + wave * M_WAVE + cd_row(r, lane);
           if (row < a.B) {
-            float x = acc[t][r] + bv;
-            if (a.activation == 0) x = fmaxf(x, 0.f);
-            else if (a.activation == 1) x = tanhf(x);
+            float x;
+            if (a.activation >= 3) {
+              const float h = a.aux[row * a.N + col];
+              x = acc[t][r] * ((a.activation == 3) ? (1.f - h * h)
+                                                   : (h > 0.f ? 1.f : 0.f));
+            } else {
+              x = acc[t][r] + bv;
+              if (a.activation == 0) x = fmaxf(x, 0.f);
+              else if (a.activation == 1) x = tanhf(x);
+            }
             if (a.heads) {
               if (col == a.N - 1) a.v[row] = x;
               else a.C[row * (a.N - 1) + col] = x;
@@ -145,10 +155,13 @@ struct DwArgs {
   const float* acts;   // [B][in]
   float* dW;           // grad + w_off  ([out][in])
   float* db;           // grad + b_off or nullptr
+  float* dW2;          // second target: delta rows >= split_row
+  float* db2;
   int64_t B;
   int out_dim, in_dim;
-  int nt;      // 32-col tiles per wave (<= MAX_NT)
-  int splits;  // K splits
+  int nt;       // 32-col tiles per wave (<= MAX_NT)
+  int splits;   // K splits
+  int split_row;  // -1: single target; else rows >= split go to dW2/db2
 };
 
 template <int NT>
@@ -186,27 +199,41 @@ __global__ void dw_mfma_kernel(DwArgs a) {
     bcol[t] = col_ok[t] ? col : 0;
   }
 
-  #pragma unroll 1
-  for (int64_t k = k0; k + 3 < k1; k += 4) {
-    // two k-pairs per iteration: loads of the second pair issue while the
-    // first pair's MFMAs run
-    const float av0 = m_ok ? a.delta[(k + k_l) * a.out_dim + mcol] : 0.f;
-    const float av1 = m_ok ? a.delta[(k + 2 + k_l) * a.out_dim + mcol] : 0.f;
-    float bv0[NT], bv1[NT];
+  // one-pair-lookahead software pipeline: the next pair's streamed loads
+  // issue before the current pair's 64-cycle MFMAs, so HBM/L2 latency
+  // hides under the matrix pipe instead of serializing with it.
+  const int64_t kend = k0 + ((k1 - k0) & ~1);
+  int64_t k = k0;
+  float av_c = 0.f;
+  float bv_c[NT];
+  if (k < kend) {
+    av_c = m_ok ? a.delta[(k + k_l) * a.out_dim + mcol] : 0.f;
     #pragma unroll
-    for (int t = 0; t < NT; ++t) {
-      bv0[t] = a.acts[(k + k_l) * a.in_dim + bcol[t]];
-      bv1[t] = a.acts[(k + 2 + k_l) * a.in_dim + bcol[t]];
-    }
-    dbacc += av0 + av1;
-    #pragma unroll
-    for (int t = 0; t < NT; ++t)
-      acc[t] = __builtin_amdgcn_mfma_f32_32x32x2f32(av0, bv0[t], acc[t], 0, 0, 0);
-    #pragma unroll
-    for (int t = 0; t < NT; ++t)
-      acc[t] = __builtin_amdgcn_mfma_f32_32x32x2f32(av1, bv1[t], acc[t], 0, 0, 0);
+    for (int t = 0; t < NT; ++t) bv_c[t] = a.acts[(k + k_l) * a.in_dim + bcol[t]];
   }
-  for (int64_t k = k0 + ((k1 - k0) & ~3); k < k1; ++k) {
+  #pragma unroll 1
+  for (; k + 2 < kend; k += 2) {
+    const float av_n = m_ok ? a.delta[(k + 2 + k_l) * a.out_dim + mcol] : 0.f;
+    float bv_n[NT];
+    #pragma unroll
+    for (int t = 0; t < NT; ++t)
+      bv_n[t] = a.acts[(k + 2 + k_l) * a.in_dim + bcol[t]];
+    dbacc += av_c;
+    #pragma unroll
+    for (int t = 0; t < NT; ++t)
+      acc[t] = __builtin_amdgcn_mfma_f32_32x32x2f32(av_c, bv_c[t], acc[t], 0, 0, 0);
+    av_c = av_n;
+    #pragma unroll
+    for (int t = 0; t < NT; ++t) bv_c[t] = bv_n[t];
+  }
+  if (k < kend) {
+    dbacc += av_c;
+    #pragma unroll
+    for (int t = 0; t < NT; ++t)
+      acc[t] = __builtin_amdgcn_mfma_f32_32x32x2f32(av_c, bv_c[t], acc[t], 0, 0, 0);
+    k += 2;
+  }
+  for (; k < k1; ++k) {
     const float av = (m_ok && k_l == 0) ? a.delta[k * a.out_dim + mcol] : 0.f;
     if (k_l == 0) dbacc += av;
     #pragma unroll
@@ -216,7 +243,7 @@ __global__ void dw_mfma_kernel(DwArgs a) {
     }
   }
 
-  // ---- scatter accumulators ----
+  // ---- scatter accumulators (optional row split for combined heads) ----
   #pragma unroll
   for (int t = 0; t < NT; ++t) {
     if (col_ok[t]) {
@@ -224,7 +251,12 @@ __global__ void dw_mfma_kernel(DwArgs a) {
       for (int r = 0; r < 16; ++r) {
         const int mrow = m0 + cd_row(r, lane);
         if (mrow < a.out_dim && acc[t][r] != 0.f) {
-          atomicAdd(&a.dW[(int64_t)mrow * a.in_dim + bcol[t]], acc[t][r]);
+          if (a.split_row >= 0 && mrow >= a.split_row) {
+            atomicAdd(&a.dW2[(int64_t)(mrow - a.split_row) * a.in_dim + bcol[t]],
+                      acc[t][r]);
+          } else {
+            atomicAdd(&a.dW[(int64_t)mrow * a.in_dim + bcol[t]], acc[t][r]);
+          }
         }
       }
     }
@@ -233,7 +265,12 @@ __global__ void dw_mfma_kernel(DwArgs a) {
   // lane+32 (same m-column), then one atomic per m from the low half.
   if (a.db != nullptr && blockIdx.y == 0) {
     const float other = __shfl(dbacc, lane ^ 32, WAVE);
-    if (lane < 32 && m_ok) atomicAdd(&a.db[mcol], dbacc + other);
+    if (lane < 32 && m_ok) {
+      if (a.split_row >= 0 && mcol >= a.split_row)
+        atomicAdd(&a.db2[mcol - a.split_row], dbacc + other);
+      else
+        atomicAdd(&a.db[mcol], dbacc + other);
+    }
   }
 }
 
@@ -283,7 +320,7 @@ __global__ void dwv_kernel(const float* __restrict__ gv,    // [B]
 
 void gemm_fwd(torch::Tensor X, torch::Tensor Wt, torch::Tensor bias,
               int64_t activation, int64_t heads, torch::Tensor C,
-              torch::Tensor v) {
+              torch::Tensor v, torch::Tensor aux) {
   // C and (for heads) v are caller-allocated so activations can land
   // directly in the backward's blob layout.
   const int64_t B = X.size(0);
@@ -291,7 +328,8 @@ void gemm_fwd(torch::Tensor X, torch::Tensor Wt, torch::Tensor bias,
   const int N = static_cast<int>(Wt.size(1));
   TORCH_CHECK(X.is_cuda() && X.is_contiguous() && Wt.is_contiguous());
   TORCH_CHECK(C.is_contiguous());
-  TORCH_CHECK(Wt.size(0) == K && bias.numel() == N);
+  TORCH_CHECK(Wt.size(0) == K);
+  TORCH_CHECK(activation >= 3 || bias.numel() == N);
   TORCH_CHECK(N <= MAX_NT * M_WAVE, "N exceeds MFMA fwd tile budget");
   if (heads) {
     TORCH_CHECK(C.numel() == B * (N - 1) && v.numel() == B);
@@ -310,6 +348,10 @@ void gemm_fwd(torch::Tensor X, torch::Tensor Wt, torch::Tensor bias,
   a.heads = static_cast<int>(heads);
   a.C = C.data_ptr<float>();
   a.v = heads ? v.data_ptr<float>() : nullptr;
+  if (activation >= 3) {
+    TORCH_CHECK(aux.numel() == B * N, "dgrad mode needs aux = fwd activations");
+    a.aux = aux.data_ptr<float>();
+  }
 
   hipStream_t stream = c10::hip::getCurrentHIPStream().stream();
   const int64_t tiles = (B + FWD_M - 1) / FWD_M;
@@ -336,7 +378,8 @@ void gemm_fwd(torch::Tensor X, torch::Tensor Wt, torch::Tensor bias,
 }
 
 void dw_mfma(torch::Tensor delta, torch::Tensor acts, torch::Tensor grad_buf,
-             int64_t w_off, int64_t b_off) {
+             int64_t w_off, int64_t b_off, int64_t split_row, int64_t w_off2,
+             int64_t b_off2) {
   const int64_t B = delta.size(0);
   const int out_dim = static_cast<int>(delta.size(1));
   const int in_dim = static_cast<int>(acts.size(1));
@@ -347,6 +390,11 @@ void dw_mfma(torch::Tensor delta, torch::Tensor acts, torch::Tensor grad_buf,
   a.acts = acts.data_ptr<float>();
   a.dW = grad_buf.data_ptr<float>() + w_off;
   a.db = (b_off >= 0) ? grad_buf.data_ptr<float>() + b_off : nullptr;
+  a.split_row = static_cast<int>(split_row);
+  a.dW2 = (split_row >= 0) ? grad_buf.data_ptr<float>() + w_off2 : nullptr;
+  a.db2 = (split_row >= 0 && b_off2 >= 0)
+              ? grad_buf.data_ptr<float>() + b_off2
+              : nullptr;
   a.B = B;
   a.out_dim = out_dim;
   a.in_dim = in_dim;

@@ -20,42 +20,15 @@
 #include <c10/hip/HIPStream.h>
 
 #include "common.h"
+#include "ppo_math.h"
 
 namespace {
 
-constexpr float LOG_2PI = 1.8378770664093453f;
-
-struct GaussRow {
-  float logp_pi, logp_old, ent;
-};
-
-// Per-sample log-probs + entropy for one row (loops over action dim).
 DEV_INLINE GaussRow gauss_row(const float* __restrict__ pdpi,
                               const float* __restrict__ pdold,
                               const float* __restrict__ act,
                               int64_t b, int A) {
-  const float* mu_pi = pdpi + (int64_t)b * 2 * A;
-  const float* ls_pi = mu_pi + A;
-  const float* mu_old = pdold + (int64_t)b * 2 * A;
-  const float* ls_old = mu_old + A;
-  const float* a = act + (int64_t)b * A;
-  float lp = 0.f, lo = 0.f, ent = 0.f;
-  for (int j = 0; j < A; ++j) {
-    const float aj = a[j];
-    const float lsp = ls_pi[j];
-    const float zp = (aj - mu_pi[j]) * __expf(-lsp);
-    lp += -0.5f * zp * zp - lsp;
-    const float lso = ls_old[j];
-    const float zo = (aj - mu_old[j]) * __expf(-lso);
-    lo += -0.5f * zo * zo - lso;
-    ent += lsp;
-  }
-  const float c = 0.5f * LOG_2PI * A;
-  GaussRow r;
-  r.logp_pi = lp - c;
-  r.logp_old = lo - c;
-  r.ent = ent + 0.5f * (LOG_2PI + 1.f) * A;
-  return r;
+  return ppo_gauss_row(pdpi, pdold, act, b, A);
 }
 
 __global__ void ppo_gauss_fwd_kernel(
@@ -218,4 +191,83 @@ std::vector<torch::Tensor> ppo_loss_gauss_bwd(
                      g_v.data_ptr<float>(), B, A, (float)clip, (float)entcoeff,
                      (float)vcoeff);
   return {g_pdflat, g_v};
+}
+
+// ---------------------------------------------------------------------------
+// Wave-per-row loss gradient: gh[b] = [dL/d pdflat | dL/d vpred].
+// Lanes j < 2A cooperate on one row (coalesced row loads, wave-reduced
+// log-prob sums), so there is no per-thread row loop to thrash L1 — the
+// input to the GEMM-based backward chain (engine _update_fused).
+// ---------------------------------------------------------------------------
+
+namespace {
+
+__launch_bounds__(256)
+__global__ void ppo_gh_kernel(
+    const float* __restrict__ pdflat, const float* __restrict__ oldflat,
+    const float* __restrict__ vpred, const float* __restrict__ oldv,
+    const float* __restrict__ act, const float* __restrict__ adv,
+    const float* __restrict__ etr, float* __restrict__ gh,  // [B][2A+1]
+    int64_t B, int A, float clip, float entcoeff, float vcoeff) {
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wave = threadIdx.x / WAVE;
+  const int P = 2 * A;
+  const int64_t waves_total = (int64_t)gridDim.x * 4;
+  const int64_t wid = (int64_t)blockIdx.x * 4 + wave;
+
+  for (int64_t b = wid; b < B; b += waves_total) {
+    float lp_part = 0.f, lo_part = 0.f, ent_part = 0.f;
+    float z = 0.f, inv_s = 0.f;
+    if (lane < P) {
+      const int jj = (lane < A) ? lane : lane - A;
+      const float mu = pdflat[b * P + jj];
+      const float ls = pdflat[b * P + A + jj];
+      const float aj = act[b * A + jj];
+      inv_s = __expf(-ls);
+      z = (aj - mu) * inv_s;
+      if (lane < A) {
+        lp_part = -0.5f * z * z - ls;
+        const float mo = oldflat[b * P + jj];
+        const float lso = oldflat[b * P + A + jj];
+        const float zo = (aj - mo) * __expf(-lso);
+        lo_part = -0.5f * zo * zo - lso;
+        ent_part = ls;
+      }
+    }
+    const float c = 0.5f * PPO_LOG_2PI * A;
+    GaussRow row;
+    row.logp_pi = __shfl(wave_reduce_sum(lp_part), 0, WAVE) - c;
+    row.logp_old = __shfl(wave_reduce_sum(lo_part), 0, WAVE) - c;
+    row.ent = __shfl(wave_reduce_sum(ent_part), 0, WAVE) +
+              0.5f * (PPO_LOG_2PI + 1.f) * A;
+    const PPORowGrads g =
+        ppo_row_grads(row, vpred[b], oldv[b], adv[b], etr[b], B, clip,
+                      entcoeff, vcoeff, 1.f);
+    if (lane < P) {
+      gh[b * (P + 1) + lane] =
+          (lane < A) ? g.g_logp * z * inv_s
+                     : g.g_logp * (z * z - 1.f) + g.g_ent;
+    }
+    if (lane == 0) gh[b * (P + 1) + P] = g.g_v;
+  }
+}
+
+}  // namespace
+
+torch::Tensor ppo_loss_gauss_gh(torch::Tensor pdflat, torch::Tensor oldflat,
+                                torch::Tensor vpred, torch::Tensor oldv,
+                                torch::Tensor act, torch::Tensor adv,
+                                torch::Tensor etr, double clip,
+                                double entcoeff, double vcoeff) {
+  const int64_t B = vpred.numel();
+  const int A = static_cast<int>(pdflat.size(1) / 2);
+  auto gh = torch::empty({B, 2 * (int64_t)A + 1}, pdflat.options());
+  hipStream_t stream = c10::hip::getCurrentHIPStream().stream();
+  hipLaunchKernelGGL(ppo_gh_kernel, dim3(2048), dim3(256), 0, stream,
+                     pdflat.data_ptr<float>(), oldflat.data_ptr<float>(),
+                     vpred.data_ptr<float>(), oldv.data_ptr<float>(),
+                     act.data_ptr<float>(), adv.data_ptr<float>(),
+                     etr.data_ptr<float>(), gh.data_ptr<float>(), B, A,
+                     (float)clip, (float)entcoeff, (float)vcoeff);
+  return gh;
 }

@@ -500,55 +500,72 @@ class DPPOEngine:
         for l in range(n_hidden):
             n = B * dims[l + 1]
             cview = acts.narrow(0, o, n).view(B, dims[l + 1])
-            ext.gemm_fwd(x, Wts[l], bs[l], act_code, 0, cview, cview)
+            ext.gemm_fwd(x, Wts[l], bs[l], act_code, 0, cview, cview, cview)
             x = cview
             a_views.append(cview)
             o += n
         P = 2 * self.act_space.shape[0]
         pdflat = torch.empty(B, P, device=states.device, dtype=states.dtype)
         v = torch.empty(B, device=states.device, dtype=states.dtype)
-        ext.gemm_fwd(x, Wth, bh, 2, 1, pdflat, v)
+        ext.gemm_fwd(x, Wth, bh, 2, 1, pdflat, v, pdflat)
         return acts, a_views, v, pdflat
 
-    def _update_fused(self, batch: RolloutBatch, l_mul: float) -> None:
-        """Fused MFMA update steps: gemm_fwd per layer -> mlp_bwd_rows ->
-        dw_mfma/dwv into the flat grad -> all-reduce -> fused Adam."""
+    def _fused_backward(self, states, acts, a_views, v, pdflat,
+                        oldflat, oldv, actions, adv, etr, clip: float) -> None:
+        """GEMM-shaped backward into the (already zeroed) flat grad:
+        wave-per-row loss grads -> dgrad GEMM chain (torch weight layouts
+        ARE the needed Wt; no transposes) -> split-K dW scatter."""
         from .ops import hip_ext
 
         ext = hip_ext()
         c = self.cfg
-        A = self.act_space.shape[0]
         offsets = [sl.start for sl in self.flat_pi.slices]
-        dims = [self.obs_space.shape[0], *c.HIDDEN_SIZES]
-        act_code = 1 if c.ACTIVATION == "tanh" else 0
-        clip = c.CLIP_PARAM * l_mul
-        B = batch.states.shape[0]
         n_hidden = len(c.HIDDEN_SIZES)
-        for _ in range(c.UPDATE_STEPS):
+        dgrad_code = 3 if c.ACTIVATION == "tanh" else 4
+        P = 2 * self.act_space.shape[0]
+        off_wv, off_bv = offsets[2 * n_hidden], offsets[2 * n_hidden + 1]
+        off_wp, off_bp = offsets[2 * n_hidden + 2], offsets[2 * n_hidden + 3]
+        dummy_bias = torch.zeros(1, device=self.device)
+
+        gh = ext.ppo_loss_gauss_gh(
+            pdflat, oldflat, v, oldv, actions, adv, etr,
+            clip, c.ENTCOEFF, c.VCOEFF,
+        )
+        with torch.no_grad():
+            # [Wp; Wv] rows == the Wt layout the dgrad GEMM needs
+            Wh_cat = torch.cat(
+                [self.pi.pi.weight, self.pi.vf.weight], dim=0
+            ).contiguous()
+        dz = [None] * n_hidden
+        delta, Wt_chain = gh, Wh_cat
+        for l in range(n_hidden - 1, -1, -1):
+            dz_l = torch.empty_like(a_views[l])
+            ext.gemm_fwd(delta, Wt_chain, dummy_bias, dgrad_code, 0,
+                         dz_l, dz_l, a_views[l])
+            dz[l] = dz_l
+            delta = dz_l
+            Wt_chain = self.pi.hidden[l].weight.detach()
+        grad = self.flat_pi.flat_grad
+        ext.dw_mfma(dz[0], states, grad, offsets[0], offsets[1], -1, -1, -1)
+        for l in range(1, n_hidden):
+            ext.dw_mfma(dz[l], a_views[l - 1], grad,
+                        offsets[2 * l], offsets[2 * l + 1], -1, -1, -1)
+        # combined heads: rows < P -> Wp/bp, row P -> Wv/bv
+        ext.dw_mfma(gh, a_views[-1], grad, off_wp, off_bp, P, off_wv, off_bv)
+
+    def _update_fused(self, batch: RolloutBatch, l_mul: float) -> None:
+        """Fused MFMA update steps: gemm_fwd xL -> GEMM-shaped backward ->
+        all-reduce -> fused Adam (~10 launches per step, no autograd)."""
+        clip = self.cfg.CLIP_PARAM * l_mul
+        for _ in range(self.cfg.UPDATE_STEPS):
             acts, a_views, v, pdflat = self._fused_forward(batch.states)
             self.flat_pi.zero_grad()
-            dz, g_pd, g_v = ext.mlp_bwd_rows(
-                self.flat_pi.flat_param.detach(), offsets, dims, act_code,
-                acts, pdflat, batch.oldflat, v, batch.oldv,
-                batch.actions, batch.adv, batch.etr,
-                clip, c.ENTCOEFF, c.VCOEFF,
+            self._fused_backward(
+                batch.states, acts, a_views, v, pdflat,
+                batch.oldflat, batch.oldv, batch.actions,
+                batch.adv, batch.etr, clip,
             )
-            dz_views, o = [], 0
-            for l in range(n_hidden):
-                n = B * dims[l + 1]
-                dz_views.append(dz.narrow(0, o, n).view(B, dims[l + 1]))
-                o += n
-            grad = self.flat_pi.flat_grad
-            ext.dw_mfma(dz_views[0], batch.states, grad, offsets[0], offsets[1])
-            for l in range(1, n_hidden):
-                ext.dw_mfma(dz_views[l], a_views[l - 1], grad,
-                            offsets[2 * l], offsets[2 * l + 1])
-            h_last = a_views[-1]
-            off_wv, off_bv = offsets[2 * n_hidden], offsets[2 * n_hidden + 1]
-            off_wp, off_bp = offsets[2 * n_hidden + 2], offsets[2 * n_hidden + 3]
-            ext.dw_mfma(g_pd, h_last, grad, off_wp, off_bp)
-            ext.dwv(g_v, h_last, grad, off_wv, off_bv)
-            self.comm.allreduce_mean_(grad)
+            self.comm.allreduce_mean_(self.flat_pi.flat_grad)
             self.optimizer.step()
     # ------------------------------------------------------------------
     def train_round(self) -> Tuple[Dict[str, float], bool]:

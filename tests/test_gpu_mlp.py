@@ -13,7 +13,6 @@ pytestmark = pytest.mark.gpu
 
 from dppo_amd.config import DPPOConfig
 from dppo_amd.distributions import DiagGaussianPdType
-from dppo_amd.ops import require_hip_ext
 from dppo_amd.ops.ppo_loss import PPOLossCoeffs, ppo_losses_ref
 from dppo_amd.parallel.comm import Comm
 from dppo_amd.trainer import DPPOEngine
@@ -49,9 +48,6 @@ def test_fused_backward_matches_autograd():
     B = 4096
     A = eng.act_space.shape[0]
     states = torch.randn(B, eng.obs_space.shape[0], device="cuda") * 0.5
-    ext = require_hip_ext()
-    offsets = [sl.start for sl in eng.flat_pi.slices]
-    dims = [eng.obs_space.shape[0], *eng.cfg.HIDDEN_SIZES]
     acts, a_views, v, pdflat = eng._fused_forward(states)
 
     pdt = DiagGaussianPdType(A)
@@ -61,7 +57,7 @@ def test_fused_backward_matches_autograd():
         actions = pdt.pdfromflat(oldflat).sample()
         adv = torch.randn(B, device="cuda")
         etr = torch.randn(B, device="cuda")
-    clip, entc, vc = 0.2, 0.01, 0.5
+    clip, entc, vc = 0.2, eng.cfg.ENTCOEFF, eng.cfg.VCOEFF
 
     # autograd reference
     eng.flat_pi.zero_grad()
@@ -72,27 +68,11 @@ def test_fused_backward_matches_autograd():
     out["total_loss"].backward()
     ref_grad = eng.flat_pi.flat_grad.clone()
 
-    # fused gradient
+    # fused GEMM-chain gradient
     eng.flat_pi.zero_grad()
-    dz, g_pd, g_v = ext.mlp_bwd_rows(
-        eng.flat_pi.flat_param.detach(), offsets, dims, 1, acts, pdflat,
-        oldflat, v, oldv, actions, adv, etr, clip, entc, vc,
-    )
+    eng._fused_backward(states, acts, a_views, v, pdflat, oldflat, oldv,
+                        actions, adv, etr, clip)
     grad = eng.flat_pi.flat_grad
-    n_hidden = len(eng.cfg.HIDDEN_SIZES)
-    dz_views, o = [], 0
-    for l in range(n_hidden):
-        n = B * dims[l + 1]
-        dz_views.append(dz.narrow(0, o, n).view(B, dims[l + 1]))
-        o += n
-    ext.dw_mfma(dz_views[0], states, grad, offsets[0], offsets[1])
-    for l in range(1, n_hidden):
-        ext.dw_mfma(dz_views[l], a_views[l - 1], grad,
-                    offsets[2 * l], offsets[2 * l + 1])
-    ext.dw_mfma(g_pd, a_views[-1], grad,
-                offsets[2 * n_hidden + 2], offsets[2 * n_hidden + 3])
-    ext.dwv(g_v, a_views[-1], grad,
-            offsets[2 * n_hidden], offsets[2 * n_hidden + 1])
 
     scale = float(ref_grad.abs().max())
     torch.testing.assert_close(grad, ref_grad, atol=scale * 2e-4 + 1e-8,
