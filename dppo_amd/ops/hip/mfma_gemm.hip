@@ -71,7 +71,7 @@ __global__ void gemm_fwd_kernel(FwdArgs a) {
   // Wt tile [BK][NT*32] staged coalesced so the MFMA B-operand comes
   // from LDS instead of a fresh L2 round trip per k-step.
   __shared__ float xs[FWD_M][BK + 1];
-  __shared__ float ws[BK][MAX_NT * M_WAVE];
+  __shared__ float ws[BK][NT * M_WAVE];
 
   const int i_l = lane & 31;   // A row within wave tile
   const int k_l = lane >> 5;   // A k within pair
@@ -96,7 +96,8 @@ __global__ void gemm_fwd_kernel(FwdArgs a) {
       for (int idx = threadIdx.x; idx < BK * NW; idx += FWD_WAVES * 64) {
         const int r = idx / NW, c = idx % NW;
         const int krow = kb + r;
-        ws[r][c] = (krow < a.K && c < a.N) ? a.Wt[(int64_t)krow * a.N + c] : 0.f;
+        ws[r][c] = (krow < a.K && c < a.N) ? a.Wt[(int64_t)krow * a.N + c]
+                                           : 0.f;
       }
       __syncthreads();
 
@@ -153,15 +154,13 @@ __global__ void gemm_fwd_kernel(FwdArgs a) {
 struct DwArgs {
   const float* delta;  // [B][out]
   const float* acts;   // [B][in]
-  float* dW;           // grad + w_off  ([out][in])
-  float* db;           // grad + b_off or nullptr
-  float* dW2;          // second target: delta rows >= split_row
-  float* db2;
+  float* slab;         // [splits][out][in] per-split partials (plain stores
+                       // — atomics measured 8.4M adds/launch and dominated)
+  float* db_slab;      // [splits][out]
   int64_t B;
   int out_dim, in_dim;
   int nt;       // 32-col tiles per wave (<= MAX_NT)
   int splits;   // K splits
-  int split_row;  // -1: single target; else rows >= split go to dW2/db2
 };
 
 template <int NT>
@@ -243,34 +242,50 @@ __global__ void dw_mfma_kernel(DwArgs a) {
     }
   }
 
-  // ---- scatter accumulators (optional row split for combined heads) ----
+  // ---- store accumulators to this split's slab (each (row,col) of a
+  // split is owned by exactly one block: no atomics, no zero-init) ----
+  float* slab = a.slab + (int64_t)split * a.out_dim * a.in_dim;
   #pragma unroll
   for (int t = 0; t < NT; ++t) {
     if (col_ok[t]) {
       #pragma unroll
       for (int r = 0; r < 16; ++r) {
         const int mrow = m0 + cd_row(r, lane);
-        if (mrow < a.out_dim && acc[t][r] != 0.f) {
-          if (a.split_row >= 0 && mrow >= a.split_row) {
-            atomicAdd(&a.dW2[(int64_t)(mrow - a.split_row) * a.in_dim + bcol[t]],
-                      acc[t][r]);
-          } else {
-            atomicAdd(&a.dW[(int64_t)mrow * a.in_dim + bcol[t]], acc[t][r]);
-          }
-        }
+        if (mrow < a.out_dim)
+          slab[(int64_t)mrow * a.in_dim + bcol[t]] = acc[t][r];
       }
     }
   }
-  // db: lane's dbacc covers rows k with k_l parity; combine lane and
-  // lane+32 (same m-column), then one atomic per m from the low half.
-  if (a.db != nullptr && blockIdx.y == 0) {
+  if (blockIdx.y == 0) {
     const float other = __shfl(dbacc, lane ^ 32, WAVE);
-    if (lane < 32 && m_ok) {
-      if (a.split_row >= 0 && mcol >= a.split_row)
-        atomicAdd(&a.db2[mcol - a.split_row], dbacc + other);
-      else
-        atomicAdd(&a.db[mcol], dbacc + other);
-    }
+    if (lane < 32 && m_ok)
+      a.db_slab[(int64_t)split * a.out_dim + mcol] = dbacc + other;
+  }
+}
+
+// Split-slab reduction into the flat grad, handling the optional
+// combined-heads row split (rows < split_row -> dW/db, rest -> dW2/db2).
+__global__ void dw_reduce_kernel(const float* __restrict__ slab,
+                                 const float* __restrict__ db_slab,
+                                 float* __restrict__ dW, float* __restrict__ db,
+                                 float* __restrict__ dW2,
+                                 float* __restrict__ db2, int64_t out_dim,
+                                 int64_t in_dim, int splits, int split_row) {
+  const int64_t n = out_dim * in_dim;
+  for (int64_t i = gidx(); i < n; i += gstride()) {
+    float acc = 0.f;
+    for (int s = 0; s < splits; ++s) acc += slab[(int64_t)s * n + i];
+    const int64_t mrow = i / in_dim;
+    if (split_row >= 0 && mrow >= split_row)
+      dW2[(mrow - split_row) * in_dim + i % in_dim] += acc;
+    else
+      dW[i] += acc;
+  }
+  for (int64_t m = gidx(); m < out_dim; m += gstride()) {
+    float acc = 0.f;
+    for (int s = 0; s < splits; ++s) acc += db_slab[(int64_t)s * out_dim + m];
+    if (split_row >= 0 && m >= split_row) db2[m - split_row] += acc;
+    else if (db != nullptr) db[m] += acc;
   }
 }
 
@@ -388,13 +403,6 @@ void dw_mfma(torch::Tensor delta, torch::Tensor acts, torch::Tensor grad_buf,
   DwArgs a{};
   a.delta = delta.data_ptr<float>();
   a.acts = acts.data_ptr<float>();
-  a.dW = grad_buf.data_ptr<float>() + w_off;
-  a.db = (b_off >= 0) ? grad_buf.data_ptr<float>() + b_off : nullptr;
-  a.split_row = static_cast<int>(split_row);
-  a.dW2 = (split_row >= 0) ? grad_buf.data_ptr<float>() + w_off2 : nullptr;
-  a.db2 = (split_row >= 0 && b_off2 >= 0)
-              ? grad_buf.data_ptr<float>() + b_off2
-              : nullptr;
   a.B = B;
   a.out_dim = out_dim;
   a.in_dim = in_dim;
@@ -407,6 +415,13 @@ void dw_mfma(torch::Tensor delta, torch::Tensor acts, torch::Tensor grad_buf,
   a.splits = std::max(1, target_blocks / std::max(1, m_tiles * n_tiles));
   a.splits = static_cast<int>(
       std::min<int64_t>(a.splits, std::max<int64_t>(1, B / 256)));
+
+  auto slab = torch::empty({(int64_t)a.splits, (int64_t)out_dim, (int64_t)in_dim},
+                           delta.options());
+  auto db_slab = torch::zeros({(int64_t)a.splits, (int64_t)out_dim},
+                              delta.options());
+  a.slab = slab.data_ptr<float>();
+  a.db_slab = db_slab.data_ptr<float>();
 
   hipStream_t stream = c10::hip::getCurrentHIPStream().stream();
   const dim3 grid(m_tiles, n_tiles, a.splits);
@@ -424,6 +439,16 @@ void dw_mfma(torch::Tensor delta, torch::Tensor acts, torch::Tensor grad_buf,
       hipLaunchKernelGGL(dw_mfma_kernel<4>, grid, dim3(WAVE), 0, stream, a);
       break;
   }
+  float* dW = grad_buf.data_ptr<float>() + w_off;
+  float* db = (b_off >= 0) ? grad_buf.data_ptr<float>() + b_off : nullptr;
+  float* dW2 = (split_row >= 0) ? grad_buf.data_ptr<float>() + w_off2 : nullptr;
+  float* db2 = (split_row >= 0 && b_off2 >= 0)
+                   ? grad_buf.data_ptr<float>() + b_off2
+                   : nullptr;
+  hipLaunchKernelGGL(dw_reduce_kernel,
+                     dim3(elementwise_grid((int64_t)out_dim * in_dim, 256)),
+                     dim3(256), 0, stream, a.slab, a.db_slab, dW, db, dW2, db2,
+                     out_dim, in_dim, a.splits, static_cast<int>(split_row));
 }
 
 void dwv(torch::Tensor gv, torch::Tensor acts, torch::Tensor grad_buf,
