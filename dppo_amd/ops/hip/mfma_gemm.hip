@@ -87,17 +87,47 @@ __global__ void gemm_fwd_kernel(FwdArgs a) {
 
     for (int kb = 0; kb < a.K; kb += BK) {
       __syncthreads();
-      for (int idx = threadIdx.x; idx < FWD_M * BK; idx += FWD_WAVES * 64) {
-        const int r = idx / BK, c = idx % BK;
+      // float4 staging (scalar element staging measured as the dominant
+      // per-call cost: 32 scalar dword loads per thread per stage)
+      constexpr int BK4 = BK / 4;
+      for (int idx = threadIdx.x; idx < FWD_M * BK4; idx += FWD_WAVES * 64) {
+        const int r = idx / BK4, c4 = (idx % BK4) * 4;
         const int64_t row = b0 + r;
-        const int col = kb + c;
-        xs[r][c] = (row < a.B && col < a.K) ? a.X[row * a.K + col] : 0.f;
+        const int col = kb + c4;
+        float4 val = make_float4(0.f, 0.f, 0.f, 0.f);
+        if (row < a.B) {
+          if (col + 3 < a.K) {
+            val = *reinterpret_cast<const float4*>(&a.X[row * a.K + col]);
+          } else {
+            float tmp[4] = {0.f, 0.f, 0.f, 0.f};
+            for (int q = 0; q < 4; ++q)
+              if (col + q < a.K) tmp[q] = a.X[row * a.K + col + q];
+            val = make_float4(tmp[0], tmp[1], tmp[2], tmp[3]);
+          }
+        }
+        xs[r][c4] = val.x;
+        xs[r][c4 + 1] = val.y;
+        xs[r][c4 + 2] = val.z;
+        xs[r][c4 + 3] = val.w;
       }
-      for (int idx = threadIdx.x; idx < BK * NW; idx += FWD_WAVES * 64) {
-        const int r = idx / NW, c = idx % NW;
+      constexpr int NW4_MAX = MAX_NT * M_WAVE / 4;
+      const int NW4 = NW / 4;
+      for (int idx = threadIdx.x; idx < BK * NW4; idx += FWD_WAVES * 64) {
+        const int r = idx / NW4, c4 = (idx % NW4) * 4;
         const int krow = kb + r;
-        ws[r][c] = (krow < a.K && c < a.N) ? a.Wt[(int64_t)krow * a.N + c]
-                                           : 0.f;
+        float4 val = make_float4(0.f, 0.f, 0.f, 0.f);
+        if (krow < a.K) {
+          if (c4 + 3 < a.N) {
+            val = *reinterpret_cast<const float4*>(
+                &a.Wt[(int64_t)krow * a.N + c4]);
+          } else {
+            float tmp[4] = {0.f, 0.f, 0.f, 0.f};
+            for (int q = 0; q < 4; ++q)
+              if (c4 + q < a.N) tmp[q] = a.Wt[(int64_t)krow * a.N + c4 + q];
+            val = make_float4(tmp[0], tmp[1], tmp[2], tmp[3]);
+          }
+        }
+        *reinterpret_cast<float4*>(&ws[r][c4]) = val;
       }
       __syncthreads();
 
@@ -265,6 +295,10 @@ __global__ void dw_mfma_kernel(DwArgs a) {
 
 // Split-slab reduction into the flat grad, handling the optional
 // combined-heads row split (rows < split_row -> dW/db, rest -> dW2/db2).
+// Parallel over (element, split-chunk): a serial full-splits loop per
+// element left the chip at <1 wave/SIMD and 480 us per call.
+constexpr int DW_RED_CHUNK = 32;
+
 __global__ void dw_reduce_kernel(const float* __restrict__ slab,
                                  const float* __restrict__ db_slab,
                                  float* __restrict__ dW, float* __restrict__ db,
@@ -272,21 +306,32 @@ __global__ void dw_reduce_kernel(const float* __restrict__ slab,
                                  float* __restrict__ db2, int64_t out_dim,
                                  int64_t in_dim, int splits, int split_row) {
   const int64_t n = out_dim * in_dim;
-  for (int64_t i = gidx(); i < n; i += gstride()) {
+  const int s0 = blockIdx.y * DW_RED_CHUNK;
+  const int s1 = min(splits, s0 + DW_RED_CHUNK);
+  const bool first = (blockIdx.y == 0);
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x) {
     float acc = 0.f;
-    for (int s = 0; s < splits; ++s) acc += slab[(int64_t)s * n + i];
+    for (int s = s0; s < s1; ++s) acc += slab[(int64_t)s * n + i];
     const int64_t mrow = i / in_dim;
-    if (split_row >= 0 && mrow >= split_row)
-      dW2[(mrow - split_row) * in_dim + i % in_dim] += acc;
-    else
-      dW[i] += acc;
+    float* dst = (split_row >= 0 && mrow >= split_row)
+                     ? &dW2[(mrow - split_row) * in_dim + i % in_dim]
+                     : &dW[i];
+    atomicAdd(dst, acc);
   }
-  for (int64_t m = gidx(); m < out_dim; m += gstride()) {
-    float acc = 0.f;
-    for (int s = 0; s < splits; ++s) acc += db_slab[(int64_t)s * out_dim + m];
-    if (split_row >= 0 && m >= split_row) db2[m - split_row] += acc;
-    else if (db != nullptr) db[m] += acc;
-  }
+  if (first) return;  // db handled below by chunked atomics too
+}
+
+__global__ void db_reduce_kernel(const float* __restrict__ db_slab,
+                                 float* __restrict__ db,
+                                 float* __restrict__ db2, int64_t out_dim,
+                                 int splits, int split_row) {
+  const int64_t m = gidx();
+  if (m >= out_dim) return;
+  float acc = 0.f;
+  for (int s = 0; s < splits; ++s) acc += db_slab[(int64_t)s * out_dim + m];
+  if (split_row >= 0 && m >= split_row) db2[m - split_row] += acc;
+  else if (db != nullptr) db[m] += acc;
 }
 
 // ---------------------------------------------------------------------------
@@ -410,8 +455,9 @@ void dw_mfma(torch::Tensor delta, torch::Tensor acts, torch::Tensor grad_buf,
   a.nt = std::min(MAX_NT, (in_dim + M_WAVE - 1) / M_WAVE);
   const int m_tiles = (out_dim + M_WAVE - 1) / M_WAVE;
   const int n_tiles = (in_dim + a.nt * M_WAVE - 1) / (a.nt * M_WAVE);
-  // enough waves to hide the streamed-operand latency (~2k waves)
-  const int target_blocks = 2048;
+  // enough waves to hide the streamed-operand latency; slab stores make
+  // extra splits nearly free (the reduce is split-chunk parallel)
+  const int target_blocks = 4096;
   a.splits = std::max(1, target_blocks / std::max(1, m_tiles * n_tiles));
   a.splits = static_cast<int>(
       std::min<int64_t>(a.splits, std::max<int64_t>(1, B / 256)));
@@ -445,10 +491,18 @@ void dw_mfma(torch::Tensor delta, torch::Tensor acts, torch::Tensor grad_buf,
   float* db2 = (split_row >= 0 && b_off2 >= 0)
                    ? grad_buf.data_ptr<float>() + b_off2
                    : nullptr;
-  hipLaunchKernelGGL(dw_reduce_kernel,
-                     dim3(elementwise_grid((int64_t)out_dim * in_dim, 256)),
-                     dim3(256), 0, stream, a.slab, a.db_slab, dW, db, dW2, db2,
-                     out_dim, in_dim, a.splits, static_cast<int>(split_row));
+  {
+    const int n_chunks = (a.splits + DW_RED_CHUNK - 1) / DW_RED_CHUNK;
+    const dim3 rgrid(elementwise_grid((int64_t)out_dim * in_dim, 256),
+                     n_chunks);
+    hipLaunchKernelGGL(dw_reduce_kernel, rgrid, dim3(256), 0, stream, a.slab,
+                       a.db_slab, dW, db, dW2, db2, out_dim, in_dim, a.splits,
+                       static_cast<int>(split_row));
+    hipLaunchKernelGGL(db_reduce_kernel,
+                       dim3((out_dim + 255) / 256), dim3(256), 0, stream,
+                       a.db_slab, db, db2, out_dim, a.splits,
+                       static_cast<int>(split_row));
+  }
 }
 
 void dwv(torch::Tensor gv, torch::Tensor acts, torch::Tensor grad_buf,
