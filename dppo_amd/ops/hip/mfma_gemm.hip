@@ -48,7 +48,7 @@ DEV_INLINE int cd_row(int reg, int lane) {
 
 struct FwdArgs {
   const float* X;     // [B][K]
-  const float* Wt;    // [K][N]
+  const float* Wt;    // [K][N] (wt_layout 0) or [N][K] torch layout (1)
   const float* bias;  // [N]
   const float* aux;   // [B][N] forward activations (dgrad modes)
   float* C;           // [B][N]      (heads: pdflat [B][N-1])
@@ -59,6 +59,7 @@ struct FwdArgs {
                       // 3 tanh-grad: C = (1-aux^2)*acc (no bias)
                       // 4 relu-grad: C = (aux>0)*acc   (no bias)
   int heads;          // if 1: last column -> v, rest -> C (pdflat)
+  int wt_layout;      // 0: Wt[K][N]; 1: torch W[N][K] (staged transposed)
 };
 
 template <int NT>
@@ -110,24 +111,50 @@ __global__ void gemm_fwd_kernel(FwdArgs a) {
         xs[r][c4 + 2] = val.z;
         xs[r][c4 + 3] = val.w;
       }
-      constexpr int NW4_MAX = MAX_NT * M_WAVE / 4;
-      const int NW4 = NW / 4;
-      for (int idx = threadIdx.x; idx < BK * NW4; idx += FWD_WAVES * 64) {
-        const int r = idx / NW4, c4 = (idx % NW4) * 4;
-        const int krow = kb + r;
-        float4 val = make_float4(0.f, 0.f, 0.f, 0.f);
-        if (krow < a.K) {
-          if (c4 + 3 < a.N) {
-            val = *reinterpret_cast<const float4*>(
-                &a.Wt[(int64_t)krow * a.N + c4]);
-          } else {
-            float tmp[4] = {0.f, 0.f, 0.f, 0.f};
-            for (int q = 0; q < 4; ++q)
-              if (c4 + q < a.N) tmp[q] = a.Wt[(int64_t)krow * a.N + c4 + q];
-            val = make_float4(tmp[0], tmp[1], tmp[2], tmp[3]);
+      if (a.wt_layout == 0) {
+        const int NW4 = NW / 4;
+        for (int idx = threadIdx.x; idx < BK * NW4; idx += FWD_WAVES * 64) {
+          const int r = idx / NW4, c4 = (idx % NW4) * 4;
+          const int krow = kb + r;
+          float4 val = make_float4(0.f, 0.f, 0.f, 0.f);
+          if (krow < a.K) {
+            if (c4 + 3 < a.N) {
+              val = *reinterpret_cast<const float4*>(
+                  &a.Wt[(int64_t)krow * a.N + c4]);
+            } else {
+              float tmp[4] = {0.f, 0.f, 0.f, 0.f};
+              for (int q = 0; q < 4; ++q)
+                if (c4 + q < a.N) tmp[q] = a.Wt[(int64_t)krow * a.N + c4 + q];
+              val = make_float4(tmp[0], tmp[1], tmp[2], tmp[3]);
+            }
           }
+          *reinterpret_cast<float4*>(&ws[r][c4]) = val;
         }
-        *reinterpret_cast<float4*>(&ws[r][c4]) = val;
+      } else {
+        // torch W[N][K]: c-major mapping keeps the row-segment reads
+        // coalesced and transposes into ws on the fly — no host-side
+        // W.t().contiguous() per update step
+        constexpr int BK4 = BK / 4;
+        for (int idx = threadIdx.x; idx < NW * BK4; idx += FWD_WAVES * 64) {
+          const int c = idx / BK4, r4 = (idx % BK4) * 4;
+          const int krow = kb + r4;
+          float4 val = make_float4(0.f, 0.f, 0.f, 0.f);
+          if (c < a.N) {
+            if (krow + 3 < a.K) {
+              val = *reinterpret_cast<const float4*>(
+                  &a.Wt[(int64_t)c * a.K + krow]);
+            } else {
+              float tmp[4] = {0.f, 0.f, 0.f, 0.f};
+              for (int q = 0; q < 4; ++q)
+                if (krow + q < a.K) tmp[q] = a.Wt[(int64_t)c * a.K + krow + q];
+              val = make_float4(tmp[0], tmp[1], tmp[2], tmp[3]);
+            }
+          }
+          ws[r4][c] = val.x;
+          ws[r4 + 1][c] = val.y;
+          ws[r4 + 2][c] = val.z;
+          ws[r4 + 3][c] = val.w;
+        }
       }
       __syncthreads();
 
@@ -326,12 +353,18 @@ __global__ void db_reduce_kernel(const float* __restrict__ db_slab,
                                  float* __restrict__ db,
                                  float* __restrict__ db2, int64_t out_dim,
                                  int splits, int split_row) {
-  const int64_t m = gidx();
+  // one wave per output column; lanes stride the splits axis
+  const int m = blockIdx.x;
   if (m >= out_dim) return;
+  const int lane = threadIdx.x;
   float acc = 0.f;
-  for (int s = 0; s < splits; ++s) acc += db_slab[(int64_t)s * out_dim + m];
-  if (split_row >= 0 && m >= split_row) db2[m - split_row] += acc;
-  else if (db != nullptr) db[m] += acc;
+  for (int s = lane; s < splits; s += WAVE)
+    acc += db_slab[(int64_t)s * out_dim + m];
+  acc = wave_reduce_sum(acc);
+  if (lane == 0) {
+    if (split_row >= 0 && m >= split_row) db2[m - split_row] += acc;
+    else if (db != nullptr) db[m] += acc;
+  }
 }
 
 // ---------------------------------------------------------------------------
@@ -380,15 +413,15 @@ __global__ void dwv_kernel(const float* __restrict__ gv,    // [B]
 
 void gemm_fwd(torch::Tensor X, torch::Tensor Wt, torch::Tensor bias,
               int64_t activation, int64_t heads, torch::Tensor C,
-              torch::Tensor v, torch::Tensor aux) {
+              torch::Tensor v, torch::Tensor aux, int64_t wt_layout) {
   // C and (for heads) v are caller-allocated so activations can land
   // directly in the backward's blob layout.
   const int64_t B = X.size(0);
   const int K = static_cast<int>(X.size(1));
-  const int N = static_cast<int>(Wt.size(1));
+  const int N = static_cast<int>(wt_layout ? Wt.size(0) : Wt.size(1));
   TORCH_CHECK(X.is_cuda() && X.is_contiguous() && Wt.is_contiguous());
   TORCH_CHECK(C.is_contiguous());
-  TORCH_CHECK(Wt.size(0) == K);
+  TORCH_CHECK((wt_layout ? Wt.size(1) : Wt.size(0)) == K);
   TORCH_CHECK(activation >= 3 || bias.numel() == N);
   TORCH_CHECK(N <= MAX_NT * M_WAVE, "N exceeds MFMA fwd tile budget");
   if (heads) {
@@ -406,6 +439,7 @@ void gemm_fwd(torch::Tensor X, torch::Tensor Wt, torch::Tensor bias,
   a.N = N;
   a.activation = static_cast<int>(activation);
   a.heads = static_cast<int>(heads);
+  a.wt_layout = static_cast<int>(wt_layout);
   a.C = C.data_ptr<float>();
   a.v = heads ? v.data_ptr<float>() : nullptr;
   if (activation >= 3) {
@@ -498,9 +532,8 @@ void dw_mfma(torch::Tensor delta, torch::Tensor acts, torch::Tensor grad_buf,
     hipLaunchKernelGGL(dw_reduce_kernel, rgrid, dim3(256), 0, stream, a.slab,
                        a.db_slab, dW, db, dW2, db2, out_dim, in_dim, a.splits,
                        static_cast<int>(split_row));
-    hipLaunchKernelGGL(db_reduce_kernel,
-                       dim3((out_dim + 255) / 256), dim3(256), 0, stream,
-                       a.db_slab, db, db2, out_dim, a.splits,
+    hipLaunchKernelGGL(db_reduce_kernel, dim3(out_dim), dim3(WAVE), 0,
+                       stream, a.db_slab, db, db2, out_dim, a.splits,
                        static_cast<int>(split_row));
   }
 }

@@ -487,11 +487,12 @@ class DPPOEngine:
         n_hidden = len(c.HIDDEN_SIZES)
         act_code = 1 if c.ACTIVATION == "tanh" else 0
         with torch.no_grad():
-            Wts = [lay.weight.t().contiguous() for lay in self.pi.hidden]
+            # torch weight layouts pass straight through: the GEMM stages
+            # [N][K] weights transposed into LDS itself (wt_layout=1)
             bs = [lay.bias.detach() for lay in self.pi.hidden]
-            Wth = torch.cat(
-                [self.pi.pi.weight.t(), self.pi.vf.weight.t()], dim=1
-            ).contiguous()
+            Wh_cat = torch.cat(
+                [self.pi.pi.weight, self.pi.vf.weight], dim=0
+            ).contiguous()  # [P+1][HL]: heads fwd (layout 1) AND dgrad Wt
             bh = torch.cat([self.pi.pi.bias, self.pi.vf.bias]).contiguous()
         total = sum(B * dims[l + 1] for l in range(n_hidden))
         acts = torch.empty(total, device=states.device, dtype=states.dtype)
@@ -500,14 +501,16 @@ class DPPOEngine:
         for l in range(n_hidden):
             n = B * dims[l + 1]
             cview = acts.narrow(0, o, n).view(B, dims[l + 1])
-            ext.gemm_fwd(x, Wts[l], bs[l], act_code, 0, cview, cview, cview)
+            ext.gemm_fwd(x, self.pi.hidden[l].weight.detach(), bs[l],
+                         act_code, 0, cview, cview, cview, 1)
             x = cview
             a_views.append(cview)
             o += n
         P = 2 * self.act_space.shape[0]
         pdflat = torch.empty(B, P, device=states.device, dtype=states.dtype)
         v = torch.empty(B, device=states.device, dtype=states.dtype)
-        ext.gemm_fwd(x, Wth, bh, 2, 1, pdflat, v, pdflat)
+        ext.gemm_fwd(x, Wh_cat, bh, 2, 1, pdflat, v, pdflat, 1)
+        self._Wh_cat = Wh_cat
         return acts, a_views, v, pdflat
 
     def _fused_backward(self, states, acts, a_views, v, pdflat,
@@ -531,17 +534,19 @@ class DPPOEngine:
             pdflat, oldflat, v, oldv, actions, adv, etr,
             clip, c.ENTCOEFF, c.VCOEFF,
         )
-        with torch.no_grad():
-            # [Wp; Wv] rows == the Wt layout the dgrad GEMM needs
-            Wh_cat = torch.cat(
-                [self.pi.pi.weight, self.pi.vf.weight], dim=0
-            ).contiguous()
+        # _fused_forward cached [Wp; Wv] — its rows are the dgrad Wt
+        Wh_cat = getattr(self, "_Wh_cat", None)
+        if Wh_cat is None:
+            with torch.no_grad():
+                Wh_cat = torch.cat(
+                    [self.pi.pi.weight, self.pi.vf.weight], dim=0
+                ).contiguous()
         dz = [None] * n_hidden
         delta, Wt_chain = gh, Wh_cat
         for l in range(n_hidden - 1, -1, -1):
             dz_l = torch.empty_like(a_views[l])
             ext.gemm_fwd(delta, Wt_chain, dummy_bias, dgrad_code, 0,
-                         dz_l, dz_l, a_views[l])
+                         dz_l, dz_l, a_views[l], 0)
             dz[l] = dz_l
             delta = dz_l
             Wt_chain = self.pi.hidden[l].weight.detach()
