@@ -201,7 +201,7 @@ __global__ void rollout_kernel(RolloutArgs a) {
       const bool is_v = heads && (u == out_dim - 1);
       const float* Wrow =
           is_v ? (a.params + off_Wv_u) : (W + (int64_t)u * in_dim);
-      #pragma unroll 2
+      #pragma unroll 4
       for (int k = k0; k + 4 <= k1; k += 4) {
         const float4 w4 = *reinterpret_cast<const float4*>(Wrow + k);
         #pragma unroll
@@ -299,20 +299,34 @@ __global__ void rollout_kernel(RolloutArgs a) {
     if (tid < nE) out_values[(int64_t)step * E + e0 + tid] = val_lds[tid];
     if (tid < ENV_TILE) racc_lds[tid] = 0.f;
 
-    // ---- env low-rank projection: wave wv handles env wv ----
-    if (wv < nE && lane < a.rank) {
+    // ---- env low-rank projection: wave wv handles env wv; lanes split
+    // as (rr, k-quarter) so all 64 lanes stream and the per-lane load
+    // chain is 4x shorter (the 16-lane serial version stalled the whole
+    // block at the next barrier) ----
+    if (wv < nE) {
       const int e = wv;
-      const int rr = lane;
-      const float* Vrow = env_Vt + (int64_t)rr * D;
+      const int rr = lane & 15;
+      const int kq = lane >> 4;  // 4 k-quarters
+      const int kq_len = ((D / 4 + 3) & ~3);
+      const int k0q = kq * kq_len;
+      const int k1q = min(D, k0q + kq_len);
       float accv = 0.f;
-      for (int k = 0; k + 4 <= D; k += 4) {
-        const float4 v4 = *reinterpret_cast<const float4*>(Vrow + k);
-        const float4 x4 =
-            *reinterpret_cast<const float4*>(&lds[X_OFF + e * MAX_D + k]);
-        accv += v4.x * x4.x + v4.y * x4.y + v4.z * x4.z + v4.w * x4.w;
+      if (rr < a.rank) {
+        const float* Vrow = env_Vt + (int64_t)rr * D;
+        int k = k0q;
+        #pragma unroll 2
+        for (; k + 4 <= k1q; k += 4) {
+          const float4 v4 = *reinterpret_cast<const float4*>(Vrow + k);
+          const float4 x4 =
+              *reinterpret_cast<const float4*>(&lds[X_OFF + e * MAX_D + k]);
+          accv += v4.x * x4.x + v4.y * x4.y + v4.z * x4.z + v4.w * x4.w;
+        }
+        for (; k < k1q; ++k) accv += Vrow[k] * lds[X_OFF + e * MAX_D + k];
       }
-      for (int k = D & ~3; k < D; ++k) accv += Vrow[k] * lds[X_OFF + e * MAX_D + k];
-      lds[XV_OFF + e * MAX_R + rr] = accv;
+      // butterfly-reduce over the k-quarter lanes (bits 4 and 5)
+      accv += __shfl_xor(accv, 16, WAVE);
+      accv += __shfl_xor(accv, 32, WAVE);
+      if (kq == 0 && rr < a.rank) lds[XV_OFF + e * MAX_R + rr] = accv;
     }
     __syncthreads();
 
