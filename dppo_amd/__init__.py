@@ -26,7 +26,7 @@ __version__ = "0.1.0"
 
 from .config import DPPOConfig, REFERENCE_DEFAULTS
 from . import spaces
-from .distributions import make_pdtype
+from .distributions import make_pdtype, validate_probtype
 from .models.mlp import Model, PolicyValueMLP
 
 __all__ = [
@@ -34,6 +34,8 @@ __all__ = [
     "REFERENCE_DEFAULTS",
     "spaces",
     "make_pdtype",
+    "validate_probtype",
+    "PPO",
     "Model",
     "PolicyValueMLP",
     "Chief",
@@ -51,4 +53,7 @@ def __getattr__(name):
     if name == "Worker":
         from .worker import Worker
         return Worker
+    if name == "PPO":
+        from .ppo import PPO
+        return PPO
     raise AttributeError(f"module 'dppo_amd' has no attribute {name!r}")

@@ -332,6 +332,43 @@ class BernoulliPdType(PdType):
 # ---------------------------------------------------------------------------
 
 
+def validate_probtype(pdtype: PdType, pdparam, n: int = 100_000,
+                      atol_scale: float = 3.0) -> None:
+    """Monte-Carlo self-check of a distribution family (the reference
+    embeds this in its distributions module, distributions.py:269-295):
+    (a) E[-logp(x)] == entropy and (b) KL(p||q) == -H(p) - E_p[log q],
+    both within `atol_scale` standard errors.  Raises AssertionError on
+    violation.  Also used by tests/test_distributions.py."""
+    import math as _math
+
+    pdparam = torch.as_tensor(pdparam, dtype=torch.float32)
+    M = pdparam.unsqueeze(0).repeat(n, 1)
+    pd = pdtype.pdfromflat(M)
+    x = pd.sample()
+    calc_logp = pd.logp(x)
+    ent = pd.entropy().mean().item()
+    mean_neglogp = -calc_logp.mean().item()
+    stderr = calc_logp.std().item() / _math.sqrt(n)
+    assert abs(ent - mean_neglogp) < atol_scale * stderr, (
+        ent, mean_neglogp, stderr)
+
+    pdparam2 = pdparam + torch.randn_like(pdparam) * 0.1
+    q = pdtype.pdfromflat(pdparam2.unsqueeze(0).repeat(n, 1))
+    kl_analytic = pd.kl(q).mean().item()
+    logq = q.logp(x)
+    kl_mc = (-ent - logq.mean()).item()
+    stderr2 = logq.std().item() / _math.sqrt(n)
+    assert abs(kl_analytic - kl_mc) < atol_scale * stderr2, (kl_analytic, kl_mc)
+
+
+def test_probtypes() -> None:
+    """Run the self-check over every family (reference distributions.py:252-266)."""
+    validate_probtype(DiagGaussianPdType(3), [-0.2, 0.3, 0.4, -0.5, 0.1, -0.5])
+    validate_probtype(CategoricalPdType(4), [-0.2, 0.3, 0.5, 0.1])
+    validate_probtype(MultiCategoricalPdType([3, 2]), [-0.1, 0.4, 0.2, 0.3, -0.2])
+    validate_probtype(BernoulliPdType(3), [-0.4, 0.2, 0.6])
+
+
 def make_pdtype(ac_space) -> PdType:
     """Space -> PdType dispatch (reference distributions.py:231-243)."""
     if isinstance(ac_space, spaces.Box):

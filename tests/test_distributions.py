@@ -20,32 +20,7 @@ from dppo_amd.distributions import (
     make_pdtype,
 )
 
-N = 100_000
-
-
-def validate_probtype(pdtype, pdparam, atol_scale=3.0):
-    """Monte-Carlo identities within 3 standard errors
-    (reference distributions.py:269-295)."""
-    pdparam = torch.as_tensor(pdparam, dtype=torch.float32)
-    M = pdparam.unsqueeze(0).repeat(N, 1)
-    pd = pdtype.pdfromflat(M)
-    x = pd.sample()
-    # identity 1: E[-logp(x)] == entropy
-    calc_logp = pd.logp(x)
-    ent = pd.entropy().mean().item()
-    mean_neglogp = -calc_logp.mean().item()
-    stderr = calc_logp.std().item() / math.sqrt(N)
-    assert abs(ent - mean_neglogp) < atol_scale * stderr, (ent, mean_neglogp, stderr)
-
-    # identity 2: KL(p||q) == -H(p) - E_p[log q]
-    pdparam2 = pdparam + torch.randn_like(pdparam) * 0.1
-    M2 = pdparam2.unsqueeze(0).repeat(N, 1)
-    q = pdtype.pdfromflat(M2)
-    kl_analytic = pd.kl(q).mean().item()
-    logq = q.logp(x)
-    kl_mc = (-ent - logq.mean()).item()
-    stderr2 = logq.std().item() / math.sqrt(N)
-    assert abs(kl_analytic - kl_mc) < atol_scale * stderr2, (kl_analytic, kl_mc)
+from dppo_amd.distributions import validate_probtype
 
 
 def test_validate_diag_gaussian():
