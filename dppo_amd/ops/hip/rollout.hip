@@ -201,7 +201,7 @@ __global__ void rollout_kernel(RolloutArgs a) {
       const bool is_v = heads && (u == out_dim - 1);
       const float* Wrow =
           is_v ? (a.params + off_Wv_u) : (W + (int64_t)u * in_dim);
-      #pragma unroll 4
+      #pragma unroll 2
       for (int k = k0; k + 4 <= k1; k += 4) {
         const float4 w4 = *reinterpret_cast<const float4*>(Wrow + k);
         #pragma unroll
