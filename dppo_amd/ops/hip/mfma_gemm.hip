@@ -68,15 +68,21 @@ __global__ void gemm_fwd_kernel(FwdArgs a) {
   const int lane = threadIdx.x & (WAVE - 1);
   const int wave = threadIdx.x / WAVE;
 
-  // X tile [FWD_M][BK] (+1 pad: the A-fragment read is a column read);
-  // Wt tile [BK][NT*32] staged coalesced so the MFMA B-operand comes
-  // from LDS instead of a fresh L2 round trip per k-step.
+  // Single-buffered LDS with a T14 (issue-early / write-late) register
+  // pipeline: the NEXT K-tile's global loads are issued before this
+  // tile's MFMA phase, and written to LDS after the barrier — HBM
+  // latency hides under the matrix pipe with no extra LDS.
   __shared__ float xs[FWD_M][BK + 1];
   __shared__ float ws[BK][NT * M_WAVE];
 
   const int i_l = lane & 31;   // A row within wave tile
   const int k_l = lane >> 5;   // A k within pair
   const int NW = NT * M_WAVE;
+
+  constexpr int XREG = FWD_M * BK / 4 / (FWD_WAVES * 64);  // = 8, exact
+  constexpr int WREG = BK * NT * M_WAVE / 4 / (FWD_WAVES * 64);  // = 2*NT, exact
+  float4 xreg[XREG];
+  float4 wreg[WREG];
 
   for (int64_t tile = blockIdx.x; tile * FWD_M < a.B; tile += gridDim.x) {
     const int64_t b0 = tile * FWD_M;
@@ -86,12 +92,12 @@ __global__ void gemm_fwd_kernel(FwdArgs a) {
       #pragma unroll
       for (int r = 0; r < 16; ++r) acc[t][r] = 0.f;
 
-    for (int kb = 0; kb < a.K; kb += BK) {
-      __syncthreads();
-      // float4 staging (scalar element staging measured as the dominant
-      // per-call cost: 32 scalar dword loads per thread per stage)
+    // ---- staging helpers (register load / LDS write split) ----
+    auto load_x = [&](int kb) {
       constexpr int BK4 = BK / 4;
-      for (int idx = threadIdx.x; idx < FWD_M * BK4; idx += FWD_WAVES * 64) {
+      #pragma unroll
+      for (int i = 0; i < XREG; ++i) {
+        const int idx = threadIdx.x + i * FWD_WAVES * 64;
         const int r = idx / BK4, c4 = (idx % BK4) * 4;
         const int64_t row = b0 + r;
         const int col = kb + c4;
@@ -106,14 +112,27 @@ __global__ void gemm_fwd_kernel(FwdArgs a) {
             val = make_float4(tmp[0], tmp[1], tmp[2], tmp[3]);
           }
         }
-        xs[r][c4] = val.x;
-        xs[r][c4 + 1] = val.y;
-        xs[r][c4 + 2] = val.z;
-        xs[r][c4 + 3] = val.w;
+        xreg[i] = val;
       }
+    };
+    auto write_x = [&]() {
+      constexpr int BK4 = BK / 4;
+      #pragma unroll
+      for (int i = 0; i < XREG; ++i) {
+        const int idx = threadIdx.x + i * FWD_WAVES * 64;
+        const int r = idx / BK4, c4 = (idx % BK4) * 4;
+        xs[r][c4] = xreg[i].x;
+        xs[r][c4 + 1] = xreg[i].y;
+        xs[r][c4 + 2] = xreg[i].z;
+        xs[r][c4 + 3] = xreg[i].w;
+      }
+    };
+    auto load_w = [&](int kb) {
       if (a.wt_layout == 0) {
         const int NW4 = NW / 4;
-        for (int idx = threadIdx.x; idx < BK * NW4; idx += FWD_WAVES * 64) {
+        #pragma unroll
+        for (int i = 0; i < WREG; ++i) {
+          const int idx = threadIdx.x + i * FWD_WAVES * 64;
           const int r = idx / NW4, c4 = (idx % NW4) * 4;
           const int krow = kb + r;
           float4 val = make_float4(0.f, 0.f, 0.f, 0.f);
@@ -128,14 +147,13 @@ __global__ void gemm_fwd_kernel(FwdArgs a) {
               val = make_float4(tmp[0], tmp[1], tmp[2], tmp[3]);
             }
           }
-          *reinterpret_cast<float4*>(&ws[r][c4]) = val;
+          wreg[i] = val;
         }
       } else {
-        // torch W[N][K]: c-major mapping keeps the row-segment reads
-        // coalesced and transposes into ws on the fly — no host-side
-        // W.t().contiguous() per update step
         constexpr int BK4 = BK / 4;
-        for (int idx = threadIdx.x; idx < NW * BK4; idx += FWD_WAVES * 64) {
+        #pragma unroll
+        for (int i = 0; i < WREG; ++i) {
+          const int idx = threadIdx.x + i * FWD_WAVES * 64;
           const int c = idx / BK4, r4 = (idx % BK4) * 4;
           const int krow = kb + r4;
           float4 val = make_float4(0.f, 0.f, 0.f, 0.f);
@@ -150,13 +168,47 @@ __global__ void gemm_fwd_kernel(FwdArgs a) {
               val = make_float4(tmp[0], tmp[1], tmp[2], tmp[3]);
             }
           }
-          ws[r4][c] = val.x;
-          ws[r4 + 1][c] = val.y;
-          ws[r4 + 2][c] = val.z;
-          ws[r4 + 3][c] = val.w;
+          wreg[i] = val;
         }
       }
-      __syncthreads();
+    };
+    auto write_w = [&]() {
+      if (a.wt_layout == 0) {
+        const int NW4 = NW / 4;
+        #pragma unroll
+        for (int i = 0; i < WREG; ++i) {
+          const int idx = threadIdx.x + i * FWD_WAVES * 64;
+          const int r = idx / NW4, c4 = (idx % NW4) * 4;
+          *reinterpret_cast<float4*>(&ws[r][c4]) = wreg[i];
+        }
+      } else {
+        constexpr int BK4 = BK / 4;
+        #pragma unroll
+        for (int i = 0; i < WREG; ++i) {
+          const int idx = threadIdx.x + i * FWD_WAVES * 64;
+          const int c = idx / BK4, r4 = (idx % BK4) * 4;
+          ws[r4][c] = wreg[i].x;
+          ws[r4 + 1][c] = wreg[i].y;
+          ws[r4 + 2][c] = wreg[i].z;
+          ws[r4 + 3][c] = wreg[i].w;
+        }
+      }
+    };
+
+    // ---- prologue: first tile ----
+    load_x(0);
+    load_w(0);
+    write_x();
+    write_w();
+    __syncthreads();
+
+    for (int kb = 0; kb < a.K; kb += BK) {
+      const int kb_next = kb + BK;
+      const bool more = kb_next < a.K;
+      if (more) {   // issue next tile's loads before the MFMA phase (T14)
+        load_x(kb_next);
+        load_w(kb_next);
+      }
 
       const int ksteps = min(BK, a.K - kb);
       #pragma unroll 4
@@ -167,6 +219,12 @@ __global__ void gemm_fwd_kernel(FwdArgs a) {
           const float bv = ws[k2 + k_l][t * M_WAVE + i_l];
           acc[t] = __builtin_amdgcn_mfma_f32_32x32x2f32(av, bv, acc[t], 0, 0, 0);
         }
+      }
+      __syncthreads();
+      if (more) {
+        write_x();
+        write_w();
+        __syncthreads();
       }
     }
 
@@ -200,6 +258,7 @@ __global__ void gemm_fwd_kernel(FwdArgs a) {
         }
       }
     }
+    __syncthreads();  // xs/ws free for the next grid-stride tile
   }
 }
 
@@ -220,11 +279,11 @@ struct DwArgs {
   int splits;   // K splits
 };
 
-template <int NT>
+template <int NT, int MT>  // MT row fragments of 32 (1 or 2)
 __launch_bounds__(64)
 __global__ void dw_mfma_kernel(DwArgs a) {
   const int lane = threadIdx.x;
-  const int m0 = blockIdx.x * M_WAVE;
+  const int m0 = blockIdx.x * (M_WAVE * MT);
   const int n_base = blockIdx.y * a.nt * M_WAVE;
   const int split = blockIdx.z;
 
@@ -234,15 +293,24 @@ __global__ void dw_mfma_kernel(DwArgs a) {
 
   const int i_l = lane & 31;
   const int k_l = lane >> 5;
-  const int mcol = m0 + i_l;          // delta column this lane reads
-  const bool m_ok = mcol < a.out_dim;
-
-  f32x16 acc[NT];
+  int mcol[MT];
+  bool m_ok[MT];
   #pragma unroll
-  for (int t = 0; t < NT; ++t)
+  for (int m = 0; m < MT; ++m) {
+    mcol[m] = m0 + m * M_WAVE + i_l;
+    m_ok[m] = mcol[m] < a.out_dim;
+  }
+
+  f32x16 acc[MT][NT];
+  #pragma unroll
+  for (int m = 0; m < MT; ++m)
     #pragma unroll
-    for (int r = 0; r < 16; ++r) acc[t][r] = 0.f;
-  float dbacc = 0.f;
+    for (int t = 0; t < NT; ++t)
+      #pragma unroll
+      for (int r = 0; r < 16; ++r) acc[m][t][r] = 0.f;
+  float dbacc[MT];
+  #pragma unroll
+  for (int m = 0; m < MT; ++m) dbacc[m] = 0.f;
 
   // column guards hoisted out of the K loop (tail tiles load col 0 and
   // discard at scatter time — a harmless re-read keeps the loop branchless)
@@ -260,42 +328,61 @@ __global__ void dw_mfma_kernel(DwArgs a) {
   // hides under the matrix pipe instead of serializing with it.
   const int64_t kend = k0 + ((k1 - k0) & ~1);
   int64_t k = k0;
-  float av_c = 0.f;
+  float av_c[MT];
   float bv_c[NT];
   if (k < kend) {
-    av_c = m_ok ? a.delta[(k + k_l) * a.out_dim + mcol] : 0.f;
+    #pragma unroll
+    for (int m = 0; m < MT; ++m)
+      av_c[m] = m_ok[m] ? a.delta[(k + k_l) * a.out_dim + mcol[m]] : 0.f;
     #pragma unroll
     for (int t = 0; t < NT; ++t) bv_c[t] = a.acts[(k + k_l) * a.in_dim + bcol[t]];
   }
   #pragma unroll 1
   for (; k + 2 < kend; k += 2) {
-    const float av_n = m_ok ? a.delta[(k + 2 + k_l) * a.out_dim + mcol] : 0.f;
+    float av_n[MT];
+    #pragma unroll
+    for (int m = 0; m < MT; ++m)
+      av_n[m] = m_ok[m] ? a.delta[(k + 2 + k_l) * a.out_dim + mcol[m]] : 0.f;
     float bv_n[NT];
     #pragma unroll
     for (int t = 0; t < NT; ++t)
       bv_n[t] = a.acts[(k + 2 + k_l) * a.in_dim + bcol[t]];
-    dbacc += av_c;
     #pragma unroll
-    for (int t = 0; t < NT; ++t)
-      acc[t] = __builtin_amdgcn_mfma_f32_32x32x2f32(av_c, bv_c[t], acc[t], 0, 0, 0);
-    av_c = av_n;
+    for (int m = 0; m < MT; ++m) {
+      dbacc[m] += av_c[m];
+      #pragma unroll
+      for (int t = 0; t < NT; ++t)
+        acc[m][t] = __builtin_amdgcn_mfma_f32_32x32x2f32(av_c[m], bv_c[t],
+                                                         acc[m][t], 0, 0, 0);
+    }
+    #pragma unroll
+    for (int m = 0; m < MT; ++m) av_c[m] = av_n[m];
     #pragma unroll
     for (int t = 0; t < NT; ++t) bv_c[t] = bv_n[t];
   }
   if (k < kend) {
-    dbacc += av_c;
     #pragma unroll
-    for (int t = 0; t < NT; ++t)
-      acc[t] = __builtin_amdgcn_mfma_f32_32x32x2f32(av_c, bv_c[t], acc[t], 0, 0, 0);
+    for (int m = 0; m < MT; ++m) {
+      dbacc[m] += av_c[m];
+      #pragma unroll
+      for (int t = 0; t < NT; ++t)
+        acc[m][t] = __builtin_amdgcn_mfma_f32_32x32x2f32(av_c[m], bv_c[t],
+                                                         acc[m][t], 0, 0, 0);
+    }
     k += 2;
   }
   for (; k < k1; ++k) {
-    const float av = (m_ok && k_l == 0) ? a.delta[k * a.out_dim + mcol] : 0.f;
-    if (k_l == 0) dbacc += av;
     #pragma unroll
-    for (int t = 0; t < NT; ++t) {
-      const float bv = (k_l == 0) ? a.acts[k * a.in_dim + bcol[t]] : 0.f;
-      acc[t] = __builtin_amdgcn_mfma_f32_32x32x2f32(av, bv, acc[t], 0, 0, 0);
+    for (int m = 0; m < MT; ++m) {
+      const float av =
+          (m_ok[m] && k_l == 0) ? a.delta[k * a.out_dim + mcol[m]] : 0.f;
+      if (k_l == 0) dbacc[m] += av;
+      #pragma unroll
+      for (int t = 0; t < NT; ++t) {
+        const float bv = (k_l == 0) ? a.acts[k * a.in_dim + bcol[t]] : 0.f;
+        acc[m][t] =
+            __builtin_amdgcn_mfma_f32_32x32x2f32(av, bv, acc[m][t], 0, 0, 0);
+      }
     }
   }
 
@@ -303,20 +390,26 @@ __global__ void dw_mfma_kernel(DwArgs a) {
   // split is owned by exactly one block: no atomics, no zero-init) ----
   float* slab = a.slab + (int64_t)split * a.out_dim * a.in_dim;
   #pragma unroll
-  for (int t = 0; t < NT; ++t) {
-    if (col_ok[t]) {
-      #pragma unroll
-      for (int r = 0; r < 16; ++r) {
-        const int mrow = m0 + cd_row(r, lane);
-        if (mrow < a.out_dim)
-          slab[(int64_t)mrow * a.in_dim + bcol[t]] = acc[t][r];
+  for (int m = 0; m < MT; ++m) {
+    #pragma unroll
+    for (int t = 0; t < NT; ++t) {
+      if (col_ok[t]) {
+        #pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          const int mrow = m0 + m * M_WAVE + cd_row(r, lane);
+          if (mrow < a.out_dim)
+            slab[(int64_t)mrow * a.in_dim + bcol[t]] = acc[m][t][r];
+        }
       }
     }
   }
   if (blockIdx.y == 0) {
-    const float other = __shfl(dbacc, lane ^ 32, WAVE);
-    if (lane < 32 && m_ok)
-      a.db_slab[(int64_t)split * a.out_dim + mcol] = dbacc + other;
+    #pragma unroll
+    for (int m = 0; m < MT; ++m) {
+      const float other = __shfl(dbacc[m], lane ^ 32, WAVE);
+      if (lane < 32 && m_ok[m])
+        a.db_slab[(int64_t)split * a.out_dim + mcol[m]] = dbacc[m] + other;
+    }
   }
 }
 
@@ -485,9 +578,11 @@ void dw_mfma(torch::Tensor delta, torch::Tensor acts, torch::Tensor grad_buf,
   a.B = B;
   a.out_dim = out_dim;
   a.in_dim = in_dim;
-  // tile shape: maximize N per wave to avoid re-reading delta
+  // tile shape: maximize N per wave to avoid re-reading delta; wide
+  // out_dim also takes 2 row fragments per wave (halves acts re-reads)
+  const int mt = (out_dim >= 64) ? 2 : 1;
   a.nt = std::min(MAX_NT, (in_dim + M_WAVE - 1) / M_WAVE);
-  const int m_tiles = (out_dim + M_WAVE - 1) / M_WAVE;
+  const int m_tiles = (out_dim + M_WAVE * mt - 1) / (M_WAVE * mt);
   const int n_tiles = (in_dim + a.nt * M_WAVE - 1) / (a.nt * M_WAVE);
   // enough waves to hide the streamed-operand latency; slab stores make
   // extra splits nearly free (the reduce is split-chunk parallel)
@@ -505,19 +600,16 @@ void dw_mfma(torch::Tensor delta, torch::Tensor acts, torch::Tensor grad_buf,
 
   hipStream_t stream = c10::hip::getCurrentHIPStream().stream();
   const dim3 grid(m_tiles, n_tiles, a.splits);
-  switch (a.nt) {
-    case 1:
-      hipLaunchKernelGGL(dw_mfma_kernel<1>, grid, dim3(WAVE), 0, stream, a);
-      break;
-    case 2:
-      hipLaunchKernelGGL(dw_mfma_kernel<2>, grid, dim3(WAVE), 0, stream, a);
-      break;
-    case 3:
-      hipLaunchKernelGGL(dw_mfma_kernel<3>, grid, dim3(WAVE), 0, stream, a);
-      break;
-    default:
-      hipLaunchKernelGGL(dw_mfma_kernel<4>, grid, dim3(WAVE), 0, stream, a);
-      break;
+  const int key = a.nt * 10 + mt;
+  switch (key) {
+    case 11: hipLaunchKernelGGL((dw_mfma_kernel<1, 1>), grid, dim3(WAVE), 0, stream, a); break;
+    case 21: hipLaunchKernelGGL((dw_mfma_kernel<2, 1>), grid, dim3(WAVE), 0, stream, a); break;
+    case 31: hipLaunchKernelGGL((dw_mfma_kernel<3, 1>), grid, dim3(WAVE), 0, stream, a); break;
+    case 41: hipLaunchKernelGGL((dw_mfma_kernel<4, 1>), grid, dim3(WAVE), 0, stream, a); break;
+    case 12: hipLaunchKernelGGL((dw_mfma_kernel<1, 2>), grid, dim3(WAVE), 0, stream, a); break;
+    case 22: hipLaunchKernelGGL((dw_mfma_kernel<2, 2>), grid, dim3(WAVE), 0, stream, a); break;
+    case 32: hipLaunchKernelGGL((dw_mfma_kernel<3, 2>), grid, dim3(WAVE), 0, stream, a); break;
+    default: hipLaunchKernelGGL((dw_mfma_kernel<4, 2>), grid, dim3(WAVE), 0, stream, a); break;
   }
   float* dW = grad_buf.data_ptr<float>() + w_off;
   float* db = (b_off >= 0) ? grad_buf.data_ptr<float>() + b_off : nullptr;
