@@ -68,21 +68,15 @@ __global__ void gemm_fwd_kernel(FwdArgs a) {
   const int lane = threadIdx.x & (WAVE - 1);
   const int wave = threadIdx.x / WAVE;
 
-  // Single-buffered LDS with a T14 (issue-early / write-late) register
-  // pipeline: the NEXT K-tile's global loads are issued before this
-  // tile's MFMA phase, and written to LDS after the barrier — HBM
-  // latency hides under the matrix pipe with no extra LDS.
+  // X tile [FWD_M][BK] (+1 pad: the A-fragment read is a column read);
+  // Wt tile [BK][NT*32] staged coalesced so the MFMA B-operand comes
+  // from LDS instead of a fresh L2 round trip per k-step.
   __shared__ float xs[FWD_M][BK + 1];
   __shared__ float ws[BK][NT * M_WAVE];
 
   const int i_l = lane & 31;   // A row within wave tile
   const int k_l = lane >> 5;   // A k within pair
   const int NW = NT * M_WAVE;
-
-  constexpr int XREG = FWD_M * BK / 4 / (FWD_WAVES * 64);  // = 8, exact
-  constexpr int WREG = BK * NT * M_WAVE / 4 / (FWD_WAVES * 64);  // = 2*NT, exact
-  float4 xreg[XREG];
-  float4 wreg[WREG];
 
   for (int64_t tile = blockIdx.x; tile * FWD_M < a.B; tile += gridDim.x) {
     const int64_t b0 = tile * FWD_M;
@@ -92,12 +86,12 @@ __global__ void gemm_fwd_kernel(FwdArgs a) {
       #pragma unroll
       for (int r = 0; r < 16; ++r) acc[t][r] = 0.f;
 
-    // ---- staging helpers (register load / LDS write split) ----
-    auto load_x = [&](int kb) {
+    for (int kb = 0; kb < a.K; kb += BK) {
+      __syncthreads();
+      // float4 staging (scalar element staging measured as the dominant
+      // per-call cost: 32 scalar dword loads per thread per stage)
       constexpr int BK4 = BK / 4;
-      #pragma unroll
-      for (int i = 0; i < XREG; ++i) {
-        const int idx = threadIdx.x + i * FWD_WAVES * 64;
+      for (int idx = threadIdx.x; idx < FWD_M * BK4; idx += FWD_WAVES * 64) {
         const int r = idx / BK4, c4 = (idx % BK4) * 4;
         const int64_t row = b0 + r;
         const int col = kb + c4;
@@ -112,27 +106,14 @@ __global__ void gemm_fwd_kernel(FwdArgs a) {
             val = make_float4(tmp[0], tmp[1], tmp[2], tmp[3]);
           }
         }
-        xreg[i] = val;
+        xs[r][c4] = val.x;
+        xs[r][c4 + 1] = val.y;
+        xs[r][c4 + 2] = val.z;
+        xs[r][c4 + 3] = val.w;
       }
-    };
-    auto write_x = [&]() {
-      constexpr int BK4 = BK / 4;
-      #pragma unroll
-      for (int i = 0; i < XREG; ++i) {
-        const int idx = threadIdx.x + i * FWD_WAVES * 64;
-        const int r = idx / BK4, c4 = (idx % BK4) * 4;
-        xs[r][c4] = xreg[i].x;
-        xs[r][c4 + 1] = xreg[i].y;
-        xs[r][c4 + 2] = xreg[i].z;
-        xs[r][c4 + 3] = xreg[i].w;
-      }
-    };
-    auto load_w = [&](int kb) {
       if (a.wt_layout == 0) {
         const int NW4 = NW / 4;
-        #pragma unroll
-        for (int i = 0; i < WREG; ++i) {
-          const int idx = threadIdx.x + i * FWD_WAVES * 64;
+        for (int idx = threadIdx.x; idx < BK * NW4; idx += FWD_WAVES * 64) {
           const int r = idx / NW4, c4 = (idx % NW4) * 4;
           const int krow = kb + r;
           float4 val = make_float4(0.f, 0.f, 0.f, 0.f);
@@ -147,13 +128,14 @@ __global__ void gemm_fwd_kernel(FwdArgs a) {
               val = make_float4(tmp[0], tmp[1], tmp[2], tmp[3]);
             }
           }
-          wreg[i] = val;
+          *reinterpret_cast<float4*>(&ws[r][c4]) = val;
         }
       } else {
+        // torch W[N][K]: c-major mapping keeps the row-segment reads
+        // coalesced and transposes into ws on the fly — no host-side
+        // W.t().contiguous() per update step
         constexpr int BK4 = BK / 4;
-        #pragma unroll
-        for (int i = 0; i < WREG; ++i) {
-          const int idx = threadIdx.x + i * FWD_WAVES * 64;
+        for (int idx = threadIdx.x; idx < NW * BK4; idx += FWD_WAVES * 64) {
           const int c = idx / BK4, r4 = (idx % BK4) * 4;
           const int krow = kb + r4;
           float4 val = make_float4(0.f, 0.f, 0.f, 0.f);
@@ -168,47 +150,13 @@ __global__ void gemm_fwd_kernel(FwdArgs a) {
               val = make_float4(tmp[0], tmp[1], tmp[2], tmp[3]);
             }
           }
-          wreg[i] = val;
+          ws[r4][c] = val.x;
+          ws[r4 + 1][c] = val.y;
+          ws[r4 + 2][c] = val.z;
+          ws[r4 + 3][c] = val.w;
         }
       }
-    };
-    auto write_w = [&]() {
-      if (a.wt_layout == 0) {
-        const int NW4 = NW / 4;
-        #pragma unroll
-        for (int i = 0; i < WREG; ++i) {
-          const int idx = threadIdx.x + i * FWD_WAVES * 64;
-          const int r = idx / NW4, c4 = (idx % NW4) * 4;
-          *reinterpret_cast<float4*>(&ws[r][c4]) = wreg[i];
-        }
-      } else {
-        constexpr int BK4 = BK / 4;
-        #pragma unroll
-        for (int i = 0; i < WREG; ++i) {
-          const int idx = threadIdx.x + i * FWD_WAVES * 64;
-          const int c = idx / BK4, r4 = (idx % BK4) * 4;
-          ws[r4][c] = wreg[i].x;
-          ws[r4 + 1][c] = wreg[i].y;
-          ws[r4 + 2][c] = wreg[i].z;
-          ws[r4 + 3][c] = wreg[i].w;
-        }
-      }
-    };
-
-    // ---- prologue: first tile ----
-    load_x(0);
-    load_w(0);
-    write_x();
-    write_w();
-    __syncthreads();
-
-    for (int kb = 0; kb < a.K; kb += BK) {
-      const int kb_next = kb + BK;
-      const bool more = kb_next < a.K;
-      if (more) {   // issue next tile's loads before the MFMA phase (T14)
-        load_x(kb_next);
-        load_w(kb_next);
-      }
+      __syncthreads();
 
       const int ksteps = min(BK, a.K - kb);
       #pragma unroll 4
@@ -219,12 +167,6 @@ __global__ void gemm_fwd_kernel(FwdArgs a) {
           const float bv = ws[k2 + k_l][t * M_WAVE + i_l];
           acc[t] = __builtin_amdgcn_mfma_f32_32x32x2f32(av, bv, acc[t], 0, 0, 0);
         }
-      }
-      __syncthreads();
-      if (more) {
-        write_x();
-        write_w();
-        __syncthreads();
       }
     }
 
@@ -258,7 +200,6 @@ __global__ void gemm_fwd_kernel(FwdArgs a) {
         }
       }
     }
-    __syncthreads();  // xs/ws free for the next grid-stride tile
   }
 }
 
@@ -580,7 +521,9 @@ void dw_mfma(torch::Tensor delta, torch::Tensor acts, torch::Tensor grad_buf,
   a.in_dim = in_dim;
   // tile shape: maximize N per wave to avoid re-reading delta; wide
   // out_dim also takes 2 row fragments per wave (halves acts re-reads)
-  const int mt = (out_dim >= 64) ? 2 : 1;
+  // MT=2 measured WORSE (128-AGPR accumulators drop occupancy to
+  // 2 waves/SIMD and the k-loop stalls); keep single row fragments.
+  const int mt = 1;
   a.nt = std::min(MAX_NT, (in_dim + M_WAVE - 1) / M_WAVE);
   const int m_tiles = (out_dim + M_WAVE * mt - 1) / (M_WAVE * mt);
   const int n_tiles = (in_dim + a.nt * M_WAVE - 1) / (a.nt * M_WAVE);
