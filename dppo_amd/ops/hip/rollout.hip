@@ -66,7 +66,12 @@ constexpr int MISC_OFF = PART_OFF + NWAVES * ENV_TILE * MAX_H;  // val, rsum, ep
 constexpr int LDS_FLOATS = MISC_OFF + 3 * ENV_TILE + ENV_TILE;  // + racc
 
 struct RolloutArgs {
-  const float* params;   // flat parameter buffer (pi)
+  const float* params;   // rollout weight blob: TRANSPOSED layer weights
+                         // (Wt[in][out]) + biases + Wv[in] + Wpt[in][2A] —
+                         // per-k row reads are then fully coalesced
+                         // (lane == unit); the torch-layout row-per-lane
+                         // reads thrashed L1 (64 lines per instruction,
+                         // 96 KB working set)
   const float* envblob;  // d | Vt[r][D] | Ut[D][r] | Bt[D][A]
   const int* horizons;   // [E]
   float* x;              // [E][D] persistent env state
@@ -188,7 +193,9 @@ __global__ void rollout_kernel(RolloutArgs a) {
   // K-split layer forward: wave wv accumulates its k-quarter of every
   // unit into the partial slab; a combine pass sums the 4 partials,
   // adds bias and applies the activation.
-  auto layer_kpart = [&](const float* W, int off_Wv_u, int in_off,
+  // W is TRANSPOSED ([in_dim][w_cols]); lane u reads column u of k-rows:
+  // one coalesced 256 B wave-read per k, streaming in k.
+  auto layer_kpart = [&](const float* W, int w_cols, int off_Wv_u, int in_off,
                          int in_stride, int in_dim, int out_dim,
                          bool heads) {
     const int kq = (((in_dim + NWAVES * 4 - 1) / (NWAVES * 4)) * 4);
@@ -199,20 +206,11 @@ __global__ void rollout_kernel(RolloutArgs a) {
       #pragma unroll
       for (int e = 0; e < ENV_TILE; ++e) acc[e] = 0.f;
       const bool is_v = heads && (u == out_dim - 1);
-      const float* Wrow =
-          is_v ? (a.params + off_Wv_u) : (W + (int64_t)u * in_dim);
-      #pragma unroll 2
-      for (int k = k0; k + 4 <= k1; k += 4) {
-        const float4 w4 = *reinterpret_cast<const float4*>(Wrow + k);
-        #pragma unroll
-        for (int e = 0; e < ENV_TILE; ++e) {
-          const float4 i4 = *reinterpret_cast<const float4*>(
-              &lds[in_off + e * in_stride + k]);
-          acc[e] += w4.x * i4.x + w4.y * i4.y + w4.z * i4.z + w4.w * i4.w;
-        }
-      }
-      for (int k = max(k0, k1 & ~3); k < k1; ++k) {
-        const float w = Wrow[k];
+      const float* Wcol = is_v ? (a.params + off_Wv_u) : (W + u);
+      const int stride = is_v ? 1 : w_cols;
+      #pragma unroll 4
+      for (int k = k0; k < k1; ++k) {
+        const float w = Wcol[(int64_t)k * stride];
         #pragma unroll
         for (int e = 0; e < ENV_TILE; ++e)
           acc[e] += w * lds[in_off + e * in_stride + k];
@@ -236,8 +234,8 @@ __global__ void rollout_kernel(RolloutArgs a) {
     for (int l = 0; l < a.n_hidden; ++l) {
       const int out_dim = a.dims[l + 1];
       const int out_off = (l & 1) ? H1_OFF : H0_OFF;
-      layer_kpart(a.params + a.off_W[l], 0, in_off, in_stride, in_dim,
-                  out_dim, false);
+      layer_kpart(a.params + a.off_W[l], out_dim, 0, in_off, in_stride,
+                  in_dim, out_dim, false);
       __syncthreads();
       const float* bias = a.params + a.off_b[l];
       for (int idx = tid; idx < ENV_TILE * out_dim; idx += NWAVES * WAVE) {
@@ -256,7 +254,7 @@ __global__ void rollout_kernel(RolloutArgs a) {
     }
 
     // ---- heads (u < P: pd params; u == P: value), no activation ----
-    layer_kpart(a.params + a.off_Wp, a.off_Wv, in_off, in_stride, in_dim,
+    layer_kpart(a.params + a.off_Wp, P, a.off_Wv, in_off, in_stride, in_dim,
                 P + 1, true);
     __syncthreads();
     for (int idx = tid; idx < ENV_TILE * (P + 1); idx += NWAVES * WAVE) {
@@ -414,8 +412,8 @@ __global__ void rollout_kernel(RolloutArgs a) {
     for (int l = 0; l < a.n_hidden; ++l) {
       const int out_dim = a.dims[l + 1];
       const int out_off = (l & 1) ? H1_OFF : H0_OFF;
-      layer_kpart(a.params + a.off_W[l], 0, in_off, in_stride, in_dim,
-                  out_dim, false);
+      layer_kpart(a.params + a.off_W[l], out_dim, 0, in_off, in_stride,
+                  in_dim, out_dim, false);
       __syncthreads();
       const float* bias = a.params + a.off_b[l];
       for (int idx = tid; idx < ENV_TILE * out_dim; idx += NWAVES * WAVE) {

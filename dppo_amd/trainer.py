@@ -231,6 +231,32 @@ class DPPOEngine:
         return self._rollout_once_eager()
 
     @torch.no_grad()
+    def _rollout_weight_blob(self):
+        """Rollout weight blob: TRANSPOSED layer weights (Wt[in][out]) +
+        biases + Wv[H] + Wpt[H][2A], so the rollout kernel's per-k W reads
+        are coalesced.  Weights are frozen for the whole rollout; the
+        transposes are a few tiny kernels per round."""
+        c = self.cfg
+        dims = [self.obs_space.shape[0], *c.HIDDEN_SIZES]
+        parts, offsets, off = [], [], 0
+
+        def push(t):
+            nonlocal off
+            flat = t.reshape(-1)
+            parts.append(flat)
+            offsets.append(off)
+            off += flat.numel()
+
+        for lay in self.pi.hidden:
+            push(lay.weight.t().contiguous())     # Wt [in][out]
+            push(lay.bias)
+        push(self.pi.vf.weight.reshape(-1))       # Wv [H]
+        push(self.pi.vf.bias)
+        push(self.pi.pi.weight.t().contiguous())  # Wpt [H][2A]
+        push(self.pi.pi.bias)
+        return torch.cat(parts), offsets, dims
+
+    @torch.no_grad()
     def _rollout_once_hip(self) -> Tuple[RolloutBatch, Dict[str, float]]:
         """One rollout iteration in a single fused kernel launch
         (ops/hip/rollout.hip): MLP forward + sampling + epsilon-greedy +
@@ -250,13 +276,10 @@ class DPPOEngine:
         ) & 0x7FFFFFFFFFFFFFFF
         low = float(self.act_space.low.flat[0])
         high = float(self.act_space.high.flat[0])
-        # weight offsets into the flat parameter buffer, in parameters()
-        # registration order: hidden (W, b)*, vf.W, vf.b, pi.W, pi.b
-        offsets = [sl.start for sl in self.flat_pi.slices]
-        dims = [self.obs_space.shape[0], *c.HIDDEN_SIZES]
+        blob, offsets, dims = self._rollout_weight_blob()
         (states, pdflats, actions, values, rewards, dones, boot_v,
          moments) = ext.rollout_run(
-            self.flat_pi.flat_param.detach(), offsets, dims,
+            blob, offsets, dims,
             1 if c.ACTIVATION == "tanh" else 0,
             env.blob, env.rank_eff, env.horizons_i32,
             float(env.NOISE), low, high, float(eps),

@@ -38,10 +38,10 @@ def run_kernel(eng, T=32, eps=0.0, noise=None, seed=1234):
         env.NOISE = noise
     low = float(eng.act_space.low.flat[0])
     high = float(eng.act_space.high.flat[0])
-    offsets = [sl.start for sl in eng.flat_pi.slices]
-    dims = [eng.obs_space.shape[0], *eng.cfg.HIDDEN_SIZES]
+    with torch.no_grad():
+        blob, offsets, dims = eng._rollout_weight_blob()
     return ext.rollout_run(
-        eng.flat_pi.flat_param.detach(), offsets, dims,
+        blob, offsets, dims,
         1 if eng.cfg.ACTIVATION == "tanh" else 0,
         env.blob, env.rank_eff, env.horizons_i32,
         float(env.NOISE), low, high, float(eps),
