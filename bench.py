@@ -43,9 +43,11 @@ PRESETS = {
         NUM_ENVS=64, MAX_EPOCH_STEPS=100, DTYPE="float32",
     ),
     # BASELINE.json config 5: wide MFMA-bound MLP
+    # BASELINE.json config 5: wide MFMA-bound MLP, bf16 compute (as named)
     "wide4096": dict(
         GAME="Wide-4096", HIDDEN_SIZES=(4096, 4096, 4096, 4096),
-        ACTIVATION="tanh", NUM_ENVS=1024, MAX_EPOCH_STEPS=16, DTYPE="float32",
+        ACTIVATION="tanh", NUM_ENVS=1024, MAX_EPOCH_STEPS=16,
+        DTYPE="bfloat16",
     ),
 }
 

@@ -31,24 +31,45 @@ class FusedFlatAdam:
         self.flat_grad = flat_grad
         self.exp_avg = torch.zeros_like(flat_param.data)
         self.exp_avg_sq = torch.zeros_like(flat_param.data)
-        self.step_count = 0
         self.betas = tuple(betas)
         self.eps = eps
         self.param_groups: List[Dict] = [
             {"lr": lr, "betas": self.betas, "eps": eps, "params": [flat_param]}
         ]
+        # device-resident step/lr/bias-correction state: kernel args are
+        # frozen under hipGraph replay, so the annealed lr and the growing
+        # step count must come from device memory (adam_step_dev).
+        dev = flat_param.device
+        self.step_dev = torch.zeros(1, device=dev, dtype=torch.int32)
+        self.lr_dev = torch.tensor([lr], device=dev, dtype=torch.float32)
+        self.coef = torch.zeros(3, device=dev, dtype=torch.float32)
+
+    @property
+    def step_count(self) -> int:
+        return int(self.step_dev.item())
+
+    @step_count.setter
+    def step_count(self, v: int) -> None:
+        self.step_dev.fill_(int(v))
 
     @torch.no_grad()
     def step(self) -> None:
-        self.step_count += 1
+        self.lr_dev.fill_(float(self.param_groups[0]["lr"]))
+        self.step_captured()
+
+    @torch.no_grad()
+    def step_captured(self) -> None:
+        """Capture-safe step: no host->device scalar traffic; the caller
+        refreshes lr_dev outside the captured region."""
         g = self.param_groups[0]
-        self._ext.adam_step(
+        self._ext.adam_step_dev(
             self.flat_param.data,
             self.flat_grad,
             self.exp_avg,
             self.exp_avg_sq,
-            self.step_count,
-            float(g["lr"]),
+            self.step_dev,
+            self.lr_dev,
+            self.coef,
             float(g["betas"][0]),
             float(g["betas"][1]),
             float(g["eps"]),

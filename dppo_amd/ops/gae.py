@@ -68,8 +68,12 @@ def gae_advantages(
     whiten: bool = True,
     eps: float = 1e-8,
     policy: str = "auto",
+    adv_out: Optional[torch.Tensor] = None,
+    etr_out: Optional[torch.Tensor] = None,
 ) -> Tuple[torch.Tensor, torch.Tensor]:
-    """GAE with HIP dispatch on CUDA tensors."""
+    """GAE with HIP dispatch on CUDA tensors.  adv_out/etr_out, when
+    given, receive the results in place (stable addresses for hipGraph
+    capture of the downstream update)."""
     from . import use_hip, hip_ext
 
     if use_hip(rewards, policy):
@@ -78,9 +82,12 @@ def gae_advantages(
         values = values.contiguous()
         dones = dones.to(rewards.dtype).contiguous()
         bootstrap_value = bootstrap_value.contiguous()
+        empty = torch.empty(0, device=rewards.device)
         adv, etr = ext.gae_scan(
             rewards, values, dones, bootstrap_value,
             float(gamma), float(lam), bool(whiten), float(eps),
+            adv_out if adv_out is not None else empty,
+            etr_out if etr_out is not None else empty,
         )
         return adv, etr
     return gae_advantages_ref(

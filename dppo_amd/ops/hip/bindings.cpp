@@ -6,7 +6,8 @@
 std::vector<torch::Tensor> gae_scan(torch::Tensor rewards, torch::Tensor values,
                                     torch::Tensor dones, torch::Tensor boot,
                                     double gamma, double lam, bool whiten,
-                                    double eps);
+                                    double eps, torch::Tensor adv_out,
+                                    torch::Tensor etr_out);
 
 torch::Tensor ppo_loss_gauss_fwd(torch::Tensor pdpi, torch::Tensor pdold,
                                  torch::Tensor vpred, torch::Tensor oldv,
@@ -18,7 +19,8 @@ torch::Tensor ppo_loss_gauss_gh(torch::Tensor pdflat, torch::Tensor oldflat,
                                 torch::Tensor vpred, torch::Tensor oldv,
                                 torch::Tensor act, torch::Tensor adv,
                                 torch::Tensor etr, double clip,
-                                double entcoeff, double vcoeff);
+                                double entcoeff, double vcoeff,
+                                torch::Tensor clip_dev);
 
 std::vector<torch::Tensor> ppo_loss_gauss_bwd(
     torch::Tensor pdpi, torch::Tensor pdold, torch::Tensor vpred,
@@ -30,6 +32,11 @@ void adam_step(torch::Tensor p, torch::Tensor g, torch::Tensor m,
                torch::Tensor v, int64_t step, double lr, double beta1,
                double beta2, double eps);
 
+void adam_step_dev(torch::Tensor p, torch::Tensor g, torch::Tensor m,
+                   torch::Tensor v, torch::Tensor step_dev,
+                   torch::Tensor lr_dev, torch::Tensor coef, double beta1,
+                   double beta2, double eps);
+
 void gemm_fwd(torch::Tensor X, torch::Tensor Wt, torch::Tensor bias,
               int64_t activation, int64_t heads, torch::Tensor C,
               torch::Tensor v, torch::Tensor aux, int64_t wt_layout);
@@ -38,16 +45,13 @@ void dw_mfma(torch::Tensor delta, torch::Tensor acts, torch::Tensor grad_buf,
              int64_t w_off, int64_t b_off, int64_t split_row, int64_t w_off2,
              int64_t b_off2);
 
-void dwv(torch::Tensor gv, torch::Tensor acts, torch::Tensor grad_buf,
-         int64_t w_off, int64_t b_off);
-
 std::vector<torch::Tensor> rollout_run(
     torch::Tensor params, std::vector<int64_t> offsets,
     std::vector<int64_t> dims, int64_t activation,
     torch::Tensor envblob, int64_t rank, torch::Tensor horizons,
     double noise, double act_low, double act_high, double eps_explore,
     torch::Tensor x, torch::Tensor t, torch::Tensor epr,
-    int64_t T, int64_t act_dim, int64_t seed);
+    int64_t T, int64_t act_dim, int64_t seed, torch::Tensor out_buf);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("gae_scan", &gae_scan,
@@ -57,6 +61,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("ppo_loss_gauss_bwd", &ppo_loss_gauss_bwd,
           "fused DiagGaussian PPO loss backward (gfx950)");
   mod.def("adam_step", &adam_step, "fused flat Adam step (gfx950)");
+  mod.def("adam_step_dev", &adam_step_dev,
+          "graph-replayable fused Adam (device step/lr) (gfx950)");
   mod.def("rollout_run", &rollout_run,
           "fused T-step rollout: MLP fwd + sample + synthetic env (gfx950)");
   mod.def("ppo_loss_gauss_gh", &ppo_loss_gauss_gh,
@@ -65,5 +71,4 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
           "MFMA f32 layer forward C=act(X@Wt+b), fused tanh (gfx950)");
   mod.def("dw_mfma", &dw_mfma,
           "MFMA f32 split-K dW += delta^T@acts into flat grad (gfx950)");
-  mod.def("dwv", &dwv, "value-head weight grad reduction (gfx950)");
 }

@@ -88,7 +88,8 @@ __global__ void gae_whiten_kernel(float* __restrict__ adv,
 std::vector<torch::Tensor> gae_scan(torch::Tensor rewards, torch::Tensor values,
                                     torch::Tensor dones, torch::Tensor boot,
                                     double gamma, double lam, bool whiten,
-                                    double eps) {
+                                    double eps, torch::Tensor adv_out,
+                                    torch::Tensor etr_out) {
   TORCH_CHECK(rewards.is_cuda() && rewards.dtype() == torch::kFloat32,
               "gae_scan: rewards must be fp32 CUDA");
   TORCH_CHECK(rewards.dim() == 2, "gae_scan: rewards must be [T, E]");
@@ -96,8 +97,12 @@ std::vector<torch::Tensor> gae_scan(torch::Tensor rewards, torch::Tensor values,
   TORCH_CHECK(values.sizes() == rewards.sizes() && dones.sizes() == rewards.sizes());
   TORCH_CHECK(boot.numel() == E);
 
-  auto adv = torch::empty_like(rewards);
-  auto etr = torch::empty_like(rewards);
+  auto adv = (adv_out.numel() == rewards.numel())
+                 ? adv_out.view_as(rewards)
+                 : torch::empty_like(rewards);
+  auto etr = (etr_out.numel() == rewards.numel())
+                 ? etr_out.view_as(rewards)
+                 : torch::empty_like(rewards);
   auto stats = torch::zeros({2}, rewards.options().dtype(torch::kFloat64));
   auto mean_inv = torch::empty({2}, rewards.options());
 

@@ -401,44 +401,6 @@ __global__ void db_reduce_kernel(const float* __restrict__ db_slab,
   }
 }
 
-// ---------------------------------------------------------------------------
-// out == 1 weight grad: dWv[n] += sum_k gv[k] * acts[k][n]; dbv += sum gv
-// ---------------------------------------------------------------------------
-
-__launch_bounds__(64)
-__global__ void dwv_kernel(const float* __restrict__ gv,    // [B]
-                           const float* __restrict__ acts,  // [B][in]
-                           float* __restrict__ dWv,         // [in]
-                           float* __restrict__ dbv,         // [1] or null
-                           int64_t B, int in_dim, int splits) {
-  const int lane = threadIdx.x;
-  const int n0 = blockIdx.x * WAVE;
-  const int split = blockIdx.y;
-  const int64_t rows_per = (B + splits - 1) / splits;
-  const int64_t k0 = split * rows_per;
-  const int64_t k1 = min(B, k0 + rows_per);
-  const int col = n0 + lane;
-  const int ccol = (col < in_dim) ? col : 0;
-  float acc[4] = {0.f, 0.f, 0.f, 0.f};
-  float bacc = 0.f;
-  int64_t k = k0;
-  for (; k + 3 < k1; k += 4) {
-    #pragma unroll
-    for (int q = 0; q < 4; ++q) {
-      const float g = gv[k + q];
-      bacc += g;
-      acc[q] += g * acts[(k + q) * in_dim + ccol];
-    }
-  }
-  for (; k < k1; ++k) {
-    const float g = gv[k];
-    bacc += g;
-    acc[0] += g * acts[k * in_dim + ccol];
-  }
-  if (col < in_dim) atomicAdd(&dWv[col], acc[0] + acc[1] + acc[2] + acc[3]);
-  if (dbv != nullptr && blockIdx.x == 0 && lane == 0) atomicAdd(dbv, bacc);
-}
-
 }  // namespace
 
 // ---------------------------------------------------------------------------
@@ -573,18 +535,3 @@ void dw_mfma(torch::Tensor delta, torch::Tensor acts, torch::Tensor grad_buf,
   }
 }
 
-void dwv(torch::Tensor gv, torch::Tensor acts, torch::Tensor grad_buf,
-         int64_t w_off, int64_t b_off) {
-  const int64_t B = gv.numel();
-  const int in_dim = static_cast<int>(acts.size(1));
-  const int n_blocks = (in_dim + WAVE - 1) / WAVE;
-  // scale splits so the launch has ~2k waves regardless of in_dim
-  const int splits = static_cast<int>(std::max<int64_t>(
-      1, std::min<int64_t>(2048 / std::max(1, n_blocks), B / 256)));
-  hipStream_t stream = c10::hip::getCurrentHIPStream().stream();
-  hipLaunchKernelGGL(dwv_kernel, dim3(n_blocks, splits), dim3(WAVE), 0, stream,
-                     gv.data_ptr<float>(), acts.data_ptr<float>(),
-                     grad_buf.data_ptr<float>() + w_off,
-                     (b_off >= 0) ? grad_buf.data_ptr<float>() + b_off : nullptr,
-                     B, in_dim, splits);
-}

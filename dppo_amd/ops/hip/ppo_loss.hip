@@ -208,7 +208,9 @@ __global__ void ppo_gh_kernel(
     const float* __restrict__ vpred, const float* __restrict__ oldv,
     const float* __restrict__ act, const float* __restrict__ adv,
     const float* __restrict__ etr, float* __restrict__ gh,  // [B][2A+1]
+    const float* __restrict__ clip_dev,  // nullptr -> use `clip` arg
     int64_t B, int A, float clip, float entcoeff, float vcoeff) {
+  if (clip_dev != nullptr) clip = clip_dev[0];
   const int lane = threadIdx.x & (WAVE - 1);
   const int wave = threadIdx.x / WAVE;
   const int P = 2 * A;
@@ -258,16 +260,19 @@ torch::Tensor ppo_loss_gauss_gh(torch::Tensor pdflat, torch::Tensor oldflat,
                                 torch::Tensor vpred, torch::Tensor oldv,
                                 torch::Tensor act, torch::Tensor adv,
                                 torch::Tensor etr, double clip,
-                                double entcoeff, double vcoeff) {
+                                double entcoeff, double vcoeff,
+                                torch::Tensor clip_dev) {
   const int64_t B = vpred.numel();
   const int A = static_cast<int>(pdflat.size(1) / 2);
   auto gh = torch::empty({B, 2 * (int64_t)A + 1}, pdflat.options());
   hipStream_t stream = c10::hip::getCurrentHIPStream().stream();
+  const float* cd =
+      (clip_dev.numel() > 0) ? clip_dev.data_ptr<float>() : nullptr;
   hipLaunchKernelGGL(ppo_gh_kernel, dim3(2048), dim3(256), 0, stream,
                      pdflat.data_ptr<float>(), oldflat.data_ptr<float>(),
                      vpred.data_ptr<float>(), oldv.data_ptr<float>(),
                      act.data_ptr<float>(), adv.data_ptr<float>(),
-                     etr.data_ptr<float>(), gh.data_ptr<float>(), B, A,
+                     etr.data_ptr<float>(), gh.data_ptr<float>(), cd, B, A,
                      (float)clip, (float)entcoeff, (float)vcoeff);
   return gh;
 }

@@ -501,7 +501,7 @@ std::vector<torch::Tensor> rollout_run(
     torch::Tensor envblob, int64_t rank, torch::Tensor horizons,
     double noise, double act_low, double act_high, double eps_explore,
     torch::Tensor x, torch::Tensor t, torch::Tensor epr,
-    int64_t T, int64_t act_dim, int64_t seed) {
+    int64_t T, int64_t act_dim, int64_t seed, torch::Tensor out_buf) {
   // offsets: [W0, b0, W1, b1, ..., Wv, bv, Wp, bp] into params
   // dims: [obs, h1, ..., hN]
   const int64_t E = x.size(0);
@@ -551,7 +551,11 @@ std::vector<torch::Tensor> rollout_run(
 
   const int P = 2 * A;
   const int64_t n_out = T * E * (D + P + A + 3) + E + 5;
-  auto out = torch::empty({n_out}, x.options());
+  // caller may provide a persistent output blob (stable addresses for
+  // hipGraph capture of the downstream update)
+  torch::Tensor out = (out_buf.numel() == n_out)
+                          ? out_buf
+                          : torch::empty({n_out}, x.options());
   auto epr_in = epr.clone();
   a.out = out.data_ptr<float>();
 

@@ -91,7 +91,10 @@ class DPPOEngine:
         self.comm = comm if comm is not None else Comm()
         self.scope = scope
         self.device = self.comm.device
-        self.dtype = cfg.torch_dtype()
+        # Rollout/batch buffers stay fp32 regardless of DTYPE;
+        # DTYPE='bfloat16' switches the update-path GEMM compute to bf16
+        # autocast (BASELINE config 5) with fp32 loss math.
+        self.dtype = torch.float32
 
         obs_space, act_space = game_spaces(cfg.GAME)
         self.obs_space, self.act_space = obs_space, act_space
@@ -277,13 +280,23 @@ class DPPOEngine:
         low = float(self.act_space.low.flat[0])
         high = float(self.act_space.high.flat[0])
         blob, offsets, dims = self._rollout_weight_blob()
+        # persistent output buffers: stable addresses let the update phase
+        # be hipGraph-captured once and replayed every round
+        A = self.act_space.shape[0]
+        D = self.obs_space.shape[0]
+        n_out = T * E * (D + 3 * A + 3) + E + 5
+        if getattr(self, "_rollout_out", None) is None or \
+                self._rollout_out.numel() != n_out:
+            self._rollout_out = torch.empty(n_out, device=self.device)
+            self._adv_buf = torch.empty(T * E, device=self.device)
+            self._etr_buf = torch.empty(T * E, device=self.device)
         (states, pdflats, actions, values, rewards, dones, boot_v,
          moments) = ext.rollout_run(
             blob, offsets, dims,
             1 if c.ACTIVATION == "tanh" else 0,
             env.blob, env.rank_eff, env.horizons_i32,
             float(env.NOISE), low, high, float(eps),
-            env.x, env.t, self.epr, T, self.act_space.shape[0], seed,
+            env.x, env.t, self.epr, T, A, seed, self._rollout_out,
         )
         self.obs = env.x  # updated in place by the kernel
 
@@ -291,6 +304,7 @@ class DPPOEngine:
             rewards, values, dones, boot_v,
             c.GAMMA, c.LAM, whiten=True, eps=c.ADV_EPS,
             policy=c.USE_HIP_KERNELS,
+            adv_out=self._adv_buf, etr_out=self._etr_buf,
         )
         obs_dim = self.obs_space.shape[0]
         P = self.pi.pdtype.param_shape()[0]
@@ -415,6 +429,12 @@ class DPPOEngine:
         the 3 extra GEMMs per update step."""
         if recorded_pi:
             v, pdflat = batch.oldv, batch.oldflat
+        elif self.cfg.DTYPE == "bfloat16" and self.device.type == "cuda":
+            # bf16 compute for the wide MFMA-bound config (BASELINE #5):
+            # rocBLAS bf16 GEMMs with fp32 loss math
+            with torch.autocast("cuda", dtype=torch.bfloat16):
+                v, pdflat = self.pi(batch.states)
+            v, pdflat = v.float(), pdflat.float()
         else:
             v, pdflat = self.pi(batch.states)
         pd = self.pi.pdtype.pdfromflat(pdflat)
@@ -485,7 +505,10 @@ class DPPOEngine:
         for g in self.optimizer.param_groups:
             g["lr"] = self.cfg.LEARNING_RATE * l_mul
         if self._can_fuse_update():
-            self._update_fused(batch, l_mul)
+            if self.cfg.USE_GRAPHS and self._can_fuse_rollout():
+                self._update_graphed(batch, l_mul)
+            else:
+                self._update_fused(batch, l_mul)
             return
         for _ in range(self.cfg.UPDATE_STEPS):
             self.flat_pi.zero_grad()
@@ -493,6 +516,75 @@ class DPPOEngine:
             losses["total_loss"].backward()
             self.comm.allreduce_mean_(self.flat_pi.flat_grad)
             self.optimizer.step()
+
+    def _clip_dev_or_empty(self) -> torch.Tensor:
+        t = getattr(self, "_clip_dev", None)
+        if t is None:
+            t = torch.empty(0, device=self.device)
+        return t
+
+    def _update_graphed(self, batch: RolloutBatch, l_mul: float) -> None:
+        """hipGraph-captured update: the whole UPDATE_STEPS pipeline
+        (forward GEMMs, loss grads, dgrad chain, dW scatter, all-reduce,
+        Adam) is captured ONCE and replayed each round.  Per-round scalars
+        (clip = CLIP_PARAM*l_mul, lr = LEARNING_RATE*l_mul, the Adam step
+        count) live in device memory, and every input tensor is a view of
+        the persistent rollout/GAE buffers, so replay sees fresh data.
+        Any capture failure falls back to the uncaptured fused path."""
+        clip = self.cfg.CLIP_PARAM * l_mul
+        lr = self.cfg.LEARNING_RATE * l_mul
+        if getattr(self, "_graph_failed", False):
+            self._update_fused(batch, l_mul)
+            return
+        if getattr(self, "_upd_graph", None) is None:
+            try:
+                self._clip_dev = torch.tensor([clip], device=self.device)
+                opt = self.optimizer
+                opt.lr_dev.fill_(lr)
+                # snapshot optimizer+param state: warmup executes real steps
+                snap = (
+                    self.flat_pi.flat_param.detach().clone(),
+                    opt.exp_avg.clone(), opt.exp_avg_sq.clone(),
+                    opt.step_dev.clone(),
+                )
+                side = torch.cuda.Stream()
+                side.wait_stream(torch.cuda.current_stream())
+                with torch.cuda.stream(side):
+                    for _ in range(2):
+                        self._update_body(batch)
+                torch.cuda.current_stream().wait_stream(side)
+                torch.cuda.synchronize()
+                with torch.no_grad():
+                    self.flat_pi.flat_param.copy_(snap[0])
+                    opt.exp_avg.copy_(snap[1])
+                    opt.exp_avg_sq.copy_(snap[2])
+                    opt.step_dev.copy_(snap[3])
+                g = torch.cuda.CUDAGraph()
+                with torch.cuda.graph(g):
+                    self._update_body(batch)
+                self._upd_graph = g
+            except Exception:  # noqa: BLE001 — capture support varies
+                self._graph_failed = True
+                self._upd_graph = None
+                self._update_fused(batch, l_mul)
+                return
+        self._clip_dev.fill_(clip)
+        self.optimizer.lr_dev.fill_(lr)
+        self._upd_graph.replay()
+
+    def _update_body(self, batch: RolloutBatch) -> None:
+        """The capture-safe UPDATE_STEPS pipeline (no host syncs, no
+        host-valued scalars: clip/lr/step come from device memory)."""
+        for _ in range(self.cfg.UPDATE_STEPS):
+            acts, a_views, v, pdflat = self._fused_forward(batch.states)
+            self.flat_pi.zero_grad()
+            self._fused_backward(
+                batch.states, acts, a_views, v, pdflat,
+                batch.oldflat, batch.oldv, batch.actions,
+                batch.adv, batch.etr, 0.0,  # clip read from _clip_dev
+            )
+            self.comm.allreduce_mean_(self.flat_pi.flat_grad)
+            self.optimizer.step_captured()
 
     def _fused_forward(self, states: torch.Tensor):
         """MFMA forward through the MLP (gemm_fwd per layer + heads).
@@ -555,7 +647,7 @@ class DPPOEngine:
 
         gh = ext.ppo_loss_gauss_gh(
             pdflat, oldflat, v, oldv, actions, adv, etr,
-            clip, c.ENTCOEFF, c.VCOEFF,
+            clip, c.ENTCOEFF, c.VCOEFF, self._clip_dev_or_empty(),
         )
         # _fused_forward cached [Wp; Wv] — its rows are the dgrad Wt
         Wh_cat = getattr(self, "_Wh_cat", None)
@@ -585,6 +677,10 @@ class DPPOEngine:
         """Fused MFMA update steps: gemm_fwd xL -> GEMM-shaped backward ->
         all-reduce -> fused Adam (~10 launches per step, no autograd)."""
         clip = self.cfg.CLIP_PARAM * l_mul
+        # a live _clip_dev overrides the kernel's clip argument; keep it
+        # coherent when falling back from the graphed path
+        if getattr(self, "_clip_dev", None) is not None:
+            self._clip_dev.fill_(clip)
         for _ in range(self.cfg.UPDATE_STEPS):
             acts, a_views, v, pdflat = self._fused_forward(batch.states)
             self.flat_pi.zero_grad()
