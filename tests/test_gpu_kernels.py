@@ -20,6 +20,10 @@ def ext():
     return require_hip_ext()
 
 
+def _e():
+    return torch.empty(0, device="cuda")
+
+
 def test_gae_scan_matches_ref(ext):
     T, E = 128, 512
     g = torch.Generator(device="cuda").manual_seed(0)
@@ -27,7 +31,7 @@ def test_gae_scan_matches_ref(ext):
     v = torch.randn(T, E, device="cuda", generator=g)
     d = (torch.rand(T, E, device="cuda", generator=g) < 0.05).float()
     boot = torch.randn(E, device="cuda", generator=g)
-    adv, etr = ext.gae_scan(r, v, d, boot, 0.99, 0.95, True, 1e-8)
+    adv, etr = ext.gae_scan(r, v, d, boot, 0.99, 0.95, True, 1e-8, _e(), _e())
     adv_ref, etr_ref = gae_advantages_ref(r, v, d, boot, 0.99, 0.95, True, 1e-8)
     torch.testing.assert_close(adv, adv_ref, atol=2e-4, rtol=2e-4)
     torch.testing.assert_close(etr, etr_ref, atol=1e-4, rtol=1e-4)
@@ -40,7 +44,7 @@ def test_gae_scan_long_rollout(ext):
     v = torch.randn(T, E, device="cuda")
     d = (torch.rand(T, E, device="cuda") < 0.01).float()
     boot = torch.randn(E, device="cuda")
-    adv, etr = ext.gae_scan(r, v, d, boot, 0.99, 0.95, False, 1e-8)
+    adv, etr = ext.gae_scan(r, v, d, boot, 0.99, 0.95, False, 1e-8, _e(), _e())
     adv_ref, etr_ref = gae_advantages_ref(r, v, d, boot, 0.99, 0.95, False)
     torch.testing.assert_close(adv, adv_ref, atol=5e-4, rtol=5e-4)
 
@@ -51,7 +55,7 @@ def test_gae_whiten_constant_guard(ext):
     v = torch.zeros(T, E, device="cuda")
     d = torch.ones(T, E, device="cuda")
     boot = torch.zeros(E, device="cuda")
-    adv, _ = ext.gae_scan(r, v, d, boot, 0.99, 0.95, True, 1e-8)
+    adv, _ = ext.gae_scan(r, v, d, boot, 0.99, 0.95, True, 1e-8, _e(), _e())
     assert torch.isfinite(adv).all()
 
 
@@ -151,3 +155,26 @@ def test_native_extension_is_loaded():
 
     assert hip_ext() is not None
     assert hasattr(_dppo_hip, "gae_scan")
+
+
+def test_adam_dev_matches_torch(ext):
+    """The graph-replayable Adam (device step/lr/bias-corrections) matches
+    torch.optim.Adam like the host-arg variant."""
+    n = 50_001
+    p = torch.randn(n, device="cuda")
+    gref = torch.randn(n, device="cuda")
+    p1 = p.clone().requires_grad_(True)
+    p1.grad = gref.clone()
+    opt = torch.optim.Adam([p1], lr=2e-3)
+    p2 = p.clone()
+    m = torch.zeros(n, device="cuda")
+    vv = torch.zeros(n, device="cuda")
+    step_dev = torch.zeros(1, device="cuda", dtype=torch.int32)
+    lr_dev = torch.tensor([2e-3], device="cuda")
+    coef = torch.zeros(3, device="cuda")
+    for _ in range(3):
+        opt.step()
+        ext.adam_step_dev(p2, gref, m, vv, step_dev, lr_dev, coef,
+                          0.9, 0.999, 1e-8)
+    assert int(step_dev.item()) == 3
+    torch.testing.assert_close(p2, p1.detach(), atol=2e-6, rtol=2e-5)
