@@ -32,7 +32,7 @@ namespace {
 
 using f32x16 = __attribute__((ext_vector_type(16))) float;
 
-constexpr int BK = 32;        // K-step per stage
+constexpr int BK = 16;        // K-step per stage
 constexpr int M_WAVE = 32;    // rows per wave tile
 constexpr int FWD_WAVES = 4;  // waves per block (each owns 32 rows)
 constexpr int FWD_M = FWD_WAVES * M_WAVE;  // 128 rows per block
