@@ -46,7 +46,7 @@
 
 namespace {
 
-constexpr int ENV_TILE = 4;
+constexpr int ENV_TILE = 8;
 constexpr int MAX_D = 512;
 constexpr int MAX_H = 128;
 constexpr int MAX_A = 32;
@@ -301,8 +301,7 @@ __global__ void rollout_kernel(RolloutArgs a) {
     // as (rr, k-quarter) so all 64 lanes stream and the per-lane load
     // chain is 4x shorter (the 16-lane serial version stalled the whole
     // block at the next barrier) ----
-    if (wv < nE) {
-      const int e = wv;
+    for (int e = wv; e < nE; e += NWAVES) {
       const int rr = lane & 15;
       const int kq = lane >> 4;  // 4 k-quarters
       const int kq_len = ((D / 4 + 3) & ~3);
@@ -431,13 +430,13 @@ __global__ void rollout_kernel(RolloutArgs a) {
       in_dim = out_dim;
     }
     // value head: wave wv reduces env wv over lanes
-    if (wv < nE) {
+    for (int e = wv; e < nE; e += NWAVES) {
       const float* Wv = a.params + a.off_Wv;
       float acc = 0.f;
       for (int k = lane; k < in_dim; k += WAVE)
-        acc += Wv[k] * lds[in_off + wv * in_stride + k];
+        acc += Wv[k] * lds[in_off + e * in_stride + k];
       const float total = wave_reduce_sum(acc);
-      if (lane == 0) out_boot[e0 + wv] = total + a.params[a.off_bv];
+      if (lane == 0) out_boot[e0 + e] = total + a.params[a.off_bv];
     }
   }
 
