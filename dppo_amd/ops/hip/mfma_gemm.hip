@@ -483,10 +483,11 @@ void dw_mfma(torch::Tensor delta, torch::Tensor acts, torch::Tensor grad_buf,
   a.in_dim = in_dim;
   // tile shape: maximize N per wave to avoid re-reading delta; wide
   // out_dim also takes 2 row fragments per wave (halves acts re-reads)
-  // MT=2 measured WORSE (128-AGPR accumulators drop occupancy to
-  // 2 waves/SIMD and the k-loop stalls); keep single row fragments.
-  const int mt = 1;
-  a.nt = std::min(MAX_NT, (in_dim + M_WAVE - 1) / M_WAVE);
+  // Wide out_dim: 2 row fragments with NT capped at 2 keeps the
+  // accumulators at 64 AGPR (4 waves/SIMD) while halving the streamed
+  // acts re-reads (MT=2 with NT=4 = 128 AGPR measured 17x stalls).
+  const int mt = (out_dim >= 64) ? 2 : 1;
+  a.nt = std::min(mt == 2 ? 2 : MAX_NT, (in_dim + M_WAVE - 1) / M_WAVE);
   const int m_tiles = (out_dim + M_WAVE * mt - 1) / (M_WAVE * mt);
   const int n_tiles = (in_dim + a.nt * M_WAVE - 1) / (a.nt * M_WAVE);
   // enough waves to hide the streamed-operand latency; slab stores make
