@@ -38,6 +38,7 @@ deadlocks.
 from __future__ import annotations
 
 import math
+import os
 from dataclasses import dataclass, field
 from typing import Dict, List, Optional, Tuple
 
@@ -505,7 +506,14 @@ class DPPOEngine:
         for g in self.optimizer.param_groups:
             g["lr"] = self.cfg.LEARNING_RATE * l_mul
         if self._can_fuse_update():
-            if self.cfg.USE_GRAPHS and self._can_fuse_rollout():
+            # Graph capture of the RCCL all-reduce is unexercised on this
+            # pool's multi-GPU boxes; keep multi-rank runs on the plain
+            # fused path unless explicitly opted in (DPPO_GRAPH_DIST=1).
+            graph_ok = self.cfg.USE_GRAPHS and self._can_fuse_rollout() and (
+                not self.comm.distributed
+                or os.environ.get("DPPO_GRAPH_DIST") == "1"
+            )
+            if graph_ok:
                 self._update_graphed(batch, l_mul)
             else:
                 self._update_fused(batch, l_mul)
