@@ -217,6 +217,9 @@ __global__ void ppo_gh_kernel(
   const int64_t waves_total = (int64_t)gridDim.x * 4;
   const int64_t wid = (int64_t)blockIdx.x * 4 + wave;
 
+  // two rows in flight per wave: each row is a chain of ~6 dependent
+  // load groups; interleaving rows hides most of that latency
+  #pragma unroll 2
   for (int64_t b = wid; b < B; b += waves_total) {
     float lp_part = 0.f, lo_part = 0.f, ent_part = 0.f;
     float z = 0.f, inv_s = 0.f;
