@@ -31,6 +31,7 @@ DEV_INLINE GaussRow gauss_row(const float* __restrict__ pdpi,
   return ppo_gauss_row(pdpi, pdold, act, b, A);
 }
 
+__launch_bounds__(256)
 __global__ void ppo_gauss_fwd_kernel(
     const float* __restrict__ pdpi, const float* __restrict__ pdold,
     const float* __restrict__ vpred, const float* __restrict__ oldv,
@@ -38,25 +39,53 @@ __global__ void ppo_gauss_fwd_kernel(
     const float* __restrict__ etr,
     double* __restrict__ acc,  // [3] {policy_min_sum, ent_sum, value_max_sum}
     int64_t B, int A, float clip) {
-  float pol = 0.f, ent = 0.f, val = 0.f;
-  for (int64_t b = gidx(); b < B; b += gstride()) {
-    const GaussRow r = gauss_row(pdpi, pdold, act, b, A);
-    const float ratio = __expf(r.logp_pi - r.logp_old);
-    const float ab = adv[b];
-    const float surr1 = ratio * ab;
-    const float rc = fminf(fmaxf(ratio, 1.f - clip), 1.f + clip);
-    const float surr2 = rc * ab;
-    pol += fminf(surr1, surr2);
-    ent += r.ent;
-    const float vb = vpred[b], ob = oldv[b], eb = etr[b];
-    const float d1 = vb - eb;
-    const float dc = fminf(fmaxf(vb - ob, -clip), clip);
-    const float d2 = ob + dc - eb;
-    val += fmaxf(d1 * d1, d2 * d2);
+  // wave-per-row (lanes cooperate on the 2A distribution columns with
+  // coalesced row loads + wave reductions): the row-per-thread version
+  // thrashed L1 across 64 concurrent 2A-float rows and ran ~10x slower.
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wave = threadIdx.x / WAVE;
+  const int P = 2 * A;
+  const int64_t waves_total = (int64_t)gridDim.x * 4;
+  const int64_t wid = (int64_t)blockIdx.x * 4 + wave;
+
+  float pol = 0.f, ent = 0.f, val = 0.f;  // lane 0 accumulates rows
+  #pragma unroll 2
+  for (int64_t b = wid; b < B; b += waves_total) {
+    float lp_part = 0.f, lo_part = 0.f, ent_part = 0.f;
+    if (lane < A) {
+      const float mu = pdpi[b * P + lane];
+      const float ls = pdpi[b * P + A + lane];
+      const float aj = act[b * A + lane];
+      const float zp = (aj - mu) * __expf(-ls);
+      lp_part = -0.5f * zp * zp - ls;
+      const float mo = pdold[b * P + lane];
+      const float lso = pdold[b * P + A + lane];
+      const float zo = (aj - mo) * __expf(-lso);
+      lo_part = -0.5f * zo * zo - lso;
+      ent_part = ls;
+    }
+    const float lp = wave_reduce_sum(lp_part);
+    const float lo = wave_reduce_sum(lo_part);
+    const float es = wave_reduce_sum(ent_part);
+    if (lane == 0) {
+      const float ratio = __expf(lp - lo);  // the logp constants cancel
+      const float ab = adv[b];
+      const float surr1 = ratio * ab;
+      const float rc = fminf(fmaxf(ratio, 1.f - clip), 1.f + clip);
+      pol += fminf(surr1, rc * ab);
+      ent += es + 0.5f * (PPO_LOG_2PI + 1.f) * A;
+      const float vb = vpred[b], ob = oldv[b], eb = etr[b];
+      const float d1 = vb - eb;
+      const float dc = fminf(fmaxf(vb - ob, -clip), clip);
+      const float d2 = ob + dc - eb;
+      val += fmaxf(d1 * d1, d2 * d2);
+    }
   }
-  wave_atomic_add(&acc[0], pol);
-  wave_atomic_add(&acc[1], ent);
-  wave_atomic_add(&acc[2], val);
+  if (lane == 0) {
+    atomicAdd(&acc[0], static_cast<double>(pol));
+    atomicAdd(&acc[1], static_cast<double>(ent));
+    atomicAdd(&acc[2], static_cast<double>(val));
+  }
 }
 
 __global__ void ppo_gauss_finalize_kernel(const double* __restrict__ acc,
@@ -158,7 +187,7 @@ torch::Tensor ppo_loss_gauss_fwd(torch::Tensor pdpi, torch::Tensor pdold,
   auto losses = torch::empty({4}, pdpi.options());
   hipStream_t stream = c10::hip::getCurrentHIPStream().stream();
   const int block = 256;
-  hipLaunchKernelGGL(ppo_gauss_fwd_kernel, dim3(elementwise_grid(B, block)),
+  hipLaunchKernelGGL(ppo_gauss_fwd_kernel, dim3(2048),
                      dim3(block), 0, stream, pdpi.data_ptr<float>(),
                      pdold.data_ptr<float>(), vpred.data_ptr<float>(),
                      oldv.data_ptr<float>(), act.data_ptr<float>(),
