@@ -53,17 +53,38 @@ constexpr int MAX_A = 32;
 constexpr int MAX_R = 32;
 constexpr int MAX_HIDDEN = 3;
 
-// LDS layout inside one shared array (float offsets)
-constexpr int X_OFF = 0;                                // [ENV_TILE][MAX_D]
-constexpr int H0_OFF = X_OFF + ENV_TILE * MAX_D;        // [ENV_TILE][MAX_H]
-constexpr int H1_OFF = H0_OFF + ENV_TILE * MAX_H;       // [ENV_TILE][MAX_H]
-constexpr int PD_OFF = H1_OFF + ENV_TILE * MAX_H;       // [ENV_TILE][2*MAX_A]
-constexpr int ACT_OFF = PD_OFF + ENV_TILE * 2 * MAX_A;  // [ENV_TILE][MAX_A]
-constexpr int XV_OFF = ACT_OFF + ENV_TILE * MAX_A;      // [ENV_TILE][MAX_R]
-constexpr int NWAVES = 4;                               // waves per block
-constexpr int PART_OFF = XV_OFF + ENV_TILE * MAX_R;     // [NWAVES][ENV_TILE][MAX_H]
-constexpr int MISC_OFF = PART_OFF + NWAVES * ENV_TILE * MAX_H;  // val, rsum, epr
-constexpr int LDS_FLOATS = MISC_OFF + 3 * ENV_TILE + ENV_TILE;  // + racc
+constexpr int NWAVES = 4;  // waves per block
+
+// The LDS image is dynamically sized to the ACTUAL dims (obs, widest
+// hidden, act): a MAX-dims static image (44 KB at the flagship shapes
+// needing only 27 KB) capped residency at 3 blocks/CU and left the
+// kernel stall-bound.  All strides are rounded to 4 floats so float4
+// LDS accesses stay 16-B aligned.
+struct LdsMap {
+  int x, h0, h1, pd, act, xv, part, val, rsum, epr, racc;  // float offsets
+  int xs, hs, total;  // x row stride, h row stride, total floats
+};
+
+DEV_INLINE LdsMap lds_map(int D, int Hmax, int A, int rank) {
+  auto r4 = [](int v) { return (v + 3) & ~3; };
+  LdsMap m;
+  m.xs = r4(D);
+  m.hs = r4(Hmax);
+  int o = 0;
+  m.x = o; o += ENV_TILE * m.xs;
+  m.h0 = o; o += ENV_TILE * m.hs;
+  m.h1 = o; o += ENV_TILE * m.hs;
+  m.pd = o; o += ENV_TILE * r4(2 * A);
+  m.act = o; o += ENV_TILE * r4(A);
+  m.xv = o; o += ENV_TILE * r4(rank);
+  m.part = o; o += NWAVES * ENV_TILE * m.hs;
+  m.val = o; o += ENV_TILE;
+  m.rsum = o; o += ENV_TILE;
+  m.epr = o; o += ENV_TILE;
+  m.racc = o; o += ENV_TILE;
+  m.total = o;
+  return m;
+}
 
 struct RolloutArgs {
   const float* params;   // rollout weight blob: TRANSPOSED layer weights
@@ -81,7 +102,7 @@ struct RolloutArgs {
   int off_W[MAX_HIDDEN], off_b[MAX_HIDDEN];
   int dims[MAX_HIDDEN + 1];
   int off_Wv, off_bv, off_Wp, off_bp;
-  int n_hidden, act_dim, activation, rank;
+  int n_hidden, act_dim, activation, rank, h_max;
   float noise, act_low, act_high, eps_explore;
   int T, E, D;
   unsigned seed;
@@ -156,11 +177,17 @@ __global__ void rollout_kernel(RolloutArgs a) {
   const int P = 2 * A;
   const int T = a.T, E = a.E;
 
-  __shared__ __attribute__((aligned(16))) float lds[LDS_FLOATS];
-  float* val_lds = &lds[MISC_OFF];
-  float* rsum_lds = &lds[MISC_OFF + ENV_TILE];
-  float* epr_lds = &lds[MISC_OFF + 2 * ENV_TILE];
-  float* racc_lds = &lds[MISC_OFF + 3 * ENV_TILE];
+  extern __shared__ __attribute__((aligned(16))) float lds[];
+  const LdsMap lm = lds_map(D, a.h_max, A, a.rank);
+  const int X_OFF = lm.x, H0_OFF = lm.h0, H1_OFF = lm.h1, PD_OFF = lm.pd;
+  const int ACT_OFF = lm.act, XV_OFF = lm.xv, PART_OFF = lm.part;
+  const int MAX_D_S = lm.xs, MAX_H_S = lm.hs;
+  const int PD_S = ((2 * A) + 3) & ~3, ACT_S = (A + 3) & ~3,
+            XV_S = (a.rank + 3) & ~3;
+  float* val_lds = &lds[lm.val];
+  float* rsum_lds = &lds[lm.rsum];
+  float* epr_lds = &lds[lm.epr];
+  float* racc_lds = &lds[lm.racc];
   __shared__ int tc_lds[ENV_TILE];
   __shared__ int done_lds[ENV_TILE];
 
@@ -181,7 +208,7 @@ __global__ void rollout_kernel(RolloutArgs a) {
   // ---- load persistent state; zero tail envs ----
   for (int e = 0; e < ENV_TILE; ++e) {
     for (int d = tid; d < D; d += NWAVES * WAVE)
-      lds[X_OFF + e * MAX_D + d] =
+      lds[X_OFF + e * MAX_D_S + d] =
           (e < nE) ? a.x[(int64_t)(e0 + e) * D + d] : 0.f;
   }
   if (tid < nE) {
@@ -217,7 +244,7 @@ __global__ void rollout_kernel(RolloutArgs a) {
       }
       #pragma unroll
       for (int e = 0; e < ENV_TILE; ++e)
-        lds[PART_OFF + (wv * ENV_TILE + e) * MAX_H + u] = acc[e];
+        lds[PART_OFF + (wv * ENV_TILE + e) * MAX_H_S + u] = acc[e];
     }
   };
 
@@ -226,11 +253,11 @@ __global__ void rollout_kernel(RolloutArgs a) {
     for (int e = 0; e < nE; ++e) {
       const int64_t base = ((int64_t)step * E + e0 + e) * D;
       for (int d = tid; d < D; d += NWAVES * WAVE)
-        out_states[base + d] = lds[X_OFF + e * MAX_D + d];
+        out_states[base + d] = lds[X_OFF + e * MAX_D_S + d];
     }
 
     // ---- policy MLP forward (K-split + combine per layer) ----
-    int in_off = X_OFF, in_stride = MAX_D, in_dim = D;
+    int in_off = X_OFF, in_stride = MAX_D_S, in_dim = D;
     for (int l = 0; l < a.n_hidden; ++l) {
       const int out_dim = a.dims[l + 1];
       const int out_off = (l & 1) ? H1_OFF : H0_OFF;
@@ -243,13 +270,13 @@ __global__ void rollout_kernel(RolloutArgs a) {
         float sum = bias[u];
         #pragma unroll
         for (int w = 0; w < NWAVES; ++w)
-          sum += lds[PART_OFF + (w * ENV_TILE + e) * MAX_H + u];
-        lds[out_off + e * MAX_H + u] =
+          sum += lds[PART_OFF + (w * ENV_TILE + e) * MAX_H_S + u];
+        lds[out_off + e * MAX_H_S + u] =
             a.activation ? tanhf(sum) : fmaxf(sum, 0.f);
       }
       __syncthreads();
       in_off = out_off;
-      in_stride = MAX_H;
+      in_stride = MAX_H_S;
       in_dim = out_dim;
     }
 
@@ -262,9 +289,9 @@ __global__ void rollout_kernel(RolloutArgs a) {
       float sum = a.params[(u == P) ? a.off_bv : a.off_bp + u];
       #pragma unroll
       for (int w = 0; w < NWAVES; ++w)
-        sum += lds[PART_OFF + (w * ENV_TILE + e) * MAX_H + u];
+        sum += lds[PART_OFF + (w * ENV_TILE + e) * MAX_H_S + u];
       if (u == P) val_lds[e] = sum;
-      else lds[PD_OFF + e * 2 * MAX_A + u] = sum;
+      else lds[PD_OFF + e * PD_S + u] = sum;
     }
     __syncthreads();
 
@@ -274,8 +301,8 @@ __global__ void rollout_kernel(RolloutArgs a) {
       const int j = tid % MAX_A;
       if (e < nE && j < A) {
         const int ge = e0 + e;
-        const float mean = lds[PD_OFF + e * 2 * MAX_A + j];
-        const float logstd = lds[PD_OFF + e * 2 * MAX_A + A + j];
+        const float mean = lds[PD_OFF + e * PD_S + j];
+        const float logstd = lds[PD_OFF + e * PD_S + A + j];
         float act = mean + __expf(logstd) * rng_normal(a.seed, ge, step, j);
         // epsilon-greedy overlay (Worker.py:149-152)
         const float u_dec = rng_uniform(a.seed, ge, step, 90001);
@@ -283,7 +310,7 @@ __global__ void rollout_kernel(RolloutArgs a) {
           const float u = rng_uniform(a.seed, ge, step, 90010 + j);
           act = a.act_low + (a.act_high - a.act_low) * u;
         }
-        lds[ACT_OFF + e * MAX_A + j] = act;
+        lds[ACT_OFF + e * ACT_S + j] = act;
       }
     }
     __syncthreads();
@@ -291,8 +318,8 @@ __global__ void rollout_kernel(RolloutArgs a) {
     // ---- write pdflat / action / value rows ----
     for (int e = 0; e < nE; ++e) {
       const int64_t row = (int64_t)step * E + e0 + e;
-      if (tid < P) out_pdflats[row * P + tid] = lds[PD_OFF + e * 2 * MAX_A + tid];
-      if (tid < A) out_actions[row * A + tid] = lds[ACT_OFF + e * MAX_A + tid];
+      if (tid < P) out_pdflats[row * P + tid] = lds[PD_OFF + e * PD_S + tid];
+      if (tid < A) out_actions[row * A + tid] = lds[ACT_OFF + e * ACT_S + tid];
     }
     if (tid < nE) out_values[(int64_t)step * E + e0 + tid] = val_lds[tid];
     if (tid < ENV_TILE) racc_lds[tid] = 0.f;
@@ -315,15 +342,15 @@ __global__ void rollout_kernel(RolloutArgs a) {
         for (; k + 4 <= k1q; k += 4) {
           const float4 v4 = *reinterpret_cast<const float4*>(Vrow + k);
           const float4 x4 =
-              *reinterpret_cast<const float4*>(&lds[X_OFF + e * MAX_D + k]);
+              *reinterpret_cast<const float4*>(&lds[X_OFF + e * MAX_D_S + k]);
           accv += v4.x * x4.x + v4.y * x4.y + v4.z * x4.z + v4.w * x4.w;
         }
-        for (; k < k1q; ++k) accv += Vrow[k] * lds[X_OFF + e * MAX_D + k];
+        for (; k < k1q; ++k) accv += Vrow[k] * lds[X_OFF + e * MAX_D_S + k];
       }
       // butterfly-reduce over the k-quarter lanes (bits 4 and 5)
       accv += __shfl_xor(accv, 16, WAVE);
       accv += __shfl_xor(accv, 32, WAVE);
-      if (kq == 0 && rr < a.rank) lds[XV_OFF + e * MAX_R + rr] = accv;
+      if (kq == 0 && rr < a.rank) lds[XV_OFF + e * XV_S + rr] = accv;
     }
     __syncthreads();
 
@@ -353,19 +380,19 @@ __global__ void rollout_kernel(RolloutArgs a) {
         const float uv = env_U[(int64_t)rr * D + d];
         #pragma unroll
         for (int e = 0; e < ENV_TILE; ++e)
-          low[e] += lds[XV_OFF + e * MAX_R + rr] * uv;
+          low[e] += lds[XV_OFF + e * XV_S + rr] * uv;
       }
       #pragma unroll 2
       for (int j = 0; j < A; ++j) {
         const float bvv = env_B[(int64_t)j * D + d];
         #pragma unroll
         for (int e = 0; e < ENV_TILE; ++e)
-          ain[e] += lds[ACT_OFF + e * MAX_A + j] * bvv;
+          ain[e] += lds[ACT_OFF + e * ACT_S + j] * bvv;
       }
       for (int e = 0; e < nE; ++e) {
-        const float xn = tanhf(lds[X_OFF + e * MAX_D + d] * dd + low[e] +
+        const float xn = tanhf(lds[X_OFF + e * MAX_D_S + d] * dd + low[e] +
                                ain[e] + a.noise * nz[e]);
-        lds[X_OFF + e * MAX_D + d] = xn;
+        lds[X_OFF + e * MAX_D_S + d] = xn;
         racc[e] += xn * xn;
       }
     }
@@ -398,7 +425,7 @@ __global__ void rollout_kernel(RolloutArgs a) {
     for (int e = 0; e < nE; ++e) {
       if (done_lds[e]) {
         for (int d = tid; d < D; d += NWAVES * WAVE)
-          lds[X_OFF + e * MAX_D + d] =
+          lds[X_OFF + e * MAX_D_S + d] =
               0.1f * rng_normal(a.seed, e0 + e, step, 5000 + d);
       }
     }
@@ -407,7 +434,7 @@ __global__ void rollout_kernel(RolloutArgs a) {
 
   // ---- bootstrap value V(x_T): trunk + value head ----
   {
-    int in_off = X_OFF, in_stride = MAX_D, in_dim = D;
+    int in_off = X_OFF, in_stride = MAX_D_S, in_dim = D;
     for (int l = 0; l < a.n_hidden; ++l) {
       const int out_dim = a.dims[l + 1];
       const int out_off = (l & 1) ? H1_OFF : H0_OFF;
@@ -420,13 +447,13 @@ __global__ void rollout_kernel(RolloutArgs a) {
         float sum = bias[u];
         #pragma unroll
         for (int w = 0; w < NWAVES; ++w)
-          sum += lds[PART_OFF + (w * ENV_TILE + e) * MAX_H + u];
-        lds[out_off + e * MAX_H + u] =
+          sum += lds[PART_OFF + (w * ENV_TILE + e) * MAX_H_S + u];
+        lds[out_off + e * MAX_H_S + u] =
             a.activation ? tanhf(sum) : fmaxf(sum, 0.f);
       }
       __syncthreads();
       in_off = out_off;
-      in_stride = MAX_H;
+      in_stride = MAX_H_S;
       in_dim = out_dim;
     }
     // value head: wave wv reduces env wv over lanes
@@ -443,7 +470,7 @@ __global__ void rollout_kernel(RolloutArgs a) {
   // ---- persist env state ----
   for (int e = 0; e < nE; ++e) {
     for (int d = tid; d < D; d += NWAVES * WAVE)
-      a.x[(int64_t)(e0 + e) * D + d] = lds[X_OFF + e * MAX_D + d];
+      a.x[(int64_t)(e0 + e) * D + d] = lds[X_OFF + e * MAX_D_S + d];
   }
   if (tid < nE) {
     a.epr[e0 + tid] = epr_lds[tid];
@@ -536,6 +563,10 @@ std::vector<torch::Tensor> rollout_run(
   a.off_Wp = static_cast<int>(offsets[2 * n_hidden + 2]);
   a.off_bp = static_cast<int>(offsets[2 * n_hidden + 3]);
   a.n_hidden = n_hidden;
+  int h_max = 0;
+  for (int l = 1; l <= n_hidden; ++l)
+    h_max = std::max(h_max, static_cast<int>(dims[l]));
+  a.h_max = h_max;
   a.act_dim = A;
   a.activation = static_cast<int>(activation);
   a.rank = static_cast<int>(rank);
@@ -560,7 +591,14 @@ std::vector<torch::Tensor> rollout_run(
 
   hipStream_t stream = c10::hip::getCurrentHIPStream().stream();
   const int grid = static_cast<int>((E + ENV_TILE - 1) / ENV_TILE);
-  hipLaunchKernelGGL(rollout_kernel, dim3(grid), dim3(NWAVES * WAVE), 0, stream, a);
+  auto r4 = [](int v) { return (v + 3) & ~3; };
+  const size_t lds_bytes =
+      (ENV_TILE * (r4((int)D) + 2 * r4(h_max) + r4(2 * A) + r4(A) +
+                   r4((int)rank) + 4) +
+       NWAVES * ENV_TILE * r4(h_max)) *
+      sizeof(float);
+  hipLaunchKernelGGL(rollout_kernel, dim3(grid), dim3(NWAVES * WAVE),
+                     lds_bytes, stream, a);
 
   // carve views out of the blob
   int64_t o = 0;
