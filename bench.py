@@ -43,6 +43,13 @@ PRESETS = {
         NUM_ENVS=64, MAX_EPOCH_STEPS=100, DTYPE="float32",
     ),
     # BASELINE.json config 5: wide MFMA-bound MLP
+    # BASELINE.json config 4: Humanoid-shaped large-batch — 65k-step
+    # rollouts (GAE HIP scan stress) with 4096-sample minibatched updates
+    "largebatch": dict(
+        GAME="Humanoid-v4", HIDDEN_SIZES=(64, 64), ACTIVATION="tanh",
+        NUM_ENVS=64, MAX_EPOCH_STEPS=65536, MINIBATCH_SIZE=4096,
+        DTYPE="float32",
+    ),
     # BASELINE.json config 5: wide MFMA-bound MLP, bf16 compute (as named)
     "wide4096": dict(
         GAME="Wide-4096", HIDDEN_SIZES=(4096, 4096, 4096, 4096),

@@ -102,6 +102,10 @@ class DPPOConfig:
     MAX_ROLLOUT_RETRIES: int = 16     # rollouts per round before a rank reports
                                       # an invalid batch (no completed episode,
                                       # Worker.py:135 push-guard analog)
+    MINIBATCH_SIZE: int = 0           # 0 = full-batch updates (the reference's
+                                      # scheme, Chief.py:64); >0 = sequential
+                                      # minibatch chunks per update step
+                                      # (BASELINE config 4's 4096-minibatch)
 
     def __post_init__(self) -> None:
         if isinstance(self.HIDDEN_SIZES, list):

@@ -505,6 +505,9 @@ class DPPOEngine:
         loss values are materialized here."""
         for g in self.optimizer.param_groups:
             g["lr"] = self.cfg.LEARNING_RATE * l_mul
+        if self.cfg.MINIBATCH_SIZE > 0:
+            self._update_minibatched(batch, l_mul)
+            return
         if self._can_fuse_update():
             # Graph capture of the RCCL all-reduce is unexercised on this
             # pool's multi-GPU boxes; keep multi-rank runs on the plain
@@ -524,6 +527,48 @@ class DPPOEngine:
             losses["total_loss"].backward()
             self.comm.allreduce_mean_(self.flat_pi.flat_grad)
             self.optimizer.step()
+
+    def _update_minibatched(self, batch: RolloutBatch, l_mul: float) -> None:
+        """Minibatched update steps (BASELINE config 4): each of the
+        UPDATE_STEPS epochs walks the batch in sequential MINIBATCH_SIZE
+        chunks with a gradient all-reduce + Adam step per chunk.  The
+        reference itself is full-batch (Chief.py:64); chunks are
+        sequential (no shuffle) so every rank takes the same number of
+        steps and collectives stay aligned."""
+        mb = self.cfg.MINIBATCH_SIZE
+        B = batch.states.shape[0]
+        fuse = self._can_fuse_update()
+        clip = self.cfg.CLIP_PARAM * l_mul
+        if fuse and getattr(self, "_clip_dev", None) is not None:
+            self._clip_dev.fill_(clip)
+        for _ in range(self.cfg.UPDATE_STEPS):
+            for o in range(0, B, mb):
+                n = min(mb, B - o)
+                sub = RolloutBatch(
+                    states=batch.states.narrow(0, o, n),
+                    actions=batch.actions.narrow(0, o, n),
+                    adv=batch.adv.narrow(0, o, n),
+                    etr=batch.etr.narrow(0, o, n),
+                    oldflat=batch.oldflat.narrow(0, o, n),
+                    oldv=batch.oldv.narrow(0, o, n),
+                    cur_lr=batch.cur_lr,
+                    ep_count=batch.ep_count, ep_sum=batch.ep_sum,
+                    ep_sumsq=batch.ep_sumsq, ep_min=batch.ep_min,
+                    ep_max=batch.ep_max, valid=batch.valid,
+                )
+                self.flat_pi.zero_grad()
+                if fuse:
+                    acts, a_views, v, pdflat = self._fused_forward(sub.states)
+                    self._fused_backward(
+                        sub.states, acts, a_views, v, pdflat,
+                        sub.oldflat, sub.oldv, sub.actions,
+                        sub.adv, sub.etr, clip,
+                    )
+                else:
+                    losses = self._losses(sub, l_mul)
+                    losses["total_loss"].backward()
+                self.comm.allreduce_mean_(self.flat_pi.flat_grad)
+                self.optimizer.step()
 
     def _clip_dev_or_empty(self) -> torch.Tensor:
         t = getattr(self, "_clip_dev", None)

@@ -133,3 +133,15 @@ def test_worker_chief_api():
     # Worker.act parity: (action, pred_v)
     act, predv = w.act(torch.randn(obs_dim).numpy())
     assert isinstance(predv, float)
+
+
+def test_minibatched_update():
+    """BASELINE config 4 scheme: MINIBATCH_SIZE chunks per update epoch."""
+    cfg = small_cfg(MINIBATCH_SIZE=32, NUM_ENVS=8, MAX_EPOCH_STEPS=16)
+    eng = DPPOEngine(cfg, comm=Comm(device="cpu"))
+    p0 = eng.flat_pi.flat_param.detach().clone()
+    stats, _ = eng.train_round()
+    assert math.isfinite(stats["total_loss"])
+    assert not torch.allclose(p0, eng.flat_pi.flat_param.detach())
+    # 16*8=128 samples / 32 per chunk * 4 epochs = 16 optimizer steps
+    assert eng.optimizer.state_dict()["state"][0]["step"] == 16
