@@ -51,7 +51,8 @@ std::vector<torch::Tensor> rollout_run(
     torch::Tensor envblob, int64_t rank, torch::Tensor horizons,
     double noise, double act_low, double act_high, double eps_explore,
     torch::Tensor x, torch::Tensor t, torch::Tensor epr,
-    int64_t T, int64_t act_dim, int64_t seed, torch::Tensor out_buf);
+    int64_t T, int64_t act_dim, int64_t seed, torch::Tensor out_buf,
+    int64_t ablate);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("gae_scan", &gae_scan,

@@ -106,6 +106,8 @@ struct RolloutArgs {
   float noise, act_low, act_high, eps_explore;
   int T, E, D;
   unsigned seed;
+  int ablate;  // perf-ablation bitmask (0 in production): 1=trunk,
+               // 2=heads, 4=sampling, 8=env dynamics, 16=buffer writes
 };
 
 // ---- counter-based 32-bit RNG --------------------------------------------
@@ -250,14 +252,17 @@ __global__ void rollout_kernel(RolloutArgs a) {
 
   for (int step = 0; step < T; ++step) {
     // ---- write current obs ----
-    for (int e = 0; e < nE; ++e) {
-      const int64_t base = ((int64_t)step * E + e0 + e) * D;
-      for (int d = tid; d < D; d += NWAVES * WAVE)
-        out_states[base + d] = lds[X_OFF + e * MAX_D_S + d];
+    if (!(a.ablate & 16)) {
+      for (int e = 0; e < nE; ++e) {
+        const int64_t base = ((int64_t)step * E + e0 + e) * D;
+        for (int d = tid; d < D; d += NWAVES * WAVE)
+          out_states[base + d] = lds[X_OFF + e * MAX_D_S + d];
+      }
     }
 
     // ---- policy MLP forward (K-split + combine per layer) ----
     int in_off = X_OFF, in_stride = MAX_D_S, in_dim = D;
+    if (!(a.ablate & 1))
     for (int l = 0; l < a.n_hidden; ++l) {
       const int out_dim = a.dims[l + 1];
       const int out_off = (l & 1) ? H1_OFF : H0_OFF;
@@ -279,24 +284,31 @@ __global__ void rollout_kernel(RolloutArgs a) {
       in_stride = MAX_H_S;
       in_dim = out_dim;
     }
+    if (a.ablate & 1) {
+      in_off = (a.n_hidden & 1) ? H0_OFF : H1_OFF;
+      in_stride = MAX_H_S;
+      in_dim = a.dims[a.n_hidden];
+    }
 
     // ---- heads (u < P: pd params; u == P: value), no activation ----
-    layer_kpart(a.params + a.off_Wp, P, a.off_Wv, in_off, in_stride, in_dim,
-                P + 1, true);
-    __syncthreads();
-    for (int idx = tid; idx < ENV_TILE * (P + 1); idx += NWAVES * WAVE) {
-      const int e = idx / (P + 1), u = idx % (P + 1);
-      float sum = a.params[(u == P) ? a.off_bv : a.off_bp + u];
-      #pragma unroll
-      for (int w = 0; w < NWAVES; ++w)
-        sum += lds[PART_OFF + (w * ENV_TILE + e) * MAX_H_S + u];
-      if (u == P) val_lds[e] = sum;
-      else lds[PD_OFF + e * PD_S + u] = sum;
+    if (!(a.ablate & 2)) {
+      layer_kpart(a.params + a.off_Wp, P, a.off_Wv, in_off, in_stride, in_dim,
+                  P + 1, true);
+      __syncthreads();
+      for (int idx = tid; idx < ENV_TILE * (P + 1); idx += NWAVES * WAVE) {
+        const int e = idx / (P + 1), u = idx % (P + 1);
+        float sum = a.params[(u == P) ? a.off_bv : a.off_bp + u];
+        #pragma unroll
+        for (int w = 0; w < NWAVES; ++w)
+          sum += lds[PART_OFF + (w * ENV_TILE + e) * MAX_H_S + u];
+        if (u == P) val_lds[e] = sum;
+        else lds[PD_OFF + e * PD_S + u] = sum;
+      }
+      __syncthreads();
     }
-    __syncthreads();
 
     // ---- sample actions (threads = (env, dim) pairs) ----
-    {
+    if (!(a.ablate & 4)) {
       const int e = tid / MAX_A;   // 8 groups x 32 dims
       const int j = tid % MAX_A;
       if (e < nE && j < A) {
@@ -316,15 +328,18 @@ __global__ void rollout_kernel(RolloutArgs a) {
     __syncthreads();
 
     // ---- write pdflat / action / value rows ----
-    for (int e = 0; e < nE; ++e) {
-      const int64_t row = (int64_t)step * E + e0 + e;
-      if (tid < P) out_pdflats[row * P + tid] = lds[PD_OFF + e * PD_S + tid];
-      if (tid < A) out_actions[row * A + tid] = lds[ACT_OFF + e * ACT_S + tid];
+    if (!(a.ablate & 16)) {
+      for (int e = 0; e < nE; ++e) {
+        const int64_t row = (int64_t)step * E + e0 + e;
+        if (tid < P) out_pdflats[row * P + tid] = lds[PD_OFF + e * PD_S + tid];
+        if (tid < A) out_actions[row * A + tid] = lds[ACT_OFF + e * ACT_S + tid];
+      }
+      if (tid < nE) out_values[(int64_t)step * E + e0 + tid] = val_lds[tid];
     }
-    if (tid < nE) out_values[(int64_t)step * E + e0 + tid] = val_lds[tid];
     if (tid < ENV_TILE) racc_lds[tid] = 0.f;
 
     // ---- env low-rank projection: wave wv handles env wv; lanes split
+    if (!(a.ablate & 8))
     // as (rr, k-quarter) so all 64 lanes stream and the per-lane load
     // chain is 4x shorter (the 16-lane serial version stalled the whole
     // block at the next barrier) ----
@@ -358,6 +373,7 @@ __global__ void rollout_kernel(RolloutArgs a) {
     float racc[ENV_TILE];
     #pragma unroll
     for (int e = 0; e < ENV_TILE; ++e) racc[e] = 0.f;
+    if (!(a.ablate & 8))
     for (int d = tid; d < D; d += NWAVES * WAVE) {
       const float dd = env_d[d];
       // one Box-Muller pair per (even env, d) feeds two envs
@@ -527,7 +543,8 @@ std::vector<torch::Tensor> rollout_run(
     torch::Tensor envblob, int64_t rank, torch::Tensor horizons,
     double noise, double act_low, double act_high, double eps_explore,
     torch::Tensor x, torch::Tensor t, torch::Tensor epr,
-    int64_t T, int64_t act_dim, int64_t seed, torch::Tensor out_buf) {
+    int64_t T, int64_t act_dim, int64_t seed, torch::Tensor out_buf,
+    int64_t ablate) {
   // offsets: [W0, b0, W1, b1, ..., Wv, bv, Wp, bp] into params
   // dims: [obs, h1, ..., hN]
   const int64_t E = x.size(0);
@@ -578,6 +595,7 @@ std::vector<torch::Tensor> rollout_run(
   a.E = static_cast<int>(E);
   a.D = static_cast<int>(D);
   a.seed = static_cast<unsigned>(seed & 0xFFFFFFFFULL);
+  a.ablate = static_cast<int>(ablate);
 
   const int P = 2 * A;
   const int64_t n_out = T * E * (D + P + A + 3) + E + 5;

@@ -297,7 +297,7 @@ class DPPOEngine:
             1 if c.ACTIVATION == "tanh" else 0,
             env.blob, env.rank_eff, env.horizons_i32,
             float(env.NOISE), low, high, float(eps),
-            env.x, env.t, self.epr, T, A, seed, self._rollout_out,
+            env.x, env.t, self.epr, T, A, seed, self._rollout_out, 0,
         )
         self.obs = env.x  # updated in place by the kernel
 

@@ -46,7 +46,7 @@ def run_kernel(eng, T=32, eps=0.0, noise=None, seed=1234):
         env.blob, env.rank_eff, env.horizons_i32,
         float(env.NOISE), low, high, float(eps),
         env.x, env.t, eng.epr, T, eng.act_space.shape[0], seed,
-        torch.empty(0, device="cuda"),
+        torch.empty(0, device="cuda"), 0,
     )
 
 
