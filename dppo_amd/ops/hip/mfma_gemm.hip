@@ -188,7 +188,7 @@ __global__ void gemm_fwd_kernel(FwdArgs a) {
             } else {
               x = acc[t][r] + bv;
               if (a.activation == 0) x = fmaxf(x, 0.f);
-              else if (a.activation == 1) x = tanhf(x);
+              else if (a.activation == 1) x = fast_tanhf(x);
             }
             if (a.heads) {
               if (col == a.N - 1) a.v[row] = x;

@@ -107,7 +107,8 @@ struct RolloutArgs {
   int T, E, D;
   unsigned seed;
   int ablate;  // perf-ablation bitmask (0 in production): 1=trunk,
-               // 2=heads, 4=sampling, 8=env dynamics, 16=buffer writes
+               // 2=heads, 4=sampling, 8=env dynamics, 16=buffer writes,
+               // 32=env noise only, 64=low/ain inner products only
 };
 
 // ---- counter-based 32-bit RNG --------------------------------------------
@@ -237,8 +238,24 @@ __global__ void rollout_kernel(RolloutArgs a) {
       const bool is_v = heads && (u == out_dim - 1);
       const float* Wcol = is_v ? (a.params + off_Wv_u) : (W + u);
       const int stride = is_v ? 1 : w_cols;
-      #pragma unroll 4
-      for (int k = k0; k < k1; ++k) {
+      // 4 k-steps per iteration: one float4 LDS broadcast per env feeds
+      // 4 FMAs (scalar b32 broadcasts measured as a main trunk cost),
+      // and the 4 independent W loads pipeline.
+      int k = k0;
+      #pragma unroll 2
+      for (; k + 4 <= k1; k += 4) {
+        const float w0 = Wcol[(int64_t)(k + 0) * stride];
+        const float w1 = Wcol[(int64_t)(k + 1) * stride];
+        const float w2 = Wcol[(int64_t)(k + 2) * stride];
+        const float w3 = Wcol[(int64_t)(k + 3) * stride];
+        #pragma unroll
+        for (int e = 0; e < ENV_TILE; ++e) {
+          const float4 x4 = *reinterpret_cast<const float4*>(
+              &lds[in_off + e * in_stride + k]);
+          acc[e] += w0 * x4.x + w1 * x4.y + w2 * x4.z + w3 * x4.w;
+        }
+      }
+      for (; k < k1; ++k) {
         const float w = Wcol[(int64_t)k * stride];
         #pragma unroll
         for (int e = 0; e < ENV_TILE; ++e)
@@ -277,7 +294,7 @@ __global__ void rollout_kernel(RolloutArgs a) {
         for (int w = 0; w < NWAVES; ++w)
           sum += lds[PART_OFF + (w * ENV_TILE + e) * MAX_H_S + u];
         lds[out_off + e * MAX_H_S + u] =
-            a.activation ? tanhf(sum) : fmaxf(sum, 0.f);
+            a.activation ? fast_tanhf(sum) : fmaxf(sum, 0.f);
       }
       __syncthreads();
       in_off = out_off;
@@ -379,10 +396,14 @@ __global__ void rollout_kernel(RolloutArgs a) {
       // one Box-Muller pair per (even env, d) feeds two envs
       float nz[ENV_TILE];
       #pragma unroll
-      for (int q = 0; q < ENV_TILE / 2; ++q) {
-        const float2 p = rng_normal2(a.seed, e0 + 2 * q, step, 1000 + d);
-        nz[2 * q] = p.x;
-        nz[2 * q + 1] = p.y;
+      for (int q = 0; q < ENV_TILE; ++q) nz[q] = 0.f;
+      if (!(a.ablate & 32)) {
+        #pragma unroll
+        for (int q = 0; q < ENV_TILE / 2; ++q) {
+          const float2 p = rng_normal2(a.seed, e0 + 2 * q, step, 1000 + d);
+          nz[2 * q] = p.x;
+          nz[2 * q + 1] = p.y;
+        }
       }
       // U[rr][d] / B[j][d] row loads: lane == d, fully coalesced, ONE
       // load per rr/j shared by all four envs (the previous transposed
@@ -391,23 +412,51 @@ __global__ void rollout_kernel(RolloutArgs a) {
       float low[ENV_TILE], ain[ENV_TILE];
       #pragma unroll
       for (int e = 0; e < ENV_TILE; ++e) low[e] = ain[e] = 0.f;
-      #pragma unroll 2
-      for (int rr = 0; rr < a.rank; ++rr) {
-        const float uv = env_U[(int64_t)rr * D + d];
-        #pragma unroll
-        for (int e = 0; e < ENV_TILE; ++e)
-          low[e] += lds[XV_OFF + e * XV_S + rr] * uv;
-      }
-      #pragma unroll 2
-      for (int j = 0; j < A; ++j) {
-        const float bvv = env_B[(int64_t)j * D + d];
-        #pragma unroll
-        for (int e = 0; e < ENV_TILE; ++e)
-          ain[e] += lds[ACT_OFF + e * ACT_S + j] * bvv;
+      if (!(a.ablate & 64)) {
+        // float4 LDS reads: 4x fewer LDS instructions than scalar
+        // broadcasts (the scalar form measured 2.5 ms of the rollout)
+        int rr = 0;
+        for (; rr + 4 <= a.rank; rr += 4) {
+          const float u0 = env_U[(int64_t)(rr + 0) * D + d];
+          const float u1 = env_U[(int64_t)(rr + 1) * D + d];
+          const float u2 = env_U[(int64_t)(rr + 2) * D + d];
+          const float u3 = env_U[(int64_t)(rr + 3) * D + d];
+          #pragma unroll
+          for (int e = 0; e < ENV_TILE; ++e) {
+            const float4 xv4 =
+                *reinterpret_cast<const float4*>(&lds[XV_OFF + e * XV_S + rr]);
+            low[e] += xv4.x * u0 + xv4.y * u1 + xv4.z * u2 + xv4.w * u3;
+          }
+        }
+        for (; rr < a.rank; ++rr) {
+          const float uv = env_U[(int64_t)rr * D + d];
+          #pragma unroll
+          for (int e = 0; e < ENV_TILE; ++e)
+            low[e] += lds[XV_OFF + e * XV_S + rr] * uv;
+        }
+        int j = 0;
+        for (; j + 4 <= A; j += 4) {
+          const float b0 = env_B[(int64_t)(j + 0) * D + d];
+          const float b1 = env_B[(int64_t)(j + 1) * D + d];
+          const float b2 = env_B[(int64_t)(j + 2) * D + d];
+          const float b3 = env_B[(int64_t)(j + 3) * D + d];
+          #pragma unroll
+          for (int e = 0; e < ENV_TILE; ++e) {
+            const float4 a4 =
+                *reinterpret_cast<const float4*>(&lds[ACT_OFF + e * ACT_S + j]);
+            ain[e] += a4.x * b0 + a4.y * b1 + a4.z * b2 + a4.w * b3;
+          }
+        }
+        for (; j < A; ++j) {
+          const float bvv = env_B[(int64_t)j * D + d];
+          #pragma unroll
+          for (int e = 0; e < ENV_TILE; ++e)
+            ain[e] += lds[ACT_OFF + e * ACT_S + j] * bvv;
+        }
       }
       for (int e = 0; e < nE; ++e) {
-        const float xn = tanhf(lds[X_OFF + e * MAX_D_S + d] * dd + low[e] +
-                               ain[e] + a.noise * nz[e]);
+        const float xn = fast_tanhf(lds[X_OFF + e * MAX_D_S + d] * dd +
+                                    low[e] + ain[e] + a.noise * nz[e]);
         lds[X_OFF + e * MAX_D_S + d] = xn;
         racc[e] += xn * xn;
       }
@@ -465,7 +514,7 @@ __global__ void rollout_kernel(RolloutArgs a) {
         for (int w = 0; w < NWAVES; ++w)
           sum += lds[PART_OFF + (w * ENV_TILE + e) * MAX_H_S + u];
         lds[out_off + e * MAX_H_S + u] =
-            a.activation ? tanhf(sum) : fmaxf(sum, 0.f);
+            a.activation ? fast_tanhf(sum) : fmaxf(sum, 0.f);
       }
       __syncthreads();
       in_off = out_off;

@@ -25,9 +25,9 @@ def run(mask):
                     eng._rollout_out, mask)
 
 for name, mask in [("full", 0), ("-trunk", 1), ("-heads", 2), ("-sample", 4),
-                   ("-env", 8), ("-writes", 16), ("-trunk-heads", 3),
-                   ("only-writes(env+trunk+heads+sample off)", 15),
-                   ("nothing(31)", 31)]:
+                   ("-env", 8), ("-writes", 16),
+                   ("-env-noise", 32), ("-env-lowain", 64),
+                   ("-env-noise-lowain", 96), ("nothing(127)", 127)]:
     for _ in range(2): run(mask)
     torch.cuda.synchronize()
     t0 = time.perf_counter()
