@@ -369,7 +369,6 @@ __global__ void dw_reduce_kernel(const float* __restrict__ slab,
   const int64_t n = out_dim * in_dim;
   const int s0 = blockIdx.y * DW_RED_CHUNK;
   const int s1 = min(splits, s0 + DW_RED_CHUNK);
-  const bool first = (blockIdx.y == 0);
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
        i += (int64_t)gridDim.x * blockDim.x) {
     float acc = 0.f;
@@ -380,7 +379,6 @@ __global__ void dw_reduce_kernel(const float* __restrict__ slab,
                      : &dW[i];
     atomicAdd(dst, acc);
   }
-  if (first) return;  // db handled below by chunked atomics too
 }
 
 __global__ void db_reduce_kernel(const float* __restrict__ db_slab,

@@ -11,8 +11,8 @@
 // T times, with env state resident in LDS, so a whole rollout costs ONE
 // launch and all policy/env weights stream from L1/L2.
 //
-// Decomposition: one block of FOUR waves (256 threads) owns ENV_TILE=4
-// envs for the full T-step loop (E/4 blocks ≈ 1024 at the bench config).
+// Decomposition: one block of FOUR waves (256 threads) owns ENV_TILE=8
+// envs for the full T-step loop (E/8 blocks ≈ 4096 at the bench config).
 // The four waves K-SPLIT each layer's dot products (wave w accumulates
 // the k-quarter of every unit, partials combined through LDS), so the
 // chip runs 4096 waves (≈4 waves/SIMD) at UNCHANGED weight traffic —

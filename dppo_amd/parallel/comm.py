@@ -26,7 +26,7 @@ from __future__ import annotations
 
 import datetime
 import os
-from typing import List, Optional, Sequence
+from typing import List, Optional
 
 import torch
 import torch.distributed as dist

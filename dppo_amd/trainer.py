@@ -39,15 +39,14 @@ from __future__ import annotations
 
 import math
 import os
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 from typing import Dict, List, Optional, Tuple
 
 import torch
 
 from .config import DPPOConfig, game_spaces
-from .distributions import make_pdtype
 from .envs.synthetic import BatchedSyntheticEnv, make_env
-from .models.mlp import PolicyValueMLP
+from .models.mlp import PolicyValueMLP  # noqa: F401 (public engine surface)
 from .ops import gae_advantages, ppo_losses, PPOLossCoeffs
 from .parallel.comm import Comm, FlatBuffers
 from .utils.logging import ScalarLogger
