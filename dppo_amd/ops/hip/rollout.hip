@@ -168,7 +168,7 @@ DEV_INLINE float u2f_mono(unsigned u) {
 
 // ---- the rollout kernel ---------------------------------------------------
 
-__launch_bounds__(NWAVES * 64)
+__launch_bounds__(NWAVES * 64, 4)  // 4 waves/SIMD: cap VGPRs at 128
 __global__ void rollout_kernel(RolloutArgs a) {
   const int tid = threadIdx.x;      // block = 4 waves of 64
   const int lane = tid & (WAVE - 1);
@@ -277,6 +277,39 @@ __global__ void rollout_kernel(RolloutArgs a) {
       }
     }
 
+    // ---- env low-rank projection, merged into the trunk phase: both
+    // only READ x, and the trunk's first barrier publishes the xv slab
+    // long before the env phase consumes it ----
+    // ---- env low-rank projection: wave wv handles env wv; lanes split
+    if (!(a.ablate & 8))
+    // as (rr, k-quarter) so all 64 lanes stream and the per-lane load
+    // chain is 4x shorter (the 16-lane serial version stalled the whole
+    // block at the next barrier) ----
+    for (int e = wv; e < nE; e += NWAVES) {
+      const int rr = lane & 15;
+      const int kq = lane >> 4;  // 4 k-quarters
+      const int kq_len = ((D / 4 + 3) & ~3);
+      const int k0q = kq * kq_len;
+      const int k1q = min(D, k0q + kq_len);
+      float accv = 0.f;
+      if (rr < a.rank) {
+        const float* Vrow = env_Vt + (int64_t)rr * D;
+        int k = k0q;
+        #pragma unroll 2
+        for (; k + 4 <= k1q; k += 4) {
+          const float4 v4 = *reinterpret_cast<const float4*>(Vrow + k);
+          const float4 x4 =
+              *reinterpret_cast<const float4*>(&lds[X_OFF + e * MAX_D_S + k]);
+          accv += v4.x * x4.x + v4.y * x4.y + v4.z * x4.z + v4.w * x4.w;
+        }
+        for (; k < k1q; ++k) accv += Vrow[k] * lds[X_OFF + e * MAX_D_S + k];
+      }
+      // butterfly-reduce over the k-quarter lanes (bits 4 and 5)
+      accv += __shfl_xor(accv, 16, WAVE);
+      accv += __shfl_xor(accv, 32, WAVE);
+      if (kq == 0 && rr < a.rank) lds[XV_OFF + e * XV_S + rr] = accv;
+    }
+
     // ---- policy MLP forward (K-split + combine per layer) ----
     int in_off = X_OFF, in_stride = MAX_D_S, in_dim = D;
     if (!(a.ablate & 1))
@@ -342,9 +375,12 @@ __global__ void rollout_kernel(RolloutArgs a) {
         lds[ACT_OFF + e * ACT_S + j] = act;
       }
     }
+    if (tid < ENV_TILE) racc_lds[tid] = 0.f;
     __syncthreads();
 
-    // ---- write pdflat / action / value rows ----
+    // ---- env state update + reward partials (threads split d) ----
+    // ---- pdflat/action/value buffer writes, folded into the env phase
+    // (their inputs were published by the sampling barrier) ----
     if (!(a.ablate & 16)) {
       for (int e = 0; e < nE; ++e) {
         const int64_t row = (int64_t)step * E + e0 + e;
@@ -353,40 +389,6 @@ __global__ void rollout_kernel(RolloutArgs a) {
       }
       if (tid < nE) out_values[(int64_t)step * E + e0 + tid] = val_lds[tid];
     }
-    if (tid < ENV_TILE) racc_lds[tid] = 0.f;
-
-    // ---- env low-rank projection: wave wv handles env wv; lanes split
-    if (!(a.ablate & 8))
-    // as (rr, k-quarter) so all 64 lanes stream and the per-lane load
-    // chain is 4x shorter (the 16-lane serial version stalled the whole
-    // block at the next barrier) ----
-    for (int e = wv; e < nE; e += NWAVES) {
-      const int rr = lane & 15;
-      const int kq = lane >> 4;  // 4 k-quarters
-      const int kq_len = ((D / 4 + 3) & ~3);
-      const int k0q = kq * kq_len;
-      const int k1q = min(D, k0q + kq_len);
-      float accv = 0.f;
-      if (rr < a.rank) {
-        const float* Vrow = env_Vt + (int64_t)rr * D;
-        int k = k0q;
-        #pragma unroll 2
-        for (; k + 4 <= k1q; k += 4) {
-          const float4 v4 = *reinterpret_cast<const float4*>(Vrow + k);
-          const float4 x4 =
-              *reinterpret_cast<const float4*>(&lds[X_OFF + e * MAX_D_S + k]);
-          accv += v4.x * x4.x + v4.y * x4.y + v4.z * x4.z + v4.w * x4.w;
-        }
-        for (; k < k1q; ++k) accv += Vrow[k] * lds[X_OFF + e * MAX_D_S + k];
-      }
-      // butterfly-reduce over the k-quarter lanes (bits 4 and 5)
-      accv += __shfl_xor(accv, 16, WAVE);
-      accv += __shfl_xor(accv, 32, WAVE);
-      if (kq == 0 && rr < a.rank) lds[XV_OFF + e * XV_S + rr] = accv;
-    }
-    __syncthreads();
-
-    // ---- env state update + reward partials (threads split d) ----
     float racc[ENV_TILE];
     #pragma unroll
     for (int e = 0; e < ENV_TILE; ++e) racc[e] = 0.f;
