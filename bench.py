@@ -35,7 +35,7 @@ PRESETS = {
     # BASELINE.json config 3 (the headline): Humanoid-shaped MLP DPPO
     "humanoid": dict(
         GAME="Humanoid-v4", HIDDEN_SIZES=(64, 64), ACTIVATION="tanh",
-        NUM_ENVS=32768, MAX_EPOCH_STEPS=64, DTYPE="float32",
+        NUM_ENVS=65536, MAX_EPOCH_STEPS=64, DTYPE="float32",
     ),
     # BASELINE.json config 2: HalfCheetah-shaped, 64 envs, 1 GPU
     "halfcheetah": dict(
