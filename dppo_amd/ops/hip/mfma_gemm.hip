@@ -11,12 +11,22 @@
 //               forward outputs.  X tiles are LDS-staged (the MFMA A
 //               fragment is a column read, lane = row); Wt streams from
 //               L2 coalesced (reused by every row block).
-//   gemm_heads: same, N = 2A+1, no activation; epilogue splits columns
+//   heads mode: same, N = 2A+1, no activation; epilogue splits columns
 //               into pdflat[B,2A] and v[B].
-//   dw_mfma:    dW[out,in] += delta^T @ acts (split-K over row blocks,
-//               fp32 atomics into the flat grad; fused db += sum(delta)).
-//   dwv:        the out==1 case (value-head weight grad) as a plain
-//               wave-per-column reduction (MFMA would waste 31/32 lanes).
+//   dgrad modes (activation 3/4): C = act'(aux) * (X @ Wt) — the
+//               backward chain reuses torch weight layouts directly.
+//   dw_mfma:    dW[out,in] += delta^T @ acts, split-K over row blocks
+//               into per-split slabs (atomics measured 8.4M adds/launch
+//               and dominated), reduced by dw_reduce/db_reduce; fused
+//               db += sum(delta); optional row split routes combined
+//               [g_pd | g_v] head deltas to Wp/bp and Wv/bv.
+//
+// Tile-size experiments are in the commit history with measurements:
+// MT=2 row fragments and NT>2 column tiles LOSE on these shapes (64-128
+// AGPR accumulators drop waves/SIMD; occupancy beats per-wave tile size
+// every time it was tried), as did T14 register staging and dual-buffer
+// staging — BK=16 single-buffer with 5-6 blocks/CU is the measured
+// optimum.
 //
 // Fragment layout (cdna guide §3, v_mfma_f32_32x32x2_f32):
 //   lane l: A[i = l&31][k = l>>5], B[k = l>>5][j = l&31]
