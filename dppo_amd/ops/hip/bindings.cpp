@@ -39,11 +39,12 @@ void adam_step_dev(torch::Tensor p, torch::Tensor g, torch::Tensor m,
 
 void gemm_fwd(torch::Tensor X, torch::Tensor Wt, torch::Tensor bias,
               int64_t activation, int64_t heads, torch::Tensor C,
-              torch::Tensor v, torch::Tensor aux, int64_t wt_layout);
+              torch::Tensor v, torch::Tensor aux, int64_t wt_layout,
+              int64_t ablate);
 
 void dw_mfma(torch::Tensor delta, torch::Tensor acts, torch::Tensor grad_buf,
              int64_t w_off, int64_t b_off, int64_t split_row, int64_t w_off2,
-             int64_t b_off2);
+             int64_t b_off2, int64_t ablate);
 
 std::vector<torch::Tensor> rollout_run(
     torch::Tensor params, std::vector<int64_t> offsets,

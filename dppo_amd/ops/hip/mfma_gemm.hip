@@ -70,6 +70,9 @@ struct FwdArgs {
                       // 4 relu-grad: C = (aux>0)*acc   (no bias)
   int heads;          // if 1: last column -> v, rest -> C (pdflat)
   int wt_layout;      // 0: Wt[K][N]; 1: torch W[N][K] (staged transposed)
+  int ablate;         // perf diagnosis only (wrong results when nonzero):
+                      // 1 skip X stage, 2 skip W stage, 4 skip MFMA,
+                      // 8 skip epilogue stores
 };
 
 template <int NT>
@@ -101,6 +104,7 @@ __global__ void gemm_fwd_kernel(FwdArgs a) {
       // float4 staging (scalar element staging measured as the dominant
       // per-call cost: 32 scalar dword loads per thread per stage)
       constexpr int BK4 = BK / 4;
+      if (!(a.ablate & 1))
       for (int idx = threadIdx.x; idx < FWD_M * BK4; idx += FWD_WAVES * 64) {
         const int r = idx / BK4, c4 = (idx % BK4) * 4;
         const int64_t row = b0 + r;
@@ -121,7 +125,9 @@ __global__ void gemm_fwd_kernel(FwdArgs a) {
         xs[r][c4 + 2] = val.z;
         xs[r][c4 + 3] = val.w;
       }
-      if (a.wt_layout == 0) {
+      if (a.ablate & 2) {
+        // skip W stage
+      } else if (a.wt_layout == 0) {
         const int NW4 = NW / 4;
         for (int idx = threadIdx.x; idx < BK * NW4; idx += FWD_WAVES * 64) {
           const int r = idx / NW4, c4 = (idx % NW4) * 4;
@@ -168,7 +174,7 @@ __global__ void gemm_fwd_kernel(FwdArgs a) {
       }
       __syncthreads();
 
-      const int ksteps = min(BK, a.K - kb);
+      const int ksteps = (a.ablate & 4) ? 0 : min(BK, a.K - kb);
       #pragma unroll 4
       for (int k2 = 0; k2 < ksteps; k2 += 2) {
         const float av = xs[wave * M_WAVE + i_l][k2 + k_l];
@@ -181,6 +187,200 @@ __global__ void gemm_fwd_kernel(FwdArgs a) {
     }
 
     // ---- epilogue: bias + activation + store ----
+    #pragma unroll
+    for (int t = 0; t < NT; ++t) {
+      const int col = t * M_WAVE + i_l;
+      if (col < a.N && !(a.ablate & 8)) {
+        const float bv = (a.activation >= 3) ? 0.f : a.bias[col];
+        #pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          const int64_t row = b0 + wave * M_WAVE + cd_row(r, lane);
+          if (row < a.B) {
+            float x;
+            if (a.activation >= 3) {
+              const float h = a.aux[row * a.N + col];
+              x = acc[t][r] * ((a.activation == 3) ? (1.f - h * h)
+                                                   : (h > 0.f ? 1.f : 0.f));
+            } else {
+              x = acc[t][r] + bv;
+              if (a.activation == 0) x = fmaxf(x, 0.f);
+              else if (a.activation == 1) x = fast_tanhf(x);
+            }
+            if (a.heads) {
+              if (col == a.N - 1) a.v[row] = x;
+              else a.C[row * (a.N - 1) + col] = x;
+            } else {
+              a.C[row * a.N + col] = x;
+            }
+          }
+        }
+      }
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Pipelined forward (T14 register staging, cdna guide "write tile t+1 AFTER
+// the barrier"): the runtime-ablation split measured the plain kernel's
+// stage / MFMA / epilogue phases strictly ADDITIVE (L1 886 us = 420 + 321
+// + 145; profiles/r01_gemm_ablation.txt) — no cross-phase overlap at 4
+// waves/SIMD.  Here each stage issues the NEXT tile's global loads into
+// registers before the barrier (plain loads survive s_barrier; guide §5),
+// so HBM latency hides under the current stage's MFMAs.
+//
+// Loads are BRANCHLESS (clamped addresses, whole-float4 validity, zero-fix
+// by select at LDS-write time — NO global access in the write phase): a
+// per-element guard branch or a write-time refetch makes hipcc emit
+// load;s_waitcnt vmcnt(0);ds_write per ELEMENT and serializes the whole
+// pipeline (first attempt measured 1.7x SLOWER than the plain kernel).
+// Whole-float4 validity requires K%4==0 (and N%4==0 for wt_layout 0);
+// other shapes take the plain kernel (host dispatch).
+// ---------------------------------------------------------------------------
+
+// NT<=2 (every shape this model family hits): force the allocation to 128
+// VGPRs = 4 waves/SIMD — the un-hinted build lands at 132 and loses a
+// whole wave to 4 registers.  NT>=3 would spill catastrophically; leave it.
+template <int NT, int PBK>
+__launch_bounds__(FWD_WAVES * 64, (NT <= 2 && PBK == 16) ? 4 : 1)
+__global__ void gemm_fwd_pipe_kernel(FwdArgs a) {
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wave = threadIdx.x / WAVE;
+
+  // PBK: K-step per stage.  32 gives the prefetch twice the MFMA cycles
+  // to hide under but costs 177 VGPR (2 waves/SIMD); 16 keeps 4
+  // waves/SIMD.  Both instantiated; dispatch measures/chooses per shape.
+  __shared__ float xs[FWD_M][PBK + 1];
+  __shared__ float ws[PBK][NT * M_WAVE];
+
+  const int i_l = lane & 31;
+  const int k_l = lane >> 5;
+  constexpr int NW = NT * M_WAVE;
+  constexpr int PBK4 = PBK / 4;
+  constexpr int XIT = (FWD_M * PBK4 + FWD_WAVES * 64 - 1) / (FWD_WAVES * 64);
+  constexpr int WIT = (PBK * NW / 4 + FWD_WAVES * 64 - 1) / (FWD_WAVES * 64);
+  constexpr int W_ITEMS = PBK * NW / 4;  // float4 slots in the W tile
+
+  float4 xreg[XIT];
+  float4 wreg[WIT];
+
+  // ---- branchless load issue for one k-stage (clamped addresses) ----
+  auto issue_x = [&](int64_t b0, int kb) {
+    #pragma unroll
+    for (int it = 0; it < XIT; ++it) {
+      const int idx = threadIdx.x + it * FWD_WAVES * 64;
+      const int r = idx / PBK4, c4 = (idx % PBK4) * 4;
+      int64_t row = b0 + r;
+      row = row < a.B ? row : a.B - 1;
+      int col = kb + c4;
+      col = col < a.K ? col : a.K - 4;   // K%4==0: col<K => col+3<K
+      xreg[it] = *reinterpret_cast<const float4*>(&a.X[row * a.K + col]);
+    }
+  };
+  auto issue_w = [&](int kb) {
+    if (a.wt_layout == 0) {
+      constexpr int NW4 = NW / 4;
+      #pragma unroll
+      for (int it = 0; it < WIT; ++it) {
+        const int idx = threadIdx.x + it * FWD_WAVES * 64;
+        if (idx >= W_ITEMS) break;
+        const int r = idx / NW4, c4 = (idx % NW4) * 4;
+        int krow = kb + r;
+        krow = krow < a.K ? krow : a.K - 1;
+        const int c = c4 < a.N ? c4 : a.N - 4;  // N%4==0
+        wreg[it] = *reinterpret_cast<const float4*>(&a.Wt[(int64_t)krow * a.N + c]);
+      }
+    } else {
+      #pragma unroll
+      for (int it = 0; it < WIT; ++it) {
+        const int idx = threadIdx.x + it * FWD_WAVES * 64;
+        if (idx >= W_ITEMS) break;
+        const int c0 = idx / PBK4, r4 = (idx % PBK4) * 4;
+        const int c = c0 < a.N ? c0 : a.N - 1;
+        int krow = kb + r4;
+        krow = krow < a.K ? krow : a.K - 4;  // K%4==0
+        wreg[it] = *reinterpret_cast<const float4*>(&a.Wt[(int64_t)c * a.K + krow]);
+      }
+    }
+  };
+  // ---- LDS write of the staged registers (selects only, no loads) ----
+  auto write_x = [&](int64_t b0, int kb) {
+    #pragma unroll
+    for (int it = 0; it < XIT; ++it) {
+      const int idx = threadIdx.x + it * FWD_WAVES * 64;
+      const int r = idx / PBK4, c4 = (idx % PBK4) * 4;
+      const bool ok = (b0 + r < a.B) & (kb + c4 < a.K);
+      const float* v = reinterpret_cast<const float*>(&xreg[it]);
+      #pragma unroll
+      for (int q = 0; q < 4; ++q) xs[r][c4 + q] = ok ? v[q] : 0.f;
+    }
+  };
+  auto write_w = [&](int kb) {
+    if (a.wt_layout == 0) {
+      constexpr int NW4 = NW / 4;
+      #pragma unroll
+      for (int it = 0; it < WIT; ++it) {
+        const int idx = threadIdx.x + it * FWD_WAVES * 64;
+        if (idx >= W_ITEMS) break;
+        const int r = idx / NW4, c4 = (idx % NW4) * 4;
+        const bool ok = (kb + r < a.K) & (c4 < a.N);
+        const float* v = reinterpret_cast<const float*>(&wreg[it]);
+        #pragma unroll
+        for (int q = 0; q < 4; ++q) ws[r][c4 + q] = ok ? v[q] : 0.f;
+      }
+    } else {
+      #pragma unroll
+      for (int it = 0; it < WIT; ++it) {
+        const int idx = threadIdx.x + it * FWD_WAVES * 64;
+        if (idx >= W_ITEMS) break;
+        const int c0 = idx / PBK4, r4 = (idx % PBK4) * 4;  // c0 < NW always
+        const bool cok = c0 < a.N;
+        const bool kok = kb + r4 < a.K;  // K%4==0: whole-float4 validity
+        const float* v = reinterpret_cast<const float*>(&wreg[it]);
+        #pragma unroll
+        for (int q = 0; q < 4; ++q)
+          ws[r4 + q][c0] = (cok & kok) ? v[q] : 0.f;
+      }
+    }
+  };
+
+  // One TILE per block (host launches grid == tile count): tile-boundary
+  // cold starts are hidden by BLOCK scheduling — a freshly placed block's
+  // prologue loads overlap the other resident blocks' MFMA phases — which
+  // costs zero registers, where an explicit cross-tile prefetch kept
+  // xreg/wreg live across the epilogue and measured +29 VGPR (137: 3
+  // waves/SIMD) and gave the occupancy back.
+  for (int64_t tile = blockIdx.x; tile * FWD_M < a.B; tile += gridDim.x) {
+    const int64_t b0 = tile * FWD_M;
+    f32x16 acc[NT];
+    #pragma unroll
+    for (int t = 0; t < NT; ++t)
+      #pragma unroll
+      for (int r = 0; r < 16; ++r) acc[t][r] = 0.f;
+
+    issue_x(b0, 0);
+    issue_w(0);
+    for (int kb = 0; kb < a.K; kb += PBK) {
+      write_x(b0, kb);
+      write_w(kb);
+      if (kb + PBK < a.K) {  // prefetch the NEXT stage before the barrier
+        issue_x(b0, kb + PBK);
+        issue_w(kb + PBK);
+      }
+      __syncthreads();
+
+      const int ksteps = min(PBK, a.K - kb);
+      #pragma unroll 4
+      for (int k2 = 0; k2 < ksteps; k2 += 2) {
+        const float av = xs[wave * M_WAVE + i_l][k2 + k_l];
+        #pragma unroll
+        for (int t = 0; t < NT; ++t) {
+          const float bv = ws[k2 + k_l][t * M_WAVE + i_l];
+          acc[t] = __builtin_amdgcn_mfma_f32_32x32x2f32(av, bv, acc[t], 0, 0, 0);
+        }
+      }
+      __syncthreads();
+    }
+
     #pragma unroll
     for (int t = 0; t < NT; ++t) {
       const int col = t * M_WAVE + i_l;
@@ -228,6 +428,7 @@ struct DwArgs {
   int out_dim, in_dim;
   int nt;       // 32-col tiles per wave (<= MAX_NT)
   int splits;   // K splits
+  int ablate;   // perf diagnosis only: 4 skip MFMA, 8 skip slab stores
 };
 
 template <int NT, int MT>  // MT row fragments of 32 (1 or 2)
@@ -298,20 +499,27 @@ __global__ void dw_mfma_kernel(DwArgs a) {
     #pragma unroll
     for (int t = 0; t < NT; ++t)
       bv_n[t] = a.acts[(k + 2 + k_l) * a.in_dim + bcol[t]];
-    #pragma unroll
-    for (int m = 0; m < MT; ++m) {
-      dbacc[m] += av_c[m];
+    if (!(a.ablate & 4)) {
       #pragma unroll
-      for (int t = 0; t < NT; ++t)
-        acc[m][t] = __builtin_amdgcn_mfma_f32_32x32x2f32(av_c[m], bv_c[t],
-                                                         acc[m][t], 0, 0, 0);
+      for (int m = 0; m < MT; ++m) {
+        dbacc[m] += av_c[m];
+        #pragma unroll
+        for (int t = 0; t < NT; ++t)
+          acc[m][t] = __builtin_amdgcn_mfma_f32_32x32x2f32(av_c[m], bv_c[t],
+                                                           acc[m][t], 0, 0, 0);
+      }
+    } else {
+      #pragma unroll
+      for (int m = 0; m < MT; ++m) dbacc[m] += av_c[m];
+      #pragma unroll
+      for (int t = 0; t < NT; ++t) dbacc[0] += bv_c[t] * 1e-38f;
     }
     #pragma unroll
     for (int m = 0; m < MT; ++m) av_c[m] = av_n[m];
     #pragma unroll
     for (int t = 0; t < NT; ++t) bv_c[t] = bv_n[t];
   }
-  if (k < kend) {
+  if (k < kend && !(a.ablate & 4)) {
     #pragma unroll
     for (int m = 0; m < MT; ++m) {
       dbacc[m] += av_c[m];
@@ -320,6 +528,8 @@ __global__ void dw_mfma_kernel(DwArgs a) {
         acc[m][t] = __builtin_amdgcn_mfma_f32_32x32x2f32(av_c[m], bv_c[t],
                                                          acc[m][t], 0, 0, 0);
     }
+    k += 2;
+  } else if (k < kend) {
     k += 2;
   }
   for (; k < k1; ++k) {
@@ -340,6 +550,7 @@ __global__ void dw_mfma_kernel(DwArgs a) {
   // ---- store accumulators to this split's slab (each (row,col) of a
   // split is owned by exactly one block: no atomics, no zero-init) ----
   float* slab = a.slab + (int64_t)split * a.out_dim * a.in_dim;
+  if (a.ablate & 8) return;
   #pragma unroll
   for (int m = 0; m < MT; ++m) {
     #pragma unroll
@@ -417,7 +628,8 @@ __global__ void db_reduce_kernel(const float* __restrict__ db_slab,
 
 void gemm_fwd(torch::Tensor X, torch::Tensor Wt, torch::Tensor bias,
               int64_t activation, int64_t heads, torch::Tensor C,
-              torch::Tensor v, torch::Tensor aux, int64_t wt_layout) {
+              torch::Tensor v, torch::Tensor aux, int64_t wt_layout,
+              int64_t ablate) {
   // C and (for heads) v are caller-allocated so activations can land
   // directly in the backward's blob layout.
   const int64_t B = X.size(0);
@@ -444,6 +656,7 @@ void gemm_fwd(torch::Tensor X, torch::Tensor Wt, torch::Tensor bias,
   a.activation = static_cast<int>(activation);
   a.heads = static_cast<int>(heads);
   a.wt_layout = static_cast<int>(wt_layout);
+  a.ablate = static_cast<int>(ablate);
   a.C = C.data_ptr<float>();
   a.v = heads ? v.data_ptr<float>() : nullptr;
   if (activation >= 3) {
@@ -453,31 +666,47 @@ void gemm_fwd(torch::Tensor X, torch::Tensor Wt, torch::Tensor bias,
 
   hipStream_t stream = c10::hip::getCurrentHIPStream().stream();
   const int64_t tiles = (B + FWD_M - 1) / FWD_M;
+  // plain kernel: capped grid + tile loop.  pipe kernel: one block per
+  // tile — block scheduling hides the per-tile staging prologue.
   const int grid = static_cast<int>(std::min<int64_t>(tiles, 4096));
+  const int grid_pipe = static_cast<int>(std::min<int64_t>(tiles, 1 << 22));
   const int NT = (N + M_WAVE - 1) / M_WAVE;
+  // T14 register-pipelined variant (see gemm_fwd_pipe_kernel).  Its
+  // branchless clamped loads need K>=4 (X / layout-1 W rows) and, for
+  // layout 0, N>=4 (W rows); tiny shapes take the plain kernel.
+  static const int pipe_env = []() {
+    const char* e = getenv("DPPO_GEMM_PIPE");
+    return e ? atoi(e) : 1;
+  }();
+  static const int pbk_env = []() {
+    const char* e = getenv("DPPO_GEMM_PBK");
+    return e ? atoi(e) : 0;  // 0 = per-shape heuristic
+  }();
+  const bool pipe = pipe_env && !a.ablate && K >= 4 && K % 4 == 0 &&
+                    (wt_layout == 1 || (N >= 4 && N % 4 == 0));
+  const int pbk = pbk_env ? pbk_env : 16;
+  #define DISPATCH_FWD(NTV)                                                  \
+    if (pipe && pbk == 32)                                                   \
+      hipLaunchKernelGGL((gemm_fwd_pipe_kernel<NTV, 32>), dim3(grid_pipe),   \
+                         dim3(FWD_WAVES * 64), 0, stream, a);                \
+    else if (pipe)                                                           \
+      hipLaunchKernelGGL((gemm_fwd_pipe_kernel<NTV, 16>), dim3(grid_pipe),   \
+                         dim3(FWD_WAVES * 64), 0, stream, a);                \
+    else                                                                     \
+      hipLaunchKernelGGL(gemm_fwd_kernel<NTV>, dim3(grid),                   \
+                         dim3(FWD_WAVES * 64), 0, stream, a)
   switch (NT) {
-    case 1:
-      hipLaunchKernelGGL(gemm_fwd_kernel<1>, dim3(grid), dim3(FWD_WAVES * 64),
-                         0, stream, a);
-      break;
-    case 2:
-      hipLaunchKernelGGL(gemm_fwd_kernel<2>, dim3(grid), dim3(FWD_WAVES * 64),
-                         0, stream, a);
-      break;
-    case 3:
-      hipLaunchKernelGGL(gemm_fwd_kernel<3>, dim3(grid), dim3(FWD_WAVES * 64),
-                         0, stream, a);
-      break;
-    default:
-      hipLaunchKernelGGL(gemm_fwd_kernel<4>, dim3(grid), dim3(FWD_WAVES * 64),
-                         0, stream, a);
-      break;
+    case 1: DISPATCH_FWD(1); break;
+    case 2: DISPATCH_FWD(2); break;
+    case 3: DISPATCH_FWD(3); break;
+    default: DISPATCH_FWD(4); break;
   }
+  #undef DISPATCH_FWD
 }
 
 void dw_mfma(torch::Tensor delta, torch::Tensor acts, torch::Tensor grad_buf,
              int64_t w_off, int64_t b_off, int64_t split_row, int64_t w_off2,
-             int64_t b_off2) {
+             int64_t b_off2, int64_t ablate) {
   const int64_t B = delta.size(0);
   const int out_dim = static_cast<int>(delta.size(1));
   const int in_dim = static_cast<int>(acts.size(1));
@@ -489,6 +718,7 @@ void dw_mfma(torch::Tensor delta, torch::Tensor acts, torch::Tensor grad_buf,
   a.B = B;
   a.out_dim = out_dim;
   a.in_dim = in_dim;
+  a.ablate = static_cast<int>(ablate);
   // tile shape: maximize N per wave to avoid re-reading delta; wide
   // out_dim also takes 2 row fragments per wave (halves acts re-reads)
   // Wide out_dim: 2 row fragments with NT capped at 2 keeps the

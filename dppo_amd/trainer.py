@@ -669,14 +669,14 @@ class DPPOEngine:
             n = B * dims[l + 1]
             cview = acts.narrow(0, o, n).view(B, dims[l + 1])
             ext.gemm_fwd(x, self.pi.hidden[l].weight.detach(), bs[l],
-                         act_code, 0, cview, cview, cview, 1)
+                         act_code, 0, cview, cview, cview, 1, 0)
             x = cview
             a_views.append(cview)
             o += n
         P = 2 * self.act_space.shape[0]
         pdflat = torch.empty(B, P, device=states.device, dtype=states.dtype)
         v = torch.empty(B, device=states.device, dtype=states.dtype)
-        ext.gemm_fwd(x, Wh_cat, bh, 2, 1, pdflat, v, pdflat, 1)
+        ext.gemm_fwd(x, Wh_cat, bh, 2, 1, pdflat, v, pdflat, 1, 0)
         self._Wh_cat = Wh_cat
         return acts, a_views, v, pdflat
 
@@ -713,17 +713,17 @@ class DPPOEngine:
         for l in range(n_hidden - 1, -1, -1):
             dz_l = torch.empty_like(a_views[l])
             ext.gemm_fwd(delta, Wt_chain, dummy_bias, dgrad_code, 0,
-                         dz_l, dz_l, a_views[l], 0)
+                         dz_l, dz_l, a_views[l], 0, 0)
             dz[l] = dz_l
             delta = dz_l
             Wt_chain = self.pi.hidden[l].weight.detach()
         grad = self.flat_pi.flat_grad
-        ext.dw_mfma(dz[0], states, grad, offsets[0], offsets[1], -1, -1, -1)
+        ext.dw_mfma(dz[0], states, grad, offsets[0], offsets[1], -1, -1, -1, 0)
         for l in range(1, n_hidden):
             ext.dw_mfma(dz[l], a_views[l - 1], grad,
-                        offsets[2 * l], offsets[2 * l + 1], -1, -1, -1)
+                        offsets[2 * l], offsets[2 * l + 1], -1, -1, -1, 0)
         # combined heads: rows < P -> Wp/bp, row P -> Wv/bv
-        ext.dw_mfma(gh, a_views[-1], grad, off_wp, off_bp, P, off_wv, off_bv)
+        ext.dw_mfma(gh, a_views[-1], grad, off_wp, off_bp, P, off_wv, off_bv, 0)
 
     def _update_fused(self, batch: RolloutBatch, l_mul: float) -> None:
         """Fused MFMA update steps: gemm_fwd xL -> GEMM-shaped backward ->

@@ -22,16 +22,37 @@ gh = torch.randn(B, P + 1, device="cuda") * 0.01
 dz2 = torch.empty(B, H, device="cuda")
 dummy = torch.zeros(1, device="cuda")
 
-def l1(): ext.gemm_fwd(X, W1, b1, 1, 0, h1, h1, h1, 1)
-def l2(): ext.gemm_fwd(h1, W2, b1, 1, 0, h2, h2, h2, 1)
-def heads(): ext.gemm_fwd(h2, Wh, bh, 2, 1, pdf, v, pdf, 1)
-def dgrad(): ext.gemm_fwd(gh, Wh, dummy, 3, 0, dz2, dz2, h2, 0)
-def dw1(): ext.dw_mfma(dz2, X, torch.zeros(H*D+H, device="cuda"), 0, H*D, -1, -1, -1)
+ABL = 0
+dwbuf = torch.zeros(H * D + H, device="cuda")
+def l1(): ext.gemm_fwd(X, W1, b1, 1, 0, h1, h1, h1, 1, ABL)
+def l2(): ext.gemm_fwd(h1, W2, b1, 1, 0, h2, h2, h2, 1, ABL)
+def heads(): ext.gemm_fwd(h2, Wh, bh, 2, 1, pdf, v, pdf, 1, ABL)
+def dgrad(): ext.gemm_fwd(gh, Wh, dummy, 3, 0, dz2, dz2, h2, 0, ABL)
+def dw1(): ext.dw_mfma(dz2, X, dwbuf, 0, H*D, -1, -1, -1, ABL)
 
-for name, fn in [("L1", l1), ("L2", l2), ("heads", heads), ("dgrad", dgrad), ("dw1", dw1)]:
-    for _ in range(3): fn()
-    torch.cuda.synchronize()
-    t0 = time.perf_counter()
-    for _ in range(10): fn()
-    torch.cuda.synchronize()
-    print(f"{name}: {(time.perf_counter()-t0)/10*1e6:.0f} us")
+# traffic per call (GB), for effective-bandwidth reporting
+GB = {"L1": (B*(D+H))*4e-9, "L2": (B*2*H)*4e-9, "heads": (B*(H+P+1))*4e-9,
+      "dgrad": (B*(P+1+2*H))*4e-9, "dw1": (B*(D+H))*4e-9}
+
+def run(label):
+    out = {}
+    for name, fn in [("L1", l1), ("L2", l2), ("heads", heads),
+                     ("dgrad", dgrad), ("dw1", dw1)]:
+        for _ in range(3): fn()
+        torch.cuda.synchronize()
+        t0 = time.perf_counter()
+        for _ in range(10): fn()
+        torch.cuda.synchronize()
+        us = (time.perf_counter() - t0) / 10 * 1e6
+        out[name] = us
+        print(f"{label:18s} {name:6s} {us:7.0f} us   {GB[name]/us*1e6:6.0f} GB/s-eff")
+    return out
+
+base = run("full")
+if "--ablate" in sys.argv:
+    for abl, label in [(1, "no-X-stage"), (2, "no-W-stage"), (3, "no-stage"),
+                       (4, "no-mfma"), (8, "no-epilogue"),
+                       (4 + 8, "stage-only")]:
+        ABL = abl
+        run(f"ablate={label}")
+    ABL = 0
