@@ -432,7 +432,9 @@ struct DwArgs {
 };
 
 template <int NT, int MT>  // MT row fragments of 32 (1 or 2)
-__launch_bounds__(64)
+// the hot <2,2> instantiation needs the allocator pressured to keep 3
+// waves/SIMD with the two-pair pipeline (unforced: 200+ VGPR, 2 waves)
+__launch_bounds__(64, (NT == 2 && MT == 2) ? 3 : 1)
 __global__ void dw_mfma_kernel(DwArgs a) {
   const int lane = threadIdx.x;
   const int m0 = blockIdx.x * (M_WAVE * MT);
@@ -475,62 +477,55 @@ __global__ void dw_mfma_kernel(DwArgs a) {
     bcol[t] = col_ok[t] ? col : 0;
   }
 
-  // one-pair-lookahead software pipeline: the next pair's streamed loads
-  // issue before the current pair's 64-cycle MFMAs, so HBM/L2 latency
-  // hides under the matrix pipe instead of serializing with it.
+  // two-pair ping-pong pipeline: TWO register sets alternate with no
+  // copies, keeping two pairs of streamed loads in flight ahead of the
+  // MFMAs (the one-pair version kept only ~16 B/wave outstanding and
+  // measured load-phase-bound: 858 of 1079 us).  MFMA order stays k-
+  // ascending -> results bitwise identical.
   const int64_t kend = k0 + ((k1 - k0) & ~1);
   int64_t k = k0;
-  float av_c[MT];
-  float bv_c[NT];
-  if (k < kend) {
+  float av_A[MT], av_B[MT];
+  float bv_A[NT], bv_B[NT];
+  auto load_pair = [&](int64_t kk, float (&av)[MT], float (&bv)[NT]) {
     #pragma unroll
     for (int m = 0; m < MT; ++m)
-      av_c[m] = m_ok[m] ? a.delta[(k + k_l) * a.out_dim + mcol[m]] : 0.f;
-    #pragma unroll
-    for (int t = 0; t < NT; ++t) bv_c[t] = a.acts[(k + k_l) * a.in_dim + bcol[t]];
-  }
-  #pragma unroll 1
-  for (; k + 2 < kend; k += 2) {
-    float av_n[MT];
-    #pragma unroll
-    for (int m = 0; m < MT; ++m)
-      av_n[m] = m_ok[m] ? a.delta[(k + 2 + k_l) * a.out_dim + mcol[m]] : 0.f;
-    float bv_n[NT];
+      av[m] = m_ok[m] ? a.delta[(kk + k_l) * a.out_dim + mcol[m]] : 0.f;
     #pragma unroll
     for (int t = 0; t < NT; ++t)
-      bv_n[t] = a.acts[(k + 2 + k_l) * a.in_dim + bcol[t]];
+      bv[t] = a.acts[(kk + k_l) * a.in_dim + bcol[t]];
+  };
+  auto fma_pair = [&](const float (&av)[MT], const float (&bv)[NT]) {
     if (!(a.ablate & 4)) {
       #pragma unroll
       for (int m = 0; m < MT; ++m) {
-        dbacc[m] += av_c[m];
+        dbacc[m] += av[m];
         #pragma unroll
         for (int t = 0; t < NT; ++t)
-          acc[m][t] = __builtin_amdgcn_mfma_f32_32x32x2f32(av_c[m], bv_c[t],
+          acc[m][t] = __builtin_amdgcn_mfma_f32_32x32x2f32(av[m], bv[t],
                                                            acc[m][t], 0, 0, 0);
       }
     } else {
       #pragma unroll
-      for (int m = 0; m < MT; ++m) dbacc[m] += av_c[m];
+      for (int m = 0; m < MT; ++m) dbacc[m] += av[m];
       #pragma unroll
-      for (int t = 0; t < NT; ++t) dbacc[0] += bv_c[t] * 1e-38f;
+      for (int t = 0; t < NT; ++t) dbacc[0] += bv[t] * 1e-38f;
     }
-    #pragma unroll
-    for (int m = 0; m < MT; ++m) av_c[m] = av_n[m];
-    #pragma unroll
-    for (int t = 0; t < NT; ++t) bv_c[t] = bv_n[t];
+  };
+  if (k < kend) load_pair(k, av_A, bv_A);
+  if (k + 2 < kend) load_pair(k + 2, av_B, bv_B);
+  #pragma unroll 1
+  for (; k + 6 < kend; k += 4) {
+    fma_pair(av_A, bv_A);
+    load_pair(k + 4, av_A, bv_A);
+    fma_pair(av_B, bv_B);
+    load_pair(k + 6, av_B, bv_B);
   }
-  if (k < kend && !(a.ablate & 4)) {
-    #pragma unroll
-    for (int m = 0; m < MT; ++m) {
-      dbacc[m] += av_c[m];
-      #pragma unroll
-      for (int t = 0; t < NT; ++t)
-        acc[m][t] = __builtin_amdgcn_mfma_f32_32x32x2f32(av_c[m], bv_c[t],
-                                                         acc[m][t], 0, 0, 0);
-    }
-    k += 2;
-  } else if (k < kend) {
-    k += 2;
+  if (k < kend) { fma_pair(av_A, bv_A); k += 2; }
+  if (k < kend) { fma_pair(av_B, bv_B); k += 2; }
+  #pragma unroll 1
+  for (; k < kend; k += 2) {
+    load_pair(k, av_A, bv_A);
+    fma_pair(av_A, bv_A);
   }
   for (; k < k1; ++k) {
     #pragma unroll
