@@ -48,39 +48,74 @@ __global__ void ppo_gauss_fwd_kernel(
   const int P = 2 * A;
   const int64_t waves_total = (int64_t)gridDim.x * 4;
   const int64_t wid = (int64_t)blockIdx.x * 4 + wave;
+  // CONTIGUOUS row chunk per wave (not wave-stride): wave-stride put the
+  // resident 8192 waves ~1.1 MB apart in 5 arrays at once — tens of
+  // thousands of concurrent DRAM streams, no row-buffer locality.
+  const int64_t per = (B + waves_total - 1) / waves_total;
+  const int64_t rb0 = wid * per;
+  const int64_t rb1 = rb0 + per < B ? rb0 + per : B;
 
   float pol = 0.f, ent = 0.f, val = 0.f;  // lane 0 accumulates rows
-  #pragma unroll 2
-  for (int64_t b = wid; b < B; b += waves_total) {
+  // three rows of operands in flight (same ping-pong as ppo_gh_kernel)
+  struct Row { float mu, ls, aj, mo, lso, vp, ov, ad, et; };
+  auto load_row = [&](int64_t b, Row& r) {
+    if (lane < A) {
+      r.mu = pdpi[b * P + lane];
+      r.ls = pdpi[b * P + A + lane];
+      r.aj = act[b * A + lane];
+      r.mo = pdold[b * P + lane];
+      r.lso = pdold[b * P + A + lane];
+    }
+    r.vp = vpred[b];
+    r.ov = oldv[b];
+    r.ad = adv[b];
+    r.et = etr[b];
+  };
+  auto compute_row = [&](const Row& r) {
     float lp_part = 0.f, lo_part = 0.f, ent_part = 0.f;
     if (lane < A) {
-      const float mu = pdpi[b * P + lane];
-      const float ls = pdpi[b * P + A + lane];
-      const float aj = act[b * A + lane];
-      const float zp = (aj - mu) * __expf(-ls);
-      lp_part = -0.5f * zp * zp - ls;
-      const float mo = pdold[b * P + lane];
-      const float lso = pdold[b * P + A + lane];
-      const float zo = (aj - mo) * __expf(-lso);
-      lo_part = -0.5f * zo * zo - lso;
-      ent_part = ls;
+      const float zp = (r.aj - r.mu) * __expf(-r.ls);
+      lp_part = -0.5f * zp * zp - r.ls;
+      const float zo = (r.aj - r.mo) * __expf(-r.lso);
+      lo_part = -0.5f * zo * zo - r.lso;
+      ent_part = r.ls;
     }
     const float lp = wave_reduce_sum(lp_part);
     const float lo = wave_reduce_sum(lo_part);
     const float es = wave_reduce_sum(ent_part);
     if (lane == 0) {
       const float ratio = __expf(lp - lo);  // the logp constants cancel
-      const float ab = adv[b];
+      const float ab = r.ad;
       const float surr1 = ratio * ab;
       const float rc = fminf(fmaxf(ratio, 1.f - clip), 1.f + clip);
       pol += fminf(surr1, rc * ab);
       ent += es + 0.5f * (PPO_LOG_2PI + 1.f) * A;
-      const float vb = vpred[b], ob = oldv[b], eb = etr[b];
-      const float d1 = vb - eb;
-      const float dc = fminf(fmaxf(vb - ob, -clip), clip);
-      const float d2 = ob + dc - eb;
+      const float d1 = r.vp - r.et;
+      const float dc = fminf(fmaxf(r.vp - r.ov, -clip), clip);
+      const float d2 = r.ov + dc - r.et;
       val += fmaxf(d1 * d1, d2 * d2);
     }
+  };
+  Row ra, rc_, re;
+  int64_t b = rb0;
+  if (b < rb1) load_row(b, ra);
+  if (b + 1 < rb1) load_row(b + 1, rc_);
+  if (b + 2 < rb1) load_row(b + 2, re);
+  #pragma unroll 1
+  for (; b + 5 < rb1; b += 3) {
+    compute_row(ra);
+    load_row(b + 3, ra);
+    compute_row(rc_);
+    load_row(b + 4, rc_);
+    compute_row(re);
+    load_row(b + 5, re);
+  }
+  if (b < rb1) { compute_row(ra); ++b; }
+  if (b < rb1) { compute_row(rc_); ++b; }
+  if (b < rb1) { compute_row(re); ++b; }
+  for (; b < rb1; ++b) {
+    load_row(b, ra);
+    compute_row(ra);
   }
   if (lane == 0) {
     atomicAdd(&acc[0], static_cast<double>(pol));
@@ -246,27 +281,42 @@ __global__ void ppo_gh_kernel(
   const int P = 2 * A;
   const int64_t waves_total = (int64_t)gridDim.x * 4;
   const int64_t wid = (int64_t)blockIdx.x * 4 + wave;
+  // contiguous chunk per wave — see the fwd kernel's comment
+  const int64_t per = (B + waves_total - 1) / waves_total;
+  const int64_t rb0 = wid * per;
+  const int64_t rb1 = rb0 + per < B ? rb0 + per : B;
 
-  // two rows in flight per wave: each row is a chain of ~6 dependent
-  // load groups; interleaving rows hides most of that latency
-  #pragma unroll 2
-  for (int64_t b = wid; b < B; b += waves_total) {
+  // Three rows of operands in flight (dw_mfma-style ping-pong register
+  // sets, no copies): one row is a ~6-deep dependent load chain, and the
+  // single-row loop measured load-latency-bound at 1.5 TB/s.
+  const int jj = (lane < A) ? lane : (lane < P ? lane - A : 0);
+  struct Row { float mu, ls, aj, mo, lso, vp, ov, ad, et; };
+  auto load_row = [&](int64_t b, Row& r) {
+    if (lane < P) {
+      r.mu = pdflat[b * P + jj];
+      r.ls = pdflat[b * P + A + jj];
+      r.aj = act[b * A + jj];
+      if (lane < A) {
+        r.mo = oldflat[b * P + jj];
+        r.lso = oldflat[b * P + A + jj];
+      }
+    }
+    r.vp = vpred[b];
+    r.ov = oldv[b];
+    r.ad = adv[b];
+    r.et = etr[b];
+  };
+  auto compute_row = [&](int64_t b, const Row& r) {
     float lp_part = 0.f, lo_part = 0.f, ent_part = 0.f;
     float z = 0.f, inv_s = 0.f;
     if (lane < P) {
-      const int jj = (lane < A) ? lane : lane - A;
-      const float mu = pdflat[b * P + jj];
-      const float ls = pdflat[b * P + A + jj];
-      const float aj = act[b * A + jj];
-      inv_s = __expf(-ls);
-      z = (aj - mu) * inv_s;
+      inv_s = __expf(-r.ls);
+      z = (r.aj - r.mu) * inv_s;
       if (lane < A) {
-        lp_part = -0.5f * z * z - ls;
-        const float mo = oldflat[b * P + jj];
-        const float lso = oldflat[b * P + A + jj];
-        const float zo = (aj - mo) * __expf(-lso);
-        lo_part = -0.5f * zo * zo - lso;
-        ent_part = ls;
+        lp_part = -0.5f * z * z - r.ls;
+        const float zo = (r.aj - r.mo) * __expf(-r.lso);
+        lo_part = -0.5f * zo * zo - r.lso;
+        ent_part = r.ls;
       }
     }
     const float c = 0.5f * PPO_LOG_2PI * A;
@@ -275,15 +325,36 @@ __global__ void ppo_gh_kernel(
     row.logp_old = __shfl(wave_reduce_sum(lo_part), 0, WAVE) - c;
     row.ent = __shfl(wave_reduce_sum(ent_part), 0, WAVE) +
               0.5f * (PPO_LOG_2PI + 1.f) * A;
-    const PPORowGrads g =
-        ppo_row_grads(row, vpred[b], oldv[b], adv[b], etr[b], B, clip,
-                      entcoeff, vcoeff, 1.f);
+    const PPORowGrads g = ppo_row_grads(row, r.vp, r.ov, r.ad, r.et, B, clip,
+                                        entcoeff, vcoeff, 1.f);
     if (lane < P) {
       gh[b * (P + 1) + lane] =
           (lane < A) ? g.g_logp * z * inv_s
                      : g.g_logp * (z * z - 1.f) + g.g_ent;
     }
     if (lane == 0) gh[b * (P + 1) + P] = g.g_v;
+  };
+
+  Row ra, rc, re;
+  int64_t b = rb0;
+  if (b < rb1) load_row(b, ra);
+  if (b + 1 < rb1) load_row(b + 1, rc);
+  if (b + 2 < rb1) load_row(b + 2, re);
+  #pragma unroll 1
+  for (; b + 5 < rb1; b += 3) {
+    compute_row(b, ra);
+    load_row(b + 3, ra);
+    compute_row(b + 1, rc);
+    load_row(b + 4, rc);
+    compute_row(b + 2, re);
+    load_row(b + 5, re);
+  }
+  if (b < rb1) { compute_row(b, ra); ++b; }
+  if (b < rb1) { compute_row(b, rc); ++b; }
+  if (b < rb1) { compute_row(b, re); ++b; }
+  for (; b < rb1; ++b) {
+    load_row(b, ra);
+    compute_row(b, ra);
   }
 }
 

@@ -30,14 +30,25 @@ def heads(): ext.gemm_fwd(h2, Wh, bh, 2, 1, pdf, v, pdf, 1, ABL)
 def dgrad(): ext.gemm_fwd(gh, Wh, dummy, 3, 0, dz2, dz2, h2, 0, ABL)
 def dw1(): ext.dw_mfma(dz2, X, dwbuf, 0, H*D, -1, -1, -1, ABL)
 
+act = torch.randn(B, A, device="cuda")
+advb = torch.randn(B, device="cuda")
+etrb = torch.randn(B, device="cuda")
+oldvb = torch.randn(B, device="cuda")
+oldf = pdf + 0.01 * torch.randn_like(pdf)
+cde = torch.empty(0, device="cuda")
+def ghk(): ext.ppo_loss_gauss_gh(pdf, oldf, v, oldvb, act, advb, etrb, 0.2, 0.01, 0.5, cde)
+def lossf(): ext.ppo_loss_gauss_fwd(pdf, oldf, v, oldvb, act, advb, etrb, 0.2, 0.01, 0.5)
+
 # traffic per call (GB), for effective-bandwidth reporting
 GB = {"L1": (B*(D+H))*4e-9, "L2": (B*2*H)*4e-9, "heads": (B*(H+P+1))*4e-9,
-      "dgrad": (B*(P+1+2*H))*4e-9, "dw1": (B*(D+H))*4e-9}
+      "dgrad": (B*(P+1+2*H))*4e-9, "dw1": (B*(D+H))*4e-9,
+      "gh": (B*(2*P+A+3+P+1))*4e-9, "lossf": (B*(2*P+A+3))*4e-9}
 
 def run(label):
     out = {}
     for name, fn in [("L1", l1), ("L2", l2), ("heads", heads),
-                     ("dgrad", dgrad), ("dw1", dw1)]:
+                     ("dgrad", dgrad), ("dw1", dw1), ("gh", ghk),
+                     ("lossf", lossf)]:
         for _ in range(3): fn()
         torch.cuda.synchronize()
         t0 = time.perf_counter()
