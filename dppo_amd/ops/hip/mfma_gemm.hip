@@ -677,7 +677,11 @@ void gemm_fwd(torch::Tensor X, torch::Tensor Wt, torch::Tensor bias,
     const char* e = getenv("DPPO_GEMM_PBK");
     return e ? atoi(e) : 0;  // 0 = per-shape heuristic
   }();
-  const bool pipe = pipe_env && !a.ablate && K >= 4 && K % 4 == 0 &&
+  // K%4!=0 is allowed for the dgrad-from-gh call (activation>=3,
+  // wt_layout 0): its X is the gh buffer, allocated with a float4 of
+  // slack, and the overhang columns multiply zeroed W rows >= K.
+  const bool pipe = pipe_env && !a.ablate && K >= 4 &&
+                    (K % 4 == 0 || (wt_layout == 0 && activation >= 3)) &&
                     (wt_layout == 1 || (N >= 4 && N % 4 == 0));
   const int pbk = pbk_env ? pbk_env : 16;
   #define DISPATCH_FWD(NTV)                                                  \

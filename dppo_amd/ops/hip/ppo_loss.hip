@@ -368,7 +368,14 @@ torch::Tensor ppo_loss_gauss_gh(torch::Tensor pdflat, torch::Tensor oldflat,
                                 torch::Tensor clip_dev) {
   const int64_t B = vpred.numel();
   const int A = static_cast<int>(pdflat.size(1) / 2);
-  auto gh = torch::empty({B, 2 * (int64_t)A + 1}, pdflat.options());
+  // one float4 of slack past the end: the pipelined dgrad GEMM reads the
+  // [B][2A+1] rows with whole-float4 loads (K=2A+1 is odd), so the last
+  // row's tail load overhangs by up to 3 floats.  The overhang elements
+  // multiply W rows >= K, which the GEMM stages as zeros.
+  const int64_t ghn = B * (2 * (int64_t)A + 1);
+  auto gh = torch::empty({ghn + 4}, pdflat.options())
+                .narrow(0, 0, ghn)
+                .view({B, 2 * (int64_t)A + 1});
   hipStream_t stream = c10::hip::getCurrentHIPStream().stream();
   const float* cd =
       (clip_dev.numel() > 0) ? clip_dev.data_ptr<float>() : nullptr;
