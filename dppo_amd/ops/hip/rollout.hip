@@ -168,7 +168,11 @@ DEV_INLINE float u2f_mono(unsigned u) {
 
 // ---- the rollout kernel ---------------------------------------------------
 
-__launch_bounds__(NWAVES * 64, 4)  // 4 waves/SIMD: cap VGPRs at 128
+// MINWAVES: waves/SIMD forced on the allocator.  4 caps VGPRs at 128 and
+// SPILLS (29 VGPR + 185 SGPR, 96 B scratch/lane measured on this source) —
+// both instantiated, dispatch measures/chooses.
+template <int MINWAVES>
+__launch_bounds__(NWAVES * 64, MINWAVES)
 __global__ void rollout_kernel(RolloutArgs a) {
   const int tid = threadIdx.x;      // block = 4 waves of 64
   const int lane = tid & (WAVE - 1);
@@ -666,8 +670,18 @@ std::vector<torch::Tensor> rollout_run(
                    r4((int)rank) + 4) +
        NWAVES * ENV_TILE * r4(h_max)) *
       sizeof(float);
-  hipLaunchKernelGGL(rollout_kernel, dim3(grid), dim3(NWAVES * WAVE),
-                     lds_bytes, stream, a);
+  // MINWAVES=4 measured 24.8 ms vs 27.7 ms at MINWAVES=3 (65k envs):
+  // the 4th wave/SIMD buys more than the 29-VGPR spill costs.
+  static const int mw = []() {
+    const char* e = getenv("DPPO_ROLLOUT_MW");
+    return e ? atoi(e) : 4;
+  }();
+  if (mw >= 4)
+    hipLaunchKernelGGL(rollout_kernel<4>, dim3(grid), dim3(NWAVES * WAVE),
+                       lds_bytes, stream, a);
+  else
+    hipLaunchKernelGGL(rollout_kernel<3>, dim3(grid), dim3(NWAVES * WAVE),
+                       lds_bytes, stream, a);
 
   // carve views out of the blob
   int64_t o = 0;

@@ -9,7 +9,7 @@ from dppo_amd.ops import require_hip_ext
 
 ext = require_hip_ext()
 cfg = DPPOConfig(GAME="Humanoid-v4", HIDDEN_SIZES=(64, 64), ACTIVATION="tanh",
-                 NUM_ENVS=32768, MAX_EPOCH_STEPS=64, EPOCH_MAX=10**6,
+                 NUM_ENVS=65536, MAX_EPOCH_STEPS=64, EPOCH_MAX=10**6,
                  STOP_EPOCH=10**6, NUM_WORKERS=1, LOG_FILE_PATH="/tmp/l",
                  DEVICE="cuda")
 eng = DPPOEngine(cfg, comm=Comm(device="cuda:0"))
