@@ -40,11 +40,25 @@ void adam_step_dev(torch::Tensor p, torch::Tensor g, torch::Tensor m,
 void gemm_fwd(torch::Tensor X, torch::Tensor Wt, torch::Tensor bias,
               int64_t activation, int64_t heads, torch::Tensor C,
               torch::Tensor v, torch::Tensor aux, int64_t wt_layout,
-              int64_t ablate);
+              int64_t ablate, int64_t ldc);
 
 void dw_mfma(torch::Tensor delta, torch::Tensor acts, torch::Tensor grad_buf,
              int64_t w_off, int64_t b_off, int64_t split_row, int64_t w_off2,
              int64_t b_off2, int64_t ablate);
+
+void rollout_sample(torch::Tensor pdflat, torch::Tensor actions,
+                    torch::Tensor xva, torch::Tensor seed_dev,
+                    torch::Tensor eps_dev, int64_t step, int64_t va_off,
+                    double act_low, double act_high);
+
+void rollout_env_step(torch::Tensor x, torch::Tensor G, torch::Tensor envd,
+                      torch::Tensor horizons, torch::Tensor t,
+                      torch::Tensor epr, torch::Tensor states_next,
+                      torch::Tensor rewards, torch::Tensor dones,
+                      torch::Tensor seed_dev, double sigma, int64_t step);
+
+torch::Tensor rollout_moments(torch::Tensor rewards, torch::Tensor dones,
+                              torch::Tensor epr_in, int64_t T, int64_t E);
 
 std::vector<torch::Tensor> rollout_run(
     torch::Tensor params, std::vector<int64_t> offsets,
@@ -67,6 +81,12 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
           "graph-replayable fused Adam (device step/lr) (gfx950)");
   mod.def("rollout_run", &rollout_run,
           "fused T-step rollout: MLP fwd + sample + synthetic env (gfx950)");
+  mod.def("rollout_sample", &rollout_sample,
+          "per-step action sampling + eps-greedy (v3 rollout) (gfx950)");
+  mod.def("rollout_env_step", &rollout_env_step,
+          "per-step synthetic env finish: tanh/reward/done/reset (gfx950)");
+  mod.def("rollout_moments", &rollout_moments,
+          "episode-reward moments from reward/done streams (gfx950)");
   mod.def("ppo_loss_gauss_gh", &ppo_loss_gauss_gh,
           "wave-per-row PPO loss gradient -> [g_pd | g_v] (gfx950)");
   mod.def("gemm_fwd", &gemm_fwd,

@@ -70,6 +70,8 @@ struct FwdArgs {
                       // 4 relu-grad: C = (aux>0)*acc   (no bias)
   int heads;          // if 1: last column -> v, rest -> C (pdflat)
   int wt_layout;      // 0: Wt[K][N]; 1: torch W[N][K] (staged transposed)
+  int ldc;            // C row stride (>= N; heads: >= N-1) — lets C land
+                      // in a column block of a wider buffer
   int ablate;         // perf diagnosis only (wrong results when nonzero):
                       // 1 skip X stage, 2 skip W stage, 4 skip MFMA,
                       // 8 skip epilogue stores
@@ -208,9 +210,9 @@ __global__ void gemm_fwd_kernel(FwdArgs a) {
             }
             if (a.heads) {
               if (col == a.N - 1) a.v[row] = x;
-              else a.C[row * (a.N - 1) + col] = x;
+              else a.C[row * a.ldc + col] = x;
             } else {
-              a.C[row * a.N + col] = x;
+              a.C[row * a.ldc + col] = x;
             }
           }
         }
@@ -259,6 +261,9 @@ __global__ void gemm_fwd_pipe_kernel(FwdArgs a) {
   constexpr int XIT = (FWD_M * PBK4 + FWD_WAVES * 64 - 1) / (FWD_WAVES * 64);
   constexpr int WIT = (PBK * NW / 4 + FWD_WAVES * 64 - 1) / (FWD_WAVES * 64);
   constexpr int W_ITEMS = PBK * NW / 4;  // float4 slots in the W tile
+  // column panel (grid.y): this block computes columns [n0, n0+NW) of N —
+  // wide-N GEMMs (the env-dynamics [E][36] @ [36][D] call) run as panels
+  const int n0 = blockIdx.y * NW;
 
   float4 xreg[XIT];
   float4 wreg[WIT];
@@ -286,7 +291,7 @@ __global__ void gemm_fwd_pipe_kernel(FwdArgs a) {
         const int r = idx / NW4, c4 = (idx % NW4) * 4;
         int krow = kb + r;
         krow = krow < a.K ? krow : a.K - 1;
-        const int c = c4 < a.N ? c4 : a.N - 4;  // N%4==0
+        const int c = n0 + c4 < a.N ? n0 + c4 : a.N - 4;  // N%4==0
         wreg[it] = *reinterpret_cast<const float4*>(&a.Wt[(int64_t)krow * a.N + c]);
       }
     } else {
@@ -295,7 +300,7 @@ __global__ void gemm_fwd_pipe_kernel(FwdArgs a) {
         const int idx = threadIdx.x + it * FWD_WAVES * 64;
         if (idx >= W_ITEMS) break;
         const int c0 = idx / PBK4, r4 = (idx % PBK4) * 4;
-        const int c = c0 < a.N ? c0 : a.N - 1;
+        const int c = n0 + c0 < a.N ? n0 + c0 : a.N - 1;
         int krow = kb + r4;
         krow = krow < a.K ? krow : a.K - 4;  // K%4==0
         wreg[it] = *reinterpret_cast<const float4*>(&a.Wt[(int64_t)c * a.K + krow]);
@@ -322,7 +327,7 @@ __global__ void gemm_fwd_pipe_kernel(FwdArgs a) {
         const int idx = threadIdx.x + it * FWD_WAVES * 64;
         if (idx >= W_ITEMS) break;
         const int r = idx / NW4, c4 = (idx % NW4) * 4;
-        const bool ok = (kb + r < a.K) & (c4 < a.N);
+        const bool ok = (kb + r < a.K) & (n0 + c4 < a.N);
         const float* v = reinterpret_cast<const float*>(&wreg[it]);
         #pragma unroll
         for (int q = 0; q < 4; ++q) ws[r][c4 + q] = ok ? v[q] : 0.f;
@@ -333,7 +338,7 @@ __global__ void gemm_fwd_pipe_kernel(FwdArgs a) {
         const int idx = threadIdx.x + it * FWD_WAVES * 64;
         if (idx >= W_ITEMS) break;
         const int c0 = idx / PBK4, r4 = (idx % PBK4) * 4;  // c0 < NW always
-        const bool cok = c0 < a.N;
+        const bool cok = n0 + c0 < a.N;
         const bool kok = kb + r4 < a.K;  // K%4==0: whole-float4 validity
         const float* v = reinterpret_cast<const float*>(&wreg[it]);
         #pragma unroll
@@ -383,7 +388,7 @@ __global__ void gemm_fwd_pipe_kernel(FwdArgs a) {
 
     #pragma unroll
     for (int t = 0; t < NT; ++t) {
-      const int col = t * M_WAVE + i_l;
+      const int col = n0 + t * M_WAVE + i_l;
       if (col < a.N) {
         const float bv = (a.activation >= 3) ? 0.f : a.bias[col];
         #pragma unroll
@@ -402,9 +407,9 @@ __global__ void gemm_fwd_pipe_kernel(FwdArgs a) {
             }
             if (a.heads) {
               if (col == a.N - 1) a.v[row] = x;
-              else a.C[row * (a.N - 1) + col] = x;
+              else a.C[row * a.ldc + col] = x;
             } else {
-              a.C[row * a.N + col] = x;
+              a.C[row * a.ldc + col] = x;
             }
           }
         }
@@ -624,9 +629,11 @@ __global__ void db_reduce_kernel(const float* __restrict__ db_slab,
 void gemm_fwd(torch::Tensor X, torch::Tensor Wt, torch::Tensor bias,
               int64_t activation, int64_t heads, torch::Tensor C,
               torch::Tensor v, torch::Tensor aux, int64_t wt_layout,
-              int64_t ablate) {
+              int64_t ablate, int64_t ldc) {
   // C and (for heads) v are caller-allocated so activations can land
-  // directly in the backward's blob layout.
+  // directly in the backward's blob layout.  ldc (0 = natural width) lets
+  // C land in a column block of a wider row-major buffer; N > 128 runs as
+  // column panels on the pipelined kernel (grid.y).
   const int64_t B = X.size(0);
   const int K = static_cast<int>(X.size(1));
   const int N = static_cast<int>(wt_layout ? Wt.size(0) : Wt.size(1));
@@ -634,11 +641,13 @@ void gemm_fwd(torch::Tensor X, torch::Tensor Wt, torch::Tensor bias,
   TORCH_CHECK(C.is_contiguous());
   TORCH_CHECK((wt_layout ? Wt.size(1) : Wt.size(0)) == K);
   TORCH_CHECK(activation >= 3 || bias.numel() == N);
-  TORCH_CHECK(N <= MAX_NT * M_WAVE, "N exceeds MFMA fwd tile budget");
+  const int ldc_eff =
+      static_cast<int>(ldc > 0 ? ldc : (heads ? N - 1 : N));
+  TORCH_CHECK(ldc_eff >= (heads ? N - 1 : N));
   if (heads) {
-    TORCH_CHECK(C.numel() == B * (N - 1) && v.numel() == B);
+    TORCH_CHECK(C.numel() >= (B - 1) * ldc_eff + N - 1 && v.numel() == B);
   } else {
-    TORCH_CHECK(C.numel() == B * N);
+    TORCH_CHECK(C.numel() >= (B - 1) * ldc_eff + N);
   }
 
   FwdArgs a{};
@@ -652,6 +661,7 @@ void gemm_fwd(torch::Tensor X, torch::Tensor Wt, torch::Tensor bias,
   a.heads = static_cast<int>(heads);
   a.wt_layout = static_cast<int>(wt_layout);
   a.ablate = static_cast<int>(ablate);
+  a.ldc = ldc_eff;
   a.C = C.data_ptr<float>();
   a.v = heads ? v.data_ptr<float>() : nullptr;
   if (activation >= 3) {
@@ -684,21 +694,31 @@ void gemm_fwd(torch::Tensor X, torch::Tensor Wt, torch::Tensor bias,
                     (K % 4 == 0 || (wt_layout == 0 && activation >= 3)) &&
                     (wt_layout == 1 || (N >= 4 && N % 4 == 0));
   const int pbk = pbk_env ? pbk_env : 16;
-  #define DISPATCH_FWD(NTV)                                                  \
+  #define DISPATCH_FWD(NTV, NPANEL)                                          \
     if (pipe && pbk == 32)                                                   \
-      hipLaunchKernelGGL((gemm_fwd_pipe_kernel<NTV, 32>), dim3(grid_pipe),   \
-                         dim3(FWD_WAVES * 64), 0, stream, a);                \
+      hipLaunchKernelGGL((gemm_fwd_pipe_kernel<NTV, 32>),                    \
+                         dim3(grid_pipe, NPANEL), dim3(FWD_WAVES * 64), 0,   \
+                         stream, a);                                         \
     else if (pipe)                                                           \
-      hipLaunchKernelGGL((gemm_fwd_pipe_kernel<NTV, 16>), dim3(grid_pipe),   \
-                         dim3(FWD_WAVES * 64), 0, stream, a);                \
+      hipLaunchKernelGGL((gemm_fwd_pipe_kernel<NTV, 16>),                    \
+                         dim3(grid_pipe, NPANEL), dim3(FWD_WAVES * 64), 0,   \
+                         stream, a);                                         \
     else                                                                     \
       hipLaunchKernelGGL(gemm_fwd_kernel<NTV>, dim3(grid),                   \
                          dim3(FWD_WAVES * 64), 0, stream, a)
-  switch (NT) {
-    case 1: DISPATCH_FWD(1); break;
-    case 2: DISPATCH_FWD(2); break;
-    case 3: DISPATCH_FWD(3); break;
-    default: DISPATCH_FWD(4); break;
+  if (N > MAX_NT * M_WAVE) {
+    // wide N: NT=2 column panels (the NT=2 pipe schedule holds 4
+    // waves/SIMD; NT=4 would drop to 2) — pipe-capable shapes only
+    TORCH_CHECK(pipe, "N > 128 needs the pipelined path (K%4==0 etc.)");
+    const int panels = (N + 2 * M_WAVE - 1) / (2 * M_WAVE);
+    DISPATCH_FWD(2, panels);
+  } else {
+    switch (NT) {
+      case 1: DISPATCH_FWD(1, 1); break;
+      case 2: DISPATCH_FWD(2, 1); break;
+      case 3: DISPATCH_FWD(3, 1); break;
+      default: DISPATCH_FWD(4, 1); break;
+    }
   }
   #undef DISPATCH_FWD
 }

@@ -590,6 +590,107 @@ __global__ void ep_moments_decode_kernel(float* __restrict__ out) {
   out[4] = (u4 == 0u) ? 0.f : u2f_mono(u4);
 }
 
+// ---------------------------------------------------------------------------
+// Per-step rollout path (v3): the T-step loop runs as a hipGraph of
+// pipelined MFMA GEMMs (mfma_gemm.hip does the MLP forward AND the env
+// dynamics' low-rank/action matmuls as [E][36] @ [36][D] panels); the two
+// kernels below supply the non-GEMM pieces.  RNG slots/order are IDENTICAL
+// to rollout_kernel, so both paths sample the same actions/noise/resets
+// for a given seed.  seed/eps live in DEVICE memory so one captured graph
+// replays across rounds (they change per round/rollout).
+// ---------------------------------------------------------------------------
+
+// thread per (env, action-dim): sample + epsilon-greedy; also writes the
+// action into the [E][xva_w] concat buffer consumed by the dynamics GEMM.
+__global__ void sample_kernel(const float* __restrict__ pdflat,  // [E][2A]
+                              float* __restrict__ actions,       // [E][A]
+                              float* __restrict__ xva,           // [E][xva_w]
+                              const long* __restrict__ seed_dev,
+                              const float* __restrict__ eps_dev,
+                              int step, int64_t E, int A, int xva_w,
+                              int va_off, float act_low, float act_high) {
+  const unsigned seed = (unsigned)*seed_dev;
+  const float eps = *eps_dev;
+  const int P = 2 * A;
+  for (int64_t i = gidx(); i < E * A; i += gstride()) {
+    const int64_t e = i / A;
+    const int j = (int)(i % A);
+    const float mean = pdflat[e * P + j];
+    const float logstd = pdflat[e * P + A + j];
+    float act = mean + __expf(logstd) * rng_normal(seed, (int)e, step, j);
+    const float u_dec = rng_uniform(seed, (int)e, step, 90001);
+    if (u_dec < eps) {
+      const float u = rng_uniform(seed, (int)e, step, 90010 + j);
+      act = act_low + (act_high - act_low) * u;
+    }
+    actions[e * A + j] = act;
+    xva[e * xva_w + va_off + j] = act;
+  }
+}
+
+// wave per env: x' = tanh(x*d + G + sigma*noise), reward, done/reset and
+// episode bookkeeping.  G = [x@V | act] @ [U; B] comes from the GEMM.
+__global__ void env_finish_kernel(
+    float* __restrict__ x,             // [E][D] in/out
+    const float* __restrict__ G,       // [E][D]
+    const float* __restrict__ envd,    // [D] diagonal
+    const int* __restrict__ horizons,  // [E]
+    int* __restrict__ t,               // [E]
+    float* __restrict__ epr,           // [E]
+    float* __restrict__ states_next,   // [E][D] out_states[step+1] or null
+    float* __restrict__ rewards,       // [E] out_rewards[step]
+    float* __restrict__ dones,         // [E] out_dones[step]
+    const long* __restrict__ seed_dev, float sigma, int step, int64_t E,
+    int D) {
+  const unsigned seed = (unsigned)*seed_dev;
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int64_t wid =
+      ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) / WAVE;
+  if (wid >= E) return;
+  const int64_t e = wid;
+  // up to 512/64 = 8 elements per lane (MAX_D = 512)
+  float xn[(MAX_D + WAVE - 1) / WAVE];
+  float ss = 0.f;
+  int nd = 0;
+  for (int d = lane; d < D; d += WAVE, ++nd) {
+    // same noise stream as the fused kernel: one Box-Muller pair per
+    // (even env, d) feeds envs 2q and 2q+1
+    float nz = 0.f;
+    if (sigma != 0.f) {
+      const float2 pr = rng_normal2(seed, (int)(e & ~1LL), step, 1000 + d);
+      nz = (e & 1) ? pr.y : pr.x;
+    }
+    const float v = fast_tanhf(x[e * D + d] * envd[d] + G[e * D + d] +
+                               sigma * nz);
+    xn[nd] = v;
+    ss += v * v;
+  }
+  ss = wave_reduce_sum(ss);
+  int done;
+  if (lane == 0) {
+    const float r = 1.0f - ss / D;
+    rewards[e] = r;
+    float ep = epr[e] + r;
+    int tc = t[e] + 1;
+    done = (tc >= horizons[e]) ? 1 : 0;
+    dones[e] = (float)done;
+    if (done) {
+      ep = 0.f;
+      tc = 0;
+    }
+    epr[e] = ep;
+    t[e] = tc;
+  }
+  done = __shfl(done, 0, WAVE);
+  nd = 0;
+  for (int d = lane; d < D; d += WAVE, ++nd) {
+    const float v =
+        done ? 0.1f * rng_normal(seed, (int)e, step, 5000 + d) : xn[nd];
+    x[e * D + d] = v;
+    if (states_next != nullptr) states_next[e * D + d] = v;
+  }
+}
+
 }  // namespace
 
 std::vector<torch::Tensor> rollout_run(
@@ -709,4 +810,61 @@ std::vector<torch::Tensor> rollout_run(
   hipLaunchKernelGGL(ep_moments_decode_kernel, dim3(1), dim3(1), 0, stream,
                      moments.data_ptr<float>());
   return {states, pdflats, actions, values, rewards, dones, boot_v, moments};
+}
+
+void rollout_sample(torch::Tensor pdflat, torch::Tensor actions,
+                    torch::Tensor xva, torch::Tensor seed_dev,
+                    torch::Tensor eps_dev, int64_t step, int64_t va_off,
+                    double act_low, double act_high) {
+  const int64_t E = pdflat.size(0);
+  const int A = static_cast<int>(pdflat.size(1) / 2);
+  TORCH_CHECK(pdflat.is_cuda() && actions.numel() == E * A);
+  TORCH_CHECK(seed_dev.dtype() == torch::kInt64 && eps_dev.dtype() == torch::kFloat32);
+  const int xva_w = static_cast<int>(xva.size(1));
+  hipStream_t stream = c10::hip::getCurrentHIPStream().stream();
+  hipLaunchKernelGGL(sample_kernel,
+                     dim3(elementwise_grid(E * A, 256)), dim3(256), 0, stream,
+                     pdflat.data_ptr<float>(), actions.data_ptr<float>(),
+                     xva.data_ptr<float>(), seed_dev.data_ptr<int64_t>(),
+                     eps_dev.data_ptr<float>(), (int)step, E, A, xva_w,
+                     (int)va_off, (float)act_low, (float)act_high);
+}
+
+void rollout_env_step(torch::Tensor x, torch::Tensor G, torch::Tensor envd,
+                      torch::Tensor horizons, torch::Tensor t,
+                      torch::Tensor epr, torch::Tensor states_next,
+                      torch::Tensor rewards, torch::Tensor dones,
+                      torch::Tensor seed_dev, double sigma, int64_t step) {
+  const int64_t E = x.size(0);
+  const int D = static_cast<int>(x.size(1));
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous() && G.numel() >= E * D);
+  TORCH_CHECK(D <= MAX_D);
+  TORCH_CHECK(horizons.dtype() == torch::kInt32 && t.dtype() == torch::kInt32);
+  hipStream_t stream = c10::hip::getCurrentHIPStream().stream();
+  const int64_t waves = E;
+  const int64_t blocks = (waves * WAVE + 255) / 256;
+  hipLaunchKernelGGL(env_finish_kernel,
+                     dim3((unsigned)std::min<int64_t>(blocks, 1 << 26)),
+                     dim3(256), 0, stream, x.data_ptr<float>(),
+                     G.data_ptr<float>(), envd.data_ptr<float>(),
+                     horizons.data_ptr<int>(), t.data_ptr<int>(),
+                     epr.data_ptr<float>(),
+                     states_next.numel() ? states_next.data_ptr<float>()
+                                         : nullptr,
+                     rewards.data_ptr<float>(), dones.data_ptr<float>(),
+                     seed_dev.data_ptr<int64_t>(), (float)sigma, (int)step, E,
+                     D);
+}
+
+torch::Tensor rollout_moments(torch::Tensor rewards, torch::Tensor dones,
+                              torch::Tensor epr_in, int64_t T, int64_t E) {
+  auto out = torch::zeros({5}, rewards.options());
+  hipStream_t stream = c10::hip::getCurrentHIPStream().stream();
+  hipLaunchKernelGGL(ep_moments_kernel, dim3(elementwise_grid(E, 256)),
+                     dim3(256), 0, stream, rewards.data_ptr<float>(),
+                     dones.data_ptr<float>(), epr_in.data_ptr<float>(),
+                     out.data_ptr<float>(), T, E);
+  hipLaunchKernelGGL(ep_moments_decode_kernel, dim3(1), dim3(1), 0, stream,
+                     out.data_ptr<float>());
+  return out;
 }

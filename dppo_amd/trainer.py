@@ -229,9 +229,26 @@ class DPPOEngine:
 
     @torch.no_grad()
     def rollout_once(self) -> Tuple[RolloutBatch, Dict[str, float]]:
+        if self._can_rollout_v3():
+            return self._rollout_once_hip_v3()
         if self._can_fuse_rollout():
             return self._rollout_once_hip()
         return self._rollout_once_eager()
+
+    def _can_rollout_v3(self) -> bool:
+        """Per-step GEMM rollout (v3): the T-step loop runs as pipelined
+        MFMA GEMMs (MLP forward + env dynamics as [E][rank+A] @ [.][D])
+        plus two small kernels (sampling, env finish).  Wins over the
+        fused whole-rollout kernel at large E, where the VALU trunk/env
+        phases bound it; the fused kernel keeps small-E latency."""
+        if not self._can_fuse_rollout():
+            return False
+        if self.obs_space.shape[0] % 4 != 0:  # pipelined-GEMM staging
+            return False
+        ov = os.environ.get("DPPO_ROLLOUT_V3")
+        if ov is not None:
+            return ov != "0"
+        return self.cfg.NUM_ENVS >= 16384
 
     @torch.no_grad()
     def _rollout_weight_blob(self):
@@ -299,7 +316,14 @@ class DPPOEngine:
             env.x, env.t, self.epr, T, A, seed, self._rollout_out, 0,
         )
         self.obs = env.x  # updated in place by the kernel
+        return self._finish_hip_rollout(
+            states, pdflats, actions, values, rewards, dones, boot_v,
+            moments, eps)
 
+    def _finish_hip_rollout(self, states, pdflats, actions, values, rewards,
+                            dones, boot_v, moments, eps):
+        c, E = self.cfg, self.cfg.NUM_ENVS
+        T = c.MAX_EPOCH_STEPS
         adv, etr = gae_advantages(
             rewards, values, dones, boot_v,
             c.GAMMA, c.LAM, whiten=True, eps=c.ADV_EPS,
@@ -321,6 +345,185 @@ class DPPOEngine:
             valid=float(moments[0]) > 0,  # the rollout's single host sync
         )
         return batch, {"exploration_rate": eps}
+
+    def _v3_buffers(self):
+        """Persistent device buffers for the per-step GEMM rollout."""
+        c, env = self.cfg, self.env
+        E, T = c.NUM_ENVS, c.MAX_EPOCH_STEPS
+        D = self.obs_space.shape[0]
+        A = self.act_space.shape[0]
+        P = 2 * A
+        H = tuple(c.HIDDEN_SIZES)
+        key = (E, T, D, A, H)
+        if getattr(self, "_v3_key", None) == key:
+            return self._v3
+        dev = self.device
+        r = env.rank_eff
+        kw = ((r + A + 3) // 4) * 4  # concat width, padded for the GEMM
+        M = torch.zeros(kw, D, device=dev)
+        M[:r] = env.U
+        M[r:r + A] = env.B
+        self._v3 = dict(
+            xva=torch.zeros(E, kw, device=dev), va_off=r, kw=kw,
+            M=M.contiguous(), V=env.V.contiguous(),
+            G=torch.empty(E, D, device=dev),
+            h=[torch.empty(E, hh, device=dev) for hh in H],
+            bz_r=torch.zeros(r, device=dev),
+            bz_D=torch.zeros(D, device=dev),
+            wh=torch.empty(P + 1, H[-1], device=dev),
+            bh=torch.empty(P + 1, device=dev),
+            pd_scratch=torch.empty(E, P, device=dev),
+            seed_dev=torch.zeros(1, dtype=torch.int64, device=dev),
+            eps_dev=torch.zeros(1, dtype=torch.float32, device=dev),
+            epr_before=torch.empty(E, device=dev),
+            empty=torch.empty(0, device=dev),
+        )
+        self._v3_key = key
+        return self._v3
+
+    def _v3_body(self, states, pdflats, actions, values, rewards, dones,
+                 boot_v):
+        """Capture-safe per-step rollout pipeline.  RNG slots match the
+        fused rollout_kernel exactly (same actions/noise/resets per seed);
+        seed/eps are read from device scalars."""
+        from .ops import hip_ext
+
+        ext = hip_ext()
+        c, env, v3 = self.cfg, self.env, self._v3
+        E, T = c.NUM_ENVS, c.MAX_EPOCH_STEPS
+        D = self.obs_space.shape[0]
+        A = self.act_space.shape[0]
+        P = 2 * A
+        act_code = 1 if c.ACTIVATION == "tanh" else 0
+        low = float(self.act_space.low.flat[0])
+        high = float(self.act_space.high.flat[0])
+        n_h = len(c.HIDDEN_SIZES)
+        wh, bh = v3["wh"], v3["bh"]
+        wh[:P].copy_(self.pi.pi.weight.detach())
+        wh[P:].copy_(self.pi.vf.weight.detach())
+        bh[:P].copy_(self.pi.pi.bias.detach())
+        bh[P:].copy_(self.pi.vf.bias.detach())
+        v3["epr_before"].copy_(self.epr)
+        states[0].reshape(-1).copy_(env.x.reshape(-1))
+        for st in range(T):
+            xin = states[st]
+            h = xin
+            for l in range(n_h):
+                hl = v3["h"][l]
+                ext.gemm_fwd(h, self.pi.hidden[l].weight.detach(),
+                             self.pi.hidden[l].bias.detach(), act_code, 0,
+                             hl, hl, hl, 1, 0, 0)
+                h = hl
+            ext.gemm_fwd(h, wh, bh, 2, 1, pdflats[st], values[st],
+                         pdflats[st], 1, 0, 0)
+            # XV = x @ V into the concat buffer's first rank columns
+            ext.gemm_fwd(xin, v3["V"], v3["bz_r"], 2, 0, v3["xva"],
+                         v3["xva"], v3["xva"], 0, 0, v3["kw"])
+            ext.rollout_sample(pdflats[st], actions[st], v3["xva"],
+                               v3["seed_dev"], v3["eps_dev"], st,
+                               v3["va_off"], low, high)
+            # G = [XV | act] @ [U; B]  (wide-N column-panel GEMM)
+            ext.gemm_fwd(v3["xva"], v3["M"], v3["bz_D"], 2, 0, v3["G"],
+                         v3["G"], v3["G"], 0, 0, 0)
+            nxt = states[st + 1] if st + 1 < T else v3["empty"]
+            ext.rollout_env_step(env.x, v3["G"], env.d, env.horizons_i32,
+                                 env.t, self.epr, nxt, rewards[st],
+                                 dones[st], v3["seed_dev"],
+                                 float(env.NOISE), st)
+        # bootstrap value V(x_T)
+        h = env.x
+        for l in range(n_h):
+            hl = v3["h"][l]
+            ext.gemm_fwd(h, self.pi.hidden[l].weight.detach(),
+                         self.pi.hidden[l].bias.detach(), act_code, 0,
+                         hl, hl, hl, 1, 0, 0)
+            h = hl
+        ext.gemm_fwd(h, wh, bh, 2, 1, v3["pd_scratch"], boot_v,
+                     v3["pd_scratch"], 1, 0, 0)
+        return ext.rollout_moments(rewards, dones, v3["epr_before"], T, E)
+
+    @torch.no_grad()
+    def _rollout_once_hip_v3(self) -> Tuple[RolloutBatch, Dict[str, float]]:
+        """Per-step GEMM rollout (see _can_rollout_v3), hipGraph-captured
+        after warmup so replay costs one launch."""
+        c, E = self.cfg, self.cfg.NUM_ENVS
+        T = c.MAX_EPOCH_STEPS
+        eps = self.exploration_rate()
+        env = self.env
+        self._rollout_counter = getattr(self, "_rollout_counter", 0) + 1
+        seed = (
+            c.SEED * 1_000_003
+            + self.comm.rank * 7_919
+            + self._rollout_counter * 104_729
+        ) & 0x7FFFFFFFFFFFFFFF
+        A = self.act_space.shape[0]
+        D = self.obs_space.shape[0]
+        P = 2 * A
+        n_out = T * E * (D + 3 * A + 3) + E + 5
+        if getattr(self, "_rollout_out", None) is None or                 self._rollout_out.numel() != n_out:
+            self._rollout_out = torch.empty(n_out, device=self.device)
+            self._adv_buf = torch.empty(T * E, device=self.device)
+            self._etr_buf = torch.empty(T * E, device=self.device)
+        out = self._rollout_out
+        o = 0
+
+        def take(shape):
+            nonlocal o
+            n = 1
+            for sd in shape:
+                n *= sd
+            v = out.narrow(0, o, n).view(shape)
+            o += n
+            return v
+
+        states = take((T, E, D))
+        pdflats = take((T, E, P))
+        actions = take((T, E, A))
+        values = take((T, E))
+        rewards = take((T, E))
+        dones = take((T, E))
+        boot_v = take((E,))
+        v3 = self._v3_buffers()
+        v3["seed_dev"].fill_(seed & 0xFFFFFFFF)
+        v3["eps_dev"].fill_(float(eps))
+        args = (states, pdflats, actions, values, rewards, dones, boot_v)
+        graph_ok = c.USE_GRAPHS and (
+            not self.comm.distributed
+            or os.environ.get("DPPO_GRAPH_DIST") == "1"
+        )
+        if not graph_ok or getattr(self, "_v3_graph_failed", False):
+            moments = self._v3_body(*args)
+        elif getattr(self, "_v3_graph", None) is None:
+            try:
+                # warmup on a side stream, then restore env/episode state
+                # so capture replays from the true starting state
+                snap = (env.x.clone(), env.t.clone(), self.epr.clone())
+                side = torch.cuda.Stream()
+                side.wait_stream(torch.cuda.current_stream())
+                with torch.cuda.stream(side):
+                    self._v3_body(*args)
+                torch.cuda.current_stream().wait_stream(side)
+                torch.cuda.synchronize()
+                env.x.copy_(snap[0])
+                env.t.copy_(snap[1])
+                self.epr.copy_(snap[2])
+                g = torch.cuda.CUDAGraph()
+                with torch.cuda.graph(g):
+                    self._v3_moments = self._v3_body(*args)
+                self._v3_graph = g
+                g.replay()
+                moments = self._v3_moments
+            except Exception:  # noqa: BLE001 — capture support varies
+                self._v3_graph_failed = True
+                self._v3_graph = None
+                moments = self._v3_body(*args)
+        else:
+            self._v3_graph.replay()
+            moments = self._v3_moments
+        self.obs = env.x
+        return self._finish_hip_rollout(
+            states, pdflats, actions, values, rewards, dones, boot_v,
+            moments, eps)
 
     @torch.no_grad()
     def _rollout_once_eager(self) -> Tuple[RolloutBatch, Dict[str, float]]:
@@ -669,14 +872,14 @@ class DPPOEngine:
             n = B * dims[l + 1]
             cview = acts.narrow(0, o, n).view(B, dims[l + 1])
             ext.gemm_fwd(x, self.pi.hidden[l].weight.detach(), bs[l],
-                         act_code, 0, cview, cview, cview, 1, 0)
+                         act_code, 0, cview, cview, cview, 1, 0, 0)
             x = cview
             a_views.append(cview)
             o += n
         P = 2 * self.act_space.shape[0]
         pdflat = torch.empty(B, P, device=states.device, dtype=states.dtype)
         v = torch.empty(B, device=states.device, dtype=states.dtype)
-        ext.gemm_fwd(x, Wh_cat, bh, 2, 1, pdflat, v, pdflat, 1, 0)
+        ext.gemm_fwd(x, Wh_cat, bh, 2, 1, pdflat, v, pdflat, 1, 0, 0)
         self._Wh_cat = Wh_cat
         return acts, a_views, v, pdflat
 
@@ -713,7 +916,7 @@ class DPPOEngine:
         for l in range(n_hidden - 1, -1, -1):
             dz_l = torch.empty_like(a_views[l])
             ext.gemm_fwd(delta, Wt_chain, dummy_bias, dgrad_code, 0,
-                         dz_l, dz_l, a_views[l], 0, 0)
+                         dz_l, dz_l, a_views[l], 0, 0, 0)
             dz[l] = dz_l
             delta = dz_l
             Wt_chain = self.pi.hidden[l].weight.detach()

@@ -160,3 +160,87 @@ def test_rollout_batch_valid_and_stats():
     assert batch.valid
     row = eng.stats_row(batch, eng.eval_losses(batch, batch.cur_lr))
     assert torch.isfinite(row).all()
+
+
+# ---- v3 per-step GEMM rollout path (trainer._rollout_once_hip_v3) ----
+
+def v3_views(eng):
+    c = eng.cfg
+    T, E = c.MAX_EPOCH_STEPS, c.NUM_ENVS
+    D = eng.obs_space.shape[0]
+    A = eng.act_space.shape[0]
+    out, o, res = eng._rollout_out, 0, []
+    for shape in [(T, E, D), (T, E, 2 * A), (T, E, A), (T, E), (T, E),
+                  (T, E), (E,)]:
+        n = 1
+        for sd in shape:
+            n *= sd
+        res.append(out.narrow(0, o, n).view(shape))
+        o += n
+    return res
+
+
+def test_v3_rollout_invariants(monkeypatch):
+    monkeypatch.setenv("DPPO_ROLLOUT_V3", "1")
+    eng = make_engine(NUM_ENVS=128, MAX_EPOCH_STEPS=16, USE_GRAPHS=False)
+    eng.env.NOISE = 0.0
+    assert eng._can_rollout_v3()
+    batch, _ = eng.rollout_once()
+    assert batch.valid
+    env = eng.env
+    states, pdflats, actions, values, rewards, dones, boot_v = v3_views(eng)
+    T = 16
+    # recorded policy outputs match an eager forward of the recorded states
+    s = states.reshape(-1, states.shape[-1])
+    with torch.no_grad():
+        v_ref, flat_ref = eng.pi(s)
+    torch.testing.assert_close(pdflats.reshape(s.shape[0], -1), flat_ref,
+                               atol=2e-4, rtol=2e-4)
+    torch.testing.assert_close(values.reshape(-1), v_ref, atol=2e-4,
+                               rtol=2e-4)
+    with torch.no_grad():
+        vb, _ = eng.pi(env.x)
+    torch.testing.assert_close(boot_v, vb, atol=2e-4, rtol=2e-4)
+    # exact env transition at noise=0
+    for t in range(T - 1):
+        pred = torch.tanh(states[t] * env.d +
+                          (states[t] @ env.V) @ env.U + actions[t] @ env.B)
+        live = dones[t] == 0
+        torch.testing.assert_close(states[t + 1][live], pred[live],
+                                   atol=3e-5, rtol=3e-5)
+        r_pred = 1.0 - pred.pow(2).mean(dim=-1)
+        torch.testing.assert_close(rewards[t], r_pred, atol=3e-5, rtol=3e-5)
+    # done schedule
+    hor = env.horizons_i32.long().cpu()
+    d_cpu = dones.cpu()
+    for e in range(8):
+        h = int(hor[e])
+        expect = torch.tensor(
+            [1.0 if (t + 1) % h == 0 else 0.0 for t in range(T)])
+        torch.testing.assert_close(d_cpu[:, e], expect)
+
+
+def test_v3_graphed_bitwise_matches_ungraphed(monkeypatch):
+    monkeypatch.setenv("DPPO_ROLLOUT_V3", "1")
+    a = make_engine(NUM_ENVS=64, MAX_EPOCH_STEPS=12, USE_GRAPHS=False)
+    b = make_engine(NUM_ENVS=64, MAX_EPOCH_STEPS=12, USE_GRAPHS=True)
+    ba, _ = a.rollout_once()
+    bb, _ = b.rollout_once()
+    assert b._v3_graph is not None or b._v3_graph_failed
+    assert torch.equal(ba.states, bb.states)
+    assert torch.equal(ba.actions, bb.actions)
+    assert torch.equal(ba.oldflat, bb.oldflat)
+    # second round replays the captured graph; engines must stay in lockstep
+    sa, _ = a.train_round()
+    sb, _ = b.train_round()
+    assert torch.equal(a.flat_pi.flat_param, b.flat_pi.flat_param)
+
+
+def test_v3_training_rounds(monkeypatch):
+    monkeypatch.setenv("DPPO_ROLLOUT_V3", "1")
+    eng = make_engine(NUM_ENVS=128, MAX_EPOCH_STEPS=16)
+    p0 = eng.flat_pi.flat_param.detach().clone()
+    for _ in range(3):
+        stats, _ = eng.train_round()
+    assert all(math.isfinite(v) for v in stats.values())
+    assert not torch.equal(p0, eng.flat_pi.flat_param.detach())

@@ -25,10 +25,10 @@ dummy = torch.zeros(1, device="cuda")
 
 ABL = 0
 dwbuf = torch.zeros(H * D + H, device="cuda")
-def l1(): ext.gemm_fwd(X, W1, b1, 1, 0, h1, h1, h1, 1, ABL)
-def l2(): ext.gemm_fwd(h1, W2, b1, 1, 0, h2, h2, h2, 1, ABL)
-def heads(): ext.gemm_fwd(h2, Wh, bh, 2, 1, pdf, v, pdf, 1, ABL)
-def dgrad(): ext.gemm_fwd(gh, Wh, dummy, 3, 0, dz2, dz2, h2, 0, ABL)
+def l1(): ext.gemm_fwd(X, W1, b1, 1, 0, h1, h1, h1, 1, ABL, 0)
+def l2(): ext.gemm_fwd(h1, W2, b1, 1, 0, h2, h2, h2, 1, ABL, 0)
+def heads(): ext.gemm_fwd(h2, Wh, bh, 2, 1, pdf, v, pdf, 1, ABL, 0)
+def dgrad(): ext.gemm_fwd(gh, Wh, dummy, 3, 0, dz2, dz2, h2, 0, ABL, 0)
 def dw1(): ext.dw_mfma(dz2, X, dwbuf, 0, H*D, -1, -1, -1, ABL)
 
 act = torch.randn(B, A, device="cuda")
