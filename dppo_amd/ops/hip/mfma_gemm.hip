@@ -251,7 +251,12 @@ __global__ void gemm_fwd_pipe_kernel(FwdArgs a) {
   // PBK: K-step per stage.  32 gives the prefetch twice the MFMA cycles
   // to hide under but costs 177 VGPR (2 waves/SIMD); 16 keeps 4
   // waves/SIMD.  Both instantiated; dispatch measures/chooses per shape.
-  __shared__ float xs[FWD_M][PBK + 1];
+  // X rows padded +4 (not +1): a 17-float row stride leaves the float4
+  // staging writes 16B-MISALIGNED, so hipcc splits them into scalar
+  // ds_writes (PMC: 5.7e8 bank-conflict cycles on 3.2e8 LDS ops).  +4
+  // keeps b128 writes; the A-fragment column reads pick up a 4-way bank
+  // conflict but there are only 8 of them per wave-stage.
+  __shared__ float xs[FWD_M][PBK + 4];
   __shared__ float ws[PBK][NT * M_WAVE];
 
   const int i_l = lane & 31;
