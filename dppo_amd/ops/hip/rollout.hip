@@ -648,22 +648,34 @@ __global__ void env_finish_kernel(
       ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) / WAVE;
   if (wid >= E) return;
   const int64_t e = wid;
-  // up to 512/64 = 8 elements per lane (MAX_D = 512)
-  float xn[(MAX_D + WAVE - 1) / WAVE];
+  // float4 per lane (the v3 gate guarantees D%4==0): the scalar version
+  // kept only ~4 B/lane in flight and ran 2x its streaming floor.  Noise
+  // slots differ from the fused kernel (one Box-Muller pair feeds d and
+  // d+1 of the SAME env, slot 1000+d/2) — each path is deterministic per
+  // seed on its own; nothing requires the two paths to share trajectories.
+  constexpr int CMAX = (MAX_D / 4 + WAVE - 1) / WAVE;  // float4 chunks/lane
+  const int nc4 = D / 4;
+  float4 xn[CMAX];
   float ss = 0.f;
-  int nd = 0;
-  for (int d = lane; d < D; d += WAVE, ++nd) {
-    // same noise stream as the fused kernel: one Box-Muller pair per
-    // (even env, d) feeds envs 2q and 2q+1
-    float nz = 0.f;
+  int it = 0;
+  for (int c = lane; c < nc4; c += WAVE, ++it) {
+    const int d = 4 * c;
+    const float4 xv = *reinterpret_cast<const float4*>(&x[e * D + d]);
+    const float4 gv = *reinterpret_cast<const float4*>(&G[e * D + d]);
+    const float4 dv = *reinterpret_cast<const float4*>(&envd[d]);
+    float nz[4] = {0.f, 0.f, 0.f, 0.f};
     if (sigma != 0.f) {
-      const float2 pr = rng_normal2(seed, (int)(e & ~1LL), step, 1000 + d);
-      nz = (e & 1) ? pr.y : pr.x;
+      const float2 p0 = rng_normal2(seed, (int)e, step, 1000 + 2 * c);
+      const float2 p1 = rng_normal2(seed, (int)e, step, 1000 + 2 * c + 1);
+      nz[0] = p0.x; nz[1] = p0.y; nz[2] = p1.x; nz[3] = p1.y;
     }
-    const float v = fast_tanhf(x[e * D + d] * envd[d] + G[e * D + d] +
-                               sigma * nz);
-    xn[nd] = v;
-    ss += v * v;
+    float4 v;
+    v.x = fast_tanhf(xv.x * dv.x + gv.x + sigma * nz[0]);
+    v.y = fast_tanhf(xv.y * dv.y + gv.y + sigma * nz[1]);
+    v.z = fast_tanhf(xv.z * dv.z + gv.z + sigma * nz[2]);
+    v.w = fast_tanhf(xv.w * dv.w + gv.w + sigma * nz[3]);
+    xn[it] = v;
+    ss += v.x * v.x + v.y * v.y + v.z * v.z + v.w * v.w;
   }
   ss = wave_reduce_sum(ss);
   int done;
@@ -682,12 +694,19 @@ __global__ void env_finish_kernel(
     t[e] = tc;
   }
   done = __shfl(done, 0, WAVE);
-  nd = 0;
-  for (int d = lane; d < D; d += WAVE, ++nd) {
-    const float v =
-        done ? 0.1f * rng_normal(seed, (int)e, step, 5000 + d) : xn[nd];
-    x[e * D + d] = v;
-    if (states_next != nullptr) states_next[e * D + d] = v;
+  it = 0;
+  for (int c = lane; c < nc4; c += WAVE, ++it) {
+    const int d = 4 * c;
+    float4 v = xn[it];
+    if (done) {
+      v.x = 0.1f * rng_normal(seed, (int)e, step, 5000 + d);
+      v.y = 0.1f * rng_normal(seed, (int)e, step, 5000 + d + 1);
+      v.z = 0.1f * rng_normal(seed, (int)e, step, 5000 + d + 2);
+      v.w = 0.1f * rng_normal(seed, (int)e, step, 5000 + d + 3);
+    }
+    *reinterpret_cast<float4*>(&x[e * D + d]) = v;
+    if (states_next != nullptr)
+      *reinterpret_cast<float4*>(&states_next[e * D + d]) = v;
   }
 }
 
@@ -838,7 +857,7 @@ void rollout_env_step(torch::Tensor x, torch::Tensor G, torch::Tensor envd,
   const int64_t E = x.size(0);
   const int D = static_cast<int>(x.size(1));
   TORCH_CHECK(x.is_cuda() && x.is_contiguous() && G.numel() >= E * D);
-  TORCH_CHECK(D <= MAX_D);
+  TORCH_CHECK(D <= MAX_D && D % 4 == 0, "env_finish_kernel is float4-wide");
   TORCH_CHECK(horizons.dtype() == torch::kInt32 && t.dtype() == torch::kInt32);
   hipStream_t stream = c10::hip::getCurrentHIPStream().stream();
   const int64_t waves = E;
