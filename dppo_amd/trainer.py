@@ -348,6 +348,18 @@ class DPPOEngine:
         )
         return batch, {"exploration_rate": eps}
 
+    def _wt_bufs(self):
+        """Persistent [in][out] transposed hidden-layer weight buffers."""
+        shapes = tuple(
+            (lay.weight.shape[1], lay.weight.shape[0]) for lay in self.pi.hidden
+        )
+        if getattr(self, "_wt_key", None) != shapes:
+            self._wt = [
+                torch.empty(sh, device=self.device) for sh in shapes
+            ]
+            self._wt_key = shapes
+        return self._wt
+
     def _v3_buffers(self):
         """Persistent device buffers for the per-step GEMM rollout."""
         c, env = self.cfg, self.env
@@ -405,6 +417,9 @@ class DPPOEngine:
         wh[P:].copy_(self.pi.vf.weight.detach())
         bh[:P].copy_(self.pi.pi.bias.detach())
         bh[P:].copy_(self.pi.vf.bias.detach())
+        wts = self._wt_bufs()
+        for l, lay in enumerate(self.pi.hidden):
+            wts[l].copy_(lay.weight.detach().t())
         v3["epr_before"].copy_(self.epr)
         states[0].reshape(-1).copy_(env.x.reshape(-1))
         for st in range(T):
@@ -412,9 +427,9 @@ class DPPOEngine:
             h = xin
             for l in range(n_h):
                 hl = v3["h"][l]
-                ext.gemm_fwd(h, self.pi.hidden[l].weight.detach(),
+                ext.gemm_fwd(h, wts[l],
                              self.pi.hidden[l].bias.detach(), act_code, 0,
-                             hl, hl, hl, 1, 0, 0)
+                             hl, hl, hl, 0, 0, 0)
                 h = hl
             ext.gemm_fwd(h, wh, bh, 2, 1, pdflats[st], values[st],
                          pdflats[st], 1, 0, 0)
@@ -436,9 +451,9 @@ class DPPOEngine:
         h = env.x
         for l in range(n_h):
             hl = v3["h"][l]
-            ext.gemm_fwd(h, self.pi.hidden[l].weight.detach(),
+            ext.gemm_fwd(h, wts[l],
                          self.pi.hidden[l].bias.detach(), act_code, 0,
-                         hl, hl, hl, 1, 0, 0)
+                         hl, hl, hl, 0, 0, 0)
             h = hl
         ext.gemm_fwd(h, wh, bh, 2, 1, v3["pd_scratch"], boot_v,
                      v3["pd_scratch"], 1, 0, 0)
@@ -489,10 +504,9 @@ class DPPOEngine:
         v3["seed_dev"].fill_(seed & 0xFFFFFFFF)
         v3["eps_dev"].fill_(float(eps))
         args = (states, pdflats, actions, values, rewards, dones, boot_v)
-        graph_ok = c.USE_GRAPHS and (
-            not self.comm.distributed
-            or os.environ.get("DPPO_GRAPH_DIST") == "1"
-        )
+        # unlike the update graph, the rollout graph contains NO
+        # collectives — capture is safe on every rank
+        graph_ok = c.USE_GRAPHS
         if not graph_ok or getattr(self, "_v3_graph_failed", False):
             moments = self._v3_body(*args)
         elif getattr(self, "_v3_graph", None) is None:
@@ -859,9 +873,17 @@ class DPPOEngine:
         n_hidden = len(c.HIDDEN_SIZES)
         act_code = 1 if c.ACTIVATION == "tanh" else 0
         with torch.no_grad():
-            # torch weight layouts pass straight through: the GEMM stages
-            # [N][K] weights transposed into LDS itself (wt_layout=1)
+            # hidden-layer weights go in PRE-TRANSPOSED ([in][out],
+            # wt_layout 0): layout-0 staging measured -16% on the big-K
+            # layer vs staging the torch [out][in] layout transposed in
+            # LDS (layout 1's per-thread scalar column writes).  The
+            # transpose copies are <100 KB and re-done per call since
+            # params change every update step (capture-safe: persistent
+            # buffers + copy_).
             bs = [lay.bias.detach() for lay in self.pi.hidden]
+            wts = self._wt_bufs()
+            for l, lay in enumerate(self.pi.hidden):
+                wts[l].copy_(lay.weight.detach().t())
             Wh_cat = torch.cat(
                 [self.pi.pi.weight, self.pi.vf.weight], dim=0
             ).contiguous()  # [P+1][HL]: heads fwd (layout 1) AND dgrad Wt
@@ -873,8 +895,8 @@ class DPPOEngine:
         for l in range(n_hidden):
             n = B * dims[l + 1]
             cview = acts.narrow(0, o, n).view(B, dims[l + 1])
-            ext.gemm_fwd(x, self.pi.hidden[l].weight.detach(), bs[l],
-                         act_code, 0, cview, cview, cview, 1, 0, 0)
+            ext.gemm_fwd(x, wts[l], bs[l],
+                         act_code, 0, cview, cview, cview, 0, 0, 0)
             x = cview
             a_views.append(cview)
             o += n

@@ -25,6 +25,10 @@ dummy = torch.zeros(1, device="cuda")
 
 ABL = 0
 dwbuf = torch.zeros(H * D + H, device="cuda")
+W1t = W1.t().contiguous()  # [D][H] layout-0 variant
+W2t = W2.t().contiguous()
+def l1t(): ext.gemm_fwd(X, W1t, b1, 1, 0, h1, h1, h1, 0, ABL, 0)
+def l2t(): ext.gemm_fwd(h1, W2t, b1, 1, 0, h2, h2, h2, 0, ABL, 0)
 def l1(): ext.gemm_fwd(X, W1, b1, 1, 0, h1, h1, h1, 1, ABL, 0)
 def l2(): ext.gemm_fwd(h1, W2, b1, 1, 0, h2, h2, h2, 1, ABL, 0)
 def heads(): ext.gemm_fwd(h2, Wh, bh, 2, 1, pdf, v, pdf, 1, ABL, 0)
@@ -41,15 +45,15 @@ def ghk(): ext.ppo_loss_gauss_gh(pdf, oldf, v, oldvb, act, advb, etrb, 0.2, 0.01
 def lossf(): ext.ppo_loss_gauss_fwd(pdf, oldf, v, oldvb, act, advb, etrb, 0.2, 0.01, 0.5)
 
 # traffic per call (GB), for effective-bandwidth reporting
-GB = {"L1": (B*(D+H))*4e-9, "L2": (B*2*H)*4e-9, "heads": (B*(H+P+1))*4e-9,
+GB = {"L1": (B*(D+H))*4e-9, "L1t": (B*(D+H))*4e-9, "L2t": (B*2*H)*4e-9, "L2": (B*2*H)*4e-9, "heads": (B*(H+P+1))*4e-9,
       "dgrad": (B*(P+1+2*H))*4e-9, "dw1": (B*(D+H))*4e-9,
       "gh": (B*(2*P+A+3+P+1))*4e-9, "lossf": (B*(2*P+A+3))*4e-9}
 
 def run(label):
     out = {}
-    for name, fn in [("L1", l1), ("L2", l2), ("heads", heads),
-                     ("dgrad", dgrad), ("dw1", dw1), ("gh", ghk),
-                     ("lossf", lossf)]:
+    for name, fn in [("L1", l1), ("L1t", l1t), ("L2", l2), ("L2t", l2t),
+                     ("heads", heads), ("dgrad", dgrad), ("dw1", dw1),
+                     ("gh", ghk), ("lossf", lossf)]:
         for _ in range(3): fn()
         torch.cuda.synchronize()
         t0 = time.perf_counter()
