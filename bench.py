@@ -57,8 +57,10 @@ PRESETS = {
     # BASELINE.json config 5: wide MFMA-bound MLP, bf16 compute (as named)
     "wide4096": dict(
         GAME="Wide-4096", HIDDEN_SIZES=(4096, 4096, 4096, 4096),
-        ACTIVATION="tanh", NUM_ENVS=1024, MAX_EPOCH_STEPS=16,
+        ACTIVATION="tanh", NUM_ENVS=4096, MAX_EPOCH_STEPS=16,
         DTYPE="bfloat16",
+        # 4096 envs: measured optimum (1024 -> 535K, 4096 -> 573K,
+        # 8192 -> 534K, 16384 -> 559K env-steps/s with the bf16 rollout).
     ),
 }
 

@@ -37,6 +37,7 @@ deadlocks.
 
 from __future__ import annotations
 
+import contextlib
 import math
 import os
 from dataclasses import dataclass
@@ -189,6 +190,14 @@ class DPPOEngine:
             n, a_dim, device=self.device, dtype=self.dtype
         )
 
+    def _amp(self):
+        """bf16 autocast context for network forwards (BASELINE config 5:
+        DTYPE='bfloat16' runs every GEMM — rollout acting, bootstrap value
+        and update-path — in bf16 with fp32 loss/distribution math)."""
+        if self.cfg.DTYPE == "bfloat16" and self.device.type == "cuda":
+            return torch.autocast("cuda", dtype=torch.bfloat16)
+        return contextlib.nullcontext()
+
     @torch.no_grad()
     def act_batch(self, obs: torch.Tensor, eps: float) -> Tuple[torch.Tensor, torch.Tensor]:
         """Sample actions + values for a batch of states, with the
@@ -196,7 +205,9 @@ class DPPOEngine:
         NOT recorded — the ratio re-evaluates oldpi on the batch later,
         exactly the reference's formulation (Worker.py:149-152 + PPO.py:31;
         SURVEY.md §7 'hard parts')."""
-        v, pdflat = self.pi(obs)
+        with self._amp():
+            v, pdflat = self.pi(obs)
+        v, pdflat = v.float(), pdflat.float()
         pd = self.pi.pdtype.pdfromflat(pdflat)
         a = pd.sample()
         if eps > 0.0:
@@ -596,7 +607,7 @@ class DPPOEngine:
             self.epr *= 1.0 - donef
         self.obs = obs
 
-        with torch.no_grad():
+        with torch.no_grad(), self._amp():
             boot_v, _ = self.pi(obs)
         adv, etr = gae_advantages(
             rewards, values, dones, boot_v.float(),
@@ -648,14 +659,11 @@ class DPPOEngine:
         the 3 extra GEMMs per update step."""
         if recorded_pi:
             v, pdflat = batch.oldv, batch.oldflat
-        elif self.cfg.DTYPE == "bfloat16" and self.device.type == "cuda":
-            # bf16 compute for the wide MFMA-bound config (BASELINE #5):
-            # rocBLAS bf16 GEMMs with fp32 loss math
-            with torch.autocast("cuda", dtype=torch.bfloat16):
+        else:
+            # bf16 config (BASELINE #5): bf16 GEMMs, fp32 loss math
+            with self._amp():
                 v, pdflat = self.pi(batch.states)
             v, pdflat = v.float(), pdflat.float()
-        else:
-            v, pdflat = self.pi(batch.states)
         pd = self.pi.pdtype.pdfromflat(pdflat)
         oldpd = self.pi.pdtype.pdfromflat(batch.oldflat)
         coeffs = PPOLossCoeffs(
