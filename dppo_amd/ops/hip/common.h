@@ -11,6 +11,7 @@
 #define DEV_INLINE __device__ __forceinline__
 
 constexpr int WAVE = 64;  // CDNA wavefront width (not 32)
+constexpr int N_CU = 256;  // MI355X: 256 CUs in 8 XCDs
 
 DEV_INLINE float wave_reduce_sum(float x) {
   #pragma unroll

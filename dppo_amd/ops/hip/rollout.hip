@@ -791,16 +791,23 @@ std::vector<torch::Tensor> rollout_run(
        NWAVES * ENV_TILE * r4(h_max)) *
       sizeof(float);
   // MINWAVES=4 measured 24.8 ms vs 27.7 ms at MINWAVES=3 (65k envs):
-  // the 4th wave/SIMD buys more than the 29-VGPR spill costs.
-  static const int mw = []() {
+  // the 4th wave/SIMD buys more than the 29-VGPR spill costs.  Below
+  // ~1 block/CU (tiny-E configs: BASELINE #2/#4) occupancy is grid-bound
+  // anyway, so MINWAVES=1 lifts the 128-VGPR cap — no spills, no
+  // per-lane scratch on the latency-critical per-step chain.
+  static const int mw_env = []() {
     const char* e = getenv("DPPO_ROLLOUT_MW");
-    return e ? atoi(e) : 4;
+    return e ? atoi(e) : 0;
   }();
+  const int mw = mw_env ? mw_env : (grid <= N_CU ? 1 : 4);
   if (mw >= 4)
     hipLaunchKernelGGL(rollout_kernel<4>, dim3(grid), dim3(NWAVES * WAVE),
                        lds_bytes, stream, a);
-  else
+  else if (mw >= 3)
     hipLaunchKernelGGL(rollout_kernel<3>, dim3(grid), dim3(NWAVES * WAVE),
+                       lds_bytes, stream, a);
+  else
+    hipLaunchKernelGGL(rollout_kernel<1>, dim3(grid), dim3(NWAVES * WAVE),
                        lds_bytes, stream, a);
 
   // carve views out of the blob

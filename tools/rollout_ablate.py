@@ -8,8 +8,10 @@ from dppo_amd.trainer import DPPOEngine
 from dppo_amd.ops import require_hip_ext
 
 ext = require_hip_ext()
+E = int(sys.argv[1]) if len(sys.argv) > 1 else 65536
+T = int(sys.argv[2]) if len(sys.argv) > 2 else 64
 cfg = DPPOConfig(GAME="Humanoid-v4", HIDDEN_SIZES=(64, 64), ACTIVATION="tanh",
-                 NUM_ENVS=65536, MAX_EPOCH_STEPS=64, EPOCH_MAX=10**6,
+                 NUM_ENVS=E, MAX_EPOCH_STEPS=T, EPOCH_MAX=10**6,
                  STOP_EPOCH=10**6, NUM_WORKERS=1, LOG_FILE_PATH="/tmp/l",
                  DEVICE="cuda")
 eng = DPPOEngine(cfg, comm=Comm(device="cuda:0"))
@@ -21,7 +23,7 @@ low, high = -1.0, 1.0
 def run(mask):
     ext.rollout_run(blob, offsets, dims, 1, env.blob, env.rank_eff,
                     env.horizons_i32, float(env.NOISE), low, high, 0.2,
-                    env.x, env.t, eng.epr, 64, 17, 1234,
+                    env.x, env.t, eng.epr, T, 17, 1234,
                     eng._rollout_out, mask)
 
 for name, mask in [("full", 0), ("-trunk", 1), ("-heads", 2), ("-sample", 4),
