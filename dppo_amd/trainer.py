@@ -767,6 +767,20 @@ class DPPOEngine:
         clip = self.cfg.CLIP_PARAM * l_mul
         if fuse and getattr(self, "_clip_dev", None) is not None:
             self._clip_dev.fill_(clip)
+        if (fuse and self.cfg.USE_GRAPHS
+                and (not self.comm.distributed
+                     or os.environ.get("DPPO_GRAPH_DIST") == "1")
+                and not getattr(self, "_mb_graph_failed", False)):
+            try:
+                self._update_minibatched_graphed(batch, l_mul)
+                return
+            except Exception:  # noqa: BLE001 — capture support varies
+                if os.environ.get("DPPO_DEBUG"):
+                    import traceback
+
+                    traceback.print_exc()
+                self._mb_graph_failed = True
+                self._mb_graphs = None
         for _ in range(self.cfg.UPDATE_STEPS):
             for o in range(0, B, mb):
                 n = min(mb, B - o)
@@ -795,6 +809,97 @@ class DPPOEngine:
                     losses["total_loss"].backward()
                 self.comm.allreduce_mean_(self.flat_pi.flat_grad)
                 self.optimizer.step()
+
+    # chunks per captured graph segment: bounds per-graph node count
+    # (~11 kernels/chunk) while keeping replay overhead negligible
+    _MB_SEG = 1024
+
+    def _update_minibatched_graphed(self, batch: RolloutBatch,
+                                    l_mul: float) -> None:
+        """hipGraph-captured minibatch epochs (BASELINE config 4): the
+        sequential chunk walk is identical every epoch and every round —
+        the rollout/GAE outputs live in persistent buffers, so chunk
+        addresses are stable — and is captured ONCE in segments of
+        _MB_SEG chunks, then replayed UPDATE_STEPS times per round.  The
+        uncaptured loop pays ~260 us of launch/alloc overhead per
+        4096-sample chunk; replay pays in-graph dispatch only.  Per-round
+        scalars (clip, lr, Adam step) are device-resident as in
+        _update_graphed.  A tail chunk (B % mb != 0) runs uncaptured
+        after each epoch's replays, preserving chunk order; Adam state is
+        fully device-side so mixing is exact."""
+        c = self.cfg
+        mb = c.MINIBATCH_SIZE
+        B = batch.states.shape[0]
+        n_full = B // mb
+        tail = B - n_full * mb
+        clip = c.CLIP_PARAM * l_mul
+        if getattr(self, "_clip_dev", None) is None:
+            self._clip_dev = torch.tensor([clip], device=self.device)
+        else:
+            self._clip_dev.fill_(clip)
+        opt = self.optimizer
+        opt.lr_dev.fill_(c.LEARNING_RATE * l_mul)
+        key = (batch.states.data_ptr(), batch.actions.data_ptr(),
+               batch.adv.data_ptr(), batch.etr.data_ptr(),
+               batch.oldflat.data_ptr(), batch.oldv.data_ptr(), B, mb)
+        if getattr(self, "_mb_graphs", None) is None or self._mb_key != key:
+            self._mb_capture(batch, key, n_full)
+        for _ in range(c.UPDATE_STEPS):
+            for g in self._mb_graphs:
+                g.replay()
+            if tail:
+                self._mb_chunk_step(batch, n_full * mb, tail)
+
+    def _mb_chunk_step(self, batch: RolloutBatch, o: int, n: int) -> None:
+        """One fused minibatch gradient step on batch[o:o+n] (capture-safe:
+        clip/lr/step all come from device memory)."""
+        s = batch.states.narrow(0, o, n)
+        acts, a_views, v, pdflat = self._fused_forward(s)
+        self.flat_pi.zero_grad()
+        self._fused_backward(
+            s, acts, a_views, v, pdflat,
+            batch.oldflat.narrow(0, o, n), batch.oldv.narrow(0, o, n),
+            batch.actions.narrow(0, o, n), batch.adv.narrow(0, o, n),
+            batch.etr.narrow(0, o, n), 0.0,  # clip read from _clip_dev
+        )
+        self.comm.allreduce_mean_(self.flat_pi.flat_grad)
+        self.optimizer.step_captured()
+
+    def _mb_capture(self, batch: RolloutBatch, key, n_full: int) -> None:
+        mb = self.cfg.MINIBATCH_SIZE
+        opt = self.optimizer
+        # warmup executes real steps (lazy inits, allocator) — snapshot
+        # and restore the full optimizer+param state around it
+        snap = (
+            self.flat_pi.flat_param.detach().clone(),
+            opt.exp_avg.clone(), opt.exp_avg_sq.clone(),
+            opt.step_dev.clone(),
+        )
+        try:
+            side = torch.cuda.Stream()
+            side.wait_stream(torch.cuda.current_stream())
+            with torch.cuda.stream(side):
+                for ci in range(min(2, n_full)):
+                    self._mb_chunk_step(batch, ci * mb, mb)
+            torch.cuda.current_stream().wait_stream(side)
+            torch.cuda.synchronize()
+        finally:
+            with torch.no_grad():
+                self.flat_pi.flat_param.copy_(snap[0])
+                opt.exp_avg.copy_(snap[1])
+                opt.exp_avg_sq.copy_(snap[2])
+                opt.step_dev.copy_(snap[3])
+        pool = torch.cuda.graph_pool_handle()
+        graphs = []
+        for s0 in range(0, n_full, self._MB_SEG):
+            s1 = min(n_full, s0 + self._MB_SEG)
+            g = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(g, pool=pool):
+                for ci in range(s0, s1):
+                    self._mb_chunk_step(batch, ci * mb, mb)
+            graphs.append(g)
+        self._mb_graphs = graphs
+        self._mb_key = key
 
     def _clip_dev_or_empty(self) -> torch.Tensor:
         t = getattr(self, "_clip_dev", None)

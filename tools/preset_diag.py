@@ -45,3 +45,7 @@ tot = sum(eng.timers.totals.values())
 print(f"phase split ({rounds} rounds, sync timers):")
 for k, v in sorted(eng.timers.totals.items(), key=lambda kv: -kv[1]):
     print(f"  {k:16s} {v / rounds * 1000:8.2f} ms/round  {100 * v / tot:5.1f}%")
+print("mb_graphs:", len(getattr(eng, "_mb_graphs", None) or []),
+      "mb_graph_failed:", getattr(eng, "_mb_graph_failed", False),
+      "upd_graph:", getattr(eng, "_upd_graph", None) is not None,
+      "v3:", eng._can_rollout_v3(), "fuse_rollout:", eng._can_fuse_rollout())
