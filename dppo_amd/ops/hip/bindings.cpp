@@ -60,6 +60,18 @@ void rollout_env_step(torch::Tensor x, torch::Tensor G, torch::Tensor envd,
 torch::Tensor rollout_moments(torch::Tensor rewards, torch::Tensor dones,
                               torch::Tensor epr_in, int64_t T, int64_t E);
 
+bool mlp_chunk_supported(int64_t D, int64_t H, int64_t A, int64_t n_hidden);
+
+void mlp_chunk_train(
+    torch::Tensor params, torch::Tensor states, torch::Tensor actions,
+    torch::Tensor adv, torch::Tensor etr, torch::Tensor oldflat,
+    torch::Tensor oldv, std::vector<int64_t> offsets,
+    std::vector<int64_t> dims, int64_t activation, torch::Tensor clip_dev,
+    double clip, double entcoeff, double vcoeff, torch::Tensor slabs,
+    torch::Tensor mom, torch::Tensor vel, torch::Tensor step_dev,
+    torch::Tensor lr_dev, torch::Tensor coef, torch::Tensor flat_grad,
+    bool fuse_adam, double beta1, double beta2, double eps);
+
 std::vector<torch::Tensor> rollout_run(
     torch::Tensor params, std::vector<int64_t> offsets,
     std::vector<int64_t> dims, int64_t activation,
@@ -93,4 +105,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
           "MFMA f32 layer forward C=act(X@Wt+b), fused tanh (gfx950)");
   mod.def("dw_mfma", &dw_mfma,
           "MFMA f32 split-K dW += delta^T@acts into flat grad (gfx950)");
+  mod.def("mlp_chunk_supported", &mlp_chunk_supported,
+          "shape eligibility for the fused chunk-step kernel (gfx950)");
+  mod.def("mlp_chunk_train", &mlp_chunk_train,
+          "fused small-MLP training chunk step: fwd+PPO grad+bwd+dW "
+          "partials + slab-reduce Adam (gfx950)");
 }

@@ -781,9 +781,15 @@ class DPPOEngine:
                     traceback.print_exc()
                 self._mb_graph_failed = True
                 self._mb_graphs = None
+        use_ck = fuse and self._can_chunk_kernel(min(mb, B))
+        if use_ck:
+            self.optimizer.lr_dev.fill_(self.cfg.LEARNING_RATE * l_mul)
         for _ in range(self.cfg.UPDATE_STEPS):
             for o in range(0, B, mb):
                 n = min(mb, B - o)
+                if use_ck:
+                    self._chunk_kernel_step(batch, o, n, clip)
+                    continue
                 sub = RolloutBatch(
                     states=batch.states.narrow(0, o, n),
                     actions=batch.actions.narrow(0, o, n),
@@ -853,6 +859,9 @@ class DPPOEngine:
     def _mb_chunk_step(self, batch: RolloutBatch, o: int, n: int) -> None:
         """One fused minibatch gradient step on batch[o:o+n] (capture-safe:
         clip/lr/step all come from device memory)."""
+        if self._can_chunk_kernel(n):
+            self._chunk_kernel_step(batch, o, n, 0.0)  # clip via _clip_dev
+            return
         s = batch.states.narrow(0, o, n)
         acts, a_views, v, pdflat = self._fused_forward(s)
         self.flat_pi.zero_grad()
@@ -900,6 +909,74 @@ class DPPOEngine:
             graphs.append(g)
         self._mb_graphs = graphs
         self._mb_key = key
+
+    # Max batch routed to the fused chunk-step kernel (mlp_train.hip).
+    # DISABLED by default (0): the kernel is numerics-exact (GPU tests
+    # force-enable it) but its first cut measured ~500 us per 4096-sample
+    # chunk — slower than the ~258 us kernel chain it replaces — so
+    # routing stays off until the latency is understood (profiles/
+    # r01_chunk_kernel_notes.md).  Tests/experiments override per
+    # instance or via DPPO_CHUNK_KERNEL_MAX_B.
+    CHUNK_KERNEL_MAX_B = int(os.environ.get("DPPO_CHUNK_KERNEL_MAX_B", "0"))
+
+    def _can_chunk_kernel(self, B: int) -> bool:
+        """Eligibility for the fused single-kernel chunk step
+        (ops/hip/mlp_train.hip): the whole fwd + PPO grad + bwd + dW +
+        Adam for one (mini)batch chunk in two kernel launches."""
+        from .ops import use_hip, hip_ext
+        from .ops.adam import FusedFlatAdam
+
+        c = self.cfg
+        if self._discrete or c.DTYPE != "float32":
+            return False
+        if not use_hip(self.device, c.USE_HIP_KERNELS):
+            return False
+        if B > self.CHUNK_KERNEL_MAX_B:
+            return False
+        if not isinstance(self.optimizer, FusedFlatAdam):
+            return False
+        hs = c.HIDDEN_SIZES
+        if len(hs) not in (1, 2) or len(set(hs)) != 1:
+            return False
+        return bool(hip_ext().mlp_chunk_supported(
+            self.obs_space.shape[0], hs[0], self.act_space.shape[0], len(hs)))
+
+    def _chunk_scratch(self) -> torch.Tensor:
+        """Per-block gradient slab scratch (<=256 blocks x padded P)."""
+        if getattr(self, "_chunk_slabs", None) is None:
+            ppad = (self.flat_pi.flat_param.numel() + 3) & ~3
+            self._chunk_slabs = torch.zeros(
+                256 * ppad, device=self.device, dtype=torch.float32)
+        return self._chunk_slabs
+
+    def _chunk_kernel_step(self, batch: RolloutBatch, o: int, n: int,
+                           clip: float) -> None:
+        """One fused chunk gradient+Adam step on batch[o:o+n].  Single
+        rank: kernel-fused Adam (capture-safe device scalars).  Multi
+        rank: the kernel writes the summed gradient, then the usual
+        all-reduce + device-state Adam step run.  A live _clip_dev
+        overrides `clip` inside the kernel (graphed callers pass 0)."""
+        from .ops import hip_ext
+
+        c = self.cfg
+        opt = self.optimizer
+        offsets = [sl.start for sl in self.flat_pi.slices]
+        dims = [self.obs_space.shape[0], *c.HIDDEN_SIZES]
+        fuse = not self.comm.distributed
+        hip_ext().mlp_chunk_train(
+            self.flat_pi.flat_param.data,
+            batch.states.narrow(0, o, n), batch.actions.narrow(0, o, n),
+            batch.adv.narrow(0, o, n), batch.etr.narrow(0, o, n),
+            batch.oldflat.narrow(0, o, n), batch.oldv.narrow(0, o, n),
+            offsets, dims, 1 if c.ACTIVATION == "tanh" else 0,
+            self._clip_dev_or_empty(), clip, c.ENTCOEFF, c.VCOEFF,
+            self._chunk_scratch(), opt.exp_avg, opt.exp_avg_sq,
+            opt.step_dev, opt.lr_dev, opt.coef,
+            self.flat_pi.flat_grad, fuse,
+            opt.betas[0], opt.betas[1], opt.eps)
+        if not fuse:
+            self.comm.allreduce_mean_(self.flat_pi.flat_grad)
+            opt.step_captured()
 
     def _clip_dev_or_empty(self) -> torch.Tensor:
         t = getattr(self, "_clip_dev", None)
@@ -959,6 +1036,11 @@ class DPPOEngine:
     def _update_body(self, batch: RolloutBatch) -> None:
         """The capture-safe UPDATE_STEPS pipeline (no host syncs, no
         host-valued scalars: clip/lr/step come from device memory)."""
+        B = batch.states.shape[0]
+        if self._can_chunk_kernel(B):
+            for _ in range(self.cfg.UPDATE_STEPS):
+                self._chunk_kernel_step(batch, 0, B, 0.0)  # clip: _clip_dev
+            return
         for _ in range(self.cfg.UPDATE_STEPS):
             acts, a_views, v, pdflat = self._fused_forward(batch.states)
             self.flat_pi.zero_grad()
@@ -1073,6 +1155,12 @@ class DPPOEngine:
         # coherent when falling back from the graphed path
         if getattr(self, "_clip_dev", None) is not None:
             self._clip_dev.fill_(clip)
+        B = batch.states.shape[0]
+        if self._can_chunk_kernel(B):
+            self.optimizer.lr_dev.fill_(self.cfg.LEARNING_RATE * l_mul)
+            for _ in range(self.cfg.UPDATE_STEPS):
+                self._chunk_kernel_step(batch, 0, B, clip)
+            return
         for _ in range(self.cfg.UPDATE_STEPS):
             acts, a_views, v, pdflat = self._fused_forward(batch.states)
             self.flat_pi.zero_grad()
