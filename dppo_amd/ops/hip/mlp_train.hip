@@ -339,10 +339,20 @@ __global__ void mlp_chunk_kernel(ChunkArgs a) {
     const int dlast = (NH == 2) ? m.dh2 : m.dh1;
     for (int i = tid; i < st * H; i += CT) {
       const int s = i / H, k = i - s * H;
-      float acc = 0.f;
-      for (int u = 0; u < P + 1; ++u)
-        acc += lds[m.wht + k * m.wsh + u] * lds[m.gh + s * (P + 1) + u];
-      lds[dlast + i] = dact(lds[hlast + i], a.activation) * acc;
+      // 4 interleaved partial sums: the single-accumulator chain pays
+      // an LDS latency per iteration
+      float a0 = 0.f, a1 = 0.f, a2 = 0.f, a3 = 0.f;
+      int u = 0;
+      for (; u + 4 <= P + 1; u += 4) {
+        a0 += lds[m.wht + k * m.wsh + u] * lds[m.gh + s * (P + 1) + u];
+        a1 += lds[m.wht + k * m.wsh + u + 1] * lds[m.gh + s * (P + 1) + u + 1];
+        a2 += lds[m.wht + k * m.wsh + u + 2] * lds[m.gh + s * (P + 1) + u + 2];
+        a3 += lds[m.wht + k * m.wsh + u + 3] * lds[m.gh + s * (P + 1) + u + 3];
+      }
+      for (; u < P + 1; ++u)
+        a0 += lds[m.wht + k * m.wsh + u] * lds[m.gh + s * (P + 1) + u];
+      lds[dlast + i] =
+          dact(lds[hlast + i], a.activation) * ((a0 + a1) + (a2 + a3));
     }
     __syncthreads();
 
@@ -350,10 +360,15 @@ __global__ void mlp_chunk_kernel(ChunkArgs a) {
     if (NH == 2) {
       for (int i = tid; i < st * H; i += CT) {
         const int s = i / H, k = i - s * H;
-        float acc = 0.f;
-        for (int u = 0; u < H; ++u)
-          acc += lds[m.w2t + k * m.ws2 + u] * lds[m.dh2 + s * H + u];
-        lds[m.dh1 + i] = dact(lds[m.h1 + i], a.activation) * acc;
+        float a0 = 0.f, a1 = 0.f, a2 = 0.f, a3 = 0.f;
+        for (int u = 0; u < H; u += 4) {
+          a0 += lds[m.w2t + k * m.ws2 + u] * lds[m.dh2 + s * H + u];
+          a1 += lds[m.w2t + k * m.ws2 + u + 1] * lds[m.dh2 + s * H + u + 1];
+          a2 += lds[m.w2t + k * m.ws2 + u + 2] * lds[m.dh2 + s * H + u + 2];
+          a3 += lds[m.w2t + k * m.ws2 + u + 3] * lds[m.dh2 + s * H + u + 3];
+        }
+        lds[m.dh1 + i] =
+            dact(lds[m.h1 + i], a.activation) * ((a0 + a1) + (a2 + a3));
       }
       __syncthreads();
     }
@@ -361,10 +376,24 @@ __global__ void mlp_chunk_kernel(ChunkArgs a) {
     // P7: dW/bias accumulation (register slices; disjoint LDS bias slots)
     for (int s = 0; s < st; ++s) {
       const float d1 = lds[m.dh1 + s * H + uo];
+      // float4 x reads + per-group guards: the per-element guarded form
+      // tripled the instruction count of the dominant dW1 term (PMC:
+      // the kernel is issue-bound, profiles/r01_chunk_kernel_notes.md)
       #pragma unroll
-      for (int j = 0; j < KC1; ++j) {
+      for (int j = 0; j < KC1; j += 4) {
         const int k = co * KC1 + j;
-        if (k < D) acc1[j] += d1 * lds[m.x + s * m.dp + k];
+        if (k + 3 < D) {
+          const float4 x4 =
+              *reinterpret_cast<const float4*>(&lds[m.x + s * m.dp + k]);
+          acc1[j] += d1 * x4.x;
+          acc1[j + 1] += d1 * x4.y;
+          acc1[j + 2] += d1 * x4.z;
+          acc1[j + 3] += d1 * x4.w;
+        } else if (k < D) {
+          #pragma unroll
+          for (int t = 0; t < 4; ++t)
+            if (k + t < D) acc1[j + t] += d1 * lds[m.x + s * m.dp + k + t];
+        }
       }
       if (NH == 2) {
         const float d2 = lds[m.dh2 + s * H + uo];
