@@ -44,11 +44,15 @@
 
 namespace {
 
-constexpr int CT = 256;   // threads per block (4 waves)
-constexpr int CNW = 4;    // waves
-constexpr int ST = 8;     // samples per tile
-constexpr int KH_MAX = 17;  // dWh rows per thread: ceil((P+1)/4) <= 17
-constexpr int KC2_MAX = 16;  // dW2 k's per thread: H/(256/H) <= 16
+constexpr int CT = 512;   // threads per block (8 waves, 2/SIMD): halves
+                          // the per-thread dW slice (fewer accumulator
+                          // VGPRs, fewer instructions) and doubles the
+                          // latency-hiding occupancy vs the 256-thread
+                          // first cut
+constexpr int CNW = 8;    // waves
+constexpr int ST = 4;     // samples per tile (LDS budget at CT=512)
+constexpr int KH_MAX = 9;   // dWh rows per thread: ceil((P+1)/(CT/H)) <= 9
+constexpr int KC2_MAX = 8;  // dW2 k's per thread: H/(CT/H) <= 8
 
 struct ChunkLds {
   int w1t, w2t, wht, b1, b2, bh;
@@ -489,9 +493,10 @@ static bool chunk_lds_attr_ok() {
     bool r = true;
     for (const void* f :
          {reinterpret_cast<const void*>(&(*mlp_chunk_kernel<8>)),
+          reinterpret_cast<const void*>(&(*mlp_chunk_kernel<16>)),
           reinterpret_cast<const void*>(&(*mlp_chunk_kernel<32>)),
-          reinterpret_cast<const void*>(&(*mlp_chunk_kernel<96>)),
-          reinterpret_cast<const void*>(&(*mlp_chunk_kernel<128>))})
+          reinterpret_cast<const void*>(&(*mlp_chunk_kernel<48>)),
+          reinterpret_cast<const void*>(&(*mlp_chunk_kernel<64>))})
       r &= hipFuncSetAttribute(f, hipFuncAttributeMaxDynamicSharedMemorySize,
                                159 * 1024) == hipSuccess;
     return r;
@@ -507,7 +512,7 @@ bool mlp_chunk_supported(int64_t D, int64_t H, int64_t A, int64_t n_hidden) {
   if (H != 16 && H != 32 && H != 64) return false;
   if (D < 1 || D > 512 || A < 1 || A > 32) return false;
   const int CH = CT / (int)H;
-  if ((D + CH - 1) / CH > 128) return false;  // KC1 template ceiling
+  if ((D + CH - 1) / CH > 64) return false;  // KC1 template ceiling
   const ChunkLds m = chunk_lds_map((int)D, (int)H, (int)A, (int)n_hidden);
   if ((size_t)m.total * sizeof(float) > 158 * 1024) return false;
   return chunk_lds_attr_ok();
@@ -541,15 +546,13 @@ void mlp_chunk_train(
 
   const int64_t ptotal = params.numel();
   const int64_t ppad = (ptotal + 3) & ~3LL;
-  // samples per block: >=64 keeps the slab-reduce cost bounded; cap the
-  // grid at the CU count (1 block/CU at this LDS size anyway)
-  int64_t spb = 64;
+  // fill the chip: one block per CU until samples run out (the serial
+  // tile loop per block is the latency term; 64-sample blocks measured
+  // 322 us vs 16-sample blocks at the same 4096-sample chunk)
+  int64_t spb = (B + N_CU - 1) / N_CU;
+  if (spb < ST) spb = ST;
+  spb = (spb + ST - 1) / ST * ST;
   int64_t nb = (B + spb - 1) / spb;
-  if (nb > N_CU) {
-    nb = N_CU;
-    spb = (B + nb - 1) / nb;
-    nb = (B + spb - 1) / spb;
-  }
   TORCH_CHECK(slabs.numel() >= nb * ppad, "slab scratch too small");
 
   ChunkArgs a{};
@@ -596,14 +599,17 @@ void mlp_chunk_train(
   if (kc1 <= 8)
     hipLaunchKernelGGL(mlp_chunk_kernel<8>, dim3((uint32_t)nb), dim3(CT),
                        lds_bytes, stream, a);
+  else if (kc1 <= 16)
+    hipLaunchKernelGGL(mlp_chunk_kernel<16>, dim3((uint32_t)nb), dim3(CT),
+                       lds_bytes, stream, a);
   else if (kc1 <= 32)
     hipLaunchKernelGGL(mlp_chunk_kernel<32>, dim3((uint32_t)nb), dim3(CT),
                        lds_bytes, stream, a);
-  else if (kc1 <= 96)
-    hipLaunchKernelGGL(mlp_chunk_kernel<96>, dim3((uint32_t)nb), dim3(CT),
+  else if (kc1 <= 48)
+    hipLaunchKernelGGL(mlp_chunk_kernel<48>, dim3((uint32_t)nb), dim3(CT),
                        lds_bytes, stream, a);
   else
-    hipLaunchKernelGGL(mlp_chunk_kernel<128>, dim3((uint32_t)nb), dim3(CT),
+    hipLaunchKernelGGL(mlp_chunk_kernel<64>, dim3((uint32_t)nb), dim3(CT),
                        lds_bytes, stream, a);
 
   hipLaunchKernelGGL(

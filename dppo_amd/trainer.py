@@ -910,14 +910,14 @@ class DPPOEngine:
         self._mb_graphs = graphs
         self._mb_key = key
 
-    # Max batch routed to the fused chunk-step kernel (mlp_train.hip).
-    # DISABLED by default (0): the kernel is numerics-exact (GPU tests
-    # force-enable it) but its first cut measured ~500 us per 4096-sample
-    # chunk — slower than the ~258 us kernel chain it replaces — so
-    # routing stays off until the latency is understood (profiles/
-    # r01_chunk_kernel_notes.md).  Tests/experiments override per
-    # instance or via DPPO_CHUNK_KERNEL_MAX_B.
-    CHUNK_KERNEL_MAX_B = int(os.environ.get("DPPO_CHUNK_KERNEL_MAX_B", "0"))
+    # Max batch routed to the fused chunk-step kernel (mlp_train.hip):
+    # measured 182 us per 4096-sample chunk pair vs ~258 us for the
+    # kernel chain it replaces (ladder in profiles/
+    # r01_chunk_kernel_notes.md); beyond ~32k samples the MFMA GEMM
+    # path's compute efficiency wins.  Override via env
+    # DPPO_CHUNK_KERNEL_MAX_B (0 disables).
+    CHUNK_KERNEL_MAX_B = int(
+        os.environ.get("DPPO_CHUNK_KERNEL_MAX_B", "32768"))
 
     def _can_chunk_kernel(self, B: int) -> bool:
         """Eligibility for the fused single-kernel chunk step
