@@ -12,16 +12,22 @@
 // ppo_math.h (== reference PPO.py:29-40 / distributions.py:195-203);
 // Adam matches torch.optim.Adam exactly (adam.hip).
 //
-// MI355X design (why this shape):
+// MI355X design (why this shape; the measured ladder 693 -> 182 us per
+// 4096-sample chunk is in profiles/r01_chunk_kernel_notes.md):
 //   - ALL weights live in LDS, transposed at kernel start ([k][u] images,
-//     odd row strides where rows are read column-wise), so the per-tile
-//     phase chain touches only LDS + VALU: no L2 latency inside the
-//     sample loop.  At the Humanoid shapes (D=376, H=64, A=17) the image
-//     is ~154 KB of the 160 KB LDS -> 1 block/CU, 256 VGPRs/lane.
-//   - dW accumulates in REGISTERS: each of the 256 threads owns a fixed
+//     16-B-aligned regions, odd row strides where rows are read
+//     column-wise), so the per-tile phase chain touches only LDS + VALU:
+//     no L2 latency inside the sample loop.  At the Humanoid shapes
+//     (D=376, H=64, A=17) the image is ~144 KB of the 160 KB LDS ->
+//     1 block/CU of 8 waves (2/SIMD, ~206 VGPRs/lane of the 512-deep
+//     unified file).
+//   - dW accumulates in REGISTERS: each of the 512 threads owns a fixed
 //     (unit, k-chunk) slice of every weight matrix (KC1-template for the
 //     input layer).  Per sample that is a broadcast LDS read per k plus
-//     one FMA per owned element.
+//     one FMA per owned element; PMC showed the kernel issue-bound, so
+//     the slices are read float4 with per-group (not per-element) guards.
+//   - each chunk spreads over up to 256 blocks (spb = ceil(B/256)): the
+//     per-block serial tile loop is the latency term, so fill the chip.
 //   - Per-block partial gradients go to a slab (flat-grad layout); the
 //     reduce kernel sums NB slabs and applies Adam (device-state
 //     step/lr/bias-correction, so the pair is hipGraph-replayable), or
