@@ -9,7 +9,7 @@ pytestmark = pytest.mark.gpu
 
 from dppo_amd.config import DPPOConfig
 from dppo_amd.parallel.comm import Comm
-from dppo_amd.trainer import DPPOEngine
+from dppo_amd.trainer import DPPOEngine, RolloutBatch
 
 
 def _cfg(**kw):
@@ -159,3 +159,52 @@ def test_chunk_kernel_single_hidden_layer():
     torch.cuda.synchronize()
     assert all(math.isfinite(v) for v in stats.values())
     assert not torch.equal(p0, eng.flat_pi.flat_param.detach())
+
+
+def test_chunk_kernel_grad_mode_matches_autograd():
+    """The chunk kernel's grad-only mode (the distributed path: summed
+    gradient into flat_grad, Adam applied separately) matches autograd
+    gradients on the same batch."""
+    from dppo_amd.ops import hip_ext
+
+    torch.manual_seed(0)
+    eng = DPPOEngine(_cfg(USE_HIP_KERNELS="always", USE_GRAPHS=False, SEED=9),
+                     comm=Comm(device="cuda:0"))
+    eng.CHUNK_KERNEL_MAX_B = 1 << 30
+    batch, _ = eng.rollout_once()
+    n = 2048
+    assert eng._can_chunk_kernel(n)
+
+    # autograd reference on the same sub-batch
+    sub = RolloutBatch(
+        states=batch.states.narrow(0, 0, n),
+        actions=batch.actions.narrow(0, 0, n),
+        adv=batch.adv.narrow(0, 0, n), etr=batch.etr.narrow(0, 0, n),
+        oldflat=batch.oldflat.narrow(0, 0, n),
+        oldv=batch.oldv.narrow(0, 0, n), cur_lr=1.0,
+        ep_count=batch.ep_count, ep_sum=batch.ep_sum,
+        ep_sumsq=batch.ep_sumsq, ep_min=batch.ep_min, ep_max=batch.ep_max,
+        valid=True,
+    )
+    eng.flat_pi.zero_grad()
+    losses = eng._losses(sub, l_mul=1.0)
+    losses["total_loss"].backward()
+    ref_grad = eng.flat_pi.flat_grad.detach().clone()
+
+    # kernel, grad-only mode (fuse_adam=False)
+    c = eng.cfg
+    opt = eng.optimizer
+    eng.flat_pi.zero_grad()
+    hip_ext().mlp_chunk_train(
+        eng.flat_pi.flat_param.data,
+        sub.states, sub.actions, sub.adv, sub.etr, sub.oldflat, sub.oldv,
+        [sl.start for sl in eng.flat_pi.slices],
+        [eng.obs_space.shape[0], *c.HIDDEN_SIZES],
+        1 if c.ACTIVATION == "tanh" else 0,
+        torch.empty(0, device=eng.device), c.CLIP_PARAM, c.ENTCOEFF,
+        c.VCOEFF, eng._chunk_scratch(), opt.exp_avg, opt.exp_avg_sq,
+        opt.step_dev, opt.lr_dev, opt.coef, eng.flat_pi.flat_grad,
+        False, opt.betas[0], opt.betas[1], opt.eps)
+    torch.cuda.synchronize()
+    torch.testing.assert_close(
+        eng.flat_pi.flat_grad, ref_grad, atol=2e-5, rtol=1e-4)
