@@ -115,3 +115,62 @@ def test_flat_bucket_allreduce(tmp_path):
         with open(tmp_path / f"ar{r}.json") as f:
             d = json.load(f)
         assert d["ok"] and d["grad_ok"]
+
+
+def _invalid_rank_worker(rank, world, port, out_dir):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    os.environ["RANK"] = str(rank)
+    os.environ["WORLD_SIZE"] = str(world)
+    from dppo_amd.parallel.comm import Comm
+    from dppo_amd.trainer import DPPOEngine
+
+    cfg = DPPOConfig(
+        GAME="Pendulum-v1", NUM_ENVS=4, MAX_EPOCH_STEPS=12, EPOCH_MAX=8,
+        STOP_EPOCH=8, LEARNING_RATE=1e-3, NUM_WORKERS=world,
+        LOG_FILE_PATH=os.path.join(out_dir, "logs"), DEVICE="cpu",
+        BROADCAST_INTERVAL=0, MAX_ROLLOUT_RETRIES=2,
+    )
+    comm = Comm(backend="gloo", device="cpu")
+    eng = DPPOEngine(cfg, comm=comm)
+    if rank == 1:
+        # rank 1 never completes an episode -> every batch invalid; the
+        # validity flag must keep all collectives aligned (SURVEY.md §7
+        # "hard parts": a rank with no completed episode still
+        # participates or the ring deadlocks)
+        eng.env.horizons.fill_(10**9)
+        eng.env.horizons_i32.fill_(10**9)
+    stats_list = []
+    for _ in range(3):
+        stats, _ = eng.train_round()
+        stats_list.append(stats)
+    pf = eng.flat_pi.flat_param.clone()
+    gathered = comm.all_gather_rows(pf)
+    with open(os.path.join(out_dir, f"inv{rank}.pkl"), "wb") as f:
+        pickle.dump(
+            {
+                "final_identical": bool(torch.equal(gathered[0], gathered[1])),
+                "best_ranks": [s["best_rank"] for s in stats_list],
+                "finite": all(
+                    all(v == v for v in s.values()) for s in stats_list),
+            },
+            f,
+        )
+    comm.shutdown()
+
+
+@pytest.mark.timeout(300)
+def test_invalid_rank_keeps_ring_alive(tmp_path):
+    """A rank that never completes an episode must neither deadlock the
+    collectives nor win the best-rank sort."""
+    mp.spawn(_invalid_rank_worker, args=(WORLD, 29767, str(tmp_path)),
+             nprocs=WORLD, join=True)
+    results = []
+    for r in range(WORLD):
+        with open(tmp_path / f"inv{r}.pkl", "rb") as f:
+            results.append(pickle.load(f))
+    for r in results:
+        assert r["final_identical"], "replicas diverged"
+        assert r["finite"]
+        assert all(b == 0.0 for b in r["best_ranks"]), \
+            "the episode-less rank must never win the sort"

@@ -145,3 +145,24 @@ def test_minibatched_update():
     assert not torch.allclose(p0, eng.flat_pi.flat_param.detach())
     # 16*8=128 samples / 32 per chunk * 4 epochs = 16 optimizer steps
     assert eng.optimizer.state_dict()["state"][0]["step"] == 16
+
+
+def test_push_guard_retry_returns_invalid_batch():
+    """Worker.py:135 push-guard analog: when no episode can complete, the
+    engine retries MAX_ROLLOUT_RETRIES times and returns the (invalid)
+    batch instead of hanging; the stats row then reports valid=0 and a
+    sort key of -inf."""
+    import math as _math
+
+    cfg = small_cfg(MAX_ROLLOUT_RETRIES=2)
+    eng = DPPOEngine(cfg, comm=Comm(device="cpu"))
+    eng.env.horizons.fill_(10**9)
+    eng.env.horizons_i32.fill_(10**9)
+    batch = eng.collect()
+    assert not batch.valid
+    row = eng.stats_row(batch, {
+        "policyLoss": 0.0, "valueLoss": 0.0,
+        "entropyLoss": 0.0, "total_loss": 0.0,
+    })
+    assert row[10] == 0.0           # valid flag
+    assert row[2] == -_math.inf     # epr_max never wins the sort
