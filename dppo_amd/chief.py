@@ -21,7 +21,6 @@ from __future__ import annotations
 from typing import Any, Dict, List, Optional
 
 import numpy as np
-import torch
 
 from .config import DPPOConfig
 from .parallel.comm import Comm

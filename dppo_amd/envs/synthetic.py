@@ -22,7 +22,7 @@ capture-safe (no host control flow) for hipGraph capture.
 from __future__ import annotations
 
 import math
-from typing import Optional, Tuple
+
 
 import torch
 

@@ -707,7 +707,8 @@ class DPPOEngine:
 
     # ------------------------------------------------------------------
     def _can_fuse_update(self) -> bool:
-        """Eligibility for the fused HIP update path (mlp_train.hip)."""
+        """Eligibility for the fused MFMA update path
+        (mfma_gemm.hip forward/dgrad/dW + ppo_loss.hip gh)."""
         from .ops import use_hip
 
         c = self.cfg
