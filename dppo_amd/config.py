@@ -161,13 +161,18 @@ class DPPOConfig:
 # ---------------------------------------------------------------------------
 
 #: name -> (obs_dim, kind, act_dim) ; kind 'discrete' -> Discrete(act_dim),
-#: 'box' -> Box(act_dim)
-GAME_SHAPES: Dict[str, Tuple[int, str, int]] = {
+#: 'box' -> Box(act_dim), 'multidiscrete' -> MultiDiscrete(act_dim tuple),
+#: 'multibinary' -> MultiBinary(act_dim) — the reference trains whatever
+#: gym action space the env exposes (make_pdtype dispatch, reference
+#: Others/distributions.py:231-243), so every family is trainable here
+GAME_SHAPES: Dict[str, Tuple[int, str, object]] = {
     "CartPole-v0": (4, "discrete", 2),
     "Pendulum-v1": (3, "box", 1),
     "HalfCheetah-v4": (17, "box", 6),
     "Humanoid-v4": (376, "box", 17),
     "Wide-4096": (4096, "box", 256),
+    "MultiLever-v0": (12, "multidiscrete", (3, 3, 4)),
+    "BitFlipper-v0": (16, "multibinary", 8),
 }
 
 
@@ -181,6 +186,10 @@ def game_spaces(game: str):
     obs_space = spaces.Box(low=-float("inf"), high=float("inf"), shape=(obs_dim,))
     if kind == "discrete":
         act_space = spaces.Discrete(act_dim)
+    elif kind == "multidiscrete":
+        act_space = spaces.MultiDiscrete(list(act_dim))
+    elif kind == "multibinary":
+        act_space = spaces.MultiBinary(act_dim)
     else:
         act_space = spaces.Box(low=-1.0, high=1.0, shape=(act_dim,))
     return obs_space, act_space

@@ -73,14 +73,29 @@ class BatchedSyntheticEnv:
         self.U = randn(r, obs_dim) * (0.3 / math.sqrt(r))
         if isinstance(act_space, spaces.Discrete):
             self.B = randn(act_space.n, obs_dim) * 0.3
-            self._discrete = True
+            self._kind = "discrete"
         elif isinstance(act_space, spaces.Box):
             self.B = randn(act_space.shape[0], obs_dim) * (
                 0.3 / math.sqrt(act_space.shape[0])
             )
-            self._discrete = False
+            self._kind = "box"
+        elif isinstance(act_space, spaces.MultiDiscrete):
+            # one embedding row per (component, choice): a_in sums the
+            # selected rows, like the Discrete path per component
+            nvec = [int(n) for n in act_space.nvec]
+            self.B = randn(sum(nvec), obs_dim) * (0.3 / math.sqrt(len(nvec)))
+            off = torch.tensor(
+                [0] + list(torch.tensor(nvec).cumsum(0)[:-1]),
+                dtype=torch.long)
+            self._md_offsets = off.to(self.device)
+            self._kind = "multidiscrete"
+        elif isinstance(act_space, spaces.MultiBinary):
+            self.B = randn(act_space.n, obs_dim) * (
+                0.3 / math.sqrt(act_space.n))
+            self._kind = "multibinary"
         else:
             raise NotImplementedError(f"synthetic env for {act_space!r}")
+        self._discrete = self._kind == "discrete"
 
         # Staggered fixed horizons in [horizon//2, 3*horizon//2) so resets
         # spread across steps and every rollout sees some done flags.
@@ -120,9 +135,13 @@ class BatchedSyntheticEnv:
         return self.x
 
     def step(self, actions: torch.Tensor):
-        if self._discrete:
+        if self._kind == "discrete":
             a_in = self.B.index_select(0, actions.long().reshape(-1))
-        else:
+        elif self._kind == "multidiscrete":
+            idx = (actions.long() + self._md_offsets).reshape(-1)  # [E*K]
+            a_in = self.B.index_select(0, idx).view(
+                self.num_envs, -1, self.obs_dim).sum(dim=1)
+        else:  # box / multibinary: linear action input
             a_in = actions.to(self.dtype) @ self.B
         noise = self.NOISE * torch.randn(
             self.num_envs, self.obs_dim,

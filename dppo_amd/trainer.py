@@ -99,7 +99,17 @@ class DPPOEngine:
 
         obs_space, act_space = game_spaces(cfg.GAME)
         self.obs_space, self.act_space = obs_space, act_space
-        self._discrete = isinstance(act_space, spaces.Discrete)
+        # action-space family (make_pdtype dispatch, reference
+        # distributions.py:231-243): all four are trainable end-to-end
+        if isinstance(act_space, spaces.Discrete):
+            self._act_kind = "discrete"
+        elif isinstance(act_space, spaces.MultiDiscrete):
+            self._act_kind = "multidiscrete"
+        elif isinstance(act_space, spaces.MultiBinary):
+            self._act_kind = "multibinary"
+        else:
+            self._act_kind = "box"
+        self._discrete = self._act_kind == "discrete"
 
         seed = cfg.SEED + 1000 * (self.comm.rank + seed_offset)
         torch.manual_seed(seed)
@@ -179,10 +189,19 @@ class DPPOEngine:
     def _random_actions(self, n: int) -> torch.Tensor:
         """Uniform random actions for the epsilon-greedy overlay
         (Worker.py:149-152; extended to Box spaces: uniform in [low, high])."""
-        if self._discrete:
+        if self._act_kind == "discrete":
             return torch.randint(
                 self.act_space.n, (n,), device=self.device, dtype=torch.long
             )
+        if self._act_kind == "multidiscrete":
+            return torch.stack(
+                [torch.randint(int(k), (n,), device=self.device,
+                               dtype=torch.long)
+                 for k in self.act_space.nvec], dim=-1)
+        if self._act_kind == "multibinary":
+            return torch.randint(
+                2, (n, self.act_space.n), device=self.device
+            ).to(self.dtype)
         low = float(self.act_space.low.flat[0])
         high = float(self.act_space.high.flat[0])
         a_dim = self.act_space.shape[0]
@@ -226,7 +245,7 @@ class DPPOEngine:
         from .ops import use_hip
 
         c = self.cfg
-        if self._discrete or c.DTYPE != "float32":
+        if self._act_kind != "box" or c.DTYPE != "float32":
             return False
         if not use_hip(self.device, c.USE_HIP_KERNELS):
             return False
@@ -565,9 +584,12 @@ class DPPOEngine:
         dev = self.device
         states = torch.empty(T, E, obs_dim, device=dev, dtype=self.dtype)
         pdflats = torch.empty(T, E, P, device=dev, dtype=self.dtype)
-        if self._discrete:
+        if self._act_kind == "discrete":
             actions = torch.empty(T, E, device=dev, dtype=torch.long)
-        else:
+        elif self._act_kind == "multidiscrete":
+            actions = torch.empty(
+                T, E, len(self.act_space.nvec), device=dev, dtype=torch.long)
+        else:  # box / multibinary: float vectors
             actions = torch.empty(
                 T, E, self.act_space.shape[0], device=dev, dtype=self.dtype
             )
@@ -712,7 +734,7 @@ class DPPOEngine:
         from .ops import use_hip
 
         c = self.cfg
-        if self._discrete or c.DTYPE != "float32":
+        if self._act_kind != "box" or c.DTYPE != "float32":
             return False
         if not use_hip(self.device, c.USE_HIP_KERNELS):
             return False
@@ -928,7 +950,7 @@ class DPPOEngine:
         from .ops.adam import FusedFlatAdam
 
         c = self.cfg
-        if self._discrete or c.DTYPE != "float32":
+        if self._act_kind != "box" or c.DTYPE != "float32":
             return False
         if not use_hip(self.device, c.USE_HIP_KERNELS):
             return False

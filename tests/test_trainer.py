@@ -166,3 +166,34 @@ def test_push_guard_retry_returns_invalid_batch():
     })
     assert row[10] == 0.0           # valid flag
     assert row[2] == -_math.inf     # epr_max never wins the sort
+
+
+@pytest.mark.parametrize("game,hidden", [
+    ("MultiLever-v0", (32,)), ("BitFlipper-v0", (32,))])
+def test_multidiscrete_multibinary_training(game, hidden):
+    """Every reference action-space family trains end-to-end
+    (make_pdtype dispatch, reference Others/distributions.py:231-243) —
+    not just Box/Discrete: MultiDiscrete and MultiBinary get synthetic
+    envs and run the full round protocol."""
+    cfg = DPPOConfig(
+        GAME=game, HIDDEN_SIZES=hidden, ACTIVATION="tanh", NUM_ENVS=8,
+        MAX_EPOCH_STEPS=16, EPOCH_MAX=100, STOP_EPOCH=100, NUM_WORKERS=1,
+        LOG_FILE_PATH="/tmp/dppo_test_logs", DEVICE="cpu", SEED=3,
+    )
+    eng = DPPOEngine(cfg, comm=Comm(device="cpu"))
+    p0 = eng.flat_pi.flat_param.detach().clone()
+    for _ in range(2):
+        stats, stop = eng.train_round()
+    assert all(v == v for v in stats.values())
+    assert not torch.equal(p0, eng.flat_pi.flat_param.detach())
+    # recorded actions respect the space
+    batch, _ = eng.rollout_once()
+    if game == "MultiLever-v0":
+        nvec = eng.act_space.nvec
+        a = batch.actions.view(-1, len(nvec))
+        assert a.dtype == torch.long
+        for k, n in enumerate(nvec):
+            assert int(a[:, k].min()) >= 0 and int(a[:, k].max()) < int(n)
+    else:
+        a = batch.actions
+        assert set(a.unique().tolist()) <= {0.0, 1.0}
