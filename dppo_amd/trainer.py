@@ -936,11 +936,14 @@ class DPPOEngine:
     # Max batch routed to the fused chunk-step kernel (mlp_train.hip):
     # measured 182 us per 4096-sample chunk pair vs ~258 us for the
     # kernel chain it replaces (ladder in profiles/
-    # r01_chunk_kernel_notes.md); beyond ~32k samples the MFMA GEMM
-    # path's compute efficiency wins.  Override via env
-    # DPPO_CHUNK_KERNEL_MAX_B (0 disables).
+    # r01_chunk_kernel_notes.md).  8192 is the largest MEASURED-faster
+    # batch regime (4096-sample config-4 chunks, 6400-sample config-2
+    # full batches); above it the chunk pair's serial tile loop grows
+    # linearly while the MFMA GEMM path's latency is ~flat, so the
+    # crossover sweep is queued (ROADMAP.md) before raising it.
+    # Override via env DPPO_CHUNK_KERNEL_MAX_B (0 disables).
     CHUNK_KERNEL_MAX_B = int(
-        os.environ.get("DPPO_CHUNK_KERNEL_MAX_B", "32768"))
+        os.environ.get("DPPO_CHUNK_KERNEL_MAX_B", "8192"))
 
     def _can_chunk_kernel(self, B: int) -> bool:
         """Eligibility for the fused single-kernel chunk step
