@@ -197,3 +197,19 @@ def test_multidiscrete_multibinary_training(game, hidden):
     else:
         a = batch.actions
         assert set(a.unique().tolist()) <= {0.0, 1.0}
+
+
+def test_eval_losses_equal_recompute():
+    """eval_losses uses the RECORDED rollout outputs instead of re-running
+    pi (trainer._losses recorded_pi=True): the reference evaluates its
+    losses while pi still equals oldpi (Worker.py:117-118 after the
+    sync at Worker.py:42), so the recorded pdflat/v must equal a true
+    forward pass bit-for-bit on the eager path."""
+    eng = DPPOEngine(small_cfg(), comm=Comm(device="cpu"))
+    batch = eng.collect()
+    fast = eng.eval_losses(batch, l_mul=1.0)
+    with torch.no_grad():
+        slow = {k: float(v)
+                for k, v in eng._losses(batch, 1.0, recorded_pi=False).items()}
+    for k in fast:
+        assert abs(fast[k] - slow[k]) < 1e-5, (k, fast[k], slow[k])
