@@ -40,6 +40,7 @@ __all__ = [
     "PolicyValueMLP",
     "Chief",
     "Worker",
+    "DPPOEngine",
     "__version__",
 ]
 
@@ -56,4 +57,7 @@ def __getattr__(name):
     if name == "PPO":
         from .ppo import PPO
         return PPO
+    if name == "DPPOEngine":
+        from .trainer import DPPOEngine
+        return DPPOEngine
     raise AttributeError(f"module 'dppo_amd' has no attribute {name!r}")
