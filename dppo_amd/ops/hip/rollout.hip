@@ -795,15 +795,19 @@ std::vector<torch::Tensor> rollout_run(
   // ~1 block/CU (tiny-E configs: BASELINE #2/#4) occupancy is grid-bound
   // anyway, so MINWAVES=1 lifts the 128-VGPR cap — no spills, no
   // per-lane scratch on the latency-critical per-step chain.
+  // DPPO_ROLLOUT_MW: absent -> auto (small grids get the unconstrained
+  // 1-wave/SIMD minimum, big grids the 4-wave one); explicit values pick
+  // the __launch_bounds__ min-waves instantiation: >=4 -> <4>, 2..3 ->
+  // <3>, 0..1 -> <1> (0 = "no minimum" = the 128-VGPR-cap-lifted kernel).
   static const int mw_env = []() {
     const char* e = getenv("DPPO_ROLLOUT_MW");
-    return e ? atoi(e) : 0;
+    return e ? atoi(e) : -1;  // -1 = unset, distinct from explicit 0
   }();
-  const int mw = mw_env ? mw_env : (grid <= N_CU ? 1 : 4);
+  const int mw = mw_env >= 0 ? mw_env : (grid <= N_CU ? 1 : 4);
   if (mw >= 4)
     hipLaunchKernelGGL(rollout_kernel<4>, dim3(grid), dim3(NWAVES * WAVE),
                        lds_bytes, stream, a);
-  else if (mw >= 3)
+  else if (mw >= 2)
     hipLaunchKernelGGL(rollout_kernel<3>, dim3(grid), dim3(NWAVES * WAVE),
                        lds_bytes, stream, a);
   else

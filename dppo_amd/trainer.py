@@ -155,13 +155,29 @@ class DPPOEngine:
         self.epr = torch.zeros(cfg.NUM_ENVS, device=self.device, dtype=torch.float32)
 
         self.CUR_EP = 0
-        self.timers = PhaseTimers(cuda_sync=False)
+        # DPPO_TIMER_SYNC=1 makes per-phase wall times attribute async GPU
+        # work correctly (adds a device sync per phase boundary; off for
+        # production — rocprof profiles are the ground truth either way).
+        self.timers = PhaseTimers(
+            cuda_sync=os.environ.get("DPPO_TIMER_SYNC", "0") == "1")
         self.logger = ScalarLogger(
             cfg.LOG_FILE_PATH, enabled=(self.comm.rank == 0), name=f"{scope}_rank0"
         )
         self._round = 0
 
     # ------------------------------------------------------------------
+    def _warn_once(self, key: str, msg: str) -> None:
+        """Rank-0 one-shot warning (graph-capture fallbacks etc. must not
+        silently de-optimize production runs — VERDICT r01 weak #5)."""
+        seen = getattr(self, "_warned", None)
+        if seen is None:
+            seen = self._warned = set()
+        if key in seen:
+            return
+        seen.add(key)
+        if self.comm.rank == 0:
+            print(f"[dppo_amd] WARNING: {msg}", flush=True)
+
     @torch.no_grad()
     def sync_oldpi(self) -> None:
         """oldpi <- pi, one flat copy (sync_pis, PPO.py:47)."""
@@ -559,7 +575,12 @@ class DPPOEngine:
                 self._v3_graph = g
                 g.replay()
                 moments = self._v3_moments
-            except Exception:  # noqa: BLE001 — capture support varies
+            except Exception as exc:  # noqa: BLE001 — capture support varies
+                self._warn_once(
+                    "v3_graph",
+                    f"rollout hipGraph capture failed ({exc!r}); "
+                    "falling back to eager per-step launches",
+                )
                 self._v3_graph_failed = True
                 self._v3_graph = None
                 moments = self._v3_body(*args)
@@ -794,16 +815,14 @@ class DPPOEngine:
                 and (not self.comm.distributed
                      or os.environ.get("DPPO_GRAPH_DIST") == "1")
                 and not getattr(self, "_mb_graph_failed", False)):
-            try:
-                self._update_minibatched_graphed(batch, l_mul)
+            # Only CAPTURE failures fall through to the uncaptured loop
+            # (capture is state-safe: warmup is snapshot/restored and a
+            # capture records, not executes).  A failure during replay or
+            # the tail chunk has already mutated params/Adam state, so it
+            # propagates instead of re-running the epochs (which would
+            # double-step the round — ADVICE r01 #1).
+            if self._update_minibatched_graphed(batch, l_mul):
                 return
-            except Exception:  # noqa: BLE001 — capture support varies
-                if os.environ.get("DPPO_DEBUG"):
-                    import traceback
-
-                    traceback.print_exc()
-                self._mb_graph_failed = True
-                self._mb_graphs = None
         use_ck = fuse and self._can_chunk_kernel(min(mb, B))
         if use_ck:
             self.optimizer.lr_dev.fill_(self.cfg.LEARNING_RATE * l_mul)
@@ -844,7 +863,7 @@ class DPPOEngine:
     _MB_SEG = 1024
 
     def _update_minibatched_graphed(self, batch: RolloutBatch,
-                                    l_mul: float) -> None:
+                                    l_mul: float) -> bool:
         """hipGraph-captured minibatch epochs (BASELINE config 4): the
         sequential chunk walk is identical every epoch and every round —
         the rollout/GAE outputs live in persistent buffers, so chunk
@@ -872,12 +891,27 @@ class DPPOEngine:
                batch.adv.data_ptr(), batch.etr.data_ptr(),
                batch.oldflat.data_ptr(), batch.oldv.data_ptr(), B, mb)
         if getattr(self, "_mb_graphs", None) is None or self._mb_key != key:
-            self._mb_capture(batch, key, n_full)
+            try:
+                self._mb_capture(batch, key, n_full)
+            except Exception as exc:  # noqa: BLE001 — capture support varies
+                if os.environ.get("DPPO_DEBUG"):
+                    import traceback
+
+                    traceback.print_exc()
+                self._warn_once(
+                    "mb_graph",
+                    f"minibatch hipGraph capture failed ({exc!r}); "
+                    "falling back to the uncaptured chunk loop",
+                )
+                self._mb_graph_failed = True
+                self._mb_graphs = None
+                return False
         for _ in range(c.UPDATE_STEPS):
             for g in self._mb_graphs:
                 g.replay()
             if tail:
                 self._mb_chunk_step(batch, n_full * mb, tail)
+        return True
 
     def _mb_chunk_step(self, batch: RolloutBatch, o: int, n: int) -> None:
         """One fused minibatch gradient step on batch[o:o+n] (capture-safe:
@@ -1050,7 +1084,12 @@ class DPPOEngine:
                 with torch.cuda.graph(g):
                     self._update_body(batch)
                 self._upd_graph = g
-            except Exception:  # noqa: BLE001 — capture support varies
+            except Exception as exc:  # noqa: BLE001 — capture support varies
+                self._warn_once(
+                    "upd_graph",
+                    f"update hipGraph capture failed ({exc!r}); "
+                    "falling back to uncaptured fused update",
+                )
                 self._graph_failed = True
                 self._upd_graph = None
                 self._update_fused(batch, l_mul)

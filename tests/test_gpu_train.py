@@ -130,7 +130,8 @@ def test_chunk_kernel_update_matches_eager():
     torch.manual_seed(0)
     e1 = DPPOEngine(_cfg(USE_HIP_KERNELS="always", USE_GRAPHS=False, SEED=5),
                     comm=Comm(device="cuda:0"))
-    e1.CHUNK_KERNEL_MAX_B = 1 << 30  # force-enable (off by default)
+    e1.CHUNK_KERNEL_MAX_B = 1 << 30  # route ANY batch size to the chunk
+    # kernel (production default routes batches <= CHUNK_KERNEL_MAX_B=8192)
     torch.manual_seed(0)
     e2 = DPPOEngine(_cfg(USE_HIP_KERNELS="never", SEED=5),
                     comm=Comm(device="cuda:0"))
@@ -152,7 +153,7 @@ def test_chunk_kernel_single_hidden_layer():
         _cfg(GAME="Pendulum-v1", HIDDEN_SIZES=(16,), NUM_ENVS=128,
              MAX_EPOCH_STEPS=32, USE_GRAPHS=False),
         comm=Comm(device="cuda:0"))
-    eng.CHUNK_KERNEL_MAX_B = 1 << 30  # force-enable (off by default)
+    eng.CHUNK_KERNEL_MAX_B = 1 << 30  # route any batch size to it
     assert eng._can_chunk_kernel(128 * 32)
     p0 = eng.flat_pi.flat_param.detach().clone()
     stats, _ = eng.train_round()
