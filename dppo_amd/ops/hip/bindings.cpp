@@ -28,6 +28,20 @@ std::vector<torch::Tensor> ppo_loss_gauss_bwd(
     torch::Tensor etr, double clip, double entcoeff, double vcoeff,
     torch::Tensor gtotal);
 
+torch::Tensor ppo_loss_cat_fwd(torch::Tensor lpi, torch::Tensor lold,
+                               torch::Tensor vpred, torch::Tensor oldv,
+                               torch::Tensor act, torch::Tensor adv,
+                               torch::Tensor etr, double clip,
+                               double entcoeff, double vcoeff);
+
+std::vector<torch::Tensor> ppo_loss_cat_bwd(
+    torch::Tensor lpi, torch::Tensor lold, torch::Tensor vpred,
+    torch::Tensor oldv, torch::Tensor act, torch::Tensor adv,
+    torch::Tensor etr, double clip, double entcoeff, double vcoeff,
+    torch::Tensor gtotal);
+
+torch::Tensor cat_sample(torch::Tensor logits, int64_t seed, int64_t ctr);
+
 void adam_step(torch::Tensor p, torch::Tensor g, torch::Tensor m,
                torch::Tensor v, int64_t step, double lr, double beta1,
                double beta2, double eps);
@@ -88,6 +102,12 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
           "fused DiagGaussian PPO loss forward (gfx950)");
   mod.def("ppo_loss_gauss_bwd", &ppo_loss_gauss_bwd,
           "fused DiagGaussian PPO loss backward (gfx950)");
+  mod.def("ppo_loss_cat_fwd", &ppo_loss_cat_fwd,
+          "fused Categorical PPO loss forward (gfx950)");
+  mod.def("ppo_loss_cat_bwd", &ppo_loss_cat_bwd,
+          "fused Categorical PPO loss backward (gfx950)");
+  mod.def("cat_sample", &cat_sample,
+          "Gumbel-max categorical sampling, counter-based RNG (gfx950)");
   mod.def("adam_step", &adam_step, "fused flat Adam step (gfx950)");
   mod.def("adam_step_dev", &adam_step_dev,
           "graph-replayable fused Adam (device step/lr) (gfx950)");

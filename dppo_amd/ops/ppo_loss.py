@@ -14,11 +14,13 @@
 where eps = CLIP_PARAM * l_mul: the l_mul anneal multiplier scales BOTH
 the Adam learning rate and the clip range (reference PPO.py:19-20).
 
-The eager path below is the autograd reference; the fused HIP kernel
-(ops/hip/ppo_loss.hip) computes the whole thing — DiagGaussian/Categorical
-logp for pi and oldpi, ratio, both clips, entropy, and the three
-block-reduced means — in one forward kernel and one analytic backward
-kernel, and is tested against this path to tolerance.
+The eager path below is the autograd reference; the fused HIP kernels
+(ops/hip/ppo_loss.hip for DiagGaussian, ops/hip/cat_loss.hip for
+Categorical — the reference's default family, distributions.py:124-159)
+each compute the whole thing — logp for pi and oldpi, ratio, both clips,
+entropy, and the three block-reduced means — in one forward kernel and
+one analytic backward kernel, and are tested against this path to
+tolerance.
 """
 
 from __future__ import annotations
@@ -113,6 +115,54 @@ class _FusedPPOLossGaussian(torch.autograd.Function):
         return (g_pdflat, None, g_v) + (None,) * 7
 
 
+class _FusedPPOLossCategorical(torch.autograd.Function):
+    """Fused Categorical PPO loss: HIP forward + analytic HIP backward
+    (ops/hip/cat_loss.hip — logsumexp CE logp, shifted-logit entropy per
+    reference distributions.py:131-153).  Gradients flow to the pi logits
+    and vpred only, like the Gaussian variant."""
+
+    @staticmethod
+    def forward(ctx, logits_pi, logits_old, vpred, oldvpred, actions, adv,
+                etr, clip_param, entcoeff, vcoeff):
+        from . import require_hip_ext
+
+        ext = require_hip_ext()
+        out = ext.ppo_loss_cat_fwd(
+            logits_pi.contiguous(), logits_old.contiguous(),
+            vpred.contiguous(), oldvpred.contiguous(),
+            actions.contiguous(), adv.contiguous(), etr.contiguous(),
+            float(clip_param), float(entcoeff), float(vcoeff),
+        )
+        ctx.save_for_backward(logits_pi, logits_old, vpred, oldvpred,
+                              actions, adv, etr)
+        ctx.coeffs = (float(clip_param), float(entcoeff), float(vcoeff))
+        return out
+
+    @staticmethod
+    def backward(ctx, grad_losses):
+        from . import require_hip_ext
+
+        ext = require_hip_ext()
+        (logits_pi, logits_old, vpred, oldvpred, actions, adv,
+         etr) = ctx.saved_tensors
+        clip_param, entcoeff, vcoeff = ctx.coeffs
+        g_total = grad_losses[3]
+        g_logits, g_v = ext.ppo_loss_cat_bwd(
+            logits_pi, logits_old, vpred, oldvpred, actions, adv, etr,
+            clip_param, entcoeff, vcoeff, g_total,
+        )
+        return (g_logits, None, g_v) + (None,) * 7
+
+
+def _loss_dict(losses: torch.Tensor) -> Dict[str, torch.Tensor]:
+    return {
+        "policyLoss": losses[0],
+        "entropyLoss": losses[1],
+        "valueLoss": losses[2],
+        "total_loss": losses[3],
+    }
+
+
 def ppo_losses(
     pd: Pd,
     oldpd: Pd,
@@ -124,7 +174,8 @@ def ppo_losses(
     coeffs: PPOLossCoeffs,
     policy: str = "auto",
 ) -> Dict[str, torch.Tensor]:
-    """PPO losses with fused-HIP dispatch for DiagGaussian policies."""
+    """PPO losses with fused-HIP dispatch for DiagGaussian and Categorical
+    policies (the other families run the eager reference)."""
     from . import use_hip
 
     if (
@@ -132,15 +183,20 @@ def ppo_losses(
         and isinstance(oldpd, DiagGaussianPd)
         and use_hip(vpred, policy)
     ):
-        losses = _FusedPPOLossGaussian.apply(
+        return _loss_dict(_FusedPPOLossGaussian.apply(
             pd.flatparam(), oldpd.flatparam().detach(),
             vpred, oldvpred.detach(), actions, adv, etr,
             coeffs.clip_param, coeffs.entcoeff, coeffs.vcoeff,
-        )
-        return {
-            "policyLoss": losses[0],
-            "entropyLoss": losses[1],
-            "valueLoss": losses[2],
-            "total_loss": losses[3],
-        }
+        ))
+    if (
+        isinstance(pd, CategoricalPd)
+        and isinstance(oldpd, CategoricalPd)
+        and pd.flatparam().shape[-1] <= 64  # one wave covers the columns
+        and use_hip(vpred, policy)
+    ):
+        return _loss_dict(_FusedPPOLossCategorical.apply(
+            pd.flatparam(), oldpd.flatparam().detach(),
+            vpred, oldvpred.detach(), actions.long(), adv, etr,
+            coeffs.clip_param, coeffs.entcoeff, coeffs.vcoeff,
+        ))
     return ppo_losses_ref(pd, oldpd, vpred, oldvpred, actions, adv, etr, coeffs)

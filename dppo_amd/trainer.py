@@ -240,11 +240,24 @@ class DPPOEngine:
         NOT recorded — the ratio re-evaluates oldpi on the batch later,
         exactly the reference's formulation (Worker.py:149-152 + PPO.py:31;
         SURVEY.md §7 'hard parts')."""
+        from .ops import use_hip
+
         with self._amp():
             v, pdflat = self.pi(obs)
         v, pdflat = v.float(), pdflat.float()
         pd = self.pi.pdtype.pdfromflat(pdflat)
-        a = pd.sample()
+        if (self._discrete and pdflat.shape[-1] <= 64
+                and use_hip(self.device, self.cfg.USE_HIP_KERNELS)):
+            # fused Gumbel-max sample kernel (cat_loss.hip) — counter-based
+            # RNG, one launch, no torch RNG round trips
+            from .ops import hip_ext
+
+            self._cat_ctr = getattr(self, "_cat_ctr", 0) + 1
+            seed = (self.cfg.SEED * 1_000_003
+                    + self.comm.rank * 7_919) & 0x7FFFFFFF
+            a = hip_ext().cat_sample(pdflat, seed, self._cat_ctr)
+        else:
+            a = pd.sample()
         if eps > 0.0:
             E = obs.shape[0]
             explore = torch.rand(E, device=self.device) < eps
