@@ -106,6 +106,13 @@ class DPPOConfig:
                                       # scheme, Chief.py:64); >0 = sequential
                                       # minibatch chunks per update step
                                       # (BASELINE config 4's 4096-minibatch)
+    BATCH_CURATION: bool = False      # reference Chief.py:33-53 parity: sort
+                                      # every rank's batch by best episode
+                                      # reward (logs[2]) and assign the top
+                                      # batches one-per-rank, so a rank can
+                                      # train on a BETTER rank's data; off =
+                                      # rank-local batches (documented
+                                      # deviation, PARITY.md §7)
 
     def __post_init__(self) -> None:
         if isinstance(self.HIDDEN_SIZES, list):

@@ -145,6 +145,15 @@ class Comm:
         dist.all_gather(outs, row)
         return torch.stack(outs, dim=0)
 
+    def send(self, t: torch.Tensor, dst: int) -> None:
+        """Blocking point-to-point send (batch-curation transport)."""
+        dist.send(t.contiguous(), dst=dst)
+
+    def recv(self, t: torch.Tensor, src: int) -> torch.Tensor:
+        """Blocking point-to-point receive into `t` (must be contiguous)."""
+        dist.recv(t, src=src)
+        return t
+
     def barrier(self) -> None:
         if self.distributed:
             dist.barrier()

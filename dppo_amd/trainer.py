@@ -1250,6 +1250,56 @@ class DPPOEngine:
             self.comm.allreduce_mean_(self.flat_pi.flat_grad)
             self.optimizer.step()
     # ------------------------------------------------------------------
+    def _curate_batches(self, batch: RolloutBatch,
+                        gathered: torch.Tensor) -> RolloutBatch:
+        """Reference batch-curation parity (Chief.py:33-53): the Chief
+        drains every worker's queue, sorts all pushed batches descending
+        by best episode reward (logs[2], the Chief.py:51 sort key) and
+        feeds the top-N one per tower — so a tower can train on a BETTER
+        worker's batch.  Here each rank holds one batch per round; the
+        sorted assignment is realized with point-to-point moves: rank i
+        trains on the batch collected by rank order[i].  When fewer than
+        world_size batches are valid the best batches are recycled
+        (order[i % V]) so every rank still trains — the reference's full
+        barrier guarantees N valid batches and never hits this case.
+
+        Every rank executes the identical schedule (gathered stats are
+        identical across ranks), so sends and recvs pair deterministically.
+        """
+        import dataclasses as _dc
+
+        W = self.comm.world_size
+        r = self.comm.rank
+        valid = gathered[:, 10] > 0.5
+        key = gathered[:, 2].clone()
+        key[~valid] = -math.inf
+        order = torch.argsort(key, descending=True, stable=True).tolist()
+        V = int(valid.sum())
+        if V == 0:
+            return batch
+        srcs = [order[i] if i < V else order[i % V] for i in range(W)]
+        if all(s == i for i, s in enumerate(srcs)):
+            return batch
+        fields = ["states", "actions", "adv", "etr", "oldflat", "oldv"]
+        orig = {f: getattr(batch, f) for f in fields}
+        for f in fields:
+            assert orig[f].is_contiguous(), f"curation needs contiguous {f}"
+        # A rank that SENDS ships a snapshot of its ORIGINAL batch
+        # (assignments are of the pre-curation batches), so a receive may
+        # land IN-PLACE in the persistent rollout buffers — keeping the
+        # hipGraph-captured update paths (keyed on stable data_ptrs) valid.
+        sends_any = any(s == r and d != r for d, s in enumerate(srcs))
+        send_copy = {f: orig[f].clone() for f in fields} if sends_any else None
+        for dst, s in enumerate(srcs):
+            if s == dst:
+                continue
+            for f in fields:
+                if r == s:
+                    self.comm.send(send_copy[f], dst)
+                elif r == dst:
+                    self.comm.recv(orig[f], s)
+        return _dc.replace(batch, valid=True)
+
     def train_round(self) -> Tuple[Dict[str, float], bool]:
         """One full synchronous round. Returns (rank0-view stats, stop)."""
         c = self.cfg
@@ -1277,6 +1327,9 @@ class DPPOEngine:
             best_cur_ep = float(self.CUR_EP)
 
         if any_valid:
+            if c.BATCH_CURATION and self.comm.distributed:
+                with self.timers.phase("curation"):
+                    batch = self._curate_batches(batch, gathered)
             with self.timers.phase("update"):
                 self.update(batch, l_mul)
 

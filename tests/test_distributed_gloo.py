@@ -241,6 +241,107 @@ def _drift_guard_worker(rank, world, port, out_dir):
     comm.shutdown()
 
 
+def _curation_worker(rank, world, port, out_dir):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    os.environ["RANK"] = str(rank)
+    os.environ["WORLD_SIZE"] = str(world)
+    from dppo_amd.parallel.comm import Comm
+    from dppo_amd.trainer import DPPOEngine
+
+    cfg = DPPOConfig(
+        GAME="Pendulum-v1", NUM_ENVS=4, MAX_EPOCH_STEPS=12, EPOCH_MAX=8,
+        STOP_EPOCH=8, LEARNING_RATE=1e-3, NUM_WORKERS=world,
+        LOG_FILE_PATH=os.path.join(out_dir, "logs"), DEVICE="cpu",
+        BROADCAST_INTERVAL=0, BATCH_CURATION=True,
+    )
+    comm = Comm(backend="gloo", device="cpu")
+    eng = DPPOEngine(cfg, comm=comm)
+    batch = eng.collect()
+    states_all = comm.all_gather_rows(batch.states.reshape(-1).clone())
+
+    results = {}
+    # Scenario A (scripted scores, the reference's sort on Chief.py:51):
+    # rank 1's batch has the higher max episode reward -> tower 0 gets
+    # rank 1's batch, tower 1 gets rank 0's (the top-N one-per-tower
+    # assignment of Chief.py:54-63 at N = world = 2).
+    g = torch.zeros(world, 11)
+    g[:, 10] = 1.0          # both valid
+    g[0, 2], g[1, 2] = 1.0, 5.0
+    eng._curate_batches(batch, g)
+    results["swap_ok"] = bool(torch.equal(
+        batch.states.reshape(-1), states_all[1 - rank]))
+
+    # Scenario B: only rank 0's (post-swap) batch is valid -> both ranks
+    # train on rank 0's batch (best-batch recycle for the invalid rank).
+    states_all2 = comm.all_gather_rows(batch.states.reshape(-1).clone())
+    g2 = torch.zeros(world, 11)
+    g2[0, 10] = 1.0
+    g2[0, 2], g2[1, 2] = 2.0, 99.0  # invalid rank never wins the sort
+    eng._curate_batches(batch, g2)
+    results["recycle_ok"] = bool(torch.equal(
+        batch.states.reshape(-1), states_all2[0]))
+
+    # Scenario C: end-to-end rounds with curation on — replicas must stay
+    # bit-identical (collective schedule aligned on every rank).
+    for _ in range(2):
+        eng.train_round()
+    pf = eng.flat_pi.flat_param.clone()
+    gathered = comm.all_gather_rows(pf)
+    results["identical"] = bool(torch.equal(gathered[0], gathered[1]))
+    with open(os.path.join(out_dir, f"cur{rank}.json"), "w") as f:
+        json.dump(results, f)
+    comm.shutdown()
+
+
+@pytest.mark.timeout(300)
+def test_batch_curation_matches_reference_decisions(tmp_path):
+    """BATCH_CURATION=True reproduces the reference Chief's drain-sort-
+    assign decisions (Chief.py:33-53) on scripted score sequences, and
+    keeps replicas aligned over full rounds."""
+    mp.spawn(_curation_worker, args=(WORLD, 29807, str(tmp_path)),
+             nprocs=WORLD, join=True)
+    for r in range(WORLD):
+        with open(tmp_path / f"cur{r}.json") as f:
+            d = json.load(f)
+        assert d["swap_ok"], "sorted batch->tower assignment wrong"
+        assert d["recycle_ok"], "best-batch recycle for invalid rank wrong"
+        assert d["identical"], "replicas diverged under curation"
+
+
+@pytest.mark.timeout(600)
+def test_bench_eight_rank_gloo():
+    """bench.py's N>1 code path end-to-end (engine per rank, stats
+    all-gather, grad all-reduce, MAX-over-ranks timing reduce) on an
+    8-rank CPU gloo group with tiny shapes."""
+    import subprocess
+
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    env = dict(os.environ)
+    env.pop("RANK", None)
+    env.pop("WORLD_SIZE", None)
+    env.pop("LOCAL_RANK", None)
+    cmd = [
+        sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+        "--nproc-per-node", "8", "--master-addr", "127.0.0.1",
+        "--master-port", "29819", os.path.join(repo, "bench.py"),
+        "--gpus", "8", "--steps", "2", "--warmup", "1",
+        "--preset", "halfcheetah", "--num-envs", "4", "--rollout", "8",
+        "--device", "cpu",
+    ]
+    out = subprocess.run(cmd, capture_output=True, text=True, timeout=540,
+                         env=env, cwd=repo)
+    assert out.returncode == 0, out.stderr[-2000:]
+    lines = [ln for ln in out.stdout.splitlines() if ln.startswith("{")]
+    assert lines, out.stdout[-2000:]
+    result = json.loads(lines[-1])
+    assert result["n_gpus"] == 8
+    assert result["config"]["parallelism"] == "dp8"
+    assert result["value"] > 0 and result["ms_per_step"] > 0
+    # whole-job aggregate: 8 ranks x E x T x steps env-steps counted
+    assert result["config"]["global_batch"] == 8 * 4 * 8
+
+
 @pytest.mark.timeout(300)
 def test_drift_guard_broadcast_heals_divergence(tmp_path):
     """BROADCAST_INTERVAL>0 periodically re-broadcasts rank-0 weights
