@@ -1,0 +1,632 @@
+// Hand-written bf16 MFMA GEMM path for the wide MLP config (gfx950).
+//
+// Replaces the round-1 rocBLAS/autocast hot path of BASELINE config 5
+// (obs=4096, 4x4096 hidden, bf16) with CDNA4-native kernels:
+//
+//   bf16_mm256      C[M,N] = epi(A[M,K] @ B[N,K]^T)  — 256x256 tile,
+//                   BK=64, 8 waves, v_mfma_f32_16x16x32_bf16, 4-phase
+//                   pipelined global_load_lds staging with counted vmcnt
+//                   across raw barriers, st_16x32 LDS swizzle, XCD-aware
+//                   bijective block remap, LDS-repacked coalesced
+//                   epilogue with fused bias+tanh / dtanh / f32-grad
+//                   stores (the cdna guide's 256^2 8-phase template,
+//                   restructured as 4 phases per K-tile x 2-tile
+//                   software pipeline with identical in-flight counts).
+//   bf16_mm_small   guarded 64x64-tile variant for the ragged heads
+//                   GEMMs (N or M = P+1 = 513), incl. heads-split and
+//                   split-row grad epilogues.
+//   bf16_transpose  64x64 LDS tile transpose with optional fused column
+//                   sums (the bias gradients ride along for free).
+//   gauss_gh_wide   wave-per-row PPO loss gradient for wide policies
+//                   (A > 32; lane-strided column loops), bf16 output
+//                   feeding the dgrad GEMM chain directly.
+//
+// The torch weight layout [out][in] IS the MFMA B-operand layout (both
+// fragments read 8 contiguous bf16 along K), so forward needs no weight
+// transposes; dgrad uses W^T copies and dW uses transposed activations/
+// deltas produced by bf16_transpose.
+//
+// Reference semantics: these kernels implement the same per-layer math
+// the reference's tf.layers.dense graph launches (reference Model.py:12-14,
+// PPO.py:46 backward) at the wide config's shapes.
+
+#include <hip/hip_runtime.h>
+#include <hip/hip_bf16.h>
+#include <torch/extension.h>
+#include <c10/hip/HIPStream.h>
+
+#include "common.h"
+#include "ppo_math.h"
+
+namespace {
+
+typedef __attribute__((ext_vector_type(8))) short short8;
+typedef __attribute__((ext_vector_type(4))) float f32x4;
+typedef __attribute__((ext_vector_type(4))) unsigned short ushort4;
+
+DEV_INLINE float bf2f(unsigned short s) {
+  return __uint_as_float(((unsigned)s) << 16);
+}
+
+DEV_INLINE unsigned short f2bf(float f) {
+  __hip_bfloat16 h = __float2bfloat16(f);  // round-to-nearest-even
+  return *reinterpret_cast<unsigned short*>(&h);
+}
+
+// ---------------------------------------------------------------------------
+// bf16_mm256: 256x256x(K)  C = epi(A @ B^T)
+// ---------------------------------------------------------------------------
+
+constexpr int BK = 64;             // K elements per tile
+constexpr int IMG = 32 * 1024;     // one [256][64] bf16 image
+constexpr int HALFB = 16 * 1024;   // [128][64] half image
+// LDS: A images at 0/IMG, B images at 2*IMG/3*IMG = 128 KiB total
+constexpr int LDS_BYTES = 4 * IMG;
+
+enum { EPI_RAW = 0, EPI_TANH_BIAS = 1, EPI_DTANH = 2, EPI_GRAD = 3 };
+
+// st_16x32 XOR swizzle: involution on byte offsets within an image
+// (bit 9 toggles every 4 rows of 128 B; XOR bit 5 spreads the
+// ds_read_b128 lane groups over four 16-B slots instead of one).
+DEV_INLINE unsigned swz(unsigned o) { return o ^ (((o >> 9) & 1u) << 5); }
+
+struct MM256Args {
+  const unsigned short* A;  // [M][K]
+  const unsigned short* B;  // [N][K]
+  unsigned short* C;        // [M][N] bf16 (epi 0/1/2)
+  const float* bias;        // [N]     (epi 1)
+  const unsigned short* aux;  // [M][N] (epi 2: dtanh factor source)
+  float* grad;              // epi 3: f32 out at grad[m*N + n]
+  long M, N, K;
+  int nbn;                  // N / 256
+  int epi;
+};
+
+// Stage one [128][64] half-image by LDS-DMA: 2 x 16 B per thread, linear
+// LDS dest, st_16x32 pre-swizzle applied to the per-lane SOURCE address
+// (guide rule 21: glds dest must stay lane-linear).
+DEV_INLINE void stage_half(const unsigned short* __restrict__ g, long row0,
+                           long k0, long ld, char* img, int half, int tid,
+                           int wave) {
+  #pragma unroll
+  for (int p = 0; p < 2; ++p) {
+    const unsigned o = (unsigned)(half * HALFB + p * 8192 + tid * 16);
+    // swz() flips only bit 5; row/col decompose from the full image offset
+    const unsigned so = swz(o);
+    const unsigned row = so >> 7;        // 128 B per row
+    const unsigned colb = so & 127u;
+    const unsigned short* src = g + (row0 + row) * ld + (k0 + (colb >> 1));
+    __builtin_amdgcn_global_load_lds(
+        (const __attribute__((address_space(1))) unsigned*)src,
+        (unsigned*)(img + half * HALFB + p * 8192 + wave * 1024), 16, 0, 0);
+  }
+}
+
+DEV_INLINE short8 frag_read(const char* img, int row, int colb) {
+  unsigned byte = (unsigned)(row * 128 + colb);
+  byte = swz(byte);
+  return *(const short8*)(img + byte);
+}
+
+__launch_bounds__(512, 2)
+__global__ void bf16_mm256_kernel(MM256Args a) {
+  __shared__ __attribute__((aligned(16))) char smem[LDS_BYTES];
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int wr = wave >> 2;   // 2 M-halves
+  const int wc = wave & 3;    // 4 N-quarters
+
+  // bijective XCD-aware remap (8 XCDs; block b runs on XCD b%8)
+  const long nwg = (long)gridDim.x;
+  const long q = nwg >> 3, r8 = nwg & 7;
+  const long xcd = blockIdx.x & 7, idx = blockIdx.x >> 3;
+  const long sw = (xcd < r8 ? xcd * (q + 1) : r8 * (q + 1) + (xcd - r8) * q)
+                  + idx;
+  const int bm = (int)(sw / a.nbn);
+  const int bn = (int)(sw % a.nbn);
+
+  char* const imgA0 = smem;
+  char* const imgA1 = smem + IMG;
+  char* const imgB0 = smem + 2 * IMG;
+  char* const imgB1 = smem + 3 * IMG;
+  const long rowA = (long)bm * 256;
+  const long rowB = (long)bn * 256;
+  const int NT = (int)(a.K / BK);
+
+  f32x4 acc[8][4] = {};
+
+  // Prologue: tile0 A+B, tile1 B; leave tile1's B in flight (vmcnt(4)).
+  stage_half(a.A, rowA, 0, a.K, imgA0, 0, tid, wave);
+  stage_half(a.A, rowA, 0, a.K, imgA0, 1, tid, wave);
+  stage_half(a.B, rowB, 0, a.K, imgB0, 0, tid, wave);
+  stage_half(a.B, rowB, 0, a.K, imgB0, 1, tid, wave);
+  if (NT > 1) {
+    stage_half(a.B, rowB, BK, a.K, imgB1, 0, tid, wave);
+    stage_half(a.B, rowB, BK, a.K, imgB1, 1, tid, wave);
+    asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+  } else {
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  }
+  __builtin_amdgcn_s_barrier();
+
+  for (int t = 0; t < NT; ++t) {
+    const char* iA = (t & 1) ? imgA1 : imgA0;
+    const char* iB = (t & 1) ? imgB1 : imgB0;
+    // B fragments: one read pass, live for the whole tile
+    short8 bfr[4][2];
+    #pragma unroll
+    for (int fn = 0; fn < 4; ++fn)
+      #pragma unroll
+      for (int ks = 0; ks < 2; ++ks)
+        bfr[fn][ks] =
+            frag_read(iB, wc * 64 + fn * 16 + (lane & 15),
+                      ks * 64 + ((lane >> 4) & 3) * 16);
+
+    #pragma unroll
+    for (int ph = 0; ph < 4; ++ph) {
+      // A fragments for this phase's two row-blocks
+      short8 afr[2][2];
+      #pragma unroll
+      for (int i = 0; i < 2; ++i)
+        #pragma unroll
+        for (int ks = 0; ks < 2; ++ks)
+          afr[i][ks] =
+              frag_read(iA, wr * 128 + (2 * ph + i) * 16 + (lane & 15),
+                        ks * 64 + ((lane >> 4) & 3) * 16);
+      // staged prefetch: ph0/1 -> A halves of t+1; ph2/3 -> B halves of t+2
+      if (ph < 2) {
+        if (t + 1 < NT)
+          stage_half(a.A, rowA, (long)(t + 1) * BK, a.K,
+                     ((t + 1) & 1) ? imgA1 : imgA0, ph, tid, wave);
+      } else {
+        if (t + 2 < NT)
+          stage_half(a.B, rowB, (long)(t + 2) * BK, a.K,
+                     (t & 1) ? imgB1 : imgB0, ph - 2, tid, wave);
+      }
+      __builtin_amdgcn_s_barrier();
+      __builtin_amdgcn_s_setprio(1);
+      #pragma unroll
+      for (int i = 0; i < 2; ++i) {
+        #pragma unroll
+        for (int fn = 0; fn < 4; ++fn) {
+          acc[2 * ph + i][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              afr[i][0], bfr[fn][0], acc[2 * ph + i][fn], 0, 0, 0);
+          acc[2 * ph + i][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              afr[i][1], bfr[fn][1], acc[2 * ph + i][fn], 0, 0, 0);
+        }
+      }
+      __builtin_amdgcn_s_setprio(0);
+      if (ph == 3) {
+        // seam: tile t+1's halves must be LANDED in every wave before
+        // anyone reads them after this barrier.  Steady state leaves the
+        // two B halves of t+2 in flight (4 glds); at the tail drain all.
+        if (t + 2 < NT)
+          asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+        else
+          asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+      }
+      __builtin_amdgcn_s_barrier();
+    }
+  }
+
+  // ---- epilogue: LDS-repack each 16-row stripe to coalesced rows ----
+  float* stripe = (float*)(smem + wave * 4096);  // [16][64] f32, wave-private
+  const long cb = (long)bn * 256 + wc * 64;
+  #pragma unroll 1
+  for (int fm = 0; fm < 8; ++fm) {
+    #pragma unroll
+    for (int fn = 0; fn < 4; ++fn)
+      #pragma unroll
+      for (int rr = 0; rr < 4; ++rr)
+        stripe[(((lane >> 4) & 3) * 4 + rr) * 64 + fn * 16 + (lane & 15)] =
+            acc[fm][fn][rr];
+    const long rb = (long)bm * 256 + wr * 128 + fm * 16;
+    #pragma unroll
+    for (int pass = 0; pass < 4; ++pass) {
+      const int row = pass * 4 + ((lane >> 4) & 3);
+      const int col0 = (lane & 15) * 4;
+      f32x4 v = *(const f32x4*)(stripe + row * 64 + col0);
+      const long gr = rb + row;
+      const long gc = cb + col0;
+      if (a.epi == EPI_TANH_BIAS) {
+        const f32x4 b4 = *(const f32x4*)(a.bias + gc);
+        #pragma unroll
+        for (int j = 0; j < 4; ++j) v[j] = fast_tanhf(v[j] + b4[j]);
+      } else if (a.epi == EPI_DTANH) {
+        const ushort4 h4 = *(const ushort4*)(a.aux + gr * a.N + gc);
+        #pragma unroll
+        for (int j = 0; j < 4; ++j) {
+          const float h = bf2f(h4[j]);
+          v[j] *= 1.f - h * h;
+        }
+      }
+      if (a.epi == EPI_GRAD) {
+        *(f32x4*)(a.grad + gr * a.N + gc) = v;
+      } else {
+        ushort4 o;
+        #pragma unroll
+        for (int j = 0; j < 4; ++j) o[j] = f2bf(v[j]);
+        *(ushort4*)(a.C + gr * a.N + gc) = o;
+      }
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// bf16_mm_small: guarded 64x64-tile GEMM for ragged heads shapes.
+//   epi 0: C[M][ldc] bf16          epi 4: heads split (C=pdflat [M][N-1],
+//   epi 2: dtanh via aux [M][N]            C2=v [M]; no activation)
+//   epi 5: grad split-row (f32 grad: rows < srow -> g1 + m*N + n,
+//          row == srow -> g2 + n; rows > srow skipped)
+// Requires K % 32 == 0 and 16-B-aligned rows (callers pad K).
+// ---------------------------------------------------------------------------
+
+struct MMSmallArgs {
+  const unsigned short* A;  // [M][K]
+  const unsigned short* B;  // [N][K]
+  unsigned short* C;
+  unsigned short* C2;       // epi 4: value column
+  const unsigned short* aux;
+  float* g1;
+  float* g2;
+  long M, N, K, Mreal, Nreal, ldc;
+  int nbn, epi, srow;
+};
+
+constexpr int SM_STRIDE = 80;  // padded row stride (bytes) of the LDS images
+
+__launch_bounds__(256)
+__global__ void bf16_mm_small_kernel(MMSmallArgs a) {
+  __shared__ __attribute__((aligned(16))) char smem[2 * 2 * 64 * SM_STRIDE];
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int bm = blockIdx.x / a.nbn;
+  const int bn = blockIdx.x % a.nbn;
+  const long r0 = (long)bm * 64;   // A rows
+  const long c0 = (long)bn * 64;   // B rows (= C cols)
+  const int row = tid >> 2;
+  const int piece = tid & 3;
+  const int NC = (int)(a.K / 32);
+
+  char* const bufA0 = smem;
+  char* const bufB0 = smem + 64 * SM_STRIDE;
+  char* const bufA1 = smem + 2 * 64 * SM_STRIDE;
+  char* const bufB1 = smem + 3 * 64 * SM_STRIDE;
+
+  auto load_guarded = [&](const unsigned short* g, long grow, long rows,
+                          long k) -> short8 {
+    short8 z = {};
+    if (grow < rows) z = *(const short8*)(g + grow * a.K + k);
+    return z;
+  };
+
+  f32x4 acc[4] = {};
+  short8 ra = load_guarded(a.A, r0 + row, a.Mreal, piece * 8);
+  short8 rb = load_guarded(a.B, c0 + row, a.Nreal, piece * 8);
+  for (int kc = 0; kc < NC; ++kc) {
+    char* wA = (kc & 1) ? bufA1 : bufA0;
+    char* wB = (kc & 1) ? bufB1 : bufB0;
+    *(short8*)(wA + row * SM_STRIDE + piece * 16) = ra;
+    *(short8*)(wB + row * SM_STRIDE + piece * 16) = rb;
+    if (kc + 1 < NC) {
+      ra = load_guarded(a.A, r0 + row, a.Mreal, (long)(kc + 1) * 32 + piece * 8);
+      rb = load_guarded(a.B, c0 + row, a.Nreal, (long)(kc + 1) * 32 + piece * 8);
+    }
+    __syncthreads();
+    const char* iA = wA;
+    const char* iB = wB;
+    short8 bf = *(const short8*)(iB + (wave * 16 + (lane & 15)) * SM_STRIDE +
+                                 ((lane >> 4) & 3) * 16);
+    #pragma unroll
+    for (int fm = 0; fm < 4; ++fm) {
+      short8 af = *(const short8*)(iA + (fm * 16 + (lane & 15)) * SM_STRIDE +
+                                   ((lane >> 4) & 3) * 16);
+      acc[fm] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, bf, acc[fm], 0, 0, 0);
+    }
+    __syncthreads();
+  }
+
+  // guarded scalar epilogue
+  #pragma unroll
+  for (int fm = 0; fm < 4; ++fm) {
+    #pragma unroll
+    for (int rr = 0; rr < 4; ++rr) {
+      const long m = r0 + fm * 16 + ((lane >> 4) & 3) * 4 + rr;
+      const long n = c0 + wave * 16 + (lane & 15);
+      if (m >= a.Mreal || n >= a.Nreal) continue;
+      float v = acc[fm][rr];
+      if (a.epi == EPI_DTANH) {
+        const float h = bf2f(a.aux[m * a.Nreal + n]);
+        v *= 1.f - h * h;
+      }
+      if (a.epi == 4) {  // heads split: col < N-1 -> pdflat, col N-1 -> v
+        if (n < a.Nreal - 1)
+          a.C[m * a.ldc + n] = f2bf(v);
+        else
+          a.C2[m] = f2bf(v);
+      } else if (a.epi == 5) {  // grad split-row
+        if (m < a.srow)
+          a.g1[m * a.Nreal + n] = v;
+        else if (m == a.srow)
+          a.g2[n] = v;
+      } else {
+        a.C[m * a.ldc + n] = f2bf(v);
+      }
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// bf16_transpose (+ fused column sums): in [R][ld_in] -> out [C][ld_out]
+// ---------------------------------------------------------------------------
+
+struct TransArgs {
+  const unsigned short* in;
+  unsigned short* out;
+  float* sums;  // optional [C] f32, += column sums of `in`
+  long R, C, ld_in, ld_out;
+  int nbc;
+};
+
+__launch_bounds__(256)
+__global__ void bf16_transpose_kernel(TransArgs a) {
+  __shared__ __attribute__((aligned(16))) unsigned short tile[64][72];
+  const int tid = threadIdx.x;
+  const long br = (long)(blockIdx.x / a.nbc) * 64;
+  const long bc = (long)(blockIdx.x % a.nbc) * 64;
+  const int row = tid >> 2;           // 0..63
+  const int seg = tid & 3;            // 16-element column segment
+
+  // load [64][64] tile (2 x short8 per thread), scatter-transpose into LDS
+  #pragma unroll
+  for (int p = 0; p < 2; ++p) {
+    const int c0 = seg * 16 + p * 8;
+    short8 v = {};
+    if (br + row < a.R) {
+      if (bc + c0 + 7 < a.C) {
+        v = *(const short8*)(a.in + (br + row) * a.ld_in + bc + c0);
+      } else {
+        #pragma unroll
+        for (int j = 0; j < 8; ++j)
+          if (bc + c0 + j < a.C)
+            v[j] = (short)a.in[(br + row) * a.ld_in + bc + c0 + j];
+      }
+    }
+    #pragma unroll
+    for (int j = 0; j < 8; ++j) tile[c0 + j][row] = (unsigned short)v[j];
+  }
+  __syncthreads();
+
+  // write out rows of the transpose (out row = column of `in`)
+  const int oc = row;  // output row index within tile (= in column)
+  float csum = 0.f;
+  #pragma unroll
+  for (int p = 0; p < 2; ++p) {
+    const int r0 = seg * 16 + p * 8;
+    short8 v = *(const short8*)(&tile[oc][r0]);
+    if (a.out != nullptr && bc + oc < a.C) {
+      if (br + r0 + 7 < a.R) {
+        *(short8*)(a.out + (bc + oc) * a.ld_out + br + r0) = v;
+      } else {
+        #pragma unroll
+        for (int j = 0; j < 8; ++j)
+          if (br + r0 + j < a.R)
+            a.out[(bc + oc) * a.ld_out + br + r0 + j] = (unsigned short)v[j];
+      }
+    }
+    if (a.sums != nullptr) {
+      #pragma unroll
+      for (int j = 0; j < 8; ++j)
+        if (br + r0 + j < a.R) csum += bf2f((unsigned short)v[j]);
+    }
+  }
+  if (a.sums != nullptr && bc + oc < a.C) {
+    // 4 threads (seg 0..3) share one output column; lane-group reduce
+    csum += __shfl_down(csum, 1, 4);
+    csum += __shfl_down(csum, 2, 4);
+    if (seg == 0) atomicAdd(&a.sums[bc + oc], csum);
+  }
+}
+
+// ---------------------------------------------------------------------------
+// gauss_gh_wide: wave-per-row PPO loss gradient, wide policies (A > 32)
+// gh[b][j] = dtotal/d pdflat[b][j] (j < 2A), gh[b][2A] = dtotal/d v[b]
+// ---------------------------------------------------------------------------
+
+struct GhWideArgs {
+  const unsigned short* pdflat;  // [B][2A] bf16 (pi outputs)
+  const float* oldflat;          // [B][2A] f32 (recorded)
+  const unsigned short* v;       // [B] bf16
+  const float* oldv;             // [B]
+  const float* act;              // [B][A]
+  const float* adv;              // [B]
+  const float* etr;              // [B]
+  unsigned short* gh;            // [B][ldgh] bf16 (zero-padded tail)
+  const float* clip_dev;         // optional device clip override
+  long B, ldgh;
+  int A;
+  float clip, entcoeff, vcoeff;
+};
+
+__launch_bounds__(256)
+__global__ void gauss_gh_wide_kernel(GhWideArgs a) {
+  const float clip = (a.clip_dev != nullptr) ? a.clip_dev[0] : a.clip;
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wave = threadIdx.x / WAVE;
+  const int P = 2 * a.A;
+  const int64_t wid = (int64_t)blockIdx.x * 4 + wave;
+  const int64_t waves_total = (int64_t)gridDim.x * 4;
+  const int64_t per = (a.B + waves_total - 1) / waves_total;
+  const int64_t rb0 = wid * per;
+  const int64_t rb1 = rb0 + per < a.B ? rb0 + per : a.B;
+
+  for (int64_t b = rb0; b < rb1; ++b) {
+    const unsigned short* mu = a.pdflat + b * P;
+    const float* mo = a.oldflat + b * P;
+    const float* aj = a.act + b * a.A;
+    float lp = 0.f, lo = 0.f;
+    for (int j = lane; j < a.A; j += WAVE) {
+      const float ls = bf2f(mu[a.A + j]);
+      const float zp = (aj[j] - bf2f(mu[j])) * __expf(-ls);
+      lp += -0.5f * zp * zp - ls;
+      const float lso = mo[a.A + j];
+      const float zo = (aj[j] - mo[j]) * __expf(-lso);
+      lo += -0.5f * zo * zo - lso;
+    }
+    const float c = 0.5f * PPO_LOG_2PI * a.A;
+    GaussRow row;
+    row.logp_pi = __shfl(wave_reduce_sum(lp), 0, WAVE) - c;
+    row.logp_old = __shfl(wave_reduce_sum(lo), 0, WAVE) - c;
+    row.ent = 0.f;  // unused by ppo_row_grads
+    const PPORowGrads g =
+        ppo_row_grads(row, bf2f(a.v[b]), a.oldv[b], a.adv[b], a.etr[b], a.B,
+                      clip, a.entcoeff, a.vcoeff, 1.f);
+    unsigned short* out = a.gh + b * a.ldgh;
+    for (int j = lane; j < a.A; j += WAVE) {
+      const float ls = bf2f(mu[a.A + j]);
+      const float inv_s = __expf(-ls);
+      const float z = (aj[j] - bf2f(mu[j])) * inv_s;
+      out[j] = f2bf(g.g_logp * z * inv_s);
+      out[a.A + j] = f2bf(g.g_logp * (z * z - 1.f) + g.g_ent);
+    }
+    if (lane == 0) out[P] = f2bf(g.g_v);
+  }
+}
+
+}  // namespace
+
+// ---------------------------------------------------------------------------
+// host wrappers
+// ---------------------------------------------------------------------------
+
+static const unsigned short* bf_ptr(const torch::Tensor& t) {
+  return reinterpret_cast<const unsigned short*>(t.data_ptr<at::BFloat16>());
+}
+static unsigned short* bf_ptr_mut(torch::Tensor& t) {
+  return reinterpret_cast<unsigned short*>(t.data_ptr<at::BFloat16>());
+}
+
+void bf16_mm256(torch::Tensor A, torch::Tensor B, torch::Tensor C,
+                int64_t epi, torch::Tensor bias, torch::Tensor aux,
+                torch::Tensor grad, int64_t grad_off) {
+  TORCH_CHECK(A.is_cuda() && A.dtype() == torch::kBFloat16 && A.dim() == 2);
+  TORCH_CHECK(B.dtype() == torch::kBFloat16 && B.dim() == 2);
+  const long M = A.size(0), K = A.size(1), N = B.size(0);
+  TORCH_CHECK(B.size(1) == K);
+  TORCH_CHECK(M % 256 == 0 && N % 256 == 0 && K % 64 == 0,
+              "bf16_mm256 needs M,N % 256 == 0 and K % 64 == 0");
+  TORCH_CHECK(A.is_contiguous() && B.is_contiguous());
+  MM256Args a{};
+  a.A = bf_ptr(A);
+  a.B = bf_ptr(B);
+  a.M = M;
+  a.N = N;
+  a.K = K;
+  a.nbn = (int)(N / 256);
+  a.epi = (int)epi;
+  if (epi == EPI_GRAD) {
+    TORCH_CHECK(grad.numel() >= grad_off + M * N);
+    a.grad = grad.data_ptr<float>() + grad_off;
+  } else {
+    TORCH_CHECK(C.dtype() == torch::kBFloat16 && C.is_contiguous() &&
+                C.numel() == M * N);
+    a.C = bf_ptr_mut(C);
+  }
+  if (epi == EPI_TANH_BIAS) {
+    TORCH_CHECK(bias.numel() == N && bias.dtype() == torch::kFloat32);
+    a.bias = bias.data_ptr<float>();
+  }
+  if (epi == EPI_DTANH) {
+    TORCH_CHECK(aux.dtype() == torch::kBFloat16 && aux.numel() == M * N);
+    a.aux = bf_ptr(aux);
+  }
+  hipStream_t stream = c10::hip::getCurrentHIPStream().stream();
+  const long grid = (M / 256) * (N / 256);
+  hipLaunchKernelGGL(bf16_mm256_kernel, dim3((unsigned)grid), dim3(512), 0,
+                     stream, a);
+}
+
+void bf16_mm_small(torch::Tensor A, torch::Tensor B, torch::Tensor C,
+                   torch::Tensor C2, torch::Tensor aux, torch::Tensor grad,
+                   int64_t g1_off, int64_t g2_off, int64_t srow, int64_t epi,
+                   int64_t m_real, int64_t n_real, int64_t ldc) {
+  TORCH_CHECK(A.is_cuda() && A.dtype() == torch::kBFloat16 && A.dim() == 2);
+  const long K = A.size(1);
+  TORCH_CHECK(B.size(1) == K && K % 32 == 0 && K % 8 == 0);
+  MMSmallArgs a{};
+  a.A = bf_ptr(A);
+  a.B = bf_ptr(B);
+  a.K = K;
+  a.Mreal = m_real > 0 ? m_real : A.size(0);
+  a.Nreal = n_real > 0 ? n_real : B.size(0);
+  a.M = (a.Mreal + 63) & ~63L;
+  a.N = (a.Nreal + 63) & ~63L;
+  a.ldc = ldc > 0 ? ldc : a.Nreal;
+  a.nbn = (int)(a.N / 64);
+  a.epi = (int)epi;
+  a.srow = (int)srow;
+  if (epi == 5) {
+    a.g1 = grad.data_ptr<float>() + g1_off;
+    a.g2 = grad.data_ptr<float>() + g2_off;
+    a.Nreal = n_real;  // grads indexed by Nreal row stride
+  } else {
+    a.C = bf_ptr_mut(C);
+    if (epi == 4) a.C2 = bf_ptr_mut(C2);
+    if (epi == EPI_DTANH) a.aux = bf_ptr(aux);
+  }
+  hipStream_t stream = c10::hip::getCurrentHIPStream().stream();
+  const long grid = (a.M / 64) * (a.N / 64);
+  hipLaunchKernelGGL(bf16_mm_small_kernel, dim3((unsigned)grid), dim3(256), 0,
+                     stream, a);
+}
+
+void bf16_transpose(torch::Tensor in, torch::Tensor out, torch::Tensor sums,
+                    int64_t sums_off, int64_t R, int64_t C, int64_t ld_in,
+                    int64_t ld_out) {
+  TORCH_CHECK(in.is_cuda() && in.dtype() == torch::kBFloat16);
+  TransArgs a{};
+  a.in = bf_ptr(in);
+  a.out = out.numel() > 0 ? bf_ptr_mut(out) : nullptr;
+  a.sums = sums.numel() > 0 ? sums.data_ptr<float>() + sums_off : nullptr;
+  a.R = R;
+  a.C = C;
+  a.ld_in = ld_in > 0 ? ld_in : C;
+  a.ld_out = ld_out > 0 ? ld_out : R;
+  const long nbr = (R + 63) / 64, nbc = (C + 63) / 64;
+  a.nbc = (int)nbc;
+  hipStream_t stream = c10::hip::getCurrentHIPStream().stream();
+  hipLaunchKernelGGL(bf16_transpose_kernel, dim3((unsigned)(nbr * nbc)),
+                     dim3(256), 0, stream, a);
+}
+
+void gauss_gh_wide(torch::Tensor pdflat_bf, torch::Tensor oldflat,
+                   torch::Tensor v_bf, torch::Tensor oldv, torch::Tensor act,
+                   torch::Tensor adv, torch::Tensor etr, torch::Tensor gh,
+                   torch::Tensor clip_dev, double clip, double entcoeff,
+                   double vcoeff) {
+  TORCH_CHECK(pdflat_bf.is_cuda() && pdflat_bf.dtype() == torch::kBFloat16);
+  TORCH_CHECK(gh.dtype() == torch::kBFloat16 && gh.dim() == 2);
+  const long B = pdflat_bf.size(0);
+  const int A = (int)(pdflat_bf.size(1) / 2);
+  TORCH_CHECK(gh.size(0) == B && gh.size(1) >= 2 * A + 1);
+  GhWideArgs a{};
+  a.pdflat = bf_ptr(pdflat_bf);
+  a.oldflat = oldflat.data_ptr<float>();
+  a.v = bf_ptr(v_bf);
+  a.oldv = oldv.data_ptr<float>();
+  a.act = act.data_ptr<float>();
+  a.adv = adv.data_ptr<float>();
+  a.etr = etr.data_ptr<float>();
+  a.gh = bf_ptr_mut(gh);
+  a.clip_dev = clip_dev.numel() > 0 ? clip_dev.data_ptr<float>() : nullptr;
+  a.B = B;
+  a.ldgh = gh.size(1);
+  a.A = A;
+  a.clip = (float)clip;
+  a.entcoeff = (float)entcoeff;
+  a.vcoeff = (float)vcoeff;
+  hipStream_t stream = c10::hip::getCurrentHIPStream().stream();
+  hipLaunchKernelGGL(gauss_gh_wide_kernel, dim3(1024), dim3(256), 0, stream, a);
+}

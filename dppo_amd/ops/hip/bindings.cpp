@@ -42,6 +42,25 @@ std::vector<torch::Tensor> ppo_loss_cat_bwd(
 
 torch::Tensor cat_sample(torch::Tensor logits, int64_t seed, int64_t ctr);
 
+void bf16_mm256(torch::Tensor A, torch::Tensor B, torch::Tensor C,
+                int64_t epi, torch::Tensor bias, torch::Tensor aux,
+                torch::Tensor grad, int64_t grad_off);
+
+void bf16_mm_small(torch::Tensor A, torch::Tensor B, torch::Tensor C,
+                   torch::Tensor C2, torch::Tensor aux, torch::Tensor grad,
+                   int64_t g1_off, int64_t g2_off, int64_t srow, int64_t epi,
+                   int64_t m_real, int64_t n_real, int64_t ldc);
+
+void bf16_transpose(torch::Tensor in, torch::Tensor out, torch::Tensor sums,
+                    int64_t sums_off, int64_t R, int64_t C, int64_t ld_in,
+                    int64_t ld_out);
+
+void gauss_gh_wide(torch::Tensor pdflat_bf, torch::Tensor oldflat,
+                   torch::Tensor v_bf, torch::Tensor oldv, torch::Tensor act,
+                   torch::Tensor adv, torch::Tensor etr, torch::Tensor gh,
+                   torch::Tensor clip_dev, double clip, double entcoeff,
+                   double vcoeff);
+
 void adam_step(torch::Tensor p, torch::Tensor g, torch::Tensor m,
                torch::Tensor v, int64_t step, double lr, double beta1,
                double beta2, double eps);
@@ -108,6 +127,15 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
           "fused Categorical PPO loss backward (gfx950)");
   mod.def("cat_sample", &cat_sample,
           "Gumbel-max categorical sampling, counter-based RNG (gfx950)");
+  mod.def("bf16_mm256", &bf16_mm256,
+          "bf16 MFMA GEMM, 256^2 tile, pipelined glds, fused epilogues "
+          "(gfx950)");
+  mod.def("bf16_mm_small", &bf16_mm_small,
+          "guarded bf16 MFMA GEMM for ragged heads shapes (gfx950)");
+  mod.def("bf16_transpose", &bf16_transpose,
+          "bf16 2D transpose with fused column sums (gfx950)");
+  mod.def("gauss_gh_wide", &gauss_gh_wide,
+          "wide-policy PPO loss gradient -> bf16 [g_pd | g_v] (gfx950)");
   mod.def("adam_step", &adam_step, "fused flat Adam step (gfx950)");
   mod.def("adam_step_dev", &adam_step_dev,
           "graph-replayable fused Adam (device step/lr) (gfx950)");

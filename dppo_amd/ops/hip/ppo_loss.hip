@@ -124,6 +124,63 @@ __global__ void ppo_gauss_fwd_kernel(
   }
 }
 
+// Wide-policy (A > 64) forward: same math, lane-strided column loop so
+// every action dim is accumulated (the pipelined kernel above maps one
+// column per lane and is dispatched only for A <= 64).
+__launch_bounds__(256)
+__global__ void ppo_gauss_fwd_wide_kernel(
+    const float* __restrict__ pdpi, const float* __restrict__ pdold,
+    const float* __restrict__ vpred, const float* __restrict__ oldv,
+    const float* __restrict__ act, const float* __restrict__ adv,
+    const float* __restrict__ etr, double* __restrict__ acc,
+    int64_t B, int A, float clip) {
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wave = threadIdx.x / WAVE;
+  const int P = 2 * A;
+  const int64_t waves_total = (int64_t)gridDim.x * 4;
+  const int64_t wid = (int64_t)blockIdx.x * 4 + wave;
+  const int64_t per = (B + waves_total - 1) / waves_total;
+  const int64_t rb0 = wid * per;
+  const int64_t rb1 = rb0 + per < B ? rb0 + per : B;
+
+  float pol = 0.f, ent = 0.f, val = 0.f;
+  for (int64_t b = rb0; b < rb1; ++b) {
+    float lp_part = 0.f, lo_part = 0.f, ent_part = 0.f;
+    const float* mu = pdpi + b * P;
+    const float* mo = pdold + b * P;
+    const float* aj = act + b * (int64_t)A;
+    for (int j = lane; j < A; j += WAVE) {
+      const float ls = mu[A + j];
+      const float zp = (aj[j] - mu[j]) * __expf(-ls);
+      lp_part += -0.5f * zp * zp - ls;
+      const float lso = mo[A + j];
+      const float zo = (aj[j] - mo[j]) * __expf(-lso);
+      lo_part += -0.5f * zo * zo - lso;
+      ent_part += ls;
+    }
+    const float lp = wave_reduce_sum(lp_part);
+    const float lo = wave_reduce_sum(lo_part);
+    const float es = wave_reduce_sum(ent_part);
+    if (lane == 0) {
+      const float ratio = __expf(lp - lo);
+      const float ab = adv[b];
+      const float surr1 = ratio * ab;
+      const float rc = fminf(fmaxf(ratio, 1.f - clip), 1.f + clip);
+      pol += fminf(surr1, rc * ab);
+      ent += es + 0.5f * (PPO_LOG_2PI + 1.f) * A;
+      const float d1 = vpred[b] - etr[b];
+      const float dc = fminf(fmaxf(vpred[b] - oldv[b], -clip), clip);
+      const float d2 = oldv[b] + dc - etr[b];
+      val += fmaxf(d1 * d1, d2 * d2);
+    }
+  }
+  if (lane == 0) {
+    atomicAdd(&acc[0], static_cast<double>(pol));
+    atomicAdd(&acc[1], static_cast<double>(ent));
+    atomicAdd(&acc[2], static_cast<double>(val));
+  }
+}
+
 __global__ void ppo_gauss_finalize_kernel(const double* __restrict__ acc,
                                           float* __restrict__ losses,  // [4]
                                           int64_t B, float entcoeff,
@@ -223,7 +280,8 @@ torch::Tensor ppo_loss_gauss_fwd(torch::Tensor pdpi, torch::Tensor pdold,
   auto losses = torch::empty({4}, pdpi.options());
   hipStream_t stream = c10::hip::getCurrentHIPStream().stream();
   const int block = 256;
-  hipLaunchKernelGGL(ppo_gauss_fwd_kernel, dim3(2048),
+  auto* fwd = (A <= WAVE) ? &ppo_gauss_fwd_kernel : &ppo_gauss_fwd_wide_kernel;
+  hipLaunchKernelGGL(fwd, dim3(2048),
                      dim3(block), 0, stream, pdpi.data_ptr<float>(),
                      pdold.data_ptr<float>(), vpred.data_ptr<float>(),
                      oldv.data_ptr<float>(), act.data_ptr<float>(),
@@ -368,6 +426,9 @@ torch::Tensor ppo_loss_gauss_gh(torch::Tensor pdflat, torch::Tensor oldflat,
                                 torch::Tensor clip_dev) {
   const int64_t B = vpred.numel();
   const int A = static_cast<int>(pdflat.size(1) / 2);
+  TORCH_CHECK(2 * A + 1 <= WAVE + 1,
+              "ppo_loss_gauss_gh is wave-per-row (2A+1 <= 65 lanes); "
+              "wide policies use the bf16 gh kernel");
   // one float4 of slack past the end: the pipelined dgrad GEMM reads the
   // [B][2A+1] rows with whole-float4 loads (K=2A+1 is odd), so the last
   // row's tail load overhangs by up to 3 floats.  The overhang elements
