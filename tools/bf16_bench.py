@@ -3,7 +3,10 @@ shapes, A/B'd against torch/rocBLAS bf16 matmul on the same random data
 (guide §5.4 rule 25: random operands, within-probe interleave)."""
 
 import argparse
+import sys
 import time
+
+sys.path.insert(0, "/root/repo")
 
 import torch
 
