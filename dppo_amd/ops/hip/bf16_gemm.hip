@@ -82,24 +82,24 @@ struct MM256Args {
   int epi;
 };
 
-// Stage one [128][64] half-image by LDS-DMA: 2 x 16 B per thread, linear
-// LDS dest, st_16x32 pre-swizzle applied to the per-lane SOURCE address
-// (guide rule 21: glds dest must stay lane-linear).
-DEV_INLINE void stage_half(const unsigned short* __restrict__ g, long row0,
-                           long k0, long ld, char* img, int half, int tid,
-                           int wave) {
-  #pragma unroll
-  for (int p = 0; p < 2; ++p) {
-    const unsigned o = (unsigned)(half * HALFB + p * 8192 + tid * 16);
-    // swz() flips only bit 5; row/col decompose from the full image offset
-    const unsigned so = swz(o);
-    const unsigned row = so >> 7;        // 128 B per row
-    const unsigned colb = so & 127u;
-    const unsigned short* src = g + (row0 + row) * ld + (k0 + (colb >> 1));
-    __builtin_amdgcn_global_load_lds(
-        (const __attribute__((address_space(1))) unsigned*)src,
-        (unsigned*)(img + half * HALFB + p * 8192 + wave * 1024), 16, 0, 0);
-  }
+// Per-thread glds source pointer for one (half, piece): linear LDS dest,
+// st_16x32 pre-swizzle applied to the SOURCE address (guide rule 21:
+// glds dest must stay lane-linear).  Row/col are constant per thread, so
+// the 64-bit address math happens ONCE; the K-loop just adds k0.
+DEV_INLINE const unsigned short* stage_src(const unsigned short* g, long row0,
+                                           long ld, int half, int p, int tid) {
+  const unsigned o = (unsigned)(half * HALFB + p * 8192 + tid * 16);
+  const unsigned so = swz(o);  // flips only bit 5 within the image
+  const unsigned row = so >> 7;        // 128 B per row
+  const unsigned colb = so & 127u;
+  return g + (row0 + row) * ld + (colb >> 1);
+}
+
+DEV_INLINE void stage_piece(const unsigned short* src, long k0, char* img,
+                            int half, int p, int wave) {
+  __builtin_amdgcn_global_load_lds(
+      (const __attribute__((address_space(1))) unsigned*)(src + k0),
+      (unsigned*)(img + half * HALFB + p * 8192 + wave * 1024), 16, 0, 0);
 }
 
 DEV_INLINE short8 frag_read(const char* img, int row, int colb) {
@@ -136,14 +136,30 @@ __global__ void bf16_mm256_kernel(MM256Args a) {
 
   f32x4 acc[8][4] = {};
 
+  // per-thread glds source pointers (constant rows/cols; advance by k0)
+  const unsigned short* sA00 = stage_src(a.A, rowA, a.K, 0, 0, tid);
+  const unsigned short* sA01 = stage_src(a.A, rowA, a.K, 0, 1, tid);
+  const unsigned short* sA10 = stage_src(a.A, rowA, a.K, 1, 0, tid);
+  const unsigned short* sA11 = stage_src(a.A, rowA, a.K, 1, 1, tid);
+  const unsigned short* sB00 = stage_src(a.B, rowB, a.K, 0, 0, tid);
+  const unsigned short* sB01 = stage_src(a.B, rowB, a.K, 0, 1, tid);
+  const unsigned short* sB10 = stage_src(a.B, rowB, a.K, 1, 0, tid);
+  const unsigned short* sB11 = stage_src(a.B, rowB, a.K, 1, 1, tid);
+
   // Prologue: tile0 A+B, tile1 B; leave tile1's B in flight (vmcnt(4)).
-  stage_half(a.A, rowA, 0, a.K, imgA0, 0, tid, wave);
-  stage_half(a.A, rowA, 0, a.K, imgA0, 1, tid, wave);
-  stage_half(a.B, rowB, 0, a.K, imgB0, 0, tid, wave);
-  stage_half(a.B, rowB, 0, a.K, imgB0, 1, tid, wave);
+  stage_piece(sA00, 0, imgA0, 0, 0, wave);
+  stage_piece(sA01, 0, imgA0, 0, 1, wave);
+  stage_piece(sA10, 0, imgA0, 1, 0, wave);
+  stage_piece(sA11, 0, imgA0, 1, 1, wave);
+  stage_piece(sB00, 0, imgB0, 0, 0, wave);
+  stage_piece(sB01, 0, imgB0, 0, 1, wave);
+  stage_piece(sB10, 0, imgB0, 1, 0, wave);
+  stage_piece(sB11, 0, imgB0, 1, 1, wave);
   if (NT > 1) {
-    stage_half(a.B, rowB, BK, a.K, imgB1, 0, tid, wave);
-    stage_half(a.B, rowB, BK, a.K, imgB1, 1, tid, wave);
+    stage_piece(sB00, BK, imgB1, 0, 0, wave);
+    stage_piece(sB01, BK, imgB1, 0, 1, wave);
+    stage_piece(sB10, BK, imgB1, 1, 0, wave);
+    stage_piece(sB11, BK, imgB1, 1, 1, wave);
     asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
   } else {
     asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
@@ -175,14 +191,30 @@ __global__ void bf16_mm256_kernel(MM256Args a) {
               frag_read(iA, wr * 128 + (2 * ph + i) * 16 + (lane & 15),
                         ks * 64 + ((lane >> 4) & 3) * 16);
       // staged prefetch: ph0/1 -> A halves of t+1; ph2/3 -> B halves of t+2
-      if (ph < 2) {
-        if (t + 1 < NT)
-          stage_half(a.A, rowA, (long)(t + 1) * BK, a.K,
-                     ((t + 1) & 1) ? imgA1 : imgA0, ph, tid, wave);
+      if (ph == 0) {
+        if (t + 1 < NT) {
+          char* img = ((t + 1) & 1) ? imgA1 : imgA0;
+          stage_piece(sA00, (long)(t + 1) * BK, img, 0, 0, wave);
+          stage_piece(sA01, (long)(t + 1) * BK, img, 0, 1, wave);
+        }
+      } else if (ph == 1) {
+        if (t + 1 < NT) {
+          char* img = ((t + 1) & 1) ? imgA1 : imgA0;
+          stage_piece(sA10, (long)(t + 1) * BK, img, 1, 0, wave);
+          stage_piece(sA11, (long)(t + 1) * BK, img, 1, 1, wave);
+        }
+      } else if (ph == 2) {
+        if (t + 2 < NT) {
+          char* img = (t & 1) ? imgB1 : imgB0;
+          stage_piece(sB00, (long)(t + 2) * BK, img, 0, 0, wave);
+          stage_piece(sB01, (long)(t + 2) * BK, img, 0, 1, wave);
+        }
       } else {
-        if (t + 2 < NT)
-          stage_half(a.B, rowB, (long)(t + 2) * BK, a.K,
-                     (t & 1) ? imgB1 : imgB0, ph - 2, tid, wave);
+        if (t + 2 < NT) {
+          char* img = (t & 1) ? imgB1 : imgB0;
+          stage_piece(sB10, (long)(t + 2) * BK, img, 1, 0, wave);
+          stage_piece(sB11, (long)(t + 2) * BK, img, 1, 1, wave);
+        }
       }
       __builtin_amdgcn_s_barrier();
       __builtin_amdgcn_s_setprio(1);
@@ -213,7 +245,10 @@ __global__ void bf16_mm256_kernel(MM256Args a) {
   // ---- epilogue: LDS-repack each 16-row stripe to coalesced rows ----
   float* stripe = (float*)(smem + wave * 4096);  // [16][64] f32, wave-private
   const long cb = (long)bn * 256 + wc * 64;
-  #pragma unroll 1
+  // full unroll: keeps acc[] indices compile-time so the accumulator
+  // array stays in registers (guide §5.4 rule 20 — a runtime index sends
+  // the whole array to scratch)
+  #pragma unroll
   for (int fm = 0; fm < 8; ++fm) {
     #pragma unroll
     for (int fn = 0; fn < 4; ++fn)
