@@ -65,10 +65,16 @@ constexpr int LDS_BYTES = 4 * IMG;
 
 enum { EPI_RAW = 0, EPI_TANH_BIAS = 1, EPI_DTANH = 2, EPI_GRAD = 3 };
 
-// st_16x32 XOR swizzle: involution on byte offsets within an image
-// (bit 9 toggles every 4 rows of 128 B; XOR bit 5 spreads the
-// ds_read_b128 lane groups over four 16-B slots instead of one).
-DEV_INLINE unsigned swz(unsigned o) { return o ^ (((o >> 9) & 1u) << 5); }
+// XOR swizzle: involution on byte offsets within an image.  Rows are
+// 128 B, so byte bits 8/9/10 are row-index bits 1/2/3; XOR-ing them into
+// the three 16-B slot-select bits (4/5/6) gives every one of the 16 rows
+// a fragment read touches a distinct 16-B slot of the 256-B bank row —
+// conflict-FREE ds_read_b128 for the [row=base+(lane&15)] access pattern
+// (the guide's 2-bit st_16x32 left a residual 2-way conflict here).
+DEV_INLINE unsigned swz(unsigned o) {
+  return o ^ (((o >> 9) & 1u) << 5) ^ (((o >> 8) & 1u) << 4) ^
+         (((o >> 10) & 1u) << 6);
+}
 
 struct MM256Args {
   const unsigned short* A;  // [M][K]
@@ -190,26 +196,26 @@ __global__ void bf16_mm256_kernel(MM256Args a) {
           afr[i][ks] =
               frag_read(iA, wr * 128 + (2 * ph + i) * 16 + (lane & 15),
                         ks * 64 + ((lane >> 4) & 3) * 16);
-      // staged prefetch: ph0/1 -> A halves of t+1; ph2/3 -> B halves of t+2
+      // Staged prefetch — every piece gets >= 3 phases of flight before
+      // its consumer's seam wait (HBM latency ~900 cyc ~ 3 phases):
+      //   ph0: BOTH A halves of t+1 (their buffer is idle during t);
+      //   ph1/ph2: B halves of t+2 (B(t)'s buffer is free after ph0's
+      //   B-fragment reads, sealed by the ph0-end barrier).
       if (ph == 0) {
         if (t + 1 < NT) {
           char* img = ((t + 1) & 1) ? imgA1 : imgA0;
           stage_piece(sA00, (long)(t + 1) * BK, img, 0, 0, wave);
           stage_piece(sA01, (long)(t + 1) * BK, img, 0, 1, wave);
-        }
-      } else if (ph == 1) {
-        if (t + 1 < NT) {
-          char* img = ((t + 1) & 1) ? imgA1 : imgA0;
           stage_piece(sA10, (long)(t + 1) * BK, img, 1, 0, wave);
           stage_piece(sA11, (long)(t + 1) * BK, img, 1, 1, wave);
         }
-      } else if (ph == 2) {
+      } else if (ph == 1) {
         if (t + 2 < NT) {
           char* img = (t & 1) ? imgB1 : imgB0;
           stage_piece(sB00, (long)(t + 2) * BK, img, 0, 0, wave);
           stage_piece(sB01, (long)(t + 2) * BK, img, 0, 1, wave);
         }
-      } else {
+      } else if (ph == 2) {
         if (t + 2 < NT) {
           char* img = (t & 1) ? imgB1 : imgB0;
           stage_piece(sB10, (long)(t + 2) * BK, img, 1, 0, wave);
