@@ -114,6 +114,7 @@ DEV_INLINE short8 frag_read(const char* img, int row, int colb) {
   return *(const short8*)(img + byte);
 }
 
+template <bool TWO_BARRIERS>
 __launch_bounds__(512, 2)
 __global__ void bf16_mm256_kernel(MM256Args a) {
   __shared__ __attribute__((aligned(16))) char smem[LDS_BYTES];
@@ -222,7 +223,13 @@ __global__ void bf16_mm256_kernel(MM256Args a) {
           stage_piece(sB11, (long)(t + 2) * BK, img, 1, 1, wave);
         }
       }
-      __builtin_amdgcn_s_barrier();
+      // TWO_BARRIERS aligns every wave's MFMA cluster (matrix||matrix on
+      // each SIMD); the single-barrier form lets a wave's reads/stages
+      // overlap its SIMD partner's MFMA segment (matrix||memory pairing,
+      // microarch §two-waves-per-SIMD item 5) — measured faster, kept as
+      // the default; correctness needs only the end-of-phase barrier
+      // (read-before-overwrite is sealed one phase ahead either way).
+      if (TWO_BARRIERS) __builtin_amdgcn_s_barrier();
       __builtin_amdgcn_s_setprio(1);
       #pragma unroll
       for (int i = 0; i < 2; ++i) {
@@ -585,8 +592,16 @@ void bf16_mm256(torch::Tensor A, torch::Tensor B, torch::Tensor C,
   }
   hipStream_t stream = c10::hip::getCurrentHIPStream().stream();
   const long grid = (M / 256) * (N / 256);
-  hipLaunchKernelGGL(bf16_mm256_kernel, dim3((unsigned)grid), dim3(512), 0,
-                     stream, a);
+  static const bool two_bar = []() {
+    const char* e = getenv("DPPO_MM256_2BAR");
+    return e != nullptr && e[0] == '1';
+  }();
+  if (two_bar)
+    hipLaunchKernelGGL(bf16_mm256_kernel<true>, dim3((unsigned)grid),
+                       dim3(512), 0, stream, a);
+  else
+    hipLaunchKernelGGL(bf16_mm256_kernel<false>, dim3((unsigned)grid),
+                       dim3(512), 0, stream, a);
 }
 
 void bf16_mm_small(torch::Tensor A, torch::Tensor B, torch::Tensor C,
