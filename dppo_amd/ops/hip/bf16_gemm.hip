@@ -316,6 +316,7 @@ struct MMSmallArgs {
   unsigned short* C;
   unsigned short* C2;       // epi 4: value column
   const unsigned short* aux;
+  const float* bias;        // optional [Nreal] f32 (epi 0/4)
   float* g1;
   float* g2;
   long M, N, K, Mreal, Nreal, ldc;
@@ -385,6 +386,7 @@ __global__ void bf16_mm_small_kernel(MMSmallArgs a) {
       const long n = c0 + wave * 16 + (lane & 15);
       if (m >= a.Mreal || n >= a.Nreal) continue;
       float v = acc[fm][rr];
+      if (a.bias != nullptr) v += a.bias[n];
       if (a.epi == EPI_DTANH) {
         const float h = bf2f(a.aux[m * a.Nreal + n]);
         v *= 1.f - h * h;
@@ -607,7 +609,8 @@ void bf16_mm256(torch::Tensor A, torch::Tensor B, torch::Tensor C,
 void bf16_mm_small(torch::Tensor A, torch::Tensor B, torch::Tensor C,
                    torch::Tensor C2, torch::Tensor aux, torch::Tensor grad,
                    int64_t g1_off, int64_t g2_off, int64_t srow, int64_t epi,
-                   int64_t m_real, int64_t n_real, int64_t ldc) {
+                   int64_t m_real, int64_t n_real, int64_t ldc,
+                   torch::Tensor bias) {
   TORCH_CHECK(A.is_cuda() && A.dtype() == torch::kBFloat16 && A.dim() == 2);
   const long K = A.size(1);
   TORCH_CHECK(B.size(1) == K && K % 32 == 0 && K % 8 == 0);
@@ -623,6 +626,10 @@ void bf16_mm_small(torch::Tensor A, torch::Tensor B, torch::Tensor C,
   a.nbn = (int)(a.N / 64);
   a.epi = (int)epi;
   a.srow = (int)srow;
+  if (bias.numel() > 0) {
+    TORCH_CHECK(bias.dtype() == torch::kFloat32 && bias.numel() >= a.Nreal);
+    a.bias = bias.data_ptr<float>();
+  }
   if (epi == 5) {
     a.g1 = grad.data_ptr<float>() + g1_off;
     a.g2 = grad.data_ptr<float>() + g2_off;

@@ -49,7 +49,8 @@ void bf16_mm256(torch::Tensor A, torch::Tensor B, torch::Tensor C,
 void bf16_mm_small(torch::Tensor A, torch::Tensor B, torch::Tensor C,
                    torch::Tensor C2, torch::Tensor aux, torch::Tensor grad,
                    int64_t g1_off, int64_t g2_off, int64_t srow, int64_t epi,
-                   int64_t m_real, int64_t n_real, int64_t ldc);
+                   int64_t m_real, int64_t n_real, int64_t ldc,
+                   torch::Tensor bias);
 
 void bf16_transpose(torch::Tensor in, torch::Tensor out, torch::Tensor sums,
                     int64_t sums_off, int64_t R, int64_t C, int64_t ld_in,

@@ -1,0 +1,225 @@
+"""Hand-written bf16 MFMA training path for the wide MLP config
+(BASELINE #5: obs=4096, 4x4096 tanh hidden, A=256, bf16).
+
+Routes every GEMM of the wide config — rollout forwards, update-step
+forwards, dgrad chain, dW — through the in-tree gfx950 kernels
+(ops/hip/bf16_gemm.hip) instead of torch autocast/rocBLAS, with the
+epilogues (bias+tanh, dtanh, bias column-sums) fused into the producing
+kernels.  Loss math stays the reference formulation: the per-row loss
+gradients come from gauss_gh_wide (exact PPO.py:29-40 analytics, bf16
+output feeding the dgrad GEMMs directly).
+
+Layout notes (see bf16_gemm.hip header): torch's [out][in] weight layout
+IS the MFMA B-operand layout, so forwards need no weight transposes;
+dgrad consumes W^T bf16 copies; dW consumes transposed deltas and
+activations produced by the fused transpose+colsum kernel (the bias
+gradients ride along with the delta transposes).
+"""
+
+from __future__ import annotations
+
+from typing import List, Optional
+
+import torch
+
+
+def _round_up(x: int, m: int) -> int:
+    return (x + m - 1) // m * m
+
+
+class WideBF16Path:
+    """Per-engine helper owning the bf16 buffers and the update pipeline."""
+
+    def __init__(self, engine):
+        from . import require_hip_ext
+
+        self.ext = require_hip_ext()
+        self.eng = engine
+        self.cfg = engine.cfg
+        self.dev = engine.device
+        self.D = engine.obs_space.shape[0]
+        self.H: List[int] = list(self.cfg.HIDDEN_SIZES)
+        self.A = engine.act_space.shape[0]
+        self.P = 2 * self.A
+        self.KP = _round_up(self.P + 1, 64)  # gh row padding (mm256 K)
+        self._weights_dirty = True
+        self._f = torch.empty(0, device=self.dev)
+        self._b = torch.empty(0, device=self.dev, dtype=torch.bfloat16)
+        # bf16 weight copies (refreshed when params change)
+        self.W_bf = [torch.empty(h_out, h_in, device=self.dev,
+                                 dtype=torch.bfloat16)
+                     for h_in, h_out in zip([self.D] + self.H[:-1], self.H)]
+        self.Wt_bf = [torch.empty(w.shape[1], w.shape[0], device=self.dev,
+                                  dtype=torch.bfloat16) for w in self.W_bf]
+        HL = self.H[-1]
+        self.whcat = torch.empty(self.P + 1, HL, device=self.dev,
+                                 dtype=torch.bfloat16)
+        self.whcatT = torch.zeros(HL, self.KP, device=self.dev,
+                                  dtype=torch.bfloat16)  # zero K padding
+        self.bhcat = torch.empty(self.P + 1, device=self.dev)
+        self._fwd_bufs = {}
+        self._upd = None
+
+    # -- weights -------------------------------------------------------
+    def mark_dirty(self) -> None:
+        self._weights_dirty = True
+
+    @torch.no_grad()
+    def refresh_weights(self) -> None:
+        pi = self.eng.pi
+        for l, lay in enumerate(pi.hidden):
+            self.W_bf[l].copy_(lay.weight.detach())
+            self.ext.bf16_transpose(
+                self.W_bf[l], self.Wt_bf[l], self._f, 0,
+                self.W_bf[l].shape[0], self.W_bf[l].shape[1],
+                self.W_bf[l].shape[1], self.W_bf[l].shape[0])
+        self.whcat[:self.P].copy_(pi.pi.weight.detach())
+        self.whcat[self.P:].copy_(pi.vf.weight.detach())
+        self.ext.bf16_transpose(self.whcat, self.whcatT, self._f, 0,
+                                self.P + 1, self.H[-1], self.H[-1], self.KP)
+        self.bhcat[:self.P].copy_(pi.pi.bias.detach())
+        self.bhcat[self.P:].copy_(pi.vf.bias.detach())
+        self._weights_dirty = False
+
+    # -- forward (rollout acting / bootstrap) --------------------------
+    def _fwd_scratch(self, M: int):
+        key = M
+        bufs = self._fwd_bufs.get(key)
+        if bufs is None:
+            bufs = {
+                "x": torch.empty(M, self.D, device=self.dev,
+                                 dtype=torch.bfloat16),
+                "h": [torch.empty(M, h, device=self.dev,
+                                  dtype=torch.bfloat16) for h in self.H],
+                "pd": torch.empty(M, self.P, device=self.dev,
+                                  dtype=torch.bfloat16),
+                "v": torch.empty(M, device=self.dev, dtype=torch.bfloat16),
+            }
+            self._fwd_bufs[key] = bufs
+        return bufs
+
+    @torch.no_grad()
+    def forward(self, obs: torch.Tensor):
+        """(v, pdflat) in f32, every GEMM on the hand bf16 kernels."""
+        if self._weights_dirty:
+            self.refresh_weights()
+        M = obs.shape[0]
+        s = self._fwd_scratch(M)
+        s["x"].copy_(obs)
+        x = s["x"]
+        pi = self.eng.pi
+        for l in range(len(self.H)):
+            self.ext.bf16_mm256(x, self.W_bf[l], s["h"][l], 1,
+                                pi.hidden[l].bias.detach(), self._b,
+                                self._f, 0)
+            x = s["h"][l]
+        self.ext.bf16_mm_small(x, self.whcat, s["pd"], s["v"], self._b,
+                               self._f, 0, 0, 0, 4, M, self.P + 1, self.P,
+                               self.bhcat)
+        return s["v"].float(), s["pd"].float()
+
+    # -- update --------------------------------------------------------
+    def _upd_bufs(self, B: int):
+        if self._upd is not None and self._upd["B"] == B:
+            return self._upd
+        dev = self.dev
+        bf = torch.bfloat16
+        u = {
+            "B": B,
+            "x": torch.empty(B, self.D, device=dev, dtype=bf),
+            "xT": torch.empty(self.D, B, device=dev, dtype=bf),
+            "h": [torch.empty(B, h, device=dev, dtype=bf) for h in self.H],
+            "hT": torch.empty(self.H[-1], B, device=dev, dtype=bf),
+            "d0": torch.empty(B, max(self.H), device=dev, dtype=bf),
+            "d1": torch.empty(B, max(self.H), device=dev, dtype=bf),
+            "dT": torch.empty(max(self.H), B, device=dev, dtype=bf),
+            "pd": torch.empty(B, self.P, device=dev, dtype=bf),
+            "v": torch.empty(B, device=dev, dtype=bf),
+            "gh": torch.zeros(B, self.KP, device=dev, dtype=bf),
+            "ghT": torch.empty(self.KP, B, device=dev, dtype=bf),
+            "bias_tmp": torch.zeros(self.KP, device=dev),
+        }
+        self._upd = u
+        return u
+
+    @torch.no_grad()
+    def update(self, batch, l_mul: float) -> None:
+        """UPDATE_STEPS repeated full-batch steps (Chief.py:64), every
+        GEMM hand-written: fwd (tanh+bias fused) -> gh -> dgrad chain
+        (dtanh fused) -> transposes (+bias colsums fused) -> dW into the
+        f32 flat grad -> all-reduce -> fused Adam."""
+        eng, ext, cfg = self.eng, self.ext, self.cfg
+        B = batch.states.shape[0]
+        u = self._upd_bufs(B)
+        clip = cfg.CLIP_PARAM * l_mul
+        nH = len(self.H)
+        flat = eng.flat_pi
+        offsets = [sl.start for sl in flat.slices]
+        w_off = [offsets[2 * l] for l in range(nH)]
+        b_off = [offsets[2 * l + 1] for l in range(nH)]
+        off_wv, off_bv = offsets[2 * nH], offsets[2 * nH + 1]
+        off_wp, off_bp = offsets[2 * nH + 2], offsets[2 * nH + 3]
+        grad = flat.flat_grad
+
+        # per-round inputs (constant across the UPDATE_STEPS epochs)
+        u["x"].copy_(batch.states)
+        ext.bf16_transpose(u["x"], u["xT"], self._f, 0, B, self.D, self.D, B)
+
+        for _ in range(cfg.UPDATE_STEPS):
+            self.refresh_weights()
+            # forward
+            x = u["x"]
+            for l in range(nH):
+                ext.bf16_mm256(x, self.W_bf[l], u["h"][l], 1,
+                               eng.pi.hidden[l].bias.detach(), self._b,
+                               self._f, 0)
+                x = u["h"][l]
+            ext.bf16_mm_small(x, self.whcat, u["pd"], u["v"], self._b,
+                              self._f, 0, 0, 0, 4, B, self.P + 1, self.P,
+                              self.bhcat)
+            # loss gradients -> gh [B][P+1] (bf16, K-padded)
+            ext.gauss_gh_wide(u["pd"], batch.oldflat, u["v"], batch.oldv,
+                              batch.actions, batch.adv, batch.etr, u["gh"],
+                              self._f, clip, cfg.ENTCOEFF, cfg.VCOEFF)
+            flat.zero_grad()
+            # heads backward: dz[nH-1] = (gh @ Whcat) * dtanh(h[-1])
+            d_cur = u["d0"].narrow(1, 0, self.H[-1]).view(B, self.H[-1])
+            ext.bf16_mm256(u["gh"], self.whcatT, d_cur, 2, self._f,
+                           u["h"][nH - 1], self._f, 0)
+            # heads dW/db: ghT (+ head bias colsums), h[-1]T
+            u["bias_tmp"].zero_()
+            ext.bf16_transpose(u["gh"], u["ghT"], u["bias_tmp"], 0,
+                               B, self.KP, self.KP, B)
+            grad[off_bp:off_bp + self.P].copy_(u["bias_tmp"][:self.P])
+            grad[off_bv:off_bv + 1].copy_(u["bias_tmp"][self.P:self.P + 1])
+            ext.bf16_transpose(u["h"][nH - 1], u["hT"], self._f, 0,
+                               B, self.H[-1], self.H[-1], B)
+            ext.bf16_mm_small(u["ghT"], u["hT"], self._b, self._b, self._b,
+                              grad, off_wp, off_wv, self.P, 5,
+                              self.P + 1, self.H[-1], 0, self._f)
+            # hidden chain
+            for l in range(nH - 1, -1, -1):
+                # dW[l] = dz[l]^T @ act[l-1]; db[l] rides the transpose
+                hl = self.H[l]
+                dT = u["dT"].narrow(0, 0, hl).view(hl, B)
+                ext.bf16_transpose(d_cur, dT, grad, b_off[l], B, hl, hl, B)
+                if l == 0:
+                    actT = u["xT"]
+                else:
+                    hprev = self.H[l - 1]
+                    actT = u["hT"].narrow(0, 0, hprev).view(hprev, B)
+                    ext.bf16_transpose(u["h"][l - 1], actT, self._f, 0,
+                                       B, hprev, hprev, B)
+                ext.bf16_mm256(dT, actT, self._b, 3, self._f, self._b,
+                               grad, w_off[l])
+                if l > 0:
+                    # dz[l-1] = (dz[l] @ W[l]) * dtanh(h[l-1])
+                    hprev = self.H[l - 1]
+                    d_nxt = (u["d1"] if d_cur.data_ptr() == u["d0"].data_ptr()
+                             else u["d0"]).narrow(1, 0, hprev).view(B, hprev)
+                    ext.bf16_mm256(d_cur, self.Wt_bf[l], d_nxt, 2, self._f,
+                                   u["h"][l - 1], self._f, 0)
+                    d_cur = d_nxt
+            eng.comm.allreduce_mean_(grad)
+            eng.optimizer.step()
+            self._weights_dirty = True

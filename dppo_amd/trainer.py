@@ -182,6 +182,48 @@ class DPPOEngine:
     def sync_oldpi(self) -> None:
         """oldpi <- pi, one flat copy (sync_pis, PPO.py:47)."""
         self.flat_old.flat_param.copy_(self.flat_pi.flat_param)
+        if getattr(self, "_wide_path", None) is not None:
+            # params may have changed since the last rollout (update,
+            # drift-guard broadcast, checkpoint restore) — refresh the
+            # bf16 weight copies once per round
+            self._wide_path.mark_dirty()
+
+    # -- wide bf16 path (BASELINE #5) ----------------------------------
+    def _can_wide_bf16(self) -> bool:
+        """Eligibility for the hand-written bf16 MFMA path (ops/wide.py +
+        ops/hip/bf16_gemm.hip): wide tanh MLP, Box policy, bf16 compute,
+        dims in 256-multiples (the 256^2-tile GEMM's shape contract)."""
+        from .ops import use_hip
+
+        c = self.cfg
+        if self._act_kind != "box" or c.DTYPE != "bfloat16":
+            return False
+        if c.ACTIVATION != "tanh" or c.MINIBATCH_SIZE != 0:
+            return False
+        if not use_hip(self.device, c.USE_HIP_KERNELS):
+            return False
+        H = c.HIDDEN_SIZES
+        if self.obs_space.shape[0] % 256 or any(h % 256 for h in H):
+            return False
+        if len(set(H)) != 1 or c.NUM_ENVS % 256:
+            return False
+        return True
+
+    def _wide(self):
+        if getattr(self, "_wide_path", None) is None:
+            from .ops.wide import WideBF16Path
+
+            self._wide_path = WideBF16Path(self)
+        return self._wide_path
+
+    def _policy_forward(self, obs: torch.Tensor):
+        """(v, pdflat) f32 for a batch of states — hand bf16 kernels on
+        the wide config, autocast/eager otherwise."""
+        if self._can_wide_bf16() and obs.shape[0] % 256 == 0:
+            return self._wide().forward(obs)
+        with self._amp():
+            v, pdflat = self.pi(obs)
+        return v.float(), pdflat.float()
 
     def exploration_rate(self) -> float:
         """Linear epsilon anneal MAX->MIN over AC_EXP_PERCENTAGE*EPOCH_MAX
@@ -242,9 +284,7 @@ class DPPOEngine:
         SURVEY.md §7 'hard parts')."""
         from .ops import use_hip
 
-        with self._amp():
-            v, pdflat = self.pi(obs)
-        v, pdflat = v.float(), pdflat.float()
+        v, pdflat = self._policy_forward(obs)
         pd = self.pi.pdtype.pdfromflat(pdflat)
         if (self._discrete and pdflat.shape[-1] <= 64
                 and use_hip(self.device, self.cfg.USE_HIP_KERNELS)):
@@ -663,8 +703,8 @@ class DPPOEngine:
             self.epr *= 1.0 - donef
         self.obs = obs
 
-        with torch.no_grad(), self._amp():
-            boot_v, _ = self.pi(obs)
+        with torch.no_grad():
+            boot_v, _ = self._policy_forward(obs)
         adv, etr = gae_advantages(
             rewards, values, dones, boot_v.float(),
             c.GAMMA, c.LAM, whiten=True, eps=c.ADV_EPS,
@@ -788,6 +828,10 @@ class DPPOEngine:
         loss values are materialized here."""
         for g in self.optimizer.param_groups:
             g["lr"] = self.cfg.LEARNING_RATE * l_mul
+        if self._can_wide_bf16():
+            # hand bf16 MFMA path (BASELINE #5): no rocBLAS/autocast GEMMs
+            self._wide().update(batch, l_mul)
+            return
         if self.cfg.MINIBATCH_SIZE > 0:
             self._update_minibatched(batch, l_mul)
             return

@@ -98,7 +98,7 @@ def test_mm_small_ragged(ext):
     M, N, K = 100, 513, 544
     A, B = _mm_case(M, N, K, seed=6, scale=0.3)
     C = torch.empty(M, N, device="cuda", dtype=torch.bfloat16)
-    ext.bf16_mm_small(A, B, C, _eb(), _eb(), _e(), 0, 0, 0, 0, M, N, N)
+    ext.bf16_mm_small(A, B, C, _eb(), _eb(), _e(), 0, 0, 0, 0, M, N, N, _e())
     ref = A.float() @ B.float().t()
     _close(C.float(), ref, K)
 
@@ -110,9 +110,10 @@ def test_mm_small_heads_split(ext):
     B_heads = B  # [P+1][K]
     pdflat = torch.empty(M, P, device="cuda", dtype=torch.bfloat16)
     v = torch.empty(M, device="cuda", dtype=torch.bfloat16)
+    bias = torch.randn(P + 1, device="cuda")
     ext.bf16_mm_small(A, B_heads, pdflat, v, _eb(), _e(), 0, 0, 0, 4,
-                      M, P + 1, P)
-    ref = A.float() @ B_heads.float().t()
+                      M, P + 1, P, bias)
+    ref = A.float() @ B_heads.float().t() + bias
     _close(pdflat.float(), ref[:, :P], K)
     _close(v.float(), ref[:, P], K)
 
@@ -124,7 +125,7 @@ def test_mm_small_grad_split(ext):
     g1_off, g2_off = 64, 64 + 512 * N
     grad = torch.zeros(g2_off + N + 8, device="cuda")
     ext.bf16_mm_small(A, B, _eb(), _eb(), _eb(), grad, g1_off, g2_off,
-                      512, 5, M, N, 0)
+                      512, 5, M, N, 0, _e())
     ref = A.float() @ B.float().t()
     _close(grad[g1_off:g1_off + 512 * N].view(512, N), ref[:512], K)
     _close(grad[g2_off:g2_off + N], ref[512], K)
