@@ -83,7 +83,9 @@ struct MM256Args {
   const float* bias;        // [N]     (epi 1)
   const unsigned short* aux;  // [M][N] (epi 2: dtanh factor source)
   float* grad;              // epi 3: f32 out at grad[m*N + n]
-  long M, N, K;
+  unsigned short* CT;       // optional transposed dual-write [N][ldt]
+  float* sums;              // optional [N] f32 += column sums (with CT)
+  long M, N, K, ldt;
   int nbn;                  // N / 256
   int epi;
 };
@@ -289,6 +291,8 @@ __global__ void bf16_mm256_kernel(MM256Args a) {
           v[j] *= 1.f - h * h;
         }
       }
+      if (a.CT != nullptr && a.epi != EPI_RAW)
+        *(f32x4*)(stripe + row * 64 + col0) = v;  // transformed write-back
       if (a.epi == EPI_GRAD) {
         *(f32x4*)(a.grad + gr * a.N + gc) = v;
       } else {
@@ -297,6 +301,28 @@ __global__ void bf16_mm256_kernel(MM256Args a) {
         for (int j = 0; j < 4; ++j) o[j] = f2bf(v[j]);
         *(ushort4*)(a.C + gr * a.N + gc) = o;
       }
+    }
+    if (a.CT != nullptr) {
+      // transposed dual-write: lane = column; its 16 transformed rows
+      // are one conflict-free b32 column walk of the stripe, stored as
+      // 32 contiguous bytes of CT[col][rb..rb+15].  Fused column sums
+      // (the bias gradient of a dgrad output) ride along.
+      float csum = 0.f;
+      ushort4 p[4];
+      #pragma unroll
+      for (int qq = 0; qq < 4; ++qq)
+        #pragma unroll
+        for (int rr = 0; rr < 4; ++rr) {
+          const float x = stripe[(qq * 4 + rr) * 64 + lane];
+          csum += x;
+          p[qq][rr] = f2bf(x);
+        }
+      unsigned short* ct = a.CT + (cb + lane) * a.ldt + rb;
+      *(ushort4*)(ct) = p[0];
+      *(ushort4*)(ct + 4) = p[1];
+      *(ushort4*)(ct + 8) = p[2];
+      *(ushort4*)(ct + 12) = p[3];
+      if (a.sums != nullptr) atomicAdd(&a.sums[cb + lane], csum);
     }
   }
 }
@@ -320,7 +346,8 @@ struct MMSmallArgs {
   float* g1;
   float* g2;
   long M, N, K, Mreal, Nreal, ldc;
-  int nbn, epi, srow;
+  int nbn, epi, srow, nsk;  // nsk: split-K slices (grid.y); epi-5 grads
+                            // become atomicAdd when nsk > 1
 };
 
 constexpr int SM_STRIDE = 80;  // padded row stride (bytes) of the LDS images
@@ -337,7 +364,10 @@ __global__ void bf16_mm_small_kernel(MMSmallArgs a) {
   const long c0 = (long)bn * 64;   // B rows (= C cols)
   const int row = tid >> 2;
   const int piece = tid & 3;
-  const int NC = (int)(a.K / 32);
+  const int NC_all = (int)(a.K / 32);
+  const int per_sk = (NC_all + a.nsk - 1) / a.nsk;
+  const int kc0 = (int)blockIdx.y * per_sk;
+  const int kc1 = min(kc0 + per_sk, NC_all);
 
   char* const bufA0 = smem;
   char* const bufB0 = smem + 64 * SM_STRIDE;
@@ -352,14 +382,14 @@ __global__ void bf16_mm_small_kernel(MMSmallArgs a) {
   };
 
   f32x4 acc[4] = {};
-  short8 ra = load_guarded(a.A, r0 + row, a.Mreal, piece * 8);
-  short8 rb = load_guarded(a.B, c0 + row, a.Nreal, piece * 8);
-  for (int kc = 0; kc < NC; ++kc) {
+  short8 ra = load_guarded(a.A, r0 + row, a.Mreal, (long)kc0 * 32 + piece * 8);
+  short8 rb = load_guarded(a.B, c0 + row, a.Nreal, (long)kc0 * 32 + piece * 8);
+  for (int kc = kc0; kc < kc1; ++kc) {
     char* wA = (kc & 1) ? bufA1 : bufA0;
     char* wB = (kc & 1) ? bufB1 : bufB0;
     *(short8*)(wA + row * SM_STRIDE + piece * 16) = ra;
     *(short8*)(wB + row * SM_STRIDE + piece * 16) = rb;
-    if (kc + 1 < NC) {
+    if (kc + 1 < kc1) {
       ra = load_guarded(a.A, r0 + row, a.Mreal, (long)(kc + 1) * 32 + piece * 8);
       rb = load_guarded(a.B, c0 + row, a.Nreal, (long)(kc + 1) * 32 + piece * 8);
     }
@@ -397,10 +427,16 @@ __global__ void bf16_mm_small_kernel(MMSmallArgs a) {
         else
           a.C2[m] = f2bf(v);
       } else if (a.epi == 5) {  // grad split-row
-        if (m < a.srow)
+        if (a.nsk > 1) {
+          if (m < a.srow)
+            atomicAdd(&a.g1[m * a.Nreal + n], v);
+          else if (m == a.srow)
+            atomicAdd(&a.g2[n], v);
+        } else if (m < a.srow) {
           a.g1[m * a.Nreal + n] = v;
-        else if (m == a.srow)
+        } else if (m == a.srow) {
           a.g2[n] = v;
+        }
       } else {
         a.C[m * a.ldc + n] = f2bf(v);
       }
@@ -560,7 +596,8 @@ static unsigned short* bf_ptr_mut(torch::Tensor& t) {
 
 void bf16_mm256(torch::Tensor A, torch::Tensor B, torch::Tensor C,
                 int64_t epi, torch::Tensor bias, torch::Tensor aux,
-                torch::Tensor grad, int64_t grad_off) {
+                torch::Tensor grad, int64_t grad_off, torch::Tensor CT,
+                int64_t ldt, torch::Tensor sums, int64_t sums_off) {
   TORCH_CHECK(A.is_cuda() && A.dtype() == torch::kBFloat16 && A.dim() == 2);
   TORCH_CHECK(B.dtype() == torch::kBFloat16 && B.dim() == 2);
   const long M = A.size(0), K = A.size(1), N = B.size(0);
@@ -592,6 +629,17 @@ void bf16_mm256(torch::Tensor A, torch::Tensor B, torch::Tensor C,
     TORCH_CHECK(aux.dtype() == torch::kBFloat16 && aux.numel() == M * N);
     a.aux = bf_ptr(aux);
   }
+  if (CT.numel() > 0) {
+    TORCH_CHECK(CT.dtype() == torch::kBFloat16 && CT.numel() >= M * N);
+    a.CT = bf_ptr_mut(CT);
+    a.ldt = ldt > 0 ? ldt : M;
+    TORCH_CHECK(a.ldt % 2 == 0);
+    if (sums.numel() > 0) {
+      TORCH_CHECK(sums.dtype() == torch::kFloat32 &&
+                  sums.numel() >= sums_off + N);
+      a.sums = sums.data_ptr<float>() + sums_off;
+    }
+  }
   hipStream_t stream = c10::hip::getCurrentHIPStream().stream();
   const long grid = (M / 256) * (N / 256);
   static const bool two_bar = []() {
@@ -612,6 +660,7 @@ void bf16_mm_small(torch::Tensor A, torch::Tensor B, torch::Tensor C,
                    int64_t m_real, int64_t n_real, int64_t ldc,
                    torch::Tensor bias) {
   TORCH_CHECK(A.is_cuda() && A.dtype() == torch::kBFloat16 && A.dim() == 2);
+  const bool splitk_ok = (epi == 5);
   const long K = A.size(1);
   TORCH_CHECK(B.size(1) == K && K % 32 == 0 && K % 8 == 0);
   MMSmallArgs a{};
@@ -641,8 +690,16 @@ void bf16_mm_small(torch::Tensor A, torch::Tensor B, torch::Tensor C,
   }
   hipStream_t stream = c10::hip::getCurrentHIPStream().stream();
   const long grid = (a.M / 64) * (a.N / 64);
-  hipLaunchKernelGGL(bf16_mm_small_kernel, dim3((unsigned)grid), dim3(256), 0,
-                     stream, a);
+  // split K across grid.y until the launch has ~4 blocks/CU (grad epilogue
+  // accumulates with atomics, so only epi 5 splits)
+  long nsk = 1;
+  if (splitk_ok) {
+    while (grid * nsk < 1024 && nsk < 32 && (a.K / 32) / (nsk * 2) >= 8)
+      nsk *= 2;
+  }
+  a.nsk = (int)nsk;
+  hipLaunchKernelGGL(bf16_mm_small_kernel, dim3((unsigned)grid, (unsigned)nsk),
+                     dim3(256), 0, stream, a);
 }
 
 void bf16_transpose(torch::Tensor in, torch::Tensor out, torch::Tensor sums,

@@ -44,7 +44,8 @@ torch::Tensor cat_sample(torch::Tensor logits, int64_t seed, int64_t ctr);
 
 void bf16_mm256(torch::Tensor A, torch::Tensor B, torch::Tensor C,
                 int64_t epi, torch::Tensor bias, torch::Tensor aux,
-                torch::Tensor grad, int64_t grad_off);
+                torch::Tensor grad, int64_t grad_off, torch::Tensor CT,
+                int64_t ldt, torch::Tensor sums, int64_t sums_off);
 
 void bf16_mm_small(torch::Tensor A, torch::Tensor B, torch::Tensor C,
                    torch::Tensor C2, torch::Tensor aux, torch::Tensor grad,

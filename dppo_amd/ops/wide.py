@@ -111,7 +111,7 @@ class WideBF16Path:
         for l in range(len(self.H)):
             self.ext.bf16_mm256(x, self.W_bf[l], s["h"][l], 1,
                                 pi.hidden[l].bias.detach(), self._b,
-                                self._f, 0)
+                                self._f, 0, self._b, 0, self._f, 0)
             x = s["h"][l]
         self.ext.bf16_mm_small(x, self.whcat, s["pd"], s["v"], self._b,
                                self._f, 0, 0, 0, 4, M, self.P + 1, self.P,
@@ -129,10 +129,12 @@ class WideBF16Path:
             "x": torch.empty(B, self.D, device=dev, dtype=bf),
             "xT": torch.empty(self.D, B, device=dev, dtype=bf),
             "h": [torch.empty(B, h, device=dev, dtype=bf) for h in self.H],
-            "hT": torch.empty(self.H[-1], B, device=dev, dtype=bf),
+            # transposed activations, dual-written by the forward GEMMs
+            "hT": [torch.empty(h, B, device=dev, dtype=bf) for h in self.H],
             "d0": torch.empty(B, max(self.H), device=dev, dtype=bf),
             "d1": torch.empty(B, max(self.H), device=dev, dtype=bf),
-            "dT": torch.empty(max(self.H), B, device=dev, dtype=bf),
+            "dT0": torch.empty(max(self.H), B, device=dev, dtype=bf),
+            "dT1": torch.empty(max(self.H), B, device=dev, dtype=bf),
             "pd": torch.empty(B, self.P, device=dev, dtype=bf),
             "v": torch.empty(B, device=dev, dtype=bf),
             "gh": torch.zeros(B, self.KP, device=dev, dtype=bf),
@@ -167,12 +169,13 @@ class WideBF16Path:
 
         for _ in range(cfg.UPDATE_STEPS):
             self.refresh_weights()
-            # forward
+            # forward; each hidden GEMM dual-writes h AND h^T from its
+            # epilogue (the dW operands), killing the transpose passes
             x = u["x"]
             for l in range(nH):
                 ext.bf16_mm256(x, self.W_bf[l], u["h"][l], 1,
                                eng.pi.hidden[l].bias.detach(), self._b,
-                               self._f, 0)
+                               self._f, 0, u["hT"][l], B, self._f, 0)
                 x = u["h"][l]
             ext.bf16_mm_small(x, self.whcat, u["pd"], u["v"], self._b,
                               self._f, 0, 0, 0, 4, B, self.P + 1, self.P,
@@ -182,44 +185,36 @@ class WideBF16Path:
                               batch.actions, batch.adv, batch.etr, u["gh"],
                               self._f, clip, cfg.ENTCOEFF, cfg.VCOEFF)
             flat.zero_grad()
-            # heads backward: dz[nH-1] = (gh @ Whcat) * dtanh(h[-1])
-            d_cur = u["d0"].narrow(1, 0, self.H[-1]).view(B, self.H[-1])
+            # heads backward: dz[nH-1] = (gh @ Whcat) * dtanh(h[-1]),
+            # dual-writing dz^T with its bias colsum -> db[nH-1]
+            d_cur = u["d0"]
+            dT_cur = u["dT0"]
             ext.bf16_mm256(u["gh"], self.whcatT, d_cur, 2, self._f,
-                           u["h"][nH - 1], self._f, 0)
-            # heads dW/db: ghT (+ head bias colsums), h[-1]T
+                           u["h"][nH - 1], self._f, 0,
+                           dT_cur, B, grad, b_off[nH - 1])
+            # heads dW/db: ghT (+ head bias colsums), split-K small GEMM
             u["bias_tmp"].zero_()
             ext.bf16_transpose(u["gh"], u["ghT"], u["bias_tmp"], 0,
                                B, self.KP, self.KP, B)
             grad[off_bp:off_bp + self.P].copy_(u["bias_tmp"][:self.P])
             grad[off_bv:off_bv + 1].copy_(u["bias_tmp"][self.P:self.P + 1])
-            ext.bf16_transpose(u["h"][nH - 1], u["hT"], self._f, 0,
-                               B, self.H[-1], self.H[-1], B)
-            ext.bf16_mm_small(u["ghT"], u["hT"], self._b, self._b, self._b,
-                              grad, off_wp, off_wv, self.P, 5,
+            ext.bf16_mm_small(u["ghT"], u["hT"][nH - 1], self._b, self._b,
+                              self._b, grad, off_wp, off_wv, self.P, 5,
                               self.P + 1, self.H[-1], 0, self._f)
-            # hidden chain
+            # hidden chain (uniform H: buffers are exact-size views)
             for l in range(nH - 1, -1, -1):
-                # dW[l] = dz[l]^T @ act[l-1]; db[l] rides the transpose
-                hl = self.H[l]
-                dT = u["dT"].narrow(0, 0, hl).view(hl, B)
-                ext.bf16_transpose(d_cur, dT, grad, b_off[l], B, hl, hl, B)
-                if l == 0:
-                    actT = u["xT"]
-                else:
-                    hprev = self.H[l - 1]
-                    actT = u["hT"].narrow(0, 0, hprev).view(hprev, B)
-                    ext.bf16_transpose(u["h"][l - 1], actT, self._f, 0,
-                                       B, hprev, hprev, B)
-                ext.bf16_mm256(dT, actT, self._b, 3, self._f, self._b,
-                               grad, w_off[l])
+                actT = u["xT"] if l == 0 else u["hT"][l - 1]
+                ext.bf16_mm256(dT_cur, actT, self._b, 3, self._f, self._b,
+                               grad, w_off[l], self._b, 0, self._f, 0)
                 if l > 0:
-                    # dz[l-1] = (dz[l] @ W[l]) * dtanh(h[l-1])
-                    hprev = self.H[l - 1]
-                    d_nxt = (u["d1"] if d_cur.data_ptr() == u["d0"].data_ptr()
-                             else u["d0"]).narrow(1, 0, hprev).view(B, hprev)
+                    # dz[l-1] = (dz[l] @ W[l]) * dtanh(h[l-1]); dual-write
+                    # dz[l-1]^T and its colsum -> db[l-1]
+                    d_nxt = u["d1"] if d_cur is u["d0"] else u["d0"]
+                    dT_nxt = u["dT1"] if dT_cur is u["dT0"] else u["dT0"]
                     ext.bf16_mm256(d_cur, self.Wt_bf[l], d_nxt, 2, self._f,
-                                   u["h"][l - 1], self._f, 0)
-                    d_cur = d_nxt
+                                   u["h"][l - 1], self._f, 0,
+                                   dT_nxt, B, grad, b_off[l - 1])
+                    d_cur, dT_cur = d_nxt, dT_nxt
             eng.comm.allreduce_mean_(grad)
             eng.optimizer.step()
             self._weights_dirty = True

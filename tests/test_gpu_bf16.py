@@ -48,7 +48,7 @@ def test_mm256_raw(ext):
     M, N, K = 512, 512, 320
     A, B = _mm_case(M, N, K, seed=1, scale=0.3)
     C = torch.empty(M, N, device="cuda", dtype=torch.bfloat16)
-    ext.bf16_mm256(A, B, C, 0, _e(), _eb(), _e(), 0)
+    ext.bf16_mm256(A, B, C, 0, _e(), _eb(), _e(), 0, _eb(), 0, _e(), 0)
     ref = A.float() @ B.float().t()
     _close(C.float(), ref, K)
 
@@ -58,7 +58,7 @@ def test_mm256_large_k(ext):
     M, N, K = 256, 256, 8192
     A, B = _mm_case(M, N, K, seed=2, scale=0.1)
     C = torch.empty(M, N, device="cuda", dtype=torch.bfloat16)
-    ext.bf16_mm256(A, B, C, 0, _e(), _eb(), _e(), 0)
+    ext.bf16_mm256(A, B, C, 0, _e(), _eb(), _e(), 0, _eb(), 0, _e(), 0)
     ref = A.float() @ B.float().t()
     _close(C.float(), ref, K)
 
@@ -68,7 +68,7 @@ def test_mm256_tanh_bias(ext):
     A, B = _mm_case(M, N, K, seed=3, scale=0.3)
     bias = torch.randn(N, device="cuda")
     C = torch.empty(M, N, device="cuda", dtype=torch.bfloat16)
-    ext.bf16_mm256(A, B, C, 1, bias, _eb(), _e(), 0)
+    ext.bf16_mm256(A, B, C, 1, bias, _eb(), _e(), 0, _eb(), 0, _e(), 0)
     ref = torch.tanh(A.float() @ B.float().t() + bias)
     torch.testing.assert_close(C.float(), ref, atol=1e-2, rtol=2e-2)
 
@@ -78,7 +78,7 @@ def test_mm256_dtanh(ext):
     A, B = _mm_case(M, N, K, seed=4, scale=0.3)
     h = torch.tanh(torch.randn(M, N, device="cuda")).bfloat16()
     C = torch.empty(M, N, device="cuda", dtype=torch.bfloat16)
-    ext.bf16_mm256(A, B, C, 2, _e(), h, _e(), 0)
+    ext.bf16_mm256(A, B, C, 2, _e(), h, _e(), 0, _eb(), 0, _e(), 0)
     ref = (A.float() @ B.float().t()) * (1.0 - h.float() ** 2)
     _close(C.float(), ref, K)
 
@@ -88,10 +88,27 @@ def test_mm256_grad_accum(ext):
     A, B = _mm_case(M, N, K, seed=5, scale=0.2)
     off = 128
     grad = torch.zeros(off + M * N + 16, device="cuda")
-    ext.bf16_mm256(A, B, _eb(), 3, _e(), _eb(), grad, off)
+    ext.bf16_mm256(A, B, _eb(), 3, _e(), _eb(), grad, off, _eb(), 0, _e(), 0)
     ref = A.float() @ B.float().t()
     _close(grad[off:off + M * N].view(M, N), ref, K)
     assert grad[:off].abs().sum() == 0
+
+
+def test_mm256_dual_write_transpose_colsum(ext):
+    """The epilogue's transposed dual-write + fused column sums (the dW
+    operand and bias-grad producers) match the straight output."""
+    M, N, K = 512, 256, 128
+    A, B = _mm_case(M, N, K, seed=12, scale=0.3)
+    h = torch.tanh(torch.randn(M, N, device="cuda")).bfloat16()
+    C = torch.empty(M, N, device="cuda", dtype=torch.bfloat16)
+    ldt = M + 64
+    CT = torch.zeros(N, ldt, device="cuda", dtype=torch.bfloat16)
+    sums = torch.zeros(N, device="cuda")
+    ext.bf16_mm256(A, B, C, 2, _e(), h, _e(), 0, CT, ldt, sums, 0)
+    ref = (A.float() @ B.float().t()) * (1.0 - h.float() ** 2)
+    _close(C.float(), ref, K)
+    torch.testing.assert_close(CT[:, :M].float(), C.float().t())
+    torch.testing.assert_close(sums, ref.sum(dim=0), atol=0.3, rtol=1e-2)
 
 
 def test_mm_small_ragged(ext):

@@ -43,7 +43,7 @@ def main():
         C = torch.empty(M, N, device="cuda", dtype=torch.bfloat16)
         fl = 2.0 * M * N * K
 
-        t_mm = bench(lambda: ext.bf16_mm256(A, B, C, 0, e, eb, e, 0),
+        t_mm = bench(lambda: ext.bf16_mm256(A, B, C, 0, e, eb, e, 0, eb, 0, e, 0),
                      args.iters)
         Bt = B.float().t().bfloat16().t().contiguous()  # same layout
         out = torch.empty(M, N, device="cuda", dtype=torch.bfloat16)
@@ -52,7 +52,7 @@ def main():
               f" | rocBLAS {t_roc*1e3:8.3f} ms = {fl/t_roc/1e12:7.1f} TF/s")
 
         t_tanh = bench(lambda: ext.bf16_mm256(
-            A, B, C, 1, torch.zeros(N, device="cuda"), eb, e, 0), args.iters)
+            A, B, C, 1, torch.zeros(N, device="cuda"), eb, e, 0, eb, 0, e, 0), args.iters)
         print(f"  +tanh-bias epilogue: {t_tanh*1e3:8.3f} ms = "
               f"{fl/t_tanh/1e12:7.1f} TF/s")
         del A, B, C, out, Bt

@@ -12,6 +12,6 @@ A = (torch.randn(M, K, device="cuda") * 0.3).bfloat16()
 B = (torch.randn(N, K, device="cuda") * 0.3).bfloat16()
 C = torch.empty(M, N, device="cuda", dtype=torch.bfloat16)
 for _ in range(int(sys.argv[1]) if len(sys.argv) > 1 else 5):
-    ext.bf16_mm256(A, B, C, 0, e, eb, e, 0)
+    ext.bf16_mm256(A, B, C, 0, e, eb, e, 0, eb, 0, e, 0)
 torch.cuda.synchronize()
 print("done")
