@@ -116,9 +116,10 @@ DEV_INLINE short8 frag_read(const char* img, int row, int colb) {
   return *(const short8*)(img + byte);
 }
 
-template <bool TWO_BARRIERS>
+template <int NPH, bool TWO_BARRIERS>
 __launch_bounds__(512, 2)
 __global__ void bf16_mm256_kernel(MM256Args a) {
+  constexpr int FMPP = 8 / NPH;  // fm-blocks computed per phase
   __shared__ __attribute__((aligned(16))) char smem[LDS_BYTES];
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -189,21 +190,22 @@ __global__ void bf16_mm256_kernel(MM256Args a) {
                       ks * 64 + ((lane >> 4) & 3) * 16);
 
     #pragma unroll
-    for (int ph = 0; ph < 4; ++ph) {
-      // A fragments for this phase's two row-blocks
-      short8 afr[2][2];
+    for (int ph = 0; ph < NPH; ++ph) {
+      // A fragments for this phase's FMPP row-blocks
+      short8 afr[FMPP][2];
       #pragma unroll
-      for (int i = 0; i < 2; ++i)
+      for (int i = 0; i < FMPP; ++i)
         #pragma unroll
         for (int ks = 0; ks < 2; ++ks)
           afr[i][ks] =
-              frag_read(iA, wr * 128 + (2 * ph + i) * 16 + (lane & 15),
+              frag_read(iA, wr * 128 + (FMPP * ph + i) * 16 + (lane & 15),
                         ks * 64 + ((lane >> 4) & 3) * 16);
-      // Staged prefetch — every piece gets >= 3 phases of flight before
-      // its consumer's seam wait (HBM latency ~900 cyc ~ 3 phases):
-      //   ph0: BOTH A halves of t+1 (their buffer is idle during t);
-      //   ph1/ph2: B halves of t+2 (B(t)'s buffer is free after ph0's
-      //   B-fragment reads, sealed by the ph0-end barrier).
+      // Staged prefetch — every piece gets >= 1 full phase of flight
+      // before its consumer's seam wait:
+      //   first phase: BOTH A halves of t+1 (their buffer is idle
+      //   during t); later phase(s): B halves of t+2 (B(t)'s buffer is
+      //   free after the first phase's B-fragment reads, sealed by the
+      //   phase-end barrier).
       if (ph == 0) {
         if (t + 1 < NT) {
           char* img = ((t + 1) & 1) ? imgA1 : imgA0;
@@ -217,6 +219,10 @@ __global__ void bf16_mm256_kernel(MM256Args a) {
           char* img = (t & 1) ? imgB1 : imgB0;
           stage_piece(sB00, (long)(t + 2) * BK, img, 0, 0, wave);
           stage_piece(sB01, (long)(t + 2) * BK, img, 0, 1, wave);
+          if (NPH == 2) {
+            stage_piece(sB10, (long)(t + 2) * BK, img, 1, 0, wave);
+            stage_piece(sB11, (long)(t + 2) * BK, img, 1, 1, wave);
+          }
         }
       } else if (ph == 2) {
         if (t + 2 < NT) {
@@ -234,17 +240,17 @@ __global__ void bf16_mm256_kernel(MM256Args a) {
       if (TWO_BARRIERS) __builtin_amdgcn_s_barrier();
       __builtin_amdgcn_s_setprio(1);
       #pragma unroll
-      for (int i = 0; i < 2; ++i) {
+      for (int i = 0; i < FMPP; ++i) {
         #pragma unroll
         for (int fn = 0; fn < 4; ++fn) {
-          acc[2 * ph + i][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              afr[i][0], bfr[fn][0], acc[2 * ph + i][fn], 0, 0, 0);
-          acc[2 * ph + i][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              afr[i][1], bfr[fn][1], acc[2 * ph + i][fn], 0, 0, 0);
+          acc[FMPP * ph + i][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              afr[i][0], bfr[fn][0], acc[FMPP * ph + i][fn], 0, 0, 0);
+          acc[FMPP * ph + i][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              afr[i][1], bfr[fn][1], acc[FMPP * ph + i][fn], 0, 0, 0);
         }
       }
       __builtin_amdgcn_s_setprio(0);
-      if (ph == 3) {
+      if (ph == NPH - 1) {
         // seam: tile t+1's halves must be LANDED in every wave before
         // anyone reads them after this barrier.  Steady state leaves the
         // two B halves of t+2 in flight (4 glds); at the tail drain all.
@@ -646,11 +652,18 @@ void bf16_mm256(torch::Tensor A, torch::Tensor B, torch::Tensor C,
     const char* e = getenv("DPPO_MM256_2BAR");
     return e != nullptr && e[0] == '1';
   }();
+  static const int nph = []() {
+    const char* e = getenv("DPPO_MM256_PH");
+    return e ? atoi(e) : 2;  // 2 phases/tile measured fastest
+  }();
   if (two_bar)
-    hipLaunchKernelGGL(bf16_mm256_kernel<true>, dim3((unsigned)grid),
+    hipLaunchKernelGGL((bf16_mm256_kernel<4, true>), dim3((unsigned)grid),
+                       dim3(512), 0, stream, a);
+  else if (nph == 2)
+    hipLaunchKernelGGL((bf16_mm256_kernel<2, false>), dim3((unsigned)grid),
                        dim3(512), 0, stream, a);
   else
-    hipLaunchKernelGGL(bf16_mm256_kernel<false>, dim3((unsigned)grid),
+    hipLaunchKernelGGL((bf16_mm256_kernel<4, false>), dim3((unsigned)grid),
                        dim3(512), 0, stream, a);
 }
 

@@ -140,8 +140,17 @@ class WideBF16Path:
             "gh": torch.zeros(B, self.KP, device=dev, dtype=bf),
             "ghT": torch.empty(self.KP, B, device=dev, dtype=bf),
             "bias_tmp": torch.zeros(self.KP, device=dev),
+            # persistent input copies: the captured update graph needs
+            # stable addresses (the eager rollout reallocates per round)
+            "oldflat": torch.empty(B, self.P, device=dev),
+            "oldv": torch.empty(B, device=dev),
+            "actions": torch.empty(B, self.A, device=dev),
+            "adv": torch.empty(B, device=dev),
+            "etr": torch.empty(B, device=dev),
+            "clip_dev": torch.zeros(1, device=dev),
         }
         self._upd = u
+        self._graph = None
         return u
 
     @torch.no_grad()
@@ -163,35 +172,109 @@ class WideBF16Path:
         off_wp, off_bp = offsets[2 * nH + 2], offsets[2 * nH + 3]
         grad = flat.flat_grad
 
-        # per-round inputs (constant across the UPDATE_STEPS epochs)
+        # per-round inputs (constant across the UPDATE_STEPS epochs),
+        # copied into persistent buffers so the captured graph sees them
         u["x"].copy_(batch.states)
         ext.bf16_transpose(u["x"], u["xT"], self._f, 0, B, self.D, self.D, B)
+        u["oldflat"].copy_(batch.oldflat)
+        u["oldv"].copy_(batch.oldv)
+        u["actions"].copy_(batch.actions)
+        u["adv"].copy_(batch.adv)
+        u["etr"].copy_(batch.etr)
+        u["clip_dev"].fill_(clip)
+        eng.optimizer.lr_dev.fill_(float(eng.optimizer.param_groups[0]["lr"]))
 
+        import os as _os
+        graph_ok = (cfg.USE_GRAPHS
+                    and (not eng.comm.distributed
+                         or _os.environ.get("DPPO_GRAPH_DIST") == "1")
+                    and not getattr(self, "_graph_failed", False))
+        if graph_ok:
+            if self._graph is None:
+                try:
+                    self._capture(u)
+                except Exception as exc:  # noqa: BLE001
+                    eng._warn_once(
+                        "wide_graph",
+                        f"wide-update hipGraph capture failed ({exc!r}); "
+                        "running uncaptured")
+                    self._graph_failed = True
+                    self._graph = None
+            if self._graph is not None:
+                self._graph.replay()
+                self._weights_dirty = True
+                return
+        self._update_body(u)
+
+    def _capture(self, u) -> None:
+        """Capture the UPDATE_STEPS pipeline once (device-state Adam,
+        clip from clip_dev; warmup executes real steps so param/optimizer
+        state is snapshot/restored around it)."""
+        eng = self.eng
+        opt = eng.optimizer
+        snap = (eng.flat_pi.flat_param.detach().clone(), opt.exp_avg.clone(),
+                opt.exp_avg_sq.clone(), opt.step_dev.clone())
+        try:
+            side = torch.cuda.Stream()
+            side.wait_stream(torch.cuda.current_stream())
+            with torch.cuda.stream(side):
+                self._update_body(u)
+            torch.cuda.current_stream().wait_stream(side)
+            torch.cuda.synchronize()
+        finally:
+            with torch.no_grad():
+                eng.flat_pi.flat_param.copy_(snap[0])
+                opt.exp_avg.copy_(snap[1])
+                opt.exp_avg_sq.copy_(snap[2])
+                opt.step_dev.copy_(snap[3])
+        g = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(g):
+            self._update_body(u)
+        self._graph = g
+
+    @torch.no_grad()
+    def _update_body(self, u) -> None:
+        eng, ext, cfg = self.eng, self.ext, self.cfg
+        B = u["B"]
+        nH = len(self.H)
+        flat = eng.flat_pi
+        offsets = [sl.start for sl in flat.slices]
+        w_off = [offsets[2 * l] for l in range(nH)]
+        b_off = [offsets[2 * l + 1] for l in range(nH)]
+        off_wv, off_bv = offsets[2 * nH], offsets[2 * nH + 1]
+        off_wp, off_bp = offsets[2 * nH + 2], offsets[2 * nH + 3]
+        grad = flat.flat_grad
         for _ in range(cfg.UPDATE_STEPS):
             self.refresh_weights()
-            # forward; each hidden GEMM dual-writes h AND h^T from its
-            # epilogue (the dW operands), killing the transpose passes
+            # forward (h^T for dW comes from the standalone transpose
+            # kernel: the epilogue dual-write measured SLOWER than the
+            # 4.4 TB/s transpose kernel — its 32-B-contiguous scattered
+            # stores cost ~0.4 ms/call vs 0.25 for the kernel)
             x = u["x"]
             for l in range(nH):
                 ext.bf16_mm256(x, self.W_bf[l], u["h"][l], 1,
                                eng.pi.hidden[l].bias.detach(), self._b,
-                               self._f, 0, u["hT"][l], B, self._f, 0)
+                               self._f, 0, self._b, 0, self._f, 0)
+                ext.bf16_transpose(u["h"][l], u["hT"][l], self._f, 0,
+                                   B, self.H[l], self.H[l], B)
                 x = u["h"][l]
             ext.bf16_mm_small(x, self.whcat, u["pd"], u["v"], self._b,
                               self._f, 0, 0, 0, 4, B, self.P + 1, self.P,
                               self.bhcat)
             # loss gradients -> gh [B][P+1] (bf16, K-padded)
-            ext.gauss_gh_wide(u["pd"], batch.oldflat, u["v"], batch.oldv,
-                              batch.actions, batch.adv, batch.etr, u["gh"],
-                              self._f, clip, cfg.ENTCOEFF, cfg.VCOEFF)
+            ext.gauss_gh_wide(u["pd"], u["oldflat"], u["v"], u["oldv"],
+                              u["actions"], u["adv"], u["etr"], u["gh"],
+                              u["clip_dev"], 0.0, cfg.ENTCOEFF, cfg.VCOEFF)
             flat.zero_grad()
-            # heads backward: dz[nH-1] = (gh @ Whcat) * dtanh(h[-1]),
-            # dual-writing dz^T with its bias colsum -> db[nH-1]
+            # heads backward: dz[nH-1] = (gh @ Whcat) * dtanh(h[-1]);
+            # its transpose (+ bias colsum -> db[nH-1]) via the kernel
             d_cur = u["d0"]
             dT_cur = u["dT0"]
             ext.bf16_mm256(u["gh"], self.whcatT, d_cur, 2, self._f,
                            u["h"][nH - 1], self._f, 0,
-                           dT_cur, B, grad, b_off[nH - 1])
+                           self._b, 0, self._f, 0)
+            ext.bf16_transpose(d_cur, dT_cur, grad, b_off[nH - 1],
+                               B, self.H[nH - 1], self.H[nH - 1], B)
             # heads dW/db: ghT (+ head bias colsums), split-K small GEMM
             u["bias_tmp"].zero_()
             ext.bf16_transpose(u["gh"], u["ghT"], u["bias_tmp"], 0,
@@ -207,14 +290,16 @@ class WideBF16Path:
                 ext.bf16_mm256(dT_cur, actT, self._b, 3, self._f, self._b,
                                grad, w_off[l], self._b, 0, self._f, 0)
                 if l > 0:
-                    # dz[l-1] = (dz[l] @ W[l]) * dtanh(h[l-1]); dual-write
-                    # dz[l-1]^T and its colsum -> db[l-1]
+                    # dz[l-1] = (dz[l] @ W[l]) * dtanh(h[l-1]); transpose
+                    # kernel writes dz[l-1]^T + colsum -> db[l-1]
                     d_nxt = u["d1"] if d_cur is u["d0"] else u["d0"]
                     dT_nxt = u["dT1"] if dT_cur is u["dT0"] else u["dT0"]
                     ext.bf16_mm256(d_cur, self.Wt_bf[l], d_nxt, 2, self._f,
                                    u["h"][l - 1], self._f, 0,
-                                   dT_nxt, B, grad, b_off[l - 1])
+                                   self._b, 0, self._f, 0)
+                    ext.bf16_transpose(d_nxt, dT_nxt, grad, b_off[l - 1],
+                                       B, self.H[l - 1], self.H[l - 1], B)
                     d_cur, dT_cur = d_nxt, dT_nxt
             eng.comm.allreduce_mean_(grad)
-            eng.optimizer.step()
+            eng.optimizer.step_captured()  # lr_dev set by the caller
             self._weights_dirty = True

@@ -55,6 +55,14 @@ def main():
             A, B, C, 1, torch.zeros(N, device="cuda"), eb, e, 0, eb, 0, e, 0), args.iters)
         print(f"  +tanh-bias epilogue: {t_tanh*1e3:8.3f} ms = "
               f"{fl/t_tanh/1e12:7.1f} TF/s")
+        CT = torch.empty(N, M, device="cuda", dtype=torch.bfloat16)
+        sums = torch.zeros(N, device="cuda")
+        t_dw = bench(lambda: ext.bf16_mm256(
+            A, B, C, 1, torch.zeros(N, device="cuda"), eb, e, 0, CT, M,
+            sums, 0), args.iters)
+        print(f"  +dual-write+colsum:  {t_dw*1e3:8.3f} ms = "
+              f"{fl/t_dw/1e12:7.1f} TF/s")
+        del CT, sums
         del A, B, C, out, Bt
         torch.cuda.empty_cache()
 
