@@ -334,6 +334,185 @@ __global__ void bf16_mm256_kernel(MM256Args a) {
 }
 
 // ---------------------------------------------------------------------------
+// 32x32x16-MFMA variant of the same pipeline: half the MFMA instruction
+// count at a higher ceiling (2382 vs 2075 TF µbench), same glds staging,
+// swizzle and phase/barrier structure.  Fragment map (guide §3):
+// A/B lane l holds 8 bf16 at [i = l&31][k = (l>>5)*8..]; C/D lane l reg r
+// covers [row = (r&3)+8*(r>>2)+4*(l>>5)][col = l&31].
+// ---------------------------------------------------------------------------
+
+typedef __attribute__((ext_vector_type(16))) float f32x16;
+
+template <int NPH>
+__launch_bounds__(512, 2)
+__global__ void bf16_mm256_kernel32(MM256Args a) {
+  constexpr int FMPP = 4 / NPH;  // 32-row fm-blocks per phase
+  __shared__ __attribute__((aligned(16))) char smem[LDS_BYTES];
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int wr = wave >> 2;
+  const int wc = wave & 3;
+
+  const long nwg = (long)gridDim.x;
+  const long q = nwg >> 3, r8 = nwg & 7;
+  const long xcd = blockIdx.x & 7, idx = blockIdx.x >> 3;
+  const long sw = (xcd < r8 ? xcd * (q + 1) : r8 * (q + 1) + (xcd - r8) * q)
+                  + idx;
+  const int bm = (int)(sw / a.nbn);
+  const int bn = (int)(sw % a.nbn);
+
+  char* const imgA0 = smem;
+  char* const imgA1 = smem + IMG;
+  char* const imgB0 = smem + 2 * IMG;
+  char* const imgB1 = smem + 3 * IMG;
+  const long rowA = (long)bm * 256;
+  const long rowB = (long)bn * 256;
+  const int NT = (int)(a.K / BK);
+
+  f32x16 acc[4][2] = {};
+
+  const unsigned short* sA00 = stage_src(a.A, rowA, a.K, 0, 0, tid);
+  const unsigned short* sA01 = stage_src(a.A, rowA, a.K, 0, 1, tid);
+  const unsigned short* sA10 = stage_src(a.A, rowA, a.K, 1, 0, tid);
+  const unsigned short* sA11 = stage_src(a.A, rowA, a.K, 1, 1, tid);
+  const unsigned short* sB00 = stage_src(a.B, rowB, a.K, 0, 0, tid);
+  const unsigned short* sB01 = stage_src(a.B, rowB, a.K, 0, 1, tid);
+  const unsigned short* sB10 = stage_src(a.B, rowB, a.K, 1, 0, tid);
+  const unsigned short* sB11 = stage_src(a.B, rowB, a.K, 1, 1, tid);
+
+  stage_piece(sA00, 0, imgA0, 0, 0, wave);
+  stage_piece(sA01, 0, imgA0, 0, 1, wave);
+  stage_piece(sA10, 0, imgA0, 1, 0, wave);
+  stage_piece(sA11, 0, imgA0, 1, 1, wave);
+  stage_piece(sB00, 0, imgB0, 0, 0, wave);
+  stage_piece(sB01, 0, imgB0, 0, 1, wave);
+  stage_piece(sB10, 0, imgB0, 1, 0, wave);
+  stage_piece(sB11, 0, imgB0, 1, 1, wave);
+  if (NT > 1) {
+    stage_piece(sB00, BK, imgB1, 0, 0, wave);
+    stage_piece(sB01, BK, imgB1, 0, 1, wave);
+    stage_piece(sB10, BK, imgB1, 1, 0, wave);
+    stage_piece(sB11, BK, imgB1, 1, 1, wave);
+    asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+  } else {
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  }
+  __builtin_amdgcn_s_barrier();
+
+  for (int t = 0; t < NT; ++t) {
+    const char* iA = (t & 1) ? imgA1 : imgA0;
+    const char* iB = (t & 1) ? imgB1 : imgB0;
+    short8 bfr[2][4];  // 2 fn-blocks x 4 k-steps of 16
+    #pragma unroll
+    for (int fn = 0; fn < 2; ++fn)
+      #pragma unroll
+      for (int ks = 0; ks < 4; ++ks)
+        bfr[fn][ks] =
+            frag_read(iB, wc * 64 + fn * 32 + (lane & 31),
+                      ks * 32 + (lane >> 5) * 16);
+
+    #pragma unroll
+    for (int ph = 0; ph < NPH; ++ph) {
+      short8 afr[FMPP][4];
+      #pragma unroll
+      for (int i = 0; i < FMPP; ++i)
+        #pragma unroll
+        for (int ks = 0; ks < 4; ++ks)
+          afr[i][ks] =
+              frag_read(iA, wr * 128 + (FMPP * ph + i) * 32 + (lane & 31),
+                        ks * 32 + (lane >> 5) * 16);
+      if (ph == 0) {
+        if (t + 1 < NT) {
+          char* img = ((t + 1) & 1) ? imgA1 : imgA0;
+          stage_piece(sA00, (long)(t + 1) * BK, img, 0, 0, wave);
+          stage_piece(sA01, (long)(t + 1) * BK, img, 0, 1, wave);
+          stage_piece(sA10, (long)(t + 1) * BK, img, 1, 0, wave);
+          stage_piece(sA11, (long)(t + 1) * BK, img, 1, 1, wave);
+        }
+      } else if (ph == 1) {
+        if (t + 2 < NT) {
+          char* img = (t & 1) ? imgB1 : imgB0;
+          stage_piece(sB00, (long)(t + 2) * BK, img, 0, 0, wave);
+          stage_piece(sB01, (long)(t + 2) * BK, img, 0, 1, wave);
+          if (NPH == 2) {
+            stage_piece(sB10, (long)(t + 2) * BK, img, 1, 0, wave);
+            stage_piece(sB11, (long)(t + 2) * BK, img, 1, 1, wave);
+          }
+        }
+      } else if (ph == 2) {
+        if (t + 2 < NT) {
+          char* img = (t & 1) ? imgB1 : imgB0;
+          stage_piece(sB10, (long)(t + 2) * BK, img, 1, 0, wave);
+          stage_piece(sB11, (long)(t + 2) * BK, img, 1, 1, wave);
+        }
+      }
+      __builtin_amdgcn_s_setprio(1);
+      #pragma unroll
+      for (int i = 0; i < FMPP; ++i) {
+        #pragma unroll
+        for (int fn = 0; fn < 2; ++fn) {
+          #pragma unroll
+          for (int ks = 0; ks < 4; ++ks)
+            acc[FMPP * ph + i][fn] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+                afr[i][ks], bfr[fn][ks], acc[FMPP * ph + i][fn], 0, 0, 0);
+        }
+      }
+      __builtin_amdgcn_s_setprio(0);
+      if (ph == NPH - 1) {
+        if (t + 2 < NT)
+          asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+        else
+          asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+      }
+      __builtin_amdgcn_s_barrier();
+    }
+  }
+
+  // epilogue: 32-row stripes through LDS (wave-private 8 KB regions)
+  float* stripe = (float*)(smem + wave * 8192);  // [32][64] f32
+  const long cb = (long)bn * 256 + wc * 64;
+  #pragma unroll
+  for (int fm = 0; fm < 4; ++fm) {
+    #pragma unroll
+    for (int fn = 0; fn < 2; ++fn)
+      #pragma unroll
+      for (int rg = 0; rg < 16; ++rg)
+        stripe[((rg & 3) + 8 * (rg >> 2) + 4 * (lane >> 5)) * 64 + fn * 32 +
+               (lane & 31)] = acc[fm][fn][rg];
+    const long rb = (long)bm * 256 + wr * 128 + fm * 32;
+    #pragma unroll
+    for (int pass = 0; pass < 8; ++pass) {
+      const int row = pass * 4 + ((lane >> 4) & 3);
+      const int col0 = (lane & 15) * 4;
+      f32x4 v = *(const f32x4*)(stripe + row * 64 + col0);
+      const long gr = rb + row;
+      const long gc = cb + col0;
+      if (a.epi == EPI_TANH_BIAS) {
+        const f32x4 b4 = *(const f32x4*)(a.bias + gc);
+        #pragma unroll
+        for (int j = 0; j < 4; ++j) v[j] = fast_tanhf(v[j] + b4[j]);
+      } else if (a.epi == EPI_DTANH) {
+        const ushort4 h4 = *(const ushort4*)(a.aux + gr * a.N + gc);
+        #pragma unroll
+        for (int j = 0; j < 4; ++j) {
+          const float h = bf2f(h4[j]);
+          v[j] *= 1.f - h * h;
+        }
+      }
+      if (a.epi == EPI_GRAD) {
+        *(f32x4*)(a.grad + gr * a.N + gc) = v;
+      } else {
+        ushort4 o;
+        #pragma unroll
+        for (int j = 0; j < 4; ++j) o[j] = f2bf(v[j]);
+        *(ushort4*)(a.C + gr * a.N + gc) = o;
+      }
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
 // bf16_mm_small: guarded 64x64-tile GEMM for ragged heads shapes.
 //   epi 0: C[M][ldc] bf16          epi 4: heads split (C=pdflat [M][N-1],
 //   epi 2: dtanh via aux [M][N]            C2=v [M]; no activation)
@@ -656,8 +835,24 @@ void bf16_mm256(torch::Tensor A, torch::Tensor B, torch::Tensor C,
     const char* e = getenv("DPPO_MM256_PH");
     return e ? atoi(e) : 2;  // 2 phases/tile measured fastest
   }();
+  static const int mfma = []() {
+    const char* e = getenv("DPPO_MM256_MFMA");
+    // 16x16x32 measured FASTER than 32x32x16 here (1160-1199 vs
+    // 1020-1047 TF/s) despite the larger µbench ceiling — the 32x32
+    // form's 4-deep chained accumulator + 16-reg C/D operands lose more
+    // than the halved instruction count buys.  Kept for ablation.
+    return e ? atoi(e) : 16;
+  }();
+  const bool want32 = (mfma == 32) && CT.numel() == 0;  // 32-variant has
+                                                        // no dual-write
   if (two_bar)
     hipLaunchKernelGGL((bf16_mm256_kernel<4, true>), dim3((unsigned)grid),
+                       dim3(512), 0, stream, a);
+  else if (want32 && nph == 2)
+    hipLaunchKernelGGL((bf16_mm256_kernel32<2>), dim3((unsigned)grid),
+                       dim3(512), 0, stream, a);
+  else if (want32)
+    hipLaunchKernelGGL((bf16_mm256_kernel32<4>), dim3((unsigned)grid),
                        dim3(512), 0, stream, a);
   else if (nph == 2)
     hipLaunchKernelGGL((bf16_mm256_kernel<2, false>), dim3((unsigned)grid),

@@ -234,6 +234,12 @@ class WideBF16Path:
 
     @torch.no_grad()
     def _update_body(self, u) -> None:
+        """UPDATE_STEPS pipeline with the memory-bound transposes and
+        heads-dW overlapped on a side stream: the 256^2 GEMMs run 2
+        waves/SIMD MFMA-bound (6 idle wave slots per SIMD and <1/3 of
+        HBM bandwidth used), so the 4.4 TB/s transpose kernel
+        co-schedules nearly for free.  Cross-stream edges are events
+        (graph capture records them)."""
         eng, ext, cfg = self.eng, self.ext, self.cfg
         B = u["B"]
         nH = len(self.H)
@@ -244,19 +250,37 @@ class WideBF16Path:
         off_wv, off_bv = offsets[2 * nH], offsets[2 * nH + 1]
         off_wp, off_bp = offsets[2 * nH + 2], offsets[2 * nH + 3]
         grad = flat.flat_grad
+        main = torch.cuda.current_stream()
+        if getattr(self, "_side", None) is None:
+            self._side = torch.cuda.Stream()
+        side = self._side
+        evs = self._evs = []  # keep alive across graph replays
+
+        def fork():
+            e = torch.cuda.Event()
+            evs.append(e)
+            e.record(main)
+            side.wait_event(e)
+
+        def join():
+            e = torch.cuda.Event()
+            evs.append(e)
+            e.record(side)
+            return e
+
         for _ in range(cfg.UPDATE_STEPS):
             self.refresh_weights()
-            # forward (h^T for dW comes from the standalone transpose
-            # kernel: the epilogue dual-write measured SLOWER than the
-            # 4.4 TB/s transpose kernel — its 32-B-contiguous scattered
-            # stores cost ~0.4 ms/call vs 0.25 for the kernel)
             x = u["x"]
+            ev_hT = [None] * nH
             for l in range(nH):
                 ext.bf16_mm256(x, self.W_bf[l], u["h"][l], 1,
                                eng.pi.hidden[l].bias.detach(), self._b,
                                self._f, 0, self._b, 0, self._f, 0)
-                ext.bf16_transpose(u["h"][l], u["hT"][l], self._f, 0,
-                                   B, self.H[l], self.H[l], B)
+                fork()
+                with torch.cuda.stream(side):
+                    ext.bf16_transpose(u["h"][l], u["hT"][l], self._f, 0,
+                                       B, self.H[l], self.H[l], B)
+                ev_hT[l] = join()
                 x = u["h"][l]
             ext.bf16_mm_small(x, self.whcat, u["pd"], u["v"], self._b,
                               self._f, 0, 0, 0, 4, B, self.P + 1, self.P,
@@ -266,40 +290,56 @@ class WideBF16Path:
                               u["actions"], u["adv"], u["etr"], u["gh"],
                               u["clip_dev"], 0.0, cfg.ENTCOEFF, cfg.VCOEFF)
             flat.zero_grad()
-            # heads backward: dz[nH-1] = (gh @ Whcat) * dtanh(h[-1]);
-            # its transpose (+ bias colsum -> db[nH-1]) via the kernel
+            # side: ghT (+ head bias colsums) and the whole heads dW —
+            # independent of the main dgrad/dW chain
+            fork()
+            with torch.cuda.stream(side):
+                u["bias_tmp"].zero_()
+                ext.bf16_transpose(u["gh"], u["ghT"], u["bias_tmp"], 0,
+                                   B, self.KP, self.KP, B)
+                grad[off_bp:off_bp + self.P].copy_(u["bias_tmp"][:self.P])
+                grad[off_bv:off_bv + 1].copy_(
+                    u["bias_tmp"][self.P:self.P + 1])
+                side.wait_event(ev_hT[nH - 1])
+                ext.bf16_mm_small(u["ghT"], u["hT"][nH - 1], self._b,
+                                  self._b, self._b, grad, off_wp, off_wv,
+                                  self.P, 5, self.P + 1, self.H[-1], 0,
+                                  self._f)
+            ev_heads = join()
+            # heads backward: dz[nH-1] = (gh @ Whcat) * dtanh(h[-1])
             d_cur = u["d0"]
             dT_cur = u["dT0"]
             ext.bf16_mm256(u["gh"], self.whcatT, d_cur, 2, self._f,
                            u["h"][nH - 1], self._f, 0,
                            self._b, 0, self._f, 0)
-            ext.bf16_transpose(d_cur, dT_cur, grad, b_off[nH - 1],
-                               B, self.H[nH - 1], self.H[nH - 1], B)
-            # heads dW/db: ghT (+ head bias colsums), split-K small GEMM
-            u["bias_tmp"].zero_()
-            ext.bf16_transpose(u["gh"], u["ghT"], u["bias_tmp"], 0,
-                               B, self.KP, self.KP, B)
-            grad[off_bp:off_bp + self.P].copy_(u["bias_tmp"][:self.P])
-            grad[off_bv:off_bv + 1].copy_(u["bias_tmp"][self.P:self.P + 1])
-            ext.bf16_mm_small(u["ghT"], u["hT"][nH - 1], self._b, self._b,
-                              self._b, grad, off_wp, off_wv, self.P, 5,
-                              self.P + 1, self.H[-1], 0, self._f)
+            fork()
+            with torch.cuda.stream(side):
+                ext.bf16_transpose(d_cur, dT_cur, grad, b_off[nH - 1],
+                                   B, self.H[nH - 1], self.H[nH - 1], B)
+            ev_dT = join()
             # hidden chain (uniform H: buffers are exact-size views)
             for l in range(nH - 1, -1, -1):
                 actT = u["xT"] if l == 0 else u["hT"][l - 1]
+                if l > 0:
+                    main.wait_event(ev_hT[l - 1])
+                main.wait_event(ev_dT)
                 ext.bf16_mm256(dT_cur, actT, self._b, 3, self._f, self._b,
                                grad, w_off[l], self._b, 0, self._f, 0)
                 if l > 0:
-                    # dz[l-1] = (dz[l] @ W[l]) * dtanh(h[l-1]); transpose
-                    # kernel writes dz[l-1]^T + colsum -> db[l-1]
+                    # dz[l-1] = (dz[l] @ W[l]) * dtanh(h[l-1]); side
+                    # transposes dz[l-1] (+ colsum -> db[l-1])
                     d_nxt = u["d1"] if d_cur is u["d0"] else u["d0"]
                     dT_nxt = u["dT1"] if dT_cur is u["dT0"] else u["dT0"]
                     ext.bf16_mm256(d_cur, self.Wt_bf[l], d_nxt, 2, self._f,
                                    u["h"][l - 1], self._f, 0,
                                    self._b, 0, self._f, 0)
-                    ext.bf16_transpose(d_nxt, dT_nxt, grad, b_off[l - 1],
-                                       B, self.H[l - 1], self.H[l - 1], B)
+                    fork()
+                    with torch.cuda.stream(side):
+                        ext.bf16_transpose(d_nxt, dT_nxt, grad, b_off[l - 1],
+                                           B, self.H[l - 1], self.H[l - 1], B)
+                    ev_dT = join()
                     d_cur, dT_cur = d_nxt, dT_nxt
+            main.wait_event(ev_heads)
             eng.comm.allreduce_mean_(grad)
             eng.optimizer.step_captured()  # lr_dev set by the caller
             self._weights_dirty = True
