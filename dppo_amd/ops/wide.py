@@ -185,7 +185,13 @@ class WideBF16Path:
         eng.optimizer.lr_dev.fill_(float(eng.optimizer.param_groups[0]["lr"]))
 
         import os as _os
+        # hipGraph capture of a multi-stream (event-forked) body segfaults
+        # in capture_end on this ROCm build, so the captured variant runs
+        # the SEQUENTIAL body and is opt-in; the default is the eager
+        # side-stream-overlapped pipeline (launch overhead is negligible
+        # against these GEMMs).
         graph_ok = (cfg.USE_GRAPHS
+                    and _os.environ.get("DPPO_WIDE_GRAPH") == "1"
                     and (not eng.comm.distributed
                          or _os.environ.get("DPPO_GRAPH_DIST") == "1")
                     and not getattr(self, "_graph_failed", False))
@@ -215,11 +221,11 @@ class WideBF16Path:
         snap = (eng.flat_pi.flat_param.detach().clone(), opt.exp_avg.clone(),
                 opt.exp_avg_sq.clone(), opt.step_dev.clone())
         try:
-            side = torch.cuda.Stream()
-            side.wait_stream(torch.cuda.current_stream())
-            with torch.cuda.stream(side):
-                self._update_body(u)
-            torch.cuda.current_stream().wait_stream(side)
+            warm = torch.cuda.Stream()
+            warm.wait_stream(torch.cuda.current_stream())
+            with torch.cuda.stream(warm):
+                self._update_body(u, overlap=False)
+            torch.cuda.current_stream().wait_stream(warm)
             torch.cuda.synchronize()
         finally:
             with torch.no_grad():
@@ -229,11 +235,11 @@ class WideBF16Path:
                 opt.step_dev.copy_(snap[3])
         g = torch.cuda.CUDAGraph()
         with torch.cuda.graph(g):
-            self._update_body(u)
+            self._update_body(u, overlap=False)
         self._graph = g
 
     @torch.no_grad()
-    def _update_body(self, u) -> None:
+    def _update_body(self, u, overlap: bool = True) -> None:
         """UPDATE_STEPS pipeline with the memory-bound transposes and
         heads-dW overlapped on a side stream: the 256^2 GEMMs run 2
         waves/SIMD MFMA-bound (6 idle wave slots per SIMD and <1/3 of
@@ -253,16 +259,30 @@ class WideBF16Path:
         main = torch.cuda.current_stream()
         if getattr(self, "_side", None) is None:
             self._side = torch.cuda.Stream()
-        side = self._side
+        side = self._side if overlap else main
         evs = self._evs = []  # keep alive across graph replays
 
+        class _NullCtx:
+            def __enter__(self):
+                return None
+
+            def __exit__(self, *a):
+                return False
+
+        side_ctx = (lambda: torch.cuda.stream(side)) if overlap \
+            else (lambda: _NullCtx())
+
         def fork():
+            if not overlap:
+                return
             e = torch.cuda.Event()
             evs.append(e)
             e.record(main)
             side.wait_event(e)
 
         def join():
+            if not overlap:
+                return None
             e = torch.cuda.Event()
             evs.append(e)
             e.record(side)
@@ -277,7 +297,7 @@ class WideBF16Path:
                                eng.pi.hidden[l].bias.detach(), self._b,
                                self._f, 0, self._b, 0, self._f, 0)
                 fork()
-                with torch.cuda.stream(side):
+                with side_ctx():
                     ext.bf16_transpose(u["h"][l], u["hT"][l], self._f, 0,
                                        B, self.H[l], self.H[l], B)
                 ev_hT[l] = join()
@@ -293,14 +313,15 @@ class WideBF16Path:
             # side: ghT (+ head bias colsums) and the whole heads dW —
             # independent of the main dgrad/dW chain
             fork()
-            with torch.cuda.stream(side):
+            with side_ctx():
                 u["bias_tmp"].zero_()
                 ext.bf16_transpose(u["gh"], u["ghT"], u["bias_tmp"], 0,
                                    B, self.KP, self.KP, B)
                 grad[off_bp:off_bp + self.P].copy_(u["bias_tmp"][:self.P])
                 grad[off_bv:off_bv + 1].copy_(
                     u["bias_tmp"][self.P:self.P + 1])
-                side.wait_event(ev_hT[nH - 1])
+                if overlap:
+                    side.wait_event(ev_hT[nH - 1])
                 ext.bf16_mm_small(u["ghT"], u["hT"][nH - 1], self._b,
                                   self._b, self._b, grad, off_wp, off_wv,
                                   self.P, 5, self.P + 1, self.H[-1], 0,
@@ -313,16 +334,17 @@ class WideBF16Path:
                            u["h"][nH - 1], self._f, 0,
                            self._b, 0, self._f, 0)
             fork()
-            with torch.cuda.stream(side):
+            with side_ctx():
                 ext.bf16_transpose(d_cur, dT_cur, grad, b_off[nH - 1],
                                    B, self.H[nH - 1], self.H[nH - 1], B)
             ev_dT = join()
             # hidden chain (uniform H: buffers are exact-size views)
             for l in range(nH - 1, -1, -1):
                 actT = u["xT"] if l == 0 else u["hT"][l - 1]
-                if l > 0:
-                    main.wait_event(ev_hT[l - 1])
-                main.wait_event(ev_dT)
+                if overlap:
+                    if l > 0:
+                        main.wait_event(ev_hT[l - 1])
+                    main.wait_event(ev_dT)
                 ext.bf16_mm256(dT_cur, actT, self._b, 3, self._f, self._b,
                                grad, w_off[l], self._b, 0, self._f, 0)
                 if l > 0:
@@ -334,12 +356,13 @@ class WideBF16Path:
                                    u["h"][l - 1], self._f, 0,
                                    self._b, 0, self._f, 0)
                     fork()
-                    with torch.cuda.stream(side):
+                    with side_ctx():
                         ext.bf16_transpose(d_nxt, dT_nxt, grad, b_off[l - 1],
                                            B, self.H[l - 1], self.H[l - 1], B)
                     ev_dT = join()
                     d_cur, dT_cur = d_nxt, dT_nxt
-            main.wait_event(ev_heads)
+            if overlap:
+                main.wait_event(ev_heads)
             eng.comm.allreduce_mean_(grad)
             eng.optimizer.step_captured()  # lr_dev set by the caller
             self._weights_dirty = True
