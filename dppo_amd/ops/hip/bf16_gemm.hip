@@ -116,7 +116,7 @@ DEV_INLINE short8 frag_read(const char* img, int row, int colb) {
   return *(const short8*)(img + byte);
 }
 
-template <int NPH, bool TWO_BARRIERS>
+template <int NPH, bool TWO_BARRIERS, bool STATIC_PRIO = false>
 __launch_bounds__(512, 2)
 __global__ void bf16_mm256_kernel(MM256Args a) {
   constexpr int FMPP = 8 / NPH;  // fm-blocks computed per phase
@@ -145,6 +145,13 @@ __global__ void bf16_mm256_kernel(MM256Args a) {
   const int NT = (int)(a.K / BK);
 
   f32x4 acc[8][4] = {};
+
+  // T5 static form: the second-dispatched half (waves 4-7) loses VALU
+  // arbitration to the older half on every segment; one wave-uniform
+  // s_setprio(1) for it, no per-cluster flips
+  if (STATIC_PRIO &&
+      __builtin_amdgcn_readfirstlane(threadIdx.x) >= 256)
+    __builtin_amdgcn_s_setprio(1);
 
   // per-thread glds source pointers (constant rows/cols; advance by k0)
   const unsigned short* sA00 = stage_src(a.A, rowA, a.K, 0, 0, tid);
@@ -238,7 +245,7 @@ __global__ void bf16_mm256_kernel(MM256Args a) {
       // the default; correctness needs only the end-of-phase barrier
       // (read-before-overwrite is sealed one phase ahead either way).
       if (TWO_BARRIERS) __builtin_amdgcn_s_barrier();
-      __builtin_amdgcn_s_setprio(1);
+      if (!STATIC_PRIO) __builtin_amdgcn_s_setprio(1);
       #pragma unroll
       for (int i = 0; i < FMPP; ++i) {
         #pragma unroll
@@ -249,7 +256,7 @@ __global__ void bf16_mm256_kernel(MM256Args a) {
               afr[i][1], bfr[fn][1], acc[FMPP * ph + i][fn], 0, 0, 0);
         }
       }
-      __builtin_amdgcn_s_setprio(0);
+      if (!STATIC_PRIO) __builtin_amdgcn_s_setprio(0);
       if (ph == NPH - 1) {
         // seam: tile t+1's halves must be LANDED in every wave before
         // anyone reads them after this barrier.  Steady state leaves the
@@ -854,10 +861,18 @@ void bf16_mm256(torch::Tensor A, torch::Tensor B, torch::Tensor C,
   else if (want32)
     hipLaunchKernelGGL((bf16_mm256_kernel32<4>), dim3((unsigned)grid),
                        dim3(512), 0, stream, a);
-  else if (nph == 2)
-    hipLaunchKernelGGL((bf16_mm256_kernel<2, false>), dim3((unsigned)grid),
-                       dim3(512), 0, stream, a);
-  else
+  else if (nph == 2) {
+    static const bool sprio = []() {
+      const char* e = getenv("DPPO_MM256_SPRIO");
+      return e != nullptr && e[0] == '1';
+    }();
+    if (sprio)
+      hipLaunchKernelGGL((bf16_mm256_kernel<2, false, true>),
+                         dim3((unsigned)grid), dim3(512), 0, stream, a);
+    else
+      hipLaunchKernelGGL((bf16_mm256_kernel<2, false>), dim3((unsigned)grid),
+                         dim3(512), 0, stream, a);
+  } else
     hipLaunchKernelGGL((bf16_mm256_kernel<4, false>), dim3((unsigned)grid),
                        dim3(512), 0, stream, a);
 }

@@ -187,11 +187,11 @@ class WideBF16Path:
         import os as _os
         # hipGraph capture of a multi-stream (event-forked) body segfaults
         # in capture_end on this ROCm build, so the captured variant runs
-        # the SEQUENTIAL body and is opt-in; the default is the eager
-        # side-stream-overlapped pipeline (launch overhead is negligible
-        # against these GEMMs).
+        # the SEQUENTIAL body (measured fastest: the side-stream overlap
+        # bought nothing — the 1-block/CU GEMMs leave no residency for
+        # co-scheduled transpose blocks).  DPPO_WIDE_GRAPH=0 opts out.
         graph_ok = (cfg.USE_GRAPHS
-                    and _os.environ.get("DPPO_WIDE_GRAPH") == "1"
+                    and _os.environ.get("DPPO_WIDE_GRAPH") != "0"
                     and (not eng.comm.distributed
                          or _os.environ.get("DPPO_GRAPH_DIST") == "1")
                     and not getattr(self, "_graph_failed", False))
@@ -239,7 +239,7 @@ class WideBF16Path:
         self._graph = g
 
     @torch.no_grad()
-    def _update_body(self, u, overlap: bool = True) -> None:
+    def _update_body(self, u, overlap: bool = False) -> None:
         """UPDATE_STEPS pipeline with the memory-bound transposes and
         heads-dW overlapped on a side stream: the 256^2 GEMMs run 2
         waves/SIMD MFMA-bound (6 idle wave slots per SIMD and <1/3 of
