@@ -645,8 +645,170 @@ class DPPOEngine:
             states, pdflats, actions, values, rewards, dones, boot_v,
             moments, eps)
 
+    # -- wide bf16 graphed rollout -------------------------------------
+    def _wide_rollout_buffers(self):
+        c = self.cfg
+        T, E = c.MAX_EPOCH_STEPS, c.NUM_ENVS
+        D = self.obs_space.shape[0]
+        A = self.act_space.shape[0]
+        P = self.pi.pdtype.param_shape()[0]
+        key = (T, E, D, A)
+        if getattr(self, "_wr_key", None) == key:
+            return self._wr
+        dev = self.device
+        self._wr = dict(
+            states=torch.empty(T, E, D, device=dev),
+            pdflats=torch.empty(T, E, P, device=dev),
+            actions=torch.empty(T, E, A, device=dev),
+            rewards=torch.empty(T, E, device=dev),
+            dones=torch.empty(T, E, device=dev),
+            values=torch.empty(T, E, device=dev),
+            boot_v=torch.empty(E, device=dev),
+            x_env=torch.empty(E, D, device=dev),
+            t_env=torch.zeros(E, device=dev, dtype=self.env.t.dtype),
+            eps_moments=[torch.zeros((), device=dev) for _ in range(5)],
+            eps_dev=torch.zeros((), device=dev),
+        )
+        self._wr_key = key
+        self._wr_graph = None
+        return self._wr
+
+    @torch.no_grad()
+    def _wide_rollout_body(self, b) -> None:
+        """Capture-safe wide rollout: T env steps with the hand bf16
+        forward, env dynamics and episode bookkeeping, reading/writing
+        only persistent buffers (env state is folded back into x_env /
+        t_env at the end so every replay continues from live state)."""
+        c, env = self.cfg, self.env
+        T, E = c.MAX_EPOCH_STEPS, c.NUM_ENVS
+        eps = b["eps_dev"]
+        (ep_count, ep_sum, ep_sumsq, ep_min, ep_max) = b["eps_moments"]
+        ep_count.zero_()
+        ep_sum.zero_()
+        ep_sumsq.zero_()
+        ep_min.fill_(math.inf)
+        ep_max.fill_(-math.inf)
+        ninf = -float("inf")
+        pinf = float("inf")
+        env.x = b["x_env"]
+        env.t = b["t_env"]
+        obs = b["x_env"]
+        wide = self._wide()
+        for t in range(T):
+            v, pdflat = wide.forward(obs)
+            pd = self.pi.pdtype.pdfromflat(pdflat)
+            a = pd.sample()
+            explore = torch.rand(E, device=self.device) < eps
+            rand_a = self._random_actions(E)
+            a = torch.where(explore.unsqueeze(-1), rand_a, a)
+            b["states"][t].copy_(obs)
+            b["actions"][t].copy_(a)
+            b["values"][t].copy_(v)
+            b["pdflats"][t].copy_(pdflat)
+            obs, r, done, _ = env.step(a)
+            b["rewards"][t].copy_(r.float())
+            donef = done.float()
+            b["dones"][t].copy_(donef)
+            self.epr += r.float()
+            ep_count += donef.sum()
+            ep_sum += (self.epr * donef).sum()
+            ep_sumsq += (self.epr.square() * donef).sum()
+            ep_min.copy_(torch.minimum(
+                ep_min, torch.where(done, self.epr, torch.full_like(self.epr, pinf)).min()))
+            ep_max.copy_(torch.maximum(
+                ep_max, torch.where(done, self.epr, torch.full_like(self.epr, ninf)).max()))
+            self.epr *= 1.0 - donef
+        boot_v, _ = wide.forward(obs)
+        b["boot_v"].copy_(boot_v)
+        # fold live env state back into the stable buffers
+        b["x_env"].copy_(obs)
+        b["t_env"].copy_(env.t)
+        env.x = b["x_env"]
+        env.t = b["t_env"]
+
+    @torch.no_grad()
+    def _rollout_once_wide_graphed(self) -> Tuple[RolloutBatch, Dict[str, float]]:
+        """Wide-config rollout as ONE hipGraph replay (the eager loop's
+        ~450 small launches per round otherwise leave the GPU idle
+        between kernels).  RNG (action sampling, eps overlay, env noise)
+        is graph-captured via registered generator states."""
+        c, E = self.cfg, self.cfg.NUM_ENVS
+        T = c.MAX_EPOCH_STEPS
+        eps = self.exploration_rate()
+        b = self._wide_rollout_buffers()
+        b["eps_dev"].fill_(eps)
+        if getattr(self, "_wr_graph", None) is None:
+            b["x_env"].copy_(self.env.x)
+            b["t_env"].copy_(self.env.t)
+            try:
+                self._wide().refresh_weights()
+                # warmup on a side stream, then restore env/episode state
+                snap = (b["x_env"].clone(), b["t_env"].clone(),
+                        self.epr.clone())
+                side = torch.cuda.Stream()
+                side.wait_stream(torch.cuda.current_stream())
+                with torch.cuda.stream(side):
+                    self._wide_rollout_body(b)
+                torch.cuda.current_stream().wait_stream(side)
+                torch.cuda.synchronize()
+                b["x_env"].copy_(snap[0])
+                b["t_env"].copy_(snap[1])
+                self.epr.copy_(snap[2])
+                g = torch.cuda.CUDAGraph()
+                g.register_generator_state(self.env._noise_gen)
+                with torch.cuda.graph(g):
+                    self._wide_rollout_body(b)
+                self._wr_graph = g
+                g.replay()
+            except Exception as exc:  # noqa: BLE001 — capture support varies
+                self._warn_once(
+                    "wide_rollout_graph",
+                    f"wide rollout hipGraph capture failed ({exc!r}); "
+                    "falling back to the eager rollout loop")
+                self._wr_graph = False
+                return self._rollout_once_eager_loop()
+        elif self._wr_graph is False:
+            return self._rollout_once_eager_loop()
+        else:
+            if self._weights_need_refresh_for_rollout():
+                self._wide().refresh_weights()
+            self._wr_graph.replay()
+        self.obs = self.env.x
+        adv, etr = gae_advantages(
+            b["rewards"], b["values"], b["dones"], b["boot_v"],
+            c.GAMMA, c.LAM, whiten=True, eps=c.ADV_EPS,
+            policy=c.USE_HIP_KERNELS,
+        )
+        P = self.pi.pdtype.param_shape()[0]
+        obs_dim = self.obs_space.shape[0]
+        m = b["eps_moments"]
+        batch = RolloutBatch(
+            states=b["states"].reshape(T * E, obs_dim),
+            actions=b["actions"].reshape(T * E, self.act_space.shape[0]),
+            adv=adv.reshape(T * E),
+            etr=etr.reshape(T * E),
+            oldflat=b["pdflats"].reshape(T * E, P),
+            oldv=b["values"].reshape(T * E),
+            cur_lr=self.current_lr_mul(),
+            ep_count=m[0], ep_sum=m[1], ep_sumsq=m[2],
+            ep_min=m[3], ep_max=m[4],
+            valid=float(m[0]) > 0,
+        )
+        return batch, {"exploration_rate": eps}
+
+    def _weights_need_refresh_for_rollout(self) -> bool:
+        w = getattr(self, "_wide_path", None)
+        return w is not None and w._weights_dirty
+
     @torch.no_grad()
     def _rollout_once_eager(self) -> Tuple[RolloutBatch, Dict[str, float]]:
+        if (self._can_wide_bf16() and self.cfg.USE_GRAPHS
+                and self.cfg.NUM_ENVS % 256 == 0):
+            return self._rollout_once_wide_graphed()
+        return self._rollout_once_eager_loop()
+
+    @torch.no_grad()
+    def _rollout_once_eager_loop(self) -> Tuple[RolloutBatch, Dict[str, float]]:
         """Collect one iteration of T = MAX_EPOCH_STEPS batched env steps
         (Worker.py:39-65), then GAE (Worker.py:82-92)."""
         c, E = self.cfg, self.cfg.NUM_ENVS
