@@ -99,24 +99,34 @@ class WideBF16Path:
         return bufs
 
     @torch.no_grad()
-    def forward(self, obs: torch.Tensor):
-        """(v, pdflat) in f32, every GEMM on the hand bf16 kernels."""
+    def forward(self, obs: torch.Tensor, out=None):
+        """(v, pdflat) in f32, every GEMM on the hand bf16 kernels.
+
+        `out = (h_views, pd_view, v_view)` redirects the layer outputs
+        into caller-owned buffers — the graphed rollout records its
+        activations straight into the update pipeline's buffers so the
+        first update step can skip its forward entirely (the parameters
+        are unchanged between rollout and step 1)."""
         if self._weights_dirty:
             self.refresh_weights()
         M = obs.shape[0]
         s = self._fwd_scratch(M)
+        if out is None:
+            hs, pd_o, v_o = s["h"], s["pd"], s["v"]
+        else:
+            hs, pd_o, v_o = out
         s["x"].copy_(obs)
         x = s["x"]
         pi = self.eng.pi
         for l in range(len(self.H)):
-            self.ext.bf16_mm256(x, self.W_bf[l], s["h"][l], 1,
+            self.ext.bf16_mm256(x, self.W_bf[l], hs[l], 1,
                                 pi.hidden[l].bias.detach(), self._b,
                                 self._f, 0, self._b, 0, self._f, 0)
-            x = s["h"][l]
-        self.ext.bf16_mm_small(x, self.whcat, s["pd"], s["v"], self._b,
+            x = hs[l]
+        self.ext.bf16_mm_small(x, self.whcat, pd_o, v_o, self._b,
                                self._f, 0, 0, 0, 4, M, self.P + 1, self.P,
                                self.bhcat)
-        return s["v"].float(), s["pd"].float()
+        return v_o.float(), pd_o.float()
 
     # -- update --------------------------------------------------------
     def _upd_bufs(self, B: int):
@@ -195,7 +205,11 @@ class WideBF16Path:
                     and (not eng.comm.distributed
                          or _os.environ.get("DPPO_GRAPH_DIST") == "1")
                     and not getattr(self, "_graph_failed", False))
+        skip_now = bool(getattr(eng, "_wide_rollout_h_valid", False))
         if graph_ok:
+            if getattr(self, "_graph_skip_first", None) is not None and \
+                    self._graph_skip_first != skip_now:
+                self._graph = None  # recapture with the new structure
             if self._graph is None:
                 try:
                     self._capture(u)
@@ -237,6 +251,8 @@ class WideBF16Path:
         with torch.cuda.graph(g):
             self._update_body(u, overlap=False)
         self._graph = g
+        self._graph_skip_first = bool(
+            getattr(self.eng, "_wide_rollout_h_valid", False))
 
     @torch.no_grad()
     def _update_body(self, u, overlap: bool = False) -> None:
@@ -288,23 +304,36 @@ class WideBF16Path:
             e.record(side)
             return e
 
-        for _ in range(cfg.UPDATE_STEPS):
+        # When the graphed rollout recorded its activations into u["h"]/
+        # u["pd"]/u["v"] (same parameters as step 1 — sync_oldpi then
+        # rollout then update, no param change in between), step 1 skips
+        # its forward+heads GEMMs entirely.
+        skip_first = bool(getattr(eng, "_wide_rollout_h_valid", False))
+        for step_i in range(cfg.UPDATE_STEPS):
             self.refresh_weights()
             x = u["x"]
             ev_hT = [None] * nH
-            for l in range(nH):
-                ext.bf16_mm256(x, self.W_bf[l], u["h"][l], 1,
-                               eng.pi.hidden[l].bias.detach(), self._b,
-                               self._f, 0, self._b, 0, self._f, 0)
-                fork()
-                with side_ctx():
-                    ext.bf16_transpose(u["h"][l], u["hT"][l], self._f, 0,
-                                       B, self.H[l], self.H[l], B)
-                ev_hT[l] = join()
-                x = u["h"][l]
-            ext.bf16_mm_small(x, self.whcat, u["pd"], u["v"], self._b,
-                              self._f, 0, 0, 0, 4, B, self.P + 1, self.P,
-                              self.bhcat)
+            if step_i == 0 and skip_first:
+                for l in range(nH):
+                    fork()
+                    with side_ctx():
+                        ext.bf16_transpose(u["h"][l], u["hT"][l], self._f,
+                                           0, B, self.H[l], self.H[l], B)
+                    ev_hT[l] = join()
+            else:
+                for l in range(nH):
+                    ext.bf16_mm256(x, self.W_bf[l], u["h"][l], 1,
+                                   eng.pi.hidden[l].bias.detach(), self._b,
+                                   self._f, 0, self._b, 0, self._f, 0)
+                    fork()
+                    with side_ctx():
+                        ext.bf16_transpose(u["h"][l], u["hT"][l], self._f,
+                                           0, B, self.H[l], self.H[l], B)
+                    ev_hT[l] = join()
+                    x = u["h"][l]
+                ext.bf16_mm_small(x, self.whcat, u["pd"], u["v"], self._b,
+                                  self._f, 0, 0, 0, 4, B, self.P + 1,
+                                  self.P, self.bhcat)
             # loss gradients -> gh [B][P+1] (bf16, K-padded)
             ext.gauss_gh_wide(u["pd"], u["oldflat"], u["v"], u["oldv"],
                               u["actions"], u["adv"], u["etr"], u["gh"],

@@ -694,8 +694,13 @@ class DPPOEngine:
         env.t = b["t_env"]
         obs = b["x_env"]
         wide = self._wide()
+        # record activations straight into the update pipeline's buffers:
+        # update step 1 runs on the same parameters and skips its forward
+        u = wide._upd_bufs(T * E)
         for t in range(T):
-            v, pdflat = wide.forward(obs)
+            outs = ([u["h"][l].narrow(0, t * E, E) for l in range(len(wide.H))],
+                    u["pd"].narrow(0, t * E, E), u["v"].narrow(0, t * E, E))
+            v, pdflat = wide.forward(obs, out=outs)
             pd = self.pi.pdtype.pdfromflat(pdflat)
             a = pd.sample()
             explore = torch.rand(E, device=self.device) < eps
@@ -773,6 +778,10 @@ class DPPOEngine:
             if self._weights_need_refresh_for_rollout():
                 self._wide().refresh_weights()
             self._wr_graph.replay()
+        # the rollout recorded h/pd/v for the whole batch into the update
+        # buffers; valid for step 1 unless curation may swap batches
+        self._wide_rollout_h_valid = not (
+            self.cfg.BATCH_CURATION and self.comm.distributed)
         self.obs = self.env.x
         adv, etr = gae_advantages(
             b["rewards"], b["values"], b["dones"], b["boot_v"],
@@ -811,6 +820,7 @@ class DPPOEngine:
     def _rollout_once_eager_loop(self) -> Tuple[RolloutBatch, Dict[str, float]]:
         """Collect one iteration of T = MAX_EPOCH_STEPS batched env steps
         (Worker.py:39-65), then GAE (Worker.py:82-92)."""
+        self._wide_rollout_h_valid = False  # no recorded activations
         c, E = self.cfg, self.cfg.NUM_ENVS
         T = c.MAX_EPOCH_STEPS
         obs_dim = self.obs_space.shape[0]
