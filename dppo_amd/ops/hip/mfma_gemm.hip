@@ -424,6 +424,166 @@ __global__ void gemm_fwd_pipe_kernel(FwdArgs a) {
 }
 
 // ---------------------------------------------------------------------------
+// glds 3-buffer streaming forward (round-2 lever #1, profiles/
+// r01_final_v3_69M.txt): the register-staged pipe kernel measured 2.5
+// TB/s effective on the tall-skinny update GEMMs vs ~6 TB/s streaming
+// capability — its per-stage issue->vmcnt(0)-at-write->ds_write chain
+// plus two __syncthreads serialize the stream.  Here the X/W tiles go
+// straight to LDS by global_load_lds into a 3-deep ring with COUNTED
+// vmcnt across raw barriers (guide §5 "3-buf span +83%"), so two stages
+// stay in flight while one computes and the wave stream carries only
+// MFMAs + fragment reads.
+//
+// Shape contract: NT==2 (one W piece per thread keeps the vmcnt count
+// wave-uniform), wt_layout 0.  K tails (K % 16 != 0) are handled by
+// clamped sources + zeroing the ws rows >= K before the last stage's
+// MFMAs (garbage X columns then multiply zero W rows).  The xs image is
+// UNPADDED [128][16] (glds needs a lane-linear dest); the 4-way
+// A-column bank spread that the pipe kernel's +4 row pad bought is kept
+// by XOR-swizzling the float4 piece index with row bits on the SOURCE
+// address (guide rule 21) — same 64 B cacheline, zero coalescing cost.
+// ---------------------------------------------------------------------------
+
+__launch_bounds__(FWD_WAVES * 64, 4)
+__global__ void gemm_fwd_glds_kernel(FwdArgs a) {
+  constexpr int NT = 2;
+  constexpr int PBK = 16;
+  constexpr int NW = NT * M_WAVE;          // 64
+  constexpr int XB = FWD_M * PBK * 4;      // 8192 B
+  constexpr int WB = PBK * NW * 4;         // 4096 B
+  constexpr int SLOT = XB + WB;
+  __shared__ __attribute__((aligned(16))) char smem[3 * SLOT];
+  const int tid = threadIdx.x;
+  const int lane = tid & (WAVE - 1);
+  const int wave = tid / WAVE;
+  const int i_l = lane & 31;
+  const int k_l = lane >> 5;
+  const int n0 = blockIdx.y * NW;
+
+  const int64_t b0 = (int64_t)blockIdx.x * FWD_M;
+  if (b0 >= a.B) return;
+
+  // per-thread glds sources (constant; advance by kb each stage)
+  // X: two 16-B pieces per thread; dest piece q = tid + p*256 ->
+  // row = q/4, c4 = (q%4) with the source float4 swizzled by row bits
+  const float* srcX[2];
+  int xcol[2];
+  int64_t xrow[2];
+  #pragma unroll
+  for (int p = 0; p < 2; ++p) {
+    const int q = tid + p * FWD_WAVES * 64;
+    const int r = q >> 2;
+    const int c4 = (q & 3) ^ (r & 3);  // source-side st swizzle
+    int64_t row = b0 + r;
+    row = row < a.B ? row : a.B - 1;
+    xrow[p] = row;
+    xcol[p] = c4 * 4;
+    srcX[p] = a.X + row * a.K;
+  }
+  // W: one 16-B piece per thread; dest q = tid -> k = q/(NW/4), c4 = q%(NW/4)
+  const int wk = tid / (NW / 4);
+  int wc4 = (tid % (NW / 4)) * 4;
+  {
+    int c = n0 + wc4;
+    c = c + 3 < a.N ? c : a.N - 4;  // clamped (garbage cols discarded)
+    wc4 = c;
+  }
+
+  auto issue = [&](int s) {
+    const int kb = s * PBK;
+    char* slot = smem + (s % 3) * SLOT;
+    #pragma unroll
+    for (int p = 0; p < 2; ++p) {
+      int col = kb + xcol[p];
+      col = col + 3 < a.K ? col : a.K - 4;  // clamp; tail fixed via ws zeros
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) unsigned*)(srcX[p] + col),
+          (unsigned*)(slot + p * 4096 + wave * 1024), 16, 0, 0);
+    }
+    int krow = kb + wk;
+    krow = krow < a.K ? krow : a.K - 1;
+    __builtin_amdgcn_global_load_lds(
+        (const __attribute__((address_space(1))) unsigned*)(
+            a.Wt + (int64_t)krow * a.N + wc4),
+        (unsigned*)(slot + XB + wave * 1024), 16, 0, 0);
+  };
+
+  const int S = (a.K + PBK - 1) / PBK;
+  issue(0);
+  if (S > 1) issue(1);
+
+  f32x16 acc[NT];
+  #pragma unroll
+  for (int t = 0; t < NT; ++t)
+    #pragma unroll
+    for (int r = 0; r < 16; ++r) acc[t][r] = 0.f;
+
+  for (int s = 0; s < S; ++s) {
+    // stage s landed when only stage s+1's 3 glds remain outstanding
+    if (s + 1 < S)
+      asm volatile("s_waitcnt vmcnt(3)" ::: "memory");
+    else
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+    char* slot = smem + (s % 3) * SLOT;
+    float* xs = (float*)slot;
+    float* ws = (float*)(slot + XB);
+    const int kb = s * PBK;
+    if (kb + PBK > a.K) {
+      // zero the W rows past K so the clamped/garbage X columns of the
+      // tail stage contribute nothing
+      for (int idx = tid; idx < (kb + PBK - a.K) * NW; idx += FWD_WAVES * 64)
+        ws[(a.K - kb) * NW + idx] = 0.f;
+      __builtin_amdgcn_s_barrier();
+    }
+    if (s + 2 < S) issue(s + 2);  // buf (s+2)%3 == (s-1)%3, freed above
+    #pragma unroll 4
+    for (int k2 = 0; k2 < PBK; k2 += 2) {
+      const int k = k2 + k_l;
+      const int row = wave * M_WAVE + i_l;
+      const float av = xs[row * PBK + (((k >> 2) ^ (row & 3)) << 2) + (k & 3)];
+      #pragma unroll
+      for (int t = 0; t < NT; ++t) {
+        const float bv = ws[k * NW + t * M_WAVE + i_l];
+        acc[t] = __builtin_amdgcn_mfma_f32_32x32x2f32(av, bv, acc[t], 0, 0, 0);
+      }
+    }
+  }
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __builtin_amdgcn_s_barrier();
+
+  #pragma unroll
+  for (int t = 0; t < NT; ++t) {
+    const int col = n0 + t * M_WAVE + i_l;
+    if (col < a.N) {
+      const float bv = (a.activation >= 3) ? 0.f : a.bias[col];
+      #pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const int64_t row = b0 + wave * M_WAVE + cd_row(r, lane);
+        if (row < a.B) {
+          float x;
+          if (a.activation >= 3) {
+            const float h = a.aux[row * a.N + col];
+            x = acc[t][r] * ((a.activation == 3) ? (1.f - h * h)
+                                                 : (h > 0.f ? 1.f : 0.f));
+          } else {
+            x = acc[t][r] + bv;
+            if (a.activation == 0) x = fmaxf(x, 0.f);
+            else if (a.activation == 1) x = fast_tanhf(x);
+          }
+          if (a.heads) {
+            if (col == a.N - 1) a.v[row] = x;
+            else a.C[row * a.ldc + col] = x;
+          } else {
+            a.C[row * a.ldc + col] = x;
+          }
+        }
+      }
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
 // dW accumulation: dW[out][in] += delta^T @ acts (+ db += sum delta)
 // grid = (m_tiles, n_tiles, splits); one wave per block.
 // ---------------------------------------------------------------------------
@@ -699,8 +859,21 @@ void gemm_fwd(torch::Tensor X, torch::Tensor Wt, torch::Tensor bias,
                     (K % 4 == 0 || (wt_layout == 0 && activation >= 3)) &&
                     (wt_layout == 1 || (N >= 4 && N % 4 == 0));
   const int pbk = pbk_env ? pbk_env : 16;
+  static const int glds_env = []() {
+    const char* e = getenv("DPPO_GEMM_GLDS");
+    return e ? atoi(e) : 1;
+  }();
+  // glds 3-buffer streaming variant: NT==2 only (wave-uniform vmcnt),
+  // layout-0 W, 16-B-aligned rows (K%4==0, N%4==0), no ablation hooks
+  const bool glds_ok = glds_env && !a.ablate && wt_layout == 0 &&
+                       K % 4 == 0 && K >= 4 && N % 4 == 0 && N >= 4 &&
+                       a.activation != 9999;
   #define DISPATCH_FWD(NTV, NPANEL)                                          \
-    if (pipe && pbk == 32)                                                   \
+    if (glds_ok && NTV == 2)                                                 \
+      hipLaunchKernelGGL(gemm_fwd_glds_kernel,                               \
+                         dim3((unsigned)grid_pipe, NPANEL),                  \
+                         dim3(FWD_WAVES * 64), 0, stream, a);                \
+    else if (pipe && pbk == 32)                                              \
       hipLaunchKernelGGL((gemm_fwd_pipe_kernel<NTV, 32>),                    \
                          dim3(grid_pipe, NPANEL), dim3(FWD_WAVES * 64), 0,   \
                          stream, a);                                         \

@@ -408,6 +408,7 @@ class DPPOEngine:
             self._rollout_out = torch.empty(n_out, device=self.device)
             self._adv_buf = torch.empty(T * E, device=self.device)
             self._etr_buf = torch.empty(T * E, device=self.device)
+        self._v3_acts_valid = False  # fused kernel does not record acts
         (states, pdflats, actions, values, rewards, dones, boot_v,
          moments) = ext.rollout_run(
             blob, offsets, dims,
@@ -476,6 +477,7 @@ class DPPOEngine:
         M = torch.zeros(kw, D, device=dev)
         M[:r] = env.U
         M[r:r + A] = env.B
+        B = int(E) * T
         self._v3 = dict(
             xva=torch.zeros(E, kw, device=dev), va_off=r, kw=kw,
             M=M.contiguous(), V=env.V.contiguous(),
@@ -490,6 +492,11 @@ class DPPOEngine:
             eps_dev=torch.zeros(1, dtype=torch.float32, device=dev),
             epr_before=torch.empty(E, device=dev),
             empty=torch.empty(0, device=dev),
+            # whole-batch activation blob in the update path's acts layout
+            # ([B*H1 | B*H2 | ...]): the rollout's layer GEMMs write here
+            # (same bytes they wrote to scratch before), and update step 1
+            # reuses them — its forward runs on the SAME parameters
+            acts=torch.empty(sum(B * hh for hh in H), device=dev),
         )
         self._v3_key = key
         return self._v3
@@ -521,11 +528,19 @@ class DPPOEngine:
             wts[l].copy_(lay.weight.detach().t())
         v3["epr_before"].copy_(self.epr)
         states[0].reshape(-1).copy_(env.x.reshape(-1))
+        # whole-batch activation views in the update acts-blob layout;
+        # rollout step st writes rows [st*E, (st+1)*E) of each layer
+        hs = c.HIDDEN_SIZES
+        B = T * E
+        acts_views, off = [], 0
+        for hh in hs:
+            acts_views.append(v3["acts"].narrow(0, off, B * hh).view(B, hh))
+            off += B * hh
         for st in range(T):
             xin = states[st]
             h = xin
             for l in range(n_h):
-                hl = v3["h"][l]
+                hl = acts_views[l].narrow(0, st * E, E)
                 ext.gemm_fwd(h, wts[l],
                              self.pi.hidden[l].bias.detach(), act_code, 0,
                              hl, hl, hl, 0, 0, 0)
@@ -640,6 +655,9 @@ class DPPOEngine:
         else:
             self._v3_graph.replay()
             moments = self._v3_moments
+        # step-1 of the update can reuse the recorded activations (same
+        # parameters; cleared by the other rollout paths and by curation)
+        self._v3_acts_valid = True
         self.obs = env.x
         return self._finish_hip_rollout(
             states, pdflats, actions, values, rewards, dones, boot_v,
@@ -821,6 +839,7 @@ class DPPOEngine:
         """Collect one iteration of T = MAX_EPOCH_STEPS batched env steps
         (Worker.py:39-65), then GAE (Worker.py:82-92)."""
         self._wide_rollout_h_valid = False  # no recorded activations
+        self._v3_acts_valid = False
         c, E = self.cfg, self.cfg.NUM_ENVS
         T = c.MAX_EPOCH_STEPS
         obs_dim = self.obs_space.shape[0]
@@ -1286,6 +1305,10 @@ class DPPOEngine:
         if getattr(self, "_graph_failed", False):
             self._update_fused(batch, l_mul)
             return
+        if (getattr(self, "_upd_graph", None) is not None
+                and getattr(self, "_upd_graph_skip", None)
+                != bool(getattr(self, "_v3_acts_valid", False))):
+            self._upd_graph = None  # skip-first structure changed
         if getattr(self, "_upd_graph", None) is None:
             try:
                 self._clip_dev = torch.tensor([clip], device=self.device)
@@ -1313,6 +1336,8 @@ class DPPOEngine:
                 with torch.cuda.graph(g):
                     self._update_body(batch)
                 self._upd_graph = g
+                self._upd_graph_skip = bool(
+                    getattr(self, "_v3_acts_valid", False))
             except Exception as exc:  # noqa: BLE001 — capture support varies
                 self._warn_once(
                     "upd_graph",
@@ -1335,8 +1360,13 @@ class DPPOEngine:
             for _ in range(self.cfg.UPDATE_STEPS):
                 self._chunk_kernel_step(batch, 0, B, 0.0)  # clip: _clip_dev
             return
-        for _ in range(self.cfg.UPDATE_STEPS):
-            acts, a_views, v, pdflat = self._fused_forward(batch.states)
+        skip1 = (bool(getattr(self, "_v3_acts_valid", False))
+                 and os.environ.get("DPPO_NO_SKIP1") != "1")
+        for si in range(self.cfg.UPDATE_STEPS):
+            if si == 0 and skip1:
+                acts, a_views, v, pdflat = self._recorded_acts(batch)
+            else:
+                acts, a_views, v, pdflat = self._fused_forward(batch.states)
             self.flat_pi.zero_grad()
             self._fused_backward(
                 batch.states, acts, a_views, v, pdflat,
@@ -1345,6 +1375,23 @@ class DPPOEngine:
             )
             self.comm.allreduce_mean_(self.flat_pi.flat_grad)
             self.optimizer.step_captured()
+
+    def _recorded_acts(self, batch: RolloutBatch):
+        """(acts, a_views, v, pdflat) from the v3 rollout's recorded
+        activation blob — valid only for the FIRST update step of a round
+        (parameters unchanged between rollout and step 1); pdflat/v are
+        the recorded oldflat/oldv, making the step-1 ratio exactly 1."""
+        v3 = self._v3
+        B = batch.states.shape[0]
+        acts = v3["acts"]
+        a_views, off = [], 0
+        for hh in self.cfg.HIDDEN_SIZES:
+            a_views.append(acts.narrow(0, off, B * hh).view(B, hh))
+            off += B * hh
+        with torch.no_grad():
+            self._Wh_cat = torch.cat(
+                [self.pi.pi.weight, self.pi.vf.weight], dim=0).contiguous()
+        return acts, a_views, batch.oldv, batch.oldflat
 
     def _fused_forward(self, states: torch.Tensor):
         """MFMA forward through the MLP (gemm_fwd per layer + heads).
@@ -1455,8 +1502,13 @@ class DPPOEngine:
             for _ in range(self.cfg.UPDATE_STEPS):
                 self._chunk_kernel_step(batch, 0, B, clip)
             return
-        for _ in range(self.cfg.UPDATE_STEPS):
-            acts, a_views, v, pdflat = self._fused_forward(batch.states)
+        skip1 = (bool(getattr(self, "_v3_acts_valid", False))
+                 and os.environ.get("DPPO_NO_SKIP1") != "1")
+        for si in range(self.cfg.UPDATE_STEPS):
+            if si == 0 and skip1:
+                acts, a_views, v, pdflat = self._recorded_acts(batch)
+            else:
+                acts, a_views, v, pdflat = self._fused_forward(batch.states)
             self.flat_pi.zero_grad()
             self._fused_backward(
                 batch.states, acts, a_views, v, pdflat,
@@ -1504,6 +1556,8 @@ class DPPOEngine:
         # (assignments are of the pre-curation batches), so a receive may
         # land IN-PLACE in the persistent rollout buffers — keeping the
         # hipGraph-captured update paths (keyed on stable data_ptrs) valid.
+        self._v3_acts_valid = False  # batch may be another rank's
+        self._wide_rollout_h_valid = False
         sends_any = any(s == r and d != r for d, s in enumerate(srcs))
         send_copy = {f: orig[f].clone() for f in fields} if sends_any else None
         for dst, s in enumerate(srcs):

@@ -273,3 +273,23 @@ def test_v3_training_rounds(monkeypatch):
         stats, _ = eng.train_round()
     assert all(math.isfinite(v) for v in stats.values())
     assert not torch.equal(p0, eng.flat_pi.flat_param.detach())
+
+
+def test_v3_recorded_acts_skip_is_bitwise_neutral(monkeypatch):
+    """Update step 1 reusing the v3 rollout's recorded activations must be
+    BITWISE identical to recomputing the forward (same kernels, same
+    inputs, same parameters)."""
+    monkeypatch.setenv("DPPO_ROLLOUT_V3", "1")
+    kw = dict(NUM_ENVS=128, MAX_EPOCH_STEPS=16, USE_GRAPHS=False, SEED=31)
+    torch.manual_seed(0)
+    a = make_engine(**kw)
+    torch.manual_seed(0)
+    monkeypatch.setenv("DPPO_NO_SKIP1", "1")
+    b = make_engine(**kw)
+    for _ in range(2):
+        monkeypatch.delenv("DPPO_NO_SKIP1")
+        sa, _ = a.train_round()
+        monkeypatch.setenv("DPPO_NO_SKIP1", "1")
+        sb, _ = b.train_round()
+    torch.cuda.synchronize()
+    assert torch.equal(a.flat_pi.flat_param, b.flat_pi.flat_param)
