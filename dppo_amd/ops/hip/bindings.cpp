@@ -86,6 +86,13 @@ void rollout_sample(torch::Tensor pdflat, torch::Tensor actions,
                     torch::Tensor eps_dev, int64_t step, int64_t va_off,
                     double act_low, double act_high);
 
+void gemm_env_step(torch::Tensor xva, torch::Tensor M, torch::Tensor x,
+                   torch::Tensor envd, torch::Tensor horizons,
+                   torch::Tensor t, torch::Tensor epr, torch::Tensor snext,
+                   torch::Tensor rewards, torch::Tensor dones,
+                   torch::Tensor rsum, torch::Tensor seed_dev, double sigma,
+                   int64_t step);
+
 void rollout_env_step(torch::Tensor x, torch::Tensor G, torch::Tensor envd,
                       torch::Tensor horizons, torch::Tensor t,
                       torch::Tensor epr, torch::Tensor states_next,
@@ -145,6 +152,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
           "fused T-step rollout: MLP fwd + sample + synthetic env (gfx950)");
   mod.def("rollout_sample", &rollout_sample,
           "per-step action sampling + eps-greedy (v3 rollout) (gfx950)");
+  mod.def("gemm_env_step", &gemm_env_step,
+          "fused G-GEMM + env transition epilogue + per-env finish "
+          "(v3 rollout; gfx950)");
   mod.def("rollout_env_step", &rollout_env_step,
           "per-step synthetic env finish: tanh/reward/done/reset (gfx950)");
   mod.def("rollout_moments", &rollout_moments,

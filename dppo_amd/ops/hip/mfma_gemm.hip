@@ -75,7 +75,47 @@ struct FwdArgs {
   int ablate;         // perf diagnosis only (wrong results when nonzero):
                       // 1 skip X stage, 2 skip W stage, 4 skip MFMA,
                       // 8 skip epilogue stores
+  // activation 5 (fused env step, v3 rollout): the G = [XV|a] @ [U;B]
+  // GEMM's epilogue computes the env transition directly —
+  // pred = tanh(x*d + acc + sigma*noise), out = done ? fresh : pred —
+  // writing the new env state and the next recorded state, and
+  // accumulating per-env pred^2 partials for the reward.  Kills the
+  // 2x[E][D] G round trip and most of env_finish_kernel (round-2 lever
+  // #3, profiles/r01_final_v3_69M.txt).  RNG slots == env_finish's.
+  float* env_x;             // [E][D] in/out
+  const float* envd;        // [D]
+  const int* horizons;      // [E]
+  const int* tcount;        // [E] (read; env_finish2 increments)
+  const long long* seed_dev;
+  float* env_rsum;          // [E] += sum(pred^2) partials (pre-zeroed)
+  float* snext;             // [E][D] states[st+1] or null
+  float sigma;
+  int step;
 };
+
+// counter-based RNG — formulas identical to rollout.hip's env path
+DEV_INLINE unsigned g_lowbias32(unsigned x) {
+  x ^= x >> 16;
+  x *= 0x7feb352dU;
+  x ^= x >> 15;
+  x *= 0x846ca68bU;
+  x ^= x >> 16;
+  return x;
+}
+DEV_INLINE float2 g_rng_normal2(unsigned seed, int env, int step, int slot) {
+  const unsigned h = g_lowbias32(seed ^ (unsigned)env * 0x9E3779B9U ^
+                                 (unsigned)step * 0x85EBCA6BU ^
+                                 (unsigned)slot * 0xC2B2AE35U);
+  const float u1 = ((h >> 16) + 1) * (1.0f / 65537.0f);
+  const float u2 = (h & 0xFFFFu) * (1.0f / 65536.0f);
+  const float r = sqrtf(-2.0f * __logf(u1));
+  float sn, cs;
+  __sincosf(6.2831853071795865f * u2, &sn, &cs);
+  return make_float2(r * cs, r * sn);
+}
+DEV_INLINE float g_rng_normal(unsigned seed, int env, int step, int slot) {
+  return g_rng_normal2(seed, env, step, slot).x;
+}
 
 template <int NT>
 __launch_bounds__(FWD_WAVES * 64)
@@ -552,6 +592,47 @@ __global__ void gemm_fwd_glds_kernel(FwdArgs a) {
   asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
   __builtin_amdgcn_s_barrier();
 
+  if (a.activation == 5) {
+    // fused env step: pred = tanh(x*d + acc + sigma*noise);
+    // out = done ? fresh-seed : pred; reward partials via one atomic per
+    // (row, 32-lane half).  RNG slots match env_finish_kernel exactly.
+    const unsigned seed = (unsigned)(*a.seed_dev);
+    #pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      const int64_t row = b0 + wave * M_WAVE + cd_row(r, lane);
+      const bool rok = row < a.B;
+      const int done =
+          rok ? (a.tcount[row] + 1 >= a.horizons[row] ? 1 : 0) : 0;
+      float part = 0.f;
+      #pragma unroll
+      for (int t = 0; t < NT; ++t) {
+        const int col = n0 + t * M_WAVE + i_l;
+        if (rok && col < a.N) {
+          const float dv = a.envd[col];
+          const float xv = a.env_x[row * a.N + col];
+          float nz = 0.f;
+          if (a.sigma != 0.f) {
+            const float2 pr =
+                g_rng_normal2(seed, (int)row, a.step, 1000 + (col >> 1));
+            nz = (col & 1) ? pr.y : pr.x;
+          }
+          const float pred = fast_tanhf(xv * dv + acc[t][r] + a.sigma * nz);
+          part += pred * pred;
+          const float out =
+              done ? 0.1f * g_rng_normal(seed, (int)row, a.step, 5000 + col)
+                   : pred;
+          a.env_x[row * a.N + col] = out;
+          if (a.snext != nullptr) a.snext[row * a.N + col] = out;
+        }
+      }
+      #pragma unroll
+      for (int off = 16; off > 0; off >>= 1)
+        part += __shfl_down(part, off, 32);
+      if (i_l == 0 && rok) atomicAdd(&a.env_rsum[row], part);
+    }
+    return;
+  }
+
   #pragma unroll
   for (int t = 0; t < NT; ++t) {
     const int col = n0 + t * M_WAVE + i_l;
@@ -580,6 +661,32 @@ __global__ void gemm_fwd_glds_kernel(FwdArgs a) {
         }
       }
     }
+  }
+}
+
+// per-env finish after the fused env-step GEMM: reward from the pred^2
+// partials, episode bookkeeping, and re-zero rsum for the next step
+__global__ void env_finish2_kernel(float* __restrict__ rsum,
+                                   int* __restrict__ t,
+                                   const int* __restrict__ horizons,
+                                   float* __restrict__ epr,
+                                   float* __restrict__ rewards,
+                                   float* __restrict__ dones, int64_t E,
+                                   int D) {
+  for (int64_t e = gidx(); e < E; e += gstride()) {
+    const float r = 1.0f - rsum[e] / (float)D;
+    rsum[e] = 0.f;
+    rewards[e] = r;
+    float ep = epr[e] + r;
+    int tc = t[e] + 1;
+    const int done = tc >= horizons[e] ? 1 : 0;
+    dones[e] = (float)done;
+    if (done) {
+      ep = 0.f;
+      tc = 0;
+    }
+    epr[e] = ep;
+    t[e] = tc;
   }
 }
 
@@ -899,6 +1006,50 @@ void gemm_fwd(torch::Tensor X, torch::Tensor Wt, torch::Tensor bias,
     }
   }
   #undef DISPATCH_FWD
+}
+
+void gemm_env_step(torch::Tensor xva, torch::Tensor M, torch::Tensor x,
+                   torch::Tensor envd, torch::Tensor horizons,
+                   torch::Tensor t, torch::Tensor epr, torch::Tensor snext,
+                   torch::Tensor rewards, torch::Tensor dones,
+                   torch::Tensor rsum, torch::Tensor seed_dev, double sigma,
+                   int64_t step) {
+  const int64_t E = xva.size(0);
+  const int K = static_cast<int>(xva.size(1));
+  const int D = static_cast<int>(M.size(1));
+  TORCH_CHECK(xva.is_contiguous() && M.is_contiguous() && x.is_contiguous());
+  TORCH_CHECK(M.size(0) == K && x.size(0) == E && x.size(1) == D);
+  TORCH_CHECK(K % 4 == 0 && D % 4 == 0, "fused env step needs 16B rows");
+  TORCH_CHECK(horizons.dtype() == torch::kInt32 && t.dtype() == torch::kInt32);
+  TORCH_CHECK(rsum.numel() >= E);
+  FwdArgs a{};
+  a.X = xva.data_ptr<float>();
+  a.Wt = M.data_ptr<float>();
+  a.B = E;
+  a.K = K;
+  a.N = D;
+  a.activation = 5;
+  a.ldc = D;
+  a.env_x = x.data_ptr<float>();
+  a.envd = envd.data_ptr<float>();
+  a.horizons = horizons.data_ptr<int>();
+  a.tcount = t.data_ptr<int>();
+  a.seed_dev = reinterpret_cast<const long long*>(seed_dev.data_ptr<int64_t>());
+  a.env_rsum = rsum.data_ptr<float>();
+  a.snext = snext.numel() ? snext.data_ptr<float>() : nullptr;
+  a.sigma = (float)sigma;
+  a.step = (int)step;
+  hipStream_t stream = c10::hip::getCurrentHIPStream().stream();
+  const int64_t tiles = (E + FWD_M - 1) / FWD_M;
+  const int panels = (D + 63) / 64;
+  hipLaunchKernelGGL(gemm_fwd_glds_kernel,
+                     dim3((unsigned)tiles, (unsigned)panels),
+                     dim3(FWD_WAVES * 64), 0, stream, a);
+  hipLaunchKernelGGL(env_finish2_kernel, dim3(elementwise_grid(E, 256)),
+                     dim3(256), 0, stream, rsum.data_ptr<float>(),
+                     t.data_ptr<int>(), horizons.data_ptr<int>(),
+                     epr.data_ptr<float>(), rewards.data_ptr<float>(),
+                     dones.data_ptr<float>(), E, D);
 }
 
 void dw_mfma(torch::Tensor delta, torch::Tensor acts, torch::Tensor grad_buf,

@@ -491,6 +491,7 @@ class DPPOEngine:
             seed_dev=torch.zeros(1, dtype=torch.int64, device=dev),
             eps_dev=torch.zeros(1, dtype=torch.float32, device=dev),
             epr_before=torch.empty(E, device=dev),
+            rsum=torch.zeros(E, device=dev),
             empty=torch.empty(0, device=dev),
             # whole-batch activation blob in the update path's acts layout
             # ([B*H1 | B*H2 | ...]): the rollout's layer GEMMs write here
@@ -553,14 +554,14 @@ class DPPOEngine:
             ext.rollout_sample(pdflats[st], actions[st], v3["xva"],
                                v3["seed_dev"], v3["eps_dev"], st,
                                v3["va_off"], low, high)
-            # G = [XV | act] @ [U; B]  (wide-N column-panel GEMM)
-            ext.gemm_fwd(v3["xva"], v3["M"], v3["bz_D"], 2, 0, v3["G"],
-                         v3["G"], v3["G"], 0, 0, 0)
+            # G = [XV | act] @ [U; B] with the env transition FUSED into
+            # the GEMM epilogue (no [E][D] G round trip, no env_finish
+            # stream pass; identical math and RNG slots)
             nxt = states[st + 1] if st + 1 < T else v3["empty"]
-            ext.rollout_env_step(env.x, v3["G"], env.d, env.horizons_i32,
-                                 env.t, self.epr, nxt, rewards[st],
-                                 dones[st], v3["seed_dev"],
-                                 float(env.NOISE), st)
+            ext.gemm_env_step(v3["xva"], v3["M"], env.x, env.d,
+                              env.horizons_i32, env.t, self.epr, nxt,
+                              rewards[st], dones[st], v3["rsum"],
+                              v3["seed_dev"], float(env.NOISE), st)
         # bootstrap value V(x_T)
         h = env.x
         for l in range(n_h):
