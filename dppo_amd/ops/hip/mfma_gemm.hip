@@ -87,7 +87,8 @@ struct FwdArgs {
   const int* horizons;      // [E]
   const int* tcount;        // [E] (read; env_finish2 increments)
   const long long* seed_dev;
-  float* env_rsum;          // [E] += sum(pred^2) partials (pre-zeroed)
+  float* env_rsum;          // [npanels][E] per-panel pred^2 partials
+                            // (plain stores — deterministic, no zeroing)
   float* snext;             // [E][D] states[st+1] or null
   float sigma;
   int step;
@@ -628,7 +629,10 @@ __global__ void gemm_fwd_glds_kernel(FwdArgs a) {
       #pragma unroll
       for (int off = 16; off > 0; off >>= 1)
         part += __shfl_down(part, off, 32);
-      if (i_l == 0 && rok) atomicAdd(&a.env_rsum[row], part);
+      // exactly one (wave, r, half) owns each (panel, row): plain store
+      // keeps the reduction deterministic (atomicAdd order is not)
+      if (i_l == 0 && rok)
+        a.env_rsum[(int64_t)blockIdx.y * a.B + row] = part;
     }
     return;
   }
@@ -666,16 +670,17 @@ __global__ void gemm_fwd_glds_kernel(FwdArgs a) {
 
 // per-env finish after the fused env-step GEMM: reward from the pred^2
 // partials, episode bookkeeping, and re-zero rsum for the next step
-__global__ void env_finish2_kernel(float* __restrict__ rsum,
+__global__ void env_finish2_kernel(const float* __restrict__ rsum,
                                    int* __restrict__ t,
                                    const int* __restrict__ horizons,
                                    float* __restrict__ epr,
                                    float* __restrict__ rewards,
                                    float* __restrict__ dones, int64_t E,
-                                   int D) {
+                                   int D, int npanels) {
   for (int64_t e = gidx(); e < E; e += gstride()) {
-    const float r = 1.0f - rsum[e] / (float)D;
-    rsum[e] = 0.f;
+    float ss = 0.f;
+    for (int p = 0; p < npanels; ++p) ss += rsum[(int64_t)p * E + e];
+    const float r = 1.0f - ss / (float)D;
     rewards[e] = r;
     float ep = epr[e] + r;
     int tc = t[e] + 1;
@@ -1021,7 +1026,8 @@ void gemm_env_step(torch::Tensor xva, torch::Tensor M, torch::Tensor x,
   TORCH_CHECK(M.size(0) == K && x.size(0) == E && x.size(1) == D);
   TORCH_CHECK(K % 4 == 0 && D % 4 == 0, "fused env step needs 16B rows");
   TORCH_CHECK(horizons.dtype() == torch::kInt32 && t.dtype() == torch::kInt32);
-  TORCH_CHECK(rsum.numel() >= E);
+  const int panels_chk = (D + 63) / 64;
+  TORCH_CHECK(rsum.numel() >= panels_chk * E);
   FwdArgs a{};
   a.X = xva.data_ptr<float>();
   a.Wt = M.data_ptr<float>();
@@ -1049,7 +1055,7 @@ void gemm_env_step(torch::Tensor xva, torch::Tensor M, torch::Tensor x,
                      dim3(256), 0, stream, rsum.data_ptr<float>(),
                      t.data_ptr<int>(), horizons.data_ptr<int>(),
                      epr.data_ptr<float>(), rewards.data_ptr<float>(),
-                     dones.data_ptr<float>(), E, D);
+                     dones.data_ptr<float>(), E, D, panels);
 }
 
 void dw_mfma(torch::Tensor delta, torch::Tensor acts, torch::Tensor grad_buf,

@@ -491,7 +491,7 @@ class DPPOEngine:
             seed_dev=torch.zeros(1, dtype=torch.int64, device=dev),
             eps_dev=torch.zeros(1, dtype=torch.float32, device=dev),
             epr_before=torch.empty(E, device=dev),
-            rsum=torch.zeros(E, device=dev),
+            rsum=torch.zeros(((D + 63) // 64) * E, device=dev),
             empty=torch.empty(0, device=dev),
             # whole-batch activation blob in the update path's acts layout
             # ([B*H1 | B*H2 | ...]): the rollout's layer GEMMs write here
