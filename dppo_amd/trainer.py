@@ -502,6 +502,10 @@ class DPPOEngine:
         self._v3_key = key
         return self._v3
 
+    @staticmethod
+    def _env_fused() -> bool:
+        return os.environ.get("DPPO_ENV_FUSED") == "1"
+
     def _v3_body(self, states, pdflats, actions, values, rewards, dones,
                  boot_v):
         """Capture-safe per-step rollout pipeline.  RNG slots match the
@@ -554,14 +558,25 @@ class DPPOEngine:
             ext.rollout_sample(pdflats[st], actions[st], v3["xva"],
                                v3["seed_dev"], v3["eps_dev"], st,
                                v3["va_off"], low, high)
-            # G = [XV | act] @ [U; B] with the env transition FUSED into
-            # the GEMM epilogue (no [E][D] G round trip, no env_finish
-            # stream pass; identical math and RNG slots)
             nxt = states[st + 1] if st + 1 < T else v3["empty"]
-            ext.gemm_env_step(v3["xva"], v3["M"], env.x, env.d,
-                              env.horizons_i32, env.t, self.epr, nxt,
-                              rewards[st], dones[st], v3["rsum"],
-                              v3["seed_dev"], float(env.NOISE), st)
+            if self._env_fused():
+                # G = [XV|act] @ [U;B] with the env transition fused into
+                # the GEMM epilogue (identical math and RNG slots).  Saves
+                # the [E][D] G round trip but pays per-ELEMENT Box-Muller
+                # (the separate env_finish shares one hash per dim pair):
+                # measured net -4ms/round at the flagship config, so the
+                # split path stays the default (DPPO_ENV_FUSED=1 opts in).
+                ext.gemm_env_step(v3["xva"], v3["M"], env.x, env.d,
+                                  env.horizons_i32, env.t, self.epr, nxt,
+                                  rewards[st], dones[st], v3["rsum"],
+                                  v3["seed_dev"], float(env.NOISE), st)
+            else:
+                ext.gemm_fwd(v3["xva"], v3["M"], v3["bz_D"], 2, 0, v3["G"],
+                             v3["G"], v3["G"], 0, 0, 0)
+                ext.rollout_env_step(env.x, v3["G"], env.d,
+                                     env.horizons_i32, env.t, self.epr, nxt,
+                                     rewards[st], dones[st], v3["seed_dev"],
+                                     float(env.NOISE), st)
         # bootstrap value V(x_T)
         h = env.x
         for l in range(n_h):
