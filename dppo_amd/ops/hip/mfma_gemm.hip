@@ -852,6 +852,160 @@ __global__ void dw_mfma_kernel(DwArgs a) {
   }
 }
 
+// ---------------------------------------------------------------------------
+// dW glds kernel (round-2 lever #2): one 8-wave block owns the WHOLE
+// [out<=64][in<=384] dW tile for one K split, with both operands staged
+// row-major by global_load_lds into a 3-deep ring with counted vmcnt
+// (same pipeline as gemm_fwd_glds_kernel).  Replaces the 1-wave
+// dw_mfma_kernel's 4-B-per-lane streams for the big flagship dW calls:
+// delta is read ONCE (the old n-tile grid re-read it 6x) and the block
+// keeps two full stages in flight.
+// Both MFMA operands are row-major-k, column-per-lane LDS reads
+// (conflict-free); K-split tails zero the delta rows past the split so
+// clamped/garbage rows contribute nothing.
+// ---------------------------------------------------------------------------
+
+constexpr int DWG_K = 16;        // k rows per stage
+constexpr int DWG_M = 64;        // out columns covered
+constexpr int DWG_N = 384;       // in columns covered (4 waves x NTW*32)
+constexpr int DWG_NTW = 3;       // 32-col tiles per wave
+constexpr int DWG_DB = DWG_K * DWG_M * 4;               // delta bytes/stage
+constexpr int DWG_AB = DWG_K * DWG_N * 4;               // acts bytes/stage
+constexpr int DWG_SLOT = DWG_DB + DWG_AB;               // 28 KiB
+constexpr int DWG_PIECES = DWG_SLOT / 16;               // 1792
+constexpr int DWG_PPT = 4;  // glds per thread per stage (2048 incl pad)
+
+__launch_bounds__(512, 2)
+__global__ void dw_glds_kernel(DwArgs a) {
+  __shared__ __attribute__((aligned(16))) char smem[3 * DWG_SLOT + 4096];
+  char* const pad = smem + 3 * DWG_SLOT;  // dummy glds target (uniformity)
+  const int tid = threadIdx.x;
+  const int lane = tid & (WAVE - 1);
+  const int wave = tid >> 6;
+  const int i_l = lane & 31;
+  const int k_l = lane >> 5;
+  const int wm = wave >> 2;               // 2 m-tiles of 32
+  const int wn = wave & 3;                // 4 n-groups of 96
+  const int split = blockIdx.x;
+
+  const int64_t rows_per = (a.B + a.splits - 1) / a.splits;
+  const int64_t k0 = (int64_t)split * rows_per;
+  const int64_t k1 = min(a.B, k0 + rows_per);
+  if (k0 >= a.B) return;
+  const int S = (int)((k1 - k0 + DWG_K - 1) / DWG_K);
+
+  // per-thread glds sources: piece q = tid + p*512; rows past B clamp to
+  // B-1 (garbage — the delta-row zeroing kills their contribution)
+  const float* base[DWG_PPT];
+  int krow_p[DWG_PPT];
+  int ld_p[DWG_PPT];
+  unsigned dsto[DWG_PPT];
+  #pragma unroll
+  for (int p = 0; p < DWG_PPT; ++p) {
+    const int q = tid + p * 512;
+    if (q < DWG_DB / 16) {               // delta piece
+      const int kq = q / (DWG_M / 4);
+      int c4 = (q % (DWG_M / 4)) * 4;
+      c4 = c4 + 3 < a.out_dim ? c4 : (a.out_dim > 4 ? a.out_dim - 4 : 0);
+      base[p] = a.delta + c4;
+      krow_p[p] = kq;
+      ld_p[p] = a.out_dim;
+      dsto[p] = q * 16;
+    } else if (q < DWG_PIECES) {         // acts piece
+      const int qq = q - DWG_DB / 16;
+      const int kq = qq / (DWG_N / 4);
+      int c4 = (qq % (DWG_N / 4)) * 4;
+      c4 = c4 + 3 < a.in_dim ? c4 : (a.in_dim > 4 ? a.in_dim - 4 : 0);
+      base[p] = a.acts + c4;
+      krow_p[p] = kq;
+      ld_p[p] = a.in_dim;
+      dsto[p] = q * 16;
+    } else {                             // pad piece: dummy, uniform vmcnt
+      base[p] = a.delta;
+      krow_p[p] = 0;
+      ld_p[p] = 0;
+      dsto[p] = 0xFFFFFFFFu;
+    }
+  }
+
+  auto issue = [&](int s) {
+    char* slot = smem + (s % 3) * DWG_SLOT;
+    #pragma unroll
+    for (int p = 0; p < DWG_PPT; ++p) {
+      int64_t row = k0 + (int64_t)s * DWG_K + krow_p[p];
+      row = row < a.B ? row : a.B - 1;
+      const float* g = base[p] + row * ld_p[p];
+      char* dst = dsto[p] == 0xFFFFFFFFu ? pad + (tid & 255) * 16
+                                         : slot + dsto[p];
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) unsigned*)g,
+          (unsigned*)dst, 16, 0, 0);
+    }
+  };
+
+  issue(0);
+  if (S > 1) issue(1);
+
+  f32x16 acc[DWG_NTW];
+  #pragma unroll
+  for (int t = 0; t < DWG_NTW; ++t)
+    #pragma unroll
+    for (int r = 0; r < 16; ++r) acc[t][r] = 0.f;
+  float dbacc = 0.f;
+  const int mcol = wm * M_WAVE + i_l;
+
+  for (int s = 0; s < S; ++s) {
+    if (s + 1 < S)
+      asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+    else
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+    char* slot = smem + (s % 3) * DWG_SLOT;
+    float* ds = (float*)slot;                    // [16][64] delta
+    float* as = (float*)(slot + DWG_DB);         // [16][384] acts
+    const int64_t kb = k0 + (int64_t)s * DWG_K;
+    const int kval = (int)((k1 - kb) < DWG_K ? (k1 - kb) : DWG_K);
+    if (kval < DWG_K) {
+      // zero delta rows past the split (kills garbage contributions)
+      for (int idx = tid; idx < (DWG_K - kval) * DWG_M; idx += 512)
+        ds[kval * DWG_M + idx] = 0.f;
+      __builtin_amdgcn_s_barrier();
+    }
+    if (s + 2 < S) issue(s + 2);
+    #pragma unroll 4
+    for (int k2 = 0; k2 < DWG_K; k2 += 2) {
+      const float av = ds[(k2 + k_l) * DWG_M + mcol];
+      dbacc += av;
+      #pragma unroll
+      for (int t = 0; t < DWG_NTW; ++t) {
+        const float bv =
+            as[(k2 + k_l) * DWG_N + wn * (DWG_NTW * M_WAVE) + t * M_WAVE + i_l];
+        acc[t] = __builtin_amdgcn_mfma_f32_32x32x2f32(av, bv, acc[t], 0, 0, 0);
+      }
+    }
+  }
+
+  // ---- store to this split's slab ----
+  float* slab = a.slab + (int64_t)split * a.out_dim * a.in_dim;
+  #pragma unroll
+  for (int t = 0; t < DWG_NTW; ++t) {
+    const int col = wn * (DWG_NTW * M_WAVE) + t * M_WAVE + i_l;
+    if (col < a.in_dim) {
+      #pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const int mrow = wm * M_WAVE + cd_row(r, lane);
+        if (mrow < a.out_dim)
+          slab[(int64_t)mrow * a.in_dim + col] = acc[t][r];
+      }
+    }
+  }
+  if (wn == 0) {
+    const float other = __shfl(dbacc, lane ^ 32, WAVE);
+    if (lane < 32 && mcol < a.out_dim)
+      a.db_slab[(int64_t)split * a.out_dim + mcol] = dbacc + other;
+  }
+}
+
 // Split-slab reduction into the flat grad, handling the optional
 // combined-heads row split (rows < split_row -> dW/db, rest -> dW2/db2).
 // Parallel over (element, split-chunk): a serial full-splits loop per
@@ -1097,6 +1251,44 @@ void dw_mfma(torch::Tensor delta, torch::Tensor acts, torch::Tensor grad_buf,
   a.db_slab = db_slab.data_ptr<float>();
 
   hipStream_t stream = c10::hip::getCurrentHIPStream().stream();
+  static const int dwglds_env = []() {
+    const char* e = getenv("DPPO_DW_GLDS");
+    return e ? atoi(e) : 1;
+  }();
+  // 8-wave LDS-staged variant: whole dW tile per block, delta read once
+  if (dwglds_env && !a.ablate && out_dim <= DWG_M && in_dim <= DWG_N &&
+      out_dim % 4 == 0 && in_dim % 4 == 0 && B >= 4096) {
+    a.splits = static_cast<int>(
+        std::min<int64_t>(1024, std::max<int64_t>(256, B / 4096)));
+    a.splits = static_cast<int>(
+        std::min<int64_t>(a.splits, std::max<int64_t>(1, B / DWG_K)));
+    auto slab2 = torch::empty(
+        {(int64_t)a.splits, (int64_t)out_dim, (int64_t)in_dim},
+        delta.options());
+    auto db_slab2 = torch::zeros({(int64_t)a.splits, (int64_t)out_dim},
+                                 delta.options());
+    a.slab = slab2.data_ptr<float>();
+    a.db_slab = db_slab2.data_ptr<float>();
+    hipLaunchKernelGGL(dw_glds_kernel, dim3((unsigned)a.splits), dim3(512),
+                       0, stream, a);
+    float* dW = grad_buf.data_ptr<float>() + w_off;
+    float* db = (b_off >= 0) ? grad_buf.data_ptr<float>() + b_off : nullptr;
+    float* dW2 =
+        (split_row >= 0) ? grad_buf.data_ptr<float>() + w_off2 : nullptr;
+    float* db2 = (split_row >= 0 && b_off2 >= 0)
+                     ? grad_buf.data_ptr<float>() + b_off2
+                     : nullptr;
+    const int n_chunks = (a.splits + DW_RED_CHUNK - 1) / DW_RED_CHUNK;
+    const dim3 rgrid(elementwise_grid((int64_t)out_dim * in_dim, 256),
+                     n_chunks);
+    hipLaunchKernelGGL(dw_reduce_kernel, rgrid, dim3(256), 0, stream, a.slab,
+                       a.db_slab, dW, db, dW2, db2, out_dim, in_dim, a.splits,
+                       static_cast<int>(split_row));
+    hipLaunchKernelGGL(db_reduce_kernel, dim3(out_dim), dim3(WAVE), 0,
+                       stream, a.db_slab, db, db2, out_dim, a.splits,
+                       static_cast<int>(split_row));
+    return;
+  }
   const dim3 grid(m_tiles, n_tiles, a.splits);
   const int key = a.nt * 10 + mt;
   switch (key) {
