@@ -1256,8 +1256,11 @@ void dw_mfma(torch::Tensor delta, torch::Tensor acts, torch::Tensor grad_buf,
     return e ? atoi(e) : 1;
   }();
   // 8-wave LDS-staged variant: whole dW tile per block, delta read once
+  // only the wide-acts calls (dW1: in=obs) — at in_dim << DWG_N most of
+  // the block's n-waves compute discarded columns and the old splits
+  // grid wins (measured: dW1 562 vs 709 us, but dW2 regressed)
   if (dwglds_env && !a.ablate && out_dim <= DWG_M && in_dim <= DWG_N &&
-      out_dim % 4 == 0 && in_dim % 4 == 0 && B >= 4096) {
+      in_dim > 192 && out_dim % 4 == 0 && in_dim % 4 == 0 && B >= 4096) {
     a.splits = static_cast<int>(
         std::min<int64_t>(1024, std::max<int64_t>(256, B / 4096)));
     a.splits = static_cast<int>(
