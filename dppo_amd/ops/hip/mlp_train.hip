@@ -133,14 +133,26 @@ DEV_INLINE float dact(float h, int activation) {
   return activation ? (1.f - h * h) : (h > 0.f ? 1.f : 0.f);
 }
 
-template <int KC1>
+// TD/TH/TA/TNH = 0 -> runtime shapes (the generic ladder); nonzero ->
+// EXACT compile-time shapes: the LDS map, every i/H-i%H index map, the
+// float4 tail guards and the per-thread slice bounds all constant-fold,
+// removing the address-arithmetic / exec-mask / SGPR-spill classes the
+// static ISA audit blamed for the issue-bound profile
+// (profiles/r01_chunk_kernel_notes.md round-2 target list).  TACT: 1 =
+// tanh folded, -1 = relu folded, 0 = runtime.
+template <int KC1, int TD = 0, int TH = 0, int TA = 0, int TNH = 0,
+          int TACT = 0>
 __launch_bounds__(CT, 1)
 __global__ void mlp_chunk_kernel(ChunkArgs a) {
   const int tid = threadIdx.x;
   const int lane = tid & (WAVE - 1);
   const int wv = tid / WAVE;
-  const int D = a.D, H = a.H, A = a.A, P = 2 * A;
-  const int NH = a.n_hidden;
+  const int D = TD ? TD : a.D;
+  const int H = TH ? TH : a.H;
+  const int A = TA ? TA : a.A;
+  const int P = 2 * A;
+  const int NH = TNH ? TNH : a.n_hidden;
+  const int activation = TACT ? (TACT > 0 ? 1 : 0) : a.activation;
   const int CH = CT / H;        // k-chunks (16/8/4 for H=16/32/64)
   const int uo = tid % H;       // owned unit (dW phases)
   const int co = tid / H;       // owned k-chunk
@@ -267,7 +279,7 @@ __global__ void mlp_chunk_kernel(ChunkArgs a) {
       float sum = lds[m.b1 + u];
       #pragma unroll
       for (int w = 0; w < CNW; ++w) sum += lds[m.part + (w * ST + s) * m.pc + u];
-      lds[m.h1 + i] = a.activation ? fast_tanhf(sum) : fmaxf(sum, 0.f);
+      lds[m.h1 + i] = activation ? fast_tanhf(sum) : fmaxf(sum, 0.f);
     }
     __syncthreads();
 
@@ -282,7 +294,7 @@ __global__ void mlp_chunk_kernel(ChunkArgs a) {
         #pragma unroll
         for (int w = 0; w < CNW; ++w)
           sum += lds[m.part + (w * ST + s) * m.pc + u];
-        lds[m.h2 + i] = a.activation ? fast_tanhf(sum) : fmaxf(sum, 0.f);
+        lds[m.h2 + i] = activation ? fast_tanhf(sum) : fmaxf(sum, 0.f);
       }
       __syncthreads();
     }
@@ -362,7 +374,7 @@ __global__ void mlp_chunk_kernel(ChunkArgs a) {
       for (; u < P + 1; ++u)
         a0 += lds[m.wht + k * m.wsh + u] * lds[m.gh + s * (P + 1) + u];
       lds[dlast + i] =
-          dact(lds[hlast + i], a.activation) * ((a0 + a1) + (a2 + a3));
+          dact(lds[hlast + i], activation) * ((a0 + a1) + (a2 + a3));
     }
     __syncthreads();
 
@@ -378,7 +390,7 @@ __global__ void mlp_chunk_kernel(ChunkArgs a) {
           a3 += lds[m.w2t + k * m.ws2 + u + 3] * lds[m.dh2 + s * H + u + 3];
         }
         lds[m.dh1 + i] =
-            dact(lds[m.h1 + i], a.activation) * ((a0 + a1) + (a2 + a3));
+            dact(lds[m.h1 + i], activation) * ((a0 + a1) + (a2 + a3));
       }
       __syncthreads();
     }
@@ -502,7 +514,13 @@ static bool chunk_lds_attr_ok() {
           reinterpret_cast<const void*>(&(*mlp_chunk_kernel<16>)),
           reinterpret_cast<const void*>(&(*mlp_chunk_kernel<32>)),
           reinterpret_cast<const void*>(&(*mlp_chunk_kernel<48>)),
-          reinterpret_cast<const void*>(&(*mlp_chunk_kernel<64>))})
+          reinterpret_cast<const void*>(&(*mlp_chunk_kernel<64>)),
+          reinterpret_cast<const void*>(
+              &(*mlp_chunk_kernel<48, 376, 64, 17, 2, 1>)),
+          reinterpret_cast<const void*>(
+              &(*mlp_chunk_kernel<4, 17, 64, 6, 2, 1>)),
+          reinterpret_cast<const void*>(
+              &(*mlp_chunk_kernel<1, 3, 16, 1, 1, -1>))})
       r &= hipFuncSetAttribute(f, hipFuncAttributeMaxDynamicSharedMemorySize,
                                159 * 1024) == hipSuccess;
     return r;
@@ -602,7 +620,17 @@ void mlp_chunk_train(
   const int CH = CT / (int)H;
   const int kc1 = (int)((D + CH - 1) / CH);
   hipStream_t stream = c10::hip::getCurrentHIPStream().stream();
-  if (kc1 <= 8)
+  // exact-trip-count specializations for the BASELINE shape families
+  if (D == 376 && H == 64 && A == 17 && n_hidden == 2 && activation == 1)
+    hipLaunchKernelGGL((mlp_chunk_kernel<48, 376, 64, 17, 2, 1>),
+                       dim3((uint32_t)nb), dim3(CT), lds_bytes, stream, a);
+  else if (D == 17 && H == 64 && A == 6 && n_hidden == 2 && activation == 1)
+    hipLaunchKernelGGL((mlp_chunk_kernel<4, 17, 64, 6, 2, 1>),
+                       dim3((uint32_t)nb), dim3(CT), lds_bytes, stream, a);
+  else if (D == 3 && H == 16 && A == 1 && n_hidden == 1 && activation == 0)
+    hipLaunchKernelGGL((mlp_chunk_kernel<1, 3, 16, 1, 1, -1>),
+                       dim3((uint32_t)nb), dim3(CT), lds_bytes, stream, a);
+  else if (kc1 <= 8)
     hipLaunchKernelGGL(mlp_chunk_kernel<8>, dim3((uint32_t)nb), dim3(CT),
                        lds_bytes, stream, a);
   else if (kc1 <= 16)
