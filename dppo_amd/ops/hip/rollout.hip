@@ -171,7 +171,12 @@ DEV_INLINE float u2f_mono(unsigned u) {
 // MINWAVES: waves/SIMD forced on the allocator.  4 caps VGPRs at 128 and
 // SPILLS (29 VGPR + 185 SGPR, 96 B scratch/lane measured on this source) —
 // both instantiated, dispatch measures/chooses.
-template <int MINWAVES>
+// TD/TH/TA/TNH/TRANK/TACT = 0 -> runtime shapes; nonzero -> exact
+// compile-time shapes (constant-folds the LDS map, index maps and trip
+// counts — the same instruction-count program as mlp_chunk_kernel;
+// profiles/r01_chunk_kernel_notes.md round-2 target list).
+template <int MINWAVES, int TD = 0, int TH = 0, int TA = 0, int TNH = 0,
+          int TRANK = 0, int TACT = 0>
 __launch_bounds__(NWAVES * 64, MINWAVES)
 __global__ void rollout_kernel(RolloutArgs a) {
   const int tid = threadIdx.x;      // block = 4 waves of 64
@@ -179,18 +184,22 @@ __global__ void rollout_kernel(RolloutArgs a) {
   const int wv = tid / WAVE;        // k-split wave index
   const int e0 = blockIdx.x * ENV_TILE;
   const int nE = min(ENV_TILE, a.E - e0);
-  const int D = a.D;
-  const int A = a.act_dim;
+  const int D = TD ? TD : a.D;
+  const int A = TA ? TA : a.act_dim;
   const int P = 2 * A;
   const int T = a.T, E = a.E;
+  const int RNK = TRANK ? TRANK : a.rank;
+  const int NHID = TNH ? TNH : a.n_hidden;
+  const int HMAX = TH ? TH : a.h_max;
+  const int ACT = TACT ? (TACT > 0 ? 1 : 0) : a.activation;
 
   extern __shared__ __attribute__((aligned(16))) float lds[];
-  const LdsMap lm = lds_map(D, a.h_max, A, a.rank);
+  const LdsMap lm = lds_map(D, HMAX, A, RNK);
   const int X_OFF = lm.x, H0_OFF = lm.h0, H1_OFF = lm.h1, PD_OFF = lm.pd;
   const int ACT_OFF = lm.act, XV_OFF = lm.xv, PART_OFF = lm.part;
   const int MAX_D_S = lm.xs, MAX_H_S = lm.hs;
   const int PD_S = ((2 * A) + 3) & ~3, ACT_S = (A + 3) & ~3,
-            XV_S = (a.rank + 3) & ~3;
+            XV_S = (RNK + 3) & ~3;
   float* val_lds = &lds[lm.val];
   float* rsum_lds = &lds[lm.rsum];
   float* epr_lds = &lds[lm.epr];
@@ -201,8 +210,8 @@ __global__ void rollout_kernel(RolloutArgs a) {
   // env blob offsets (d | Vt[r][D] | U[r][D] | B[A][D])
   const float* env_d = a.envblob;
   const float* env_Vt = env_d + D;
-  const float* env_U = env_Vt + (int64_t)a.rank * D;
-  const float* env_B = env_U + (int64_t)a.rank * D;
+  const float* env_U = env_Vt + (int64_t)RNK * D;
+  const float* env_B = env_U + (int64_t)RNK * D;
   // output blob offsets
   float* out_states = a.out;
   float* out_pdflats = out_states + (int64_t)T * E * D;
@@ -296,7 +305,7 @@ __global__ void rollout_kernel(RolloutArgs a) {
       const int k0q = kq * kq_len;
       const int k1q = min(D, k0q + kq_len);
       float accv = 0.f;
-      if (rr < a.rank) {
+      if (rr < RNK) {
         const float* Vrow = env_Vt + (int64_t)rr * D;
         int k = k0q;
         #pragma unroll 2
@@ -311,14 +320,14 @@ __global__ void rollout_kernel(RolloutArgs a) {
       // butterfly-reduce over the k-quarter lanes (bits 4 and 5)
       accv += __shfl_xor(accv, 16, WAVE);
       accv += __shfl_xor(accv, 32, WAVE);
-      if (kq == 0 && rr < a.rank) lds[XV_OFF + e * XV_S + rr] = accv;
+      if (kq == 0 && rr < RNK) lds[XV_OFF + e * XV_S + rr] = accv;
     }
 
     // ---- policy MLP forward (K-split + combine per layer) ----
     int in_off = X_OFF, in_stride = MAX_D_S, in_dim = D;
     if (!(a.ablate & 1))
-    for (int l = 0; l < a.n_hidden; ++l) {
-      const int out_dim = a.dims[l + 1];
+    for (int l = 0; l < NHID; ++l) {
+      const int out_dim = TH ? TH : a.dims[l + 1];
       const int out_off = (l & 1) ? H1_OFF : H0_OFF;
       layer_kpart(a.params + a.off_W[l], out_dim, 0, in_off, in_stride,
                   in_dim, out_dim, false);
@@ -331,7 +340,7 @@ __global__ void rollout_kernel(RolloutArgs a) {
         for (int w = 0; w < NWAVES; ++w)
           sum += lds[PART_OFF + (w * ENV_TILE + e) * MAX_H_S + u];
         lds[out_off + e * MAX_H_S + u] =
-            a.activation ? fast_tanhf(sum) : fmaxf(sum, 0.f);
+            ACT ? fast_tanhf(sum) : fmaxf(sum, 0.f);
       }
       __syncthreads();
       in_off = out_off;
@@ -339,9 +348,9 @@ __global__ void rollout_kernel(RolloutArgs a) {
       in_dim = out_dim;
     }
     if (a.ablate & 1) {
-      in_off = (a.n_hidden & 1) ? H0_OFF : H1_OFF;
+      in_off = (NHID & 1) ? H0_OFF : H1_OFF;
       in_stride = MAX_H_S;
-      in_dim = a.dims[a.n_hidden];
+      in_dim = TH ? TH : a.dims[NHID];
     }
 
     // ---- heads (u < P: pd params; u == P: value), no activation ----
@@ -422,7 +431,7 @@ __global__ void rollout_kernel(RolloutArgs a) {
         // float4 LDS reads: 4x fewer LDS instructions than scalar
         // broadcasts (the scalar form measured 2.5 ms of the rollout)
         int rr = 0;
-        for (; rr + 4 <= a.rank; rr += 4) {
+        for (; rr + 4 <= RNK; rr += 4) {
           const float u0 = env_U[(int64_t)(rr + 0) * D + d];
           const float u1 = env_U[(int64_t)(rr + 1) * D + d];
           const float u2 = env_U[(int64_t)(rr + 2) * D + d];
@@ -434,7 +443,7 @@ __global__ void rollout_kernel(RolloutArgs a) {
             low[e] += xv4.x * u0 + xv4.y * u1 + xv4.z * u2 + xv4.w * u3;
           }
         }
-        for (; rr < a.rank; ++rr) {
+        for (; rr < RNK; ++rr) {
           const float uv = env_U[(int64_t)rr * D + d];
           #pragma unroll
           for (int e = 0; e < ENV_TILE; ++e)
@@ -506,8 +515,8 @@ __global__ void rollout_kernel(RolloutArgs a) {
   // ---- bootstrap value V(x_T): trunk + value head ----
   {
     int in_off = X_OFF, in_stride = MAX_D_S, in_dim = D;
-    for (int l = 0; l < a.n_hidden; ++l) {
-      const int out_dim = a.dims[l + 1];
+    for (int l = 0; l < NHID; ++l) {
+      const int out_dim = TH ? TH : a.dims[l + 1];
       const int out_off = (l & 1) ? H1_OFF : H0_OFF;
       layer_kpart(a.params + a.off_W[l], out_dim, 0, in_off, in_stride,
                   in_dim, out_dim, false);
@@ -520,7 +529,7 @@ __global__ void rollout_kernel(RolloutArgs a) {
         for (int w = 0; w < NWAVES; ++w)
           sum += lds[PART_OFF + (w * ENV_TILE + e) * MAX_H_S + u];
         lds[out_off + e * MAX_H_S + u] =
-            a.activation ? fast_tanhf(sum) : fmaxf(sum, 0.f);
+            ACT ? fast_tanhf(sum) : fmaxf(sum, 0.f);
       }
       __syncthreads();
       in_off = out_off;
@@ -804,15 +813,42 @@ std::vector<torch::Tensor> rollout_run(
     return e ? atoi(e) : -1;  // -1 = unset, distinct from explicit 0
   }();
   const int mw = mw_env >= 0 ? mw_env : (grid <= N_CU ? 1 : 4);
-  if (mw >= 4)
-    hipLaunchKernelGGL(rollout_kernel<4>, dim3(grid), dim3(NWAVES * WAVE),
-                       lds_bytes, stream, a);
-  else if (mw >= 2)
+  // exact-shape specializations for the BASELINE families (index maps,
+  // trip counts and the LDS map constant-folded)
+  const bool hc_shape = (a.D == 17 && a.h_max == 64 && a.act_dim == 6 &&
+                         a.n_hidden == 2 && a.rank == 16 &&
+                         a.activation == 1);
+  const bool hum_shape = (a.D == 376 && a.h_max == 64 && a.act_dim == 17 &&
+                          a.n_hidden == 2 && a.rank == 16 &&
+                          a.activation == 1);
+  if (mw >= 4) {
+    if (hc_shape)
+      hipLaunchKernelGGL((rollout_kernel<4, 17, 64, 6, 2, 16, 1>),
+                         dim3(grid), dim3(NWAVES * WAVE), lds_bytes, stream,
+                         a);
+    else if (hum_shape)
+      hipLaunchKernelGGL((rollout_kernel<4, 376, 64, 17, 2, 16, 1>),
+                         dim3(grid), dim3(NWAVES * WAVE), lds_bytes, stream,
+                         a);
+    else
+      hipLaunchKernelGGL(rollout_kernel<4>, dim3(grid), dim3(NWAVES * WAVE),
+                         lds_bytes, stream, a);
+  } else if (mw >= 2) {
     hipLaunchKernelGGL(rollout_kernel<3>, dim3(grid), dim3(NWAVES * WAVE),
                        lds_bytes, stream, a);
-  else
-    hipLaunchKernelGGL(rollout_kernel<1>, dim3(grid), dim3(NWAVES * WAVE),
-                       lds_bytes, stream, a);
+  } else {
+    if (hc_shape)
+      hipLaunchKernelGGL((rollout_kernel<1, 17, 64, 6, 2, 16, 1>),
+                         dim3(grid), dim3(NWAVES * WAVE), lds_bytes, stream,
+                         a);
+    else if (hum_shape)
+      hipLaunchKernelGGL((rollout_kernel<1, 376, 64, 17, 2, 16, 1>),
+                         dim3(grid), dim3(NWAVES * WAVE), lds_bytes, stream,
+                         a);
+    else
+      hipLaunchKernelGGL(rollout_kernel<1>, dim3(grid), dim3(NWAVES * WAVE),
+                         lds_bytes, stream, a);
+  }
 
   // carve views out of the blob
   int64_t o = 0;
