@@ -183,7 +183,9 @@ __global__ void rollout_kernel(RolloutArgs a) {
   const int lane = tid & (WAVE - 1);
   const int wv = tid / WAVE;        // k-split wave index
   const int e0 = blockIdx.x * ENV_TILE;
-  const int nE = min(ENV_TILE, a.E - e0);
+  // specialized variants are dispatched only when E % ENV_TILE == 0:
+  // nE folds and every per-env tail guard disappears
+  const int nE = TD ? ENV_TILE : min(ENV_TILE, a.E - e0);
   const int D = TD ? TD : a.D;
   const int A = TA ? TA : a.act_dim;
   const int P = 2 * A;
@@ -264,7 +266,7 @@ __global__ void rollout_kernel(RolloutArgs a) {
         #pragma unroll
         for (int e = 0; e < ENV_TILE; ++e) {
           const float4 x4 = *reinterpret_cast<const float4*>(
-              &lds[in_off + e * in_stride + k]);
+              __builtin_assume_aligned(&lds[in_off + e * in_stride + k], 16));
           acc[e] += w0 * x4.x + w1 * x4.y + w2 * x4.z + w3 * x4.w;
         }
       }
@@ -312,7 +314,8 @@ __global__ void rollout_kernel(RolloutArgs a) {
         for (; k + 4 <= k1q; k += 4) {
           const float4 v4 = *reinterpret_cast<const float4*>(Vrow + k);
           const float4 x4 =
-              *reinterpret_cast<const float4*>(&lds[X_OFF + e * MAX_D_S + k]);
+              *reinterpret_cast<const float4*>(
+              __builtin_assume_aligned(&lds[X_OFF + e * MAX_D_S + k], 16));
           accv += v4.x * x4.x + v4.y * x4.y + v4.z * x4.z + v4.w * x4.w;
         }
         for (; k < k1q; ++k) accv += Vrow[k] * lds[X_OFF + e * MAX_D_S + k];
@@ -439,7 +442,8 @@ __global__ void rollout_kernel(RolloutArgs a) {
           #pragma unroll
           for (int e = 0; e < ENV_TILE; ++e) {
             const float4 xv4 =
-                *reinterpret_cast<const float4*>(&lds[XV_OFF + e * XV_S + rr]);
+                *reinterpret_cast<const float4*>(
+              __builtin_assume_aligned(&lds[XV_OFF + e * XV_S + rr], 16));
             low[e] += xv4.x * u0 + xv4.y * u1 + xv4.z * u2 + xv4.w * u3;
           }
         }
@@ -458,7 +462,8 @@ __global__ void rollout_kernel(RolloutArgs a) {
           #pragma unroll
           for (int e = 0; e < ENV_TILE; ++e) {
             const float4 a4 =
-                *reinterpret_cast<const float4*>(&lds[ACT_OFF + e * ACT_S + j]);
+                *reinterpret_cast<const float4*>(
+              __builtin_assume_aligned(&lds[ACT_OFF + e * ACT_S + j], 16));
             ain[e] += a4.x * b0 + a4.y * b1 + a4.z * b2 + a4.w * b3;
           }
         }
@@ -815,12 +820,13 @@ std::vector<torch::Tensor> rollout_run(
   const int mw = mw_env >= 0 ? mw_env : (grid <= N_CU ? 1 : 4);
   // exact-shape specializations for the BASELINE families (index maps,
   // trip counts and the LDS map constant-folded)
-  const bool hc_shape = (a.D == 17 && a.h_max == 64 && a.act_dim == 6 &&
-                         a.n_hidden == 2 && a.rank == 16 &&
-                         a.activation == 1);
-  const bool hum_shape = (a.D == 376 && a.h_max == 64 && a.act_dim == 17 &&
-                          a.n_hidden == 2 && a.rank == 16 &&
-                          a.activation == 1);
+  const bool full_tiles = (a.E % ENV_TILE) == 0;  // nE folds to 8
+  const bool hc_shape = full_tiles && (a.D == 17 && a.h_max == 64 &&
+                         a.act_dim == 6 && a.n_hidden == 2 &&
+                         a.rank == 16 && a.activation == 1);
+  const bool hum_shape = full_tiles && (a.D == 376 && a.h_max == 64 &&
+                          a.act_dim == 17 && a.n_hidden == 2 &&
+                          a.rank == 16 && a.activation == 1);
   if (mw >= 4) {
     if (hc_shape)
       hipLaunchKernelGGL((rollout_kernel<4, 17, 64, 6, 2, 16, 1>),
