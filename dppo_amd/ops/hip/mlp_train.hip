@@ -228,7 +228,7 @@ __global__ void mlp_chunk_kernel(ChunkArgs a) {
         for (int s = 0; s < ST; ++s) {
           if (s < st) {
             const float4 x4 = *reinterpret_cast<const float4*>(
-                &lds[in_off + s * in_stride + k]);
+              __builtin_assume_aligned(&lds[in_off + s * in_stride + k], 16));
             acc[s] += w0 * x4.x + w1 * x4.y + w2 * x4.z + w3 * x4.w;
           }
         }
@@ -406,7 +406,8 @@ __global__ void mlp_chunk_kernel(ChunkArgs a) {
         const int k = co * KC1 + j;
         if (k + 3 < D) {
           const float4 x4 =
-              *reinterpret_cast<const float4*>(&lds[m.x + s * m.dp + k]);
+              *reinterpret_cast<const float4*>(
+              __builtin_assume_aligned(&lds[m.x + s * m.dp + k], 16));
           acc1[j] += d1 * x4.x;
           acc1[j + 1] += d1 * x4.y;
           acc1[j + 2] += d1 * x4.z;
