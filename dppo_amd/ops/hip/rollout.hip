@@ -65,23 +65,23 @@ struct LdsMap {
   int xs, hs, total;  // x row stride, h row stride, total floats
 };
 
-DEV_INLINE LdsMap lds_map(int D, int Hmax, int A, int rank) {
+DEV_INLINE LdsMap lds_map(int D, int Hmax, int A, int rank, int et) {
   auto r4 = [](int v) { return (v + 3) & ~3; };
   LdsMap m;
   m.xs = r4(D);
   m.hs = r4(Hmax);
   int o = 0;
-  m.x = o; o += ENV_TILE * m.xs;
-  m.h0 = o; o += ENV_TILE * m.hs;
-  m.h1 = o; o += ENV_TILE * m.hs;
-  m.pd = o; o += ENV_TILE * r4(2 * A);
-  m.act = o; o += ENV_TILE * r4(A);
-  m.xv = o; o += ENV_TILE * r4(rank);
-  m.part = o; o += NWAVES * ENV_TILE * m.hs;
-  m.val = o; o += ENV_TILE;
-  m.rsum = o; o += ENV_TILE;
-  m.epr = o; o += ENV_TILE;
-  m.racc = o; o += ENV_TILE;
+  m.x = o; o += et * m.xs;
+  m.h0 = o; o += et * m.hs;
+  m.h1 = o; o += et * m.hs;
+  m.pd = o; o += et * r4(2 * A);
+  m.act = o; o += et * r4(A);
+  m.xv = o; o += et * r4(rank);
+  m.part = o; o += NWAVES * et * m.hs;
+  m.val = o; o += et;
+  m.rsum = o; o += et;
+  m.epr = o; o += et;
+  m.racc = o; o += et;
   m.total = o;
   return m;
 }
@@ -175,17 +175,17 @@ DEV_INLINE float u2f_mono(unsigned u) {
 // compile-time shapes (constant-folds the LDS map, index maps and trip
 // counts — the same instruction-count program as mlp_chunk_kernel;
 // profiles/r01_chunk_kernel_notes.md round-2 target list).
-template <int MINWAVES, int TD = 0, int TH = 0, int TA = 0, int TNH = 0,
-          int TRANK = 0, int TACT = 0>
+template <int MINWAVES, int ET = ENV_TILE, int TD = 0, int TH = 0,
+          int TA = 0, int TNH = 0, int TRANK = 0, int TACT = 0>
 __launch_bounds__(NWAVES * 64, MINWAVES)
 __global__ void rollout_kernel(RolloutArgs a) {
   const int tid = threadIdx.x;      // block = 4 waves of 64
   const int lane = tid & (WAVE - 1);
   const int wv = tid / WAVE;        // k-split wave index
-  const int e0 = blockIdx.x * ENV_TILE;
-  // specialized variants are dispatched only when E % ENV_TILE == 0:
+  const int e0 = blockIdx.x * ET;
+  // specialized variants are dispatched only when E % ET == 0:
   // nE folds and every per-env tail guard disappears
-  const int nE = TD ? ENV_TILE : min(ENV_TILE, a.E - e0);
+  const int nE = TD ? ET : min(ET, a.E - e0);
   const int D = TD ? TD : a.D;
   const int A = TA ? TA : a.act_dim;
   const int P = 2 * A;
@@ -196,7 +196,7 @@ __global__ void rollout_kernel(RolloutArgs a) {
   const int ACT = TACT ? (TACT > 0 ? 1 : 0) : a.activation;
 
   extern __shared__ __attribute__((aligned(16))) float lds[];
-  const LdsMap lm = lds_map(D, HMAX, A, RNK);
+  const LdsMap lm = lds_map(D, HMAX, A, RNK, ET);
   const int X_OFF = lm.x, H0_OFF = lm.h0, H1_OFF = lm.h1, PD_OFF = lm.pd;
   const int ACT_OFF = lm.act, XV_OFF = lm.xv, PART_OFF = lm.part;
   const int MAX_D_S = lm.xs, MAX_H_S = lm.hs;
@@ -206,8 +206,8 @@ __global__ void rollout_kernel(RolloutArgs a) {
   float* rsum_lds = &lds[lm.rsum];
   float* epr_lds = &lds[lm.epr];
   float* racc_lds = &lds[lm.racc];
-  __shared__ int tc_lds[ENV_TILE];
-  __shared__ int done_lds[ENV_TILE];
+  __shared__ int tc_lds[ET];
+  __shared__ int done_lds[ET];
 
   // env blob offsets (d | Vt[r][D] | U[r][D] | B[A][D])
   const float* env_d = a.envblob;
@@ -224,7 +224,7 @@ __global__ void rollout_kernel(RolloutArgs a) {
   float* out_boot = out_dones + (int64_t)T * E;
 
   // ---- load persistent state; zero tail envs ----
-  for (int e = 0; e < ENV_TILE; ++e) {
+  for (int e = 0; e < ET; ++e) {
     for (int d = tid; d < D; d += NWAVES * WAVE)
       lds[X_OFF + e * MAX_D_S + d] =
           (e < nE) ? a.x[(int64_t)(e0 + e) * D + d] : 0.f;
@@ -247,9 +247,9 @@ __global__ void rollout_kernel(RolloutArgs a) {
     const int k0 = wv * kq;
     const int k1 = min(in_dim, k0 + kq);
     for (int u = lane; u < out_dim; u += WAVE) {
-      float acc[ENV_TILE];
+      float acc[ET];
       #pragma unroll
-      for (int e = 0; e < ENV_TILE; ++e) acc[e] = 0.f;
+      for (int e = 0; e < ET; ++e) acc[e] = 0.f;
       const bool is_v = heads && (u == out_dim - 1);
       const float* Wcol = is_v ? (a.params + off_Wv_u) : (W + u);
       const int stride = is_v ? 1 : w_cols;
@@ -264,7 +264,7 @@ __global__ void rollout_kernel(RolloutArgs a) {
         const float w2 = Wcol[(int64_t)(k + 2) * stride];
         const float w3 = Wcol[(int64_t)(k + 3) * stride];
         #pragma unroll
-        for (int e = 0; e < ENV_TILE; ++e) {
+        for (int e = 0; e < ET; ++e) {
           const float4 x4 = *reinterpret_cast<const float4*>(
               __builtin_assume_aligned(&lds[in_off + e * in_stride + k], 16));
           acc[e] += w0 * x4.x + w1 * x4.y + w2 * x4.z + w3 * x4.w;
@@ -273,12 +273,12 @@ __global__ void rollout_kernel(RolloutArgs a) {
       for (; k < k1; ++k) {
         const float w = Wcol[(int64_t)k * stride];
         #pragma unroll
-        for (int e = 0; e < ENV_TILE; ++e)
+        for (int e = 0; e < ET; ++e)
           acc[e] += w * lds[in_off + e * in_stride + k];
       }
       #pragma unroll
-      for (int e = 0; e < ENV_TILE; ++e)
-        lds[PART_OFF + (wv * ENV_TILE + e) * MAX_H_S + u] = acc[e];
+      for (int e = 0; e < ET; ++e)
+        lds[PART_OFF + (wv * ET + e) * MAX_H_S + u] = acc[e];
     }
   };
 
@@ -336,12 +336,12 @@ __global__ void rollout_kernel(RolloutArgs a) {
                   in_dim, out_dim, false);
       __syncthreads();
       const float* bias = a.params + a.off_b[l];
-      for (int idx = tid; idx < ENV_TILE * out_dim; idx += NWAVES * WAVE) {
+      for (int idx = tid; idx < ET * out_dim; idx += NWAVES * WAVE) {
         const int e = idx / out_dim, u = idx % out_dim;
         float sum = bias[u];
         #pragma unroll
         for (int w = 0; w < NWAVES; ++w)
-          sum += lds[PART_OFF + (w * ENV_TILE + e) * MAX_H_S + u];
+          sum += lds[PART_OFF + (w * ET + e) * MAX_H_S + u];
         lds[out_off + e * MAX_H_S + u] =
             ACT ? fast_tanhf(sum) : fmaxf(sum, 0.f);
       }
@@ -361,12 +361,12 @@ __global__ void rollout_kernel(RolloutArgs a) {
       layer_kpart(a.params + a.off_Wp, P, a.off_Wv, in_off, in_stride, in_dim,
                   P + 1, true);
       __syncthreads();
-      for (int idx = tid; idx < ENV_TILE * (P + 1); idx += NWAVES * WAVE) {
+      for (int idx = tid; idx < ET * (P + 1); idx += NWAVES * WAVE) {
         const int e = idx / (P + 1), u = idx % (P + 1);
         float sum = a.params[(u == P) ? a.off_bv : a.off_bp + u];
         #pragma unroll
         for (int w = 0; w < NWAVES; ++w)
-          sum += lds[PART_OFF + (w * ENV_TILE + e) * MAX_H_S + u];
+          sum += lds[PART_OFF + (w * ET + e) * MAX_H_S + u];
         if (u == P) val_lds[e] = sum;
         else lds[PD_OFF + e * PD_S + u] = sum;
       }
@@ -391,7 +391,7 @@ __global__ void rollout_kernel(RolloutArgs a) {
         lds[ACT_OFF + e * ACT_S + j] = act;
       }
     }
-    if (tid < ENV_TILE) racc_lds[tid] = 0.f;
+    if (tid < ET) racc_lds[tid] = 0.f;
     __syncthreads();
 
     // ---- env state update + reward partials (threads split d) ----
@@ -405,19 +405,19 @@ __global__ void rollout_kernel(RolloutArgs a) {
       }
       if (tid < nE) out_values[(int64_t)step * E + e0 + tid] = val_lds[tid];
     }
-    float racc[ENV_TILE];
+    float racc[ET];
     #pragma unroll
-    for (int e = 0; e < ENV_TILE; ++e) racc[e] = 0.f;
+    for (int e = 0; e < ET; ++e) racc[e] = 0.f;
     if (!(a.ablate & 8))
     for (int d = tid; d < D; d += NWAVES * WAVE) {
       const float dd = env_d[d];
       // one Box-Muller pair per (even env, d) feeds two envs
-      float nz[ENV_TILE];
+      float nz[ET];
       #pragma unroll
-      for (int q = 0; q < ENV_TILE; ++q) nz[q] = 0.f;
+      for (int q = 0; q < ET; ++q) nz[q] = 0.f;
       if (!(a.ablate & 32)) {
         #pragma unroll
-        for (int q = 0; q < ENV_TILE / 2; ++q) {
+        for (int q = 0; q < ET / 2; ++q) {
           const float2 p = rng_normal2(a.seed, e0 + 2 * q, step, 1000 + d);
           nz[2 * q] = p.x;
           nz[2 * q + 1] = p.y;
@@ -427,9 +427,9 @@ __global__ void rollout_kernel(RolloutArgs a) {
       // load per rr/j shared by all four envs (the previous transposed
       // per-d rows cost 33 uncoalesced loads per env per d — the kernel
       // measured VMEM-wait bound, SQ_WAIT_ANY 27x SQ_BUSY).
-      float low[ENV_TILE], ain[ENV_TILE];
+      float low[ET], ain[ET];
       #pragma unroll
-      for (int e = 0; e < ENV_TILE; ++e) low[e] = ain[e] = 0.f;
+      for (int e = 0; e < ET; ++e) low[e] = ain[e] = 0.f;
       if (!(a.ablate & 64)) {
         // float4 LDS reads: 4x fewer LDS instructions than scalar
         // broadcasts (the scalar form measured 2.5 ms of the rollout)
@@ -440,7 +440,7 @@ __global__ void rollout_kernel(RolloutArgs a) {
           const float u2 = env_U[(int64_t)(rr + 2) * D + d];
           const float u3 = env_U[(int64_t)(rr + 3) * D + d];
           #pragma unroll
-          for (int e = 0; e < ENV_TILE; ++e) {
+          for (int e = 0; e < ET; ++e) {
             const float4 xv4 =
                 *reinterpret_cast<const float4*>(
               __builtin_assume_aligned(&lds[XV_OFF + e * XV_S + rr], 16));
@@ -450,7 +450,7 @@ __global__ void rollout_kernel(RolloutArgs a) {
         for (; rr < RNK; ++rr) {
           const float uv = env_U[(int64_t)rr * D + d];
           #pragma unroll
-          for (int e = 0; e < ENV_TILE; ++e)
+          for (int e = 0; e < ET; ++e)
             low[e] += lds[XV_OFF + e * XV_S + rr] * uv;
         }
         int j = 0;
@@ -460,7 +460,7 @@ __global__ void rollout_kernel(RolloutArgs a) {
           const float b2 = env_B[(int64_t)(j + 2) * D + d];
           const float b3 = env_B[(int64_t)(j + 3) * D + d];
           #pragma unroll
-          for (int e = 0; e < ENV_TILE; ++e) {
+          for (int e = 0; e < ET; ++e) {
             const float4 a4 =
                 *reinterpret_cast<const float4*>(
               __builtin_assume_aligned(&lds[ACT_OFF + e * ACT_S + j], 16));
@@ -470,7 +470,7 @@ __global__ void rollout_kernel(RolloutArgs a) {
         for (; j < A; ++j) {
           const float bvv = env_B[(int64_t)j * D + d];
           #pragma unroll
-          for (int e = 0; e < ENV_TILE; ++e)
+          for (int e = 0; e < ET; ++e)
             ain[e] += lds[ACT_OFF + e * ACT_S + j] * bvv;
         }
       }
@@ -482,7 +482,7 @@ __global__ void rollout_kernel(RolloutArgs a) {
       }
     }
     #pragma unroll
-    for (int e = 0; e < ENV_TILE; ++e) {
+    for (int e = 0; e < ET; ++e) {
       const float w = wave_reduce_sum(racc[e]);
       if (lane == 0 && e < nE) atomicAdd(&racc_lds[e], w);
     }
@@ -527,12 +527,12 @@ __global__ void rollout_kernel(RolloutArgs a) {
                   in_dim, out_dim, false);
       __syncthreads();
       const float* bias = a.params + a.off_b[l];
-      for (int idx = tid; idx < ENV_TILE * out_dim; idx += NWAVES * WAVE) {
+      for (int idx = tid; idx < ET * out_dim; idx += NWAVES * WAVE) {
         const int e = idx / out_dim, u = idx % out_dim;
         float sum = bias[u];
         #pragma unroll
         for (int w = 0; w < NWAVES; ++w)
-          sum += lds[PART_OFF + (w * ENV_TILE + e) * MAX_H_S + u];
+          sum += lds[PART_OFF + (w * ET + e) * MAX_H_S + u];
         lds[out_off + e * MAX_H_S + u] =
             ACT ? fast_tanhf(sum) : fmaxf(sum, 0.f);
       }
@@ -797,12 +797,18 @@ std::vector<torch::Tensor> rollout_run(
   a.out = out.data_ptr<float>();
 
   hipStream_t stream = c10::hip::getCurrentHIPStream().stream();
-  const int grid = static_cast<int>((E + ENV_TILE - 1) / ENV_TILE);
+  // tiny-E: 2 envs per block -> 4x the blocks (E=64 fills 32 CUs instead
+  // of 8) AND a ~4x shorter per-phase issue chain (the per-lane env
+  // unrolls shrink); large E keeps the 8-env tile (ILP beats occupancy
+  // once the grid fills the chip)
+  const int et = ((E + ENV_TILE - 1) / ENV_TILE < 128 && E % 2 == 0) ? 2
+                                                                     : ENV_TILE;
+  const int grid = static_cast<int>((E + et - 1) / et);
   auto r4 = [](int v) { return (v + 3) & ~3; };
   const size_t lds_bytes =
-      (ENV_TILE * (r4((int)D) + 2 * r4(h_max) + r4(2 * A) + r4(A) +
-                   r4((int)rank) + 4) +
-       NWAVES * ENV_TILE * r4(h_max)) *
+      (et * (r4((int)D) + 2 * r4(h_max) + r4(2 * A) + r4(A) +
+             r4((int)rank) + 4) +
+       NWAVES * et * r4(h_max)) *
       sizeof(float);
   // MINWAVES=4 measured 24.8 ms vs 27.7 ms at MINWAVES=3 (65k envs):
   // the 4th wave/SIMD buys more than the 29-VGPR spill costs.  Below
@@ -820,7 +826,7 @@ std::vector<torch::Tensor> rollout_run(
   const int mw = mw_env >= 0 ? mw_env : (grid <= N_CU ? 1 : 4);
   // exact-shape specializations for the BASELINE families (index maps,
   // trip counts and the LDS map constant-folded)
-  const bool full_tiles = (a.E % ENV_TILE) == 0;  // nE folds to 8
+  const bool full_tiles = (a.E % et) == 0;  // nE folds to the tile
   const bool hc_shape = full_tiles && (a.D == 17 && a.h_max == 64 &&
                          a.act_dim == 6 && a.n_hidden == 2 &&
                          a.rank == 16 && a.activation == 1);
@@ -828,29 +834,47 @@ std::vector<torch::Tensor> rollout_run(
                           a.act_dim == 17 && a.n_hidden == 2 &&
                           a.rank == 16 && a.activation == 1);
   if (mw >= 4) {
-    if (hc_shape)
-      hipLaunchKernelGGL((rollout_kernel<4, 17, 64, 6, 2, 16, 1>),
+    if (hc_shape && et == 8)
+      hipLaunchKernelGGL((rollout_kernel<4, 8, 17, 64, 6, 2, 16, 1>),
                          dim3(grid), dim3(NWAVES * WAVE), lds_bytes, stream,
                          a);
-    else if (hum_shape)
-      hipLaunchKernelGGL((rollout_kernel<4, 376, 64, 17, 2, 16, 1>),
+    else if (hum_shape && et == 8)
+      hipLaunchKernelGGL((rollout_kernel<4, 8, 376, 64, 17, 2, 16, 1>),
                          dim3(grid), dim3(NWAVES * WAVE), lds_bytes, stream,
                          a);
+    else if (et == 2)
+      hipLaunchKernelGGL((rollout_kernel<4, 2>), dim3(grid),
+                         dim3(NWAVES * WAVE), lds_bytes, stream, a);
     else
       hipLaunchKernelGGL(rollout_kernel<4>, dim3(grid), dim3(NWAVES * WAVE),
                          lds_bytes, stream, a);
   } else if (mw >= 2) {
-    hipLaunchKernelGGL(rollout_kernel<3>, dim3(grid), dim3(NWAVES * WAVE),
-                       lds_bytes, stream, a);
+    if (et == 2)
+      hipLaunchKernelGGL((rollout_kernel<3, 2>), dim3(grid),
+                         dim3(NWAVES * WAVE), lds_bytes, stream, a);
+    else
+      hipLaunchKernelGGL(rollout_kernel<3>, dim3(grid), dim3(NWAVES * WAVE),
+                         lds_bytes, stream, a);
   } else {
-    if (hc_shape)
-      hipLaunchKernelGGL((rollout_kernel<1, 17, 64, 6, 2, 16, 1>),
+    if (hc_shape && et == 2)
+      hipLaunchKernelGGL((rollout_kernel<1, 2, 17, 64, 6, 2, 16, 1>),
+                         dim3(grid), dim3(NWAVES * WAVE), lds_bytes, stream,
+                         a);
+    else if (hum_shape && et == 2)
+      hipLaunchKernelGGL((rollout_kernel<1, 2, 376, 64, 17, 2, 16, 1>),
+                         dim3(grid), dim3(NWAVES * WAVE), lds_bytes, stream,
+                         a);
+    else if (hc_shape)
+      hipLaunchKernelGGL((rollout_kernel<1, 8, 17, 64, 6, 2, 16, 1>),
                          dim3(grid), dim3(NWAVES * WAVE), lds_bytes, stream,
                          a);
     else if (hum_shape)
-      hipLaunchKernelGGL((rollout_kernel<1, 376, 64, 17, 2, 16, 1>),
+      hipLaunchKernelGGL((rollout_kernel<1, 8, 376, 64, 17, 2, 16, 1>),
                          dim3(grid), dim3(NWAVES * WAVE), lds_bytes, stream,
                          a);
+    else if (et == 2)
+      hipLaunchKernelGGL((rollout_kernel<1, 2>), dim3(grid),
+                         dim3(NWAVES * WAVE), lds_bytes, stream, a);
     else
       hipLaunchKernelGGL(rollout_kernel<1>, dim3(grid), dim3(NWAVES * WAVE),
                          lds_bytes, stream, a);
