@@ -367,6 +367,17 @@ class WideBF16Path:
                 ext.bf16_transpose(d_cur, dT_cur, grad, b_off[nH - 1],
                                    B, self.H[nH - 1], self.H[nH - 1], B)
             ev_dT = join()
+            # heads gradients are the LAST contiguous flat-grad region
+            # (vf.w | vf.b | pi.w | pi.b): all-reduce them as their own
+            # bucket as soon as they are complete, overlapping the rest
+            # of the backward (the wide config's ~276 MB f32
+            # grads are bandwidth-bound — bucket-per-layer overlap; the
+            # tiny flagship grads keep the single fused bucket)
+            per_layer_ar = eng.comm.distributed
+            if per_layer_ar:
+                if overlap:
+                    main.wait_event(ev_heads)
+                eng.comm.allreduce_mean_(grad[off_wv:])
             # hidden chain (uniform H: buffers are exact-size views)
             for l in range(nH - 1, -1, -1):
                 actT = u["xT"] if l == 0 else u["hT"][l - 1]
@@ -376,6 +387,11 @@ class WideBF16Path:
                     main.wait_event(ev_dT)
                 ext.bf16_mm256(dT_cur, actT, self._b, 3, self._f, self._b,
                                grad, w_off[l], self._b, 0, self._f, 0)
+                if per_layer_ar and l > 0:
+                    # W_l and b_l are adjacent in the flat layout; b_l's
+                    # colsum landed with dz_l's transpose, dW_l just now
+                    eng.comm.allreduce_mean_(
+                        grad[w_off[l]:b_off[l] + self.H[l]])
                 if l > 0:
                     # dz[l-1] = (dz[l] @ W[l]) * dtanh(h[l-1]); side
                     # transposes dz[l-1] (+ colsum -> db[l-1])
@@ -390,8 +406,13 @@ class WideBF16Path:
                                            B, self.H[l - 1], self.H[l - 1], B)
                     ev_dT = join()
                     d_cur, dT_cur = d_nxt, dT_nxt
-            if overlap:
+            if overlap and not per_layer_ar:
                 main.wait_event(ev_heads)
-            eng.comm.allreduce_mean_(grad)
+            if per_layer_ar:
+                # last bucket: layer 0's W/b (its bias colsum rides dz_0's
+                # transpose which precedes dW_0 in stream order)
+                eng.comm.allreduce_mean_(grad[w_off[0]:b_off[0] + self.H[0]])
+            else:
+                eng.comm.allreduce_mean_(grad)
             eng.optimizer.step_captured()  # lr_dev set by the caller
             self._weights_dirty = True
