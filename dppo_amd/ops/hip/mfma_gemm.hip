@@ -866,17 +866,22 @@ __global__ void dw_mfma_kernel(DwArgs a) {
 // ---------------------------------------------------------------------------
 
 constexpr int DWG_K = 16;        // k rows per stage
-constexpr int DWG_M = 64;        // out columns covered
-constexpr int DWG_N = 384;       // in columns covered (4 waves x NTW*32)
-constexpr int DWG_NTW = 3;       // 32-col tiles per wave
-constexpr int DWG_DB = DWG_K * DWG_M * 4;               // delta bytes/stage
-constexpr int DWG_AB = DWG_K * DWG_N * 4;               // acts bytes/stage
-constexpr int DWG_SLOT = DWG_DB + DWG_AB;               // 28 KiB
-constexpr int DWG_PIECES = DWG_SLOT / 16;               // 1792
-constexpr int DWG_PPT = 4;  // glds per thread per stage (2048 incl pad)
 
-__launch_bounds__(512, 2)
+// Generic over the block geometry: MW m-wave-tiles x NWV n-wave-groups
+// of NTW 32-col tiles (threads = MW*NWV*64).  <2,4,3> covers the wide
+// dW1 tile [64][384]; <2,2,1> covers the narrow [64][64] shapes (dW2,
+// padded heads dW) without idling 3/4 of the block.
+template <int MW, int NWV, int NTW>
+__launch_bounds__(MW * NWV * 64, 2)
 __global__ void dw_glds_kernel(DwArgs a) {
+  constexpr int THREADS = MW * NWV * 64;
+  constexpr int DWG_M = MW * 32;
+  constexpr int DWG_N = NWV * NTW * 32;
+  constexpr int DWG_DB = DWG_K * DWG_M * 4;
+  constexpr int DWG_AB = DWG_K * DWG_N * 4;
+  constexpr int DWG_SLOT = DWG_DB + DWG_AB;
+  constexpr int DWG_PIECES = DWG_SLOT / 16;
+  constexpr int DWG_PPT = (DWG_PIECES + THREADS - 1) / THREADS;
   __shared__ __attribute__((aligned(16))) char smem[3 * DWG_SLOT + 4096];
   char* const pad = smem + 3 * DWG_SLOT;  // dummy glds target (uniformity)
   const int tid = threadIdx.x;
@@ -884,8 +889,8 @@ __global__ void dw_glds_kernel(DwArgs a) {
   const int wave = tid >> 6;
   const int i_l = lane & 31;
   const int k_l = lane >> 5;
-  const int wm = wave >> 2;               // 2 m-tiles of 32
-  const int wn = wave & 3;                // 4 n-groups of 96
+  const int wm = wave / NWV;              // m-tile of 32
+  const int wn = wave % NWV;              // n-group of NTW*32
   const int split = blockIdx.x;
 
   const int64_t rows_per = (a.B + a.splits - 1) / a.splits;
@@ -894,15 +899,15 @@ __global__ void dw_glds_kernel(DwArgs a) {
   if (k0 >= a.B) return;
   const int S = (int)((k1 - k0 + DWG_K - 1) / DWG_K);
 
-  // per-thread glds sources: piece q = tid + p*512; rows past B clamp to
-  // B-1 (garbage — the delta-row zeroing kills their contribution)
+  // per-thread glds sources: piece q = tid + p*THREADS; rows past B
+  // clamp to B-1 (garbage — the delta-row zeroing kills them)
   const float* base[DWG_PPT];
   int krow_p[DWG_PPT];
   int ld_p[DWG_PPT];
   unsigned dsto[DWG_PPT];
   #pragma unroll
   for (int p = 0; p < DWG_PPT; ++p) {
-    const int q = tid + p * 512;
+    const int q = tid + p * THREADS;
     if (q < DWG_DB / 16) {               // delta piece
       const int kq = q / (DWG_M / 4);
       int c4 = (q % (DWG_M / 4)) * 4;
@@ -946,9 +951,9 @@ __global__ void dw_glds_kernel(DwArgs a) {
   issue(0);
   if (S > 1) issue(1);
 
-  f32x16 acc[DWG_NTW];
+  f32x16 acc[NTW];
   #pragma unroll
-  for (int t = 0; t < DWG_NTW; ++t)
+  for (int t = 0; t < NTW; ++t)
     #pragma unroll
     for (int r = 0; r < 16; ++r) acc[t][r] = 0.f;
   float dbacc = 0.f;
@@ -956,30 +961,30 @@ __global__ void dw_glds_kernel(DwArgs a) {
 
   for (int s = 0; s < S; ++s) {
     if (s + 1 < S)
-      asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+      asm volatile("s_waitcnt vmcnt(%0)" :: "i"(DWG_PPT) : "memory");
     else
       asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
     __builtin_amdgcn_s_barrier();
     char* slot = smem + (s % 3) * DWG_SLOT;
-    float* ds = (float*)slot;                    // [16][64] delta
-    float* as = (float*)(slot + DWG_DB);         // [16][384] acts
+    float* ds = (float*)slot;                    // [16][DWG_M] delta
+    float* as = (float*)(slot + DWG_DB);         // [16][DWG_N] acts
     const int64_t kb = k0 + (int64_t)s * DWG_K;
     const int kval = (int)((k1 - kb) < DWG_K ? (k1 - kb) : DWG_K);
     if (kval < DWG_K) {
-      // zero delta rows past the split (kills garbage contributions)
-      for (int idx = tid; idx < (DWG_K - kval) * DWG_M; idx += 512)
+      // zero the delta rows past the split (kills garbage contributions)
+      for (int idx = tid; idx < (DWG_K - kval) * DWG_M; idx += THREADS)
         ds[kval * DWG_M + idx] = 0.f;
       __builtin_amdgcn_s_barrier();
     }
-    if (s + 2 < S) issue(s + 2);
+    if (s + 2 < S) issue(s + 2);  // buf (s+2)%3 == (s-1)%3, freed above
     #pragma unroll 4
     for (int k2 = 0; k2 < DWG_K; k2 += 2) {
       const float av = ds[(k2 + k_l) * DWG_M + mcol];
       dbacc += av;
       #pragma unroll
-      for (int t = 0; t < DWG_NTW; ++t) {
+      for (int t = 0; t < NTW; ++t) {
         const float bv =
-            as[(k2 + k_l) * DWG_N + wn * (DWG_NTW * M_WAVE) + t * M_WAVE + i_l];
+            as[(k2 + k_l) * DWG_N + wn * (NTW * M_WAVE) + t * M_WAVE + i_l];
         acc[t] = __builtin_amdgcn_mfma_f32_32x32x2f32(av, bv, acc[t], 0, 0, 0);
       }
     }
@@ -988,8 +993,8 @@ __global__ void dw_glds_kernel(DwArgs a) {
   // ---- store to this split's slab ----
   float* slab = a.slab + (int64_t)split * a.out_dim * a.in_dim;
   #pragma unroll
-  for (int t = 0; t < DWG_NTW; ++t) {
-    const int col = wn * (DWG_NTW * M_WAVE) + t * M_WAVE + i_l;
+  for (int t = 0; t < NTW; ++t) {
+    const int col = wn * (NTW * M_WAVE) + t * M_WAVE + i_l;
     if (col < a.in_dim) {
       #pragma unroll
       for (int r = 0; r < 16; ++r) {
@@ -1256,11 +1261,13 @@ void dw_mfma(torch::Tensor delta, torch::Tensor acts, torch::Tensor grad_buf,
     return e ? atoi(e) : 1;
   }();
   // 8-wave LDS-staged variant: whole dW tile per block, delta read once
-  // only the wide-acts calls (dW1: in=obs) — at in_dim << DWG_N most of
-  // the block's n-waves compute discarded columns and the old splits
-  // grid wins (measured: dW1 562 vs 709 us, but dW2 regressed)
-  if (dwglds_env && !a.ablate && out_dim <= DWG_M && in_dim <= DWG_N &&
-      in_dim > 192 && out_dim % 4 == 0 && in_dim % 4 == 0 && B >= 4096) {
+  // <2,4,3> covers wide-acts tiles (dW1: in=obs); <2,2,1> covers the
+  // narrow [<=64][<=64] shapes (dW2, padded heads dW)
+  const bool glds_wide = in_dim > 192 && in_dim <= 384;
+  const bool glds_narrow = in_dim <= 64 && out_dim <= 64;
+  if (dwglds_env && !a.ablate && out_dim <= 64 &&
+      (glds_wide || glds_narrow) && out_dim % 4 == 0 && in_dim % 4 == 0 &&
+      B >= 4096) {
     a.splits = static_cast<int>(
         std::min<int64_t>(1024, std::max<int64_t>(256, B / 4096)));
     a.splits = static_cast<int>(
@@ -1272,8 +1279,12 @@ void dw_mfma(torch::Tensor delta, torch::Tensor acts, torch::Tensor grad_buf,
                                  delta.options());
     a.slab = slab2.data_ptr<float>();
     a.db_slab = db_slab2.data_ptr<float>();
-    hipLaunchKernelGGL(dw_glds_kernel, dim3((unsigned)a.splits), dim3(512),
-                       0, stream, a);
+    if (glds_wide)
+      hipLaunchKernelGGL((dw_glds_kernel<2, 4, 3>), dim3((unsigned)a.splits),
+                         dim3(512), 0, stream, a);
+    else
+      hipLaunchKernelGGL((dw_glds_kernel<2, 2, 1>), dim3((unsigned)a.splits),
+                         dim3(256), 0, stream, a);
     float* dW = grad_buf.data_ptr<float>() + w_off;
     float* db = (b_off >= 0) ? grad_buf.data_ptr<float>() + b_off : nullptr;
     float* dW2 =

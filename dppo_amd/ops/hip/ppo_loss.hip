@@ -330,9 +330,9 @@ __global__ void ppo_gh_kernel(
     const float* __restrict__ pdflat, const float* __restrict__ oldflat,
     const float* __restrict__ vpred, const float* __restrict__ oldv,
     const float* __restrict__ act, const float* __restrict__ adv,
-    const float* __restrict__ etr, float* __restrict__ gh,  // [B][2A+1]
+    const float* __restrict__ etr, float* __restrict__ gh,  // [B][ldgh]
     const float* __restrict__ clip_dev,  // nullptr -> use `clip` arg
-    int64_t B, int A, float clip, float entcoeff, float vcoeff) {
+    int64_t B, int A, int ldgh, float clip, float entcoeff, float vcoeff) {
   if (clip_dev != nullptr) clip = clip_dev[0];
   const int lane = threadIdx.x & (WAVE - 1);
   const int wave = threadIdx.x / WAVE;
@@ -386,11 +386,14 @@ __global__ void ppo_gh_kernel(
     const PPORowGrads g = ppo_row_grads(row, r.vp, r.ov, r.ad, r.et, B, clip,
                                         entcoeff, vcoeff, 1.f);
     if (lane < P) {
-      gh[b * (P + 1) + lane] =
+      gh[b * ldgh + lane] =
           (lane < A) ? g.g_logp * z * inv_s
                      : g.g_logp * (z * z - 1.f) + g.g_ent;
     }
-    if (lane == 0) gh[b * (P + 1) + P] = g.g_v;
+    if (lane == 0) gh[b * ldgh + P] = g.g_v;
+    // zero the pad columns (P+1..ldgh): downstream GEMM/dW treat gh as
+    // a [B][ldgh] operand and the pad must contribute exactly nothing
+    if (lane > P && lane < ldgh) gh[b * ldgh + lane] = 0.f;
   };
 
   Row ra, rc, re;
@@ -426,17 +429,14 @@ torch::Tensor ppo_loss_gauss_gh(torch::Tensor pdflat, torch::Tensor oldflat,
                                 torch::Tensor clip_dev) {
   const int64_t B = vpred.numel();
   const int A = static_cast<int>(pdflat.size(1) / 2);
-  TORCH_CHECK(2 * A + 1 <= WAVE + 1,
-              "ppo_loss_gauss_gh is wave-per-row (2A+1 <= 65 lanes); "
+  TORCH_CHECK(2 * A + 2 <= WAVE,
+              "ppo_loss_gauss_gh is wave-per-row (2A+1 <= 63 lanes); "
               "wide policies use the bf16 gh kernel");
-  // one float4 of slack past the end: the pipelined dgrad GEMM reads the
-  // [B][2A+1] rows with whole-float4 loads (K=2A+1 is odd), so the last
-  // row's tail load overhangs by up to 3 floats.  The overhang elements
-  // multiply W rows >= K, which the GEMM stages as zeros.
-  const int64_t ghn = B * (2 * (int64_t)A + 1);
-  auto gh = torch::empty({ghn + 4}, pdflat.options())
-                .narrow(0, 0, ghn)
-                .view({B, 2 * (int64_t)A + 1});
+  // row stride padded to a float4 multiple: 16-B-aligned rows let the
+  // dgrad GEMM and the glds dW kernel consume gh directly (the pad
+  // columns are zeroed in-kernel and contribute nothing downstream)
+  const int ldgh = (2 * A + 1 + 3) & ~3;
+  auto gh = torch::empty({B, (int64_t)ldgh}, pdflat.options());
   hipStream_t stream = c10::hip::getCurrentHIPStream().stream();
   const float* cd =
       (clip_dev.numel() > 0) ? clip_dev.data_ptr<float>() : nullptr;
@@ -445,6 +445,6 @@ torch::Tensor ppo_loss_gauss_gh(torch::Tensor pdflat, torch::Tensor oldflat,
                      vpred.data_ptr<float>(), oldv.data_ptr<float>(),
                      act.data_ptr<float>(), adv.data_ptr<float>(),
                      etr.data_ptr<float>(), gh.data_ptr<float>(), cd, B, A,
-                     (float)clip, (float)entcoeff, (float)vcoeff);
+                     ldgh, (float)clip, (float)entcoeff, (float)vcoeff);
   return gh;
 }

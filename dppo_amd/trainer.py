@@ -1405,8 +1405,12 @@ class DPPOEngine:
             a_views.append(acts.narrow(0, off, B * hh).view(B, hh))
             off += B * hh
         with torch.no_grad():
-            self._Wh_cat = torch.cat(
+            Wh_cat = torch.cat(
                 [self.pi.pi.weight, self.pi.vf.weight], dim=0).contiguous()
+            ldp = (Wh_cat.shape[0] + 3) & ~3
+            Wh_pad = torch.zeros(ldp, Wh_cat.shape[1], device=self.device)
+            Wh_pad[:Wh_cat.shape[0]] = Wh_cat
+            self._Wh_pad = Wh_pad
         return acts, a_views, batch.oldv, batch.oldflat
 
     def _fused_forward(self, states: torch.Tensor):
@@ -1438,7 +1442,13 @@ class DPPOEngine:
                 wts[l].copy_(lay.weight.detach().t())
             Wh_cat = torch.cat(
                 [self.pi.pi.weight, self.pi.vf.weight], dim=0
-            ).contiguous()  # [P+1][HL]: heads fwd (layout 1) AND dgrad Wt
+            ).contiguous()  # [P+1][HL]: heads fwd (layout 1)
+            # dgrad Wt padded to gh's float4 row stride (zero rows: the
+            # gh pad columns multiply them and contribute nothing)
+            ldp = (Wh_cat.shape[0] + 3) & ~3
+            Wh_pad = torch.zeros(ldp, Wh_cat.shape[1],
+                                 device=Wh_cat.device)
+            Wh_pad[:Wh_cat.shape[0]] = Wh_cat
             bh = torch.cat([self.pi.pi.bias, self.pi.vf.bias]).contiguous()
         total = sum(B * dims[l + 1] for l in range(n_hidden))
         acts = torch.empty(total, device=states.device, dtype=states.dtype)
@@ -1456,7 +1466,7 @@ class DPPOEngine:
         pdflat = torch.empty(B, P, device=states.device, dtype=states.dtype)
         v = torch.empty(B, device=states.device, dtype=states.dtype)
         ext.gemm_fwd(x, Wh_cat, bh, 2, 1, pdflat, v, pdflat, 1, 0, 0)
-        self._Wh_cat = Wh_cat
+        self._Wh_pad = Wh_pad
         return acts, a_views, v, pdflat
 
     def _fused_backward(self, states, acts, a_views, v, pdflat,
@@ -1480,15 +1490,19 @@ class DPPOEngine:
             pdflat, oldflat, v, oldv, actions, adv, etr,
             clip, c.ENTCOEFF, c.VCOEFF, self._clip_dev_or_empty(),
         )
-        # _fused_forward cached [Wp; Wv] — its rows are the dgrad Wt
-        Wh_cat = getattr(self, "_Wh_cat", None)
-        if Wh_cat is None:
+        # _fused_forward cached the PADDED [Wp; Wv; 0...] — its rows are
+        # the dgrad Wt matching gh's float4-padded row stride
+        Wh_pad = getattr(self, "_Wh_pad", None)
+        if Wh_pad is None or Wh_pad.shape[0] != gh.shape[1]:
             with torch.no_grad():
                 Wh_cat = torch.cat(
                     [self.pi.pi.weight, self.pi.vf.weight], dim=0
                 ).contiguous()
+                Wh_pad = torch.zeros(gh.shape[1], Wh_cat.shape[1],
+                                     device=self.device)
+                Wh_pad[:Wh_cat.shape[0]] = Wh_cat
         dz = [None] * n_hidden
-        delta, Wt_chain = gh, Wh_cat
+        delta, Wt_chain = gh, Wh_pad
         for l in range(n_hidden - 1, -1, -1):
             dz_l = torch.empty_like(a_views[l])
             ext.gemm_fwd(delta, Wt_chain, dummy_bias, dgrad_code, 0,
