@@ -1217,6 +1217,27 @@ void gemm_env_step(torch::Tensor xva, torch::Tensor M, torch::Tensor x,
                      dones.data_ptr<float>(), E, D, panels);
 }
 
+// Grow-only scratch cache for the dW split slabs.  Allocating them per
+// call fragmented the caching allocator against hipGraph private pools
+// (tests with several captured engines measured SECONDS per eager dW
+// call in hipMalloc/hipFree); a persistent buffer also gives captures a
+// stable address.  Correctness: every (split, element) of the used
+// region is overwritten each call (splits are clamped so k0 < B).
+static torch::Tensor& dw_scratch(int idx, int64_t need,
+                                 const torch::TensorOptions& opt) {
+  static torch::Tensor cache[2];
+  static std::vector<torch::Tensor> retired;  // graphs captured against an
+                                              // old buffer must keep it
+                                              // alive (grow-only, bounded
+                                              // by the few distinct sizes)
+  torch::Tensor& t = cache[idx];
+  if (!t.defined() || t.numel() < need) {
+    if (t.defined()) retired.push_back(t);
+    t = torch::empty({need}, opt);
+  }
+  return t;
+}
+
 void dw_mfma(torch::Tensor delta, torch::Tensor acts, torch::Tensor grad_buf,
              int64_t w_off, int64_t b_off, int64_t split_row, int64_t w_off2,
              int64_t b_off2, int64_t ablate) {
@@ -1248,10 +1269,13 @@ void dw_mfma(torch::Tensor delta, torch::Tensor acts, torch::Tensor grad_buf,
   a.splits = static_cast<int>(
       std::min<int64_t>(a.splits, std::max<int64_t>(1, B / 256)));
 
-  auto slab = torch::empty({(int64_t)a.splits, (int64_t)out_dim, (int64_t)in_dim},
-                           delta.options());
-  auto db_slab = torch::zeros({(int64_t)a.splits, (int64_t)out_dim},
-                              delta.options());
+  auto& slab = dw_scratch(0, (int64_t)a.splits * out_dim * in_dim,
+                          delta.options());
+  auto& db_slab = dw_scratch(1, (int64_t)a.splits * out_dim,
+                             delta.options());
+  // the old kernels' split grid also overwrites every used element, but
+  // a db tail beyond this call's splits must not leak: zero the region
+  db_slab.narrow(0, 0, (int64_t)a.splits * out_dim).zero_();
   a.slab = slab.data_ptr<float>();
   a.db_slab = db_slab.data_ptr<float>();
 
@@ -1272,11 +1296,10 @@ void dw_mfma(torch::Tensor delta, torch::Tensor acts, torch::Tensor grad_buf,
         std::min<int64_t>(1024, std::max<int64_t>(256, B / 4096)));
     a.splits = static_cast<int>(
         std::min<int64_t>(a.splits, std::max<int64_t>(1, B / DWG_K)));
-    auto slab2 = torch::empty(
-        {(int64_t)a.splits, (int64_t)out_dim, (int64_t)in_dim},
-        delta.options());
-    auto db_slab2 = torch::zeros({(int64_t)a.splits, (int64_t)out_dim},
-                                 delta.options());
+    auto& slab2 = dw_scratch(0, (int64_t)a.splits * out_dim * in_dim,
+                             delta.options());
+    auto& db_slab2 = dw_scratch(1, (int64_t)a.splits * out_dim,
+                                delta.options());
     a.slab = slab2.data_ptr<float>();
     a.db_slab = db_slab2.data_ptr<float>();
     if (glds_wide)
