@@ -293,3 +293,24 @@ def test_v3_recorded_acts_skip_is_bitwise_neutral(monkeypatch):
         sb, _ = b.train_round()
     torch.cuda.synchronize()
     assert torch.equal(a.flat_pi.flat_param, b.flat_pi.flat_param)
+
+
+def test_env_fused_epilogue_matches_default(monkeypatch):
+    """The DPPO_ENV_FUSED=1 path (env transition fused into the G-GEMM
+    epilogue, gemm_env_step) shares the default path's math and RNG slots
+    exactly — same seed must give bitwise-equal trajectories."""
+    monkeypatch.setenv("DPPO_ROLLOUT_V3", "1")
+    kw = dict(NUM_ENVS=64, MAX_EPOCH_STEPS=12, USE_GRAPHS=False, SEED=41)
+    torch.manual_seed(0)
+    a = make_engine(**kw)
+    monkeypatch.setenv("DPPO_ENV_FUSED", "1")
+    torch.manual_seed(0)
+    b = make_engine(**kw)
+    ba, _ = a.rollout_once()
+    monkeypatch.delenv("DPPO_ENV_FUSED")
+    bb, _ = b.rollout_once()
+    assert torch.equal(ba.states, bb.states)
+    assert torch.equal(ba.actions, bb.actions)
+    sa, sb = v3_views(a), v3_views(b)
+    assert torch.equal(sa[4], sb[4])  # rewards
+    assert torch.equal(sa[5], sb[5])  # dones
