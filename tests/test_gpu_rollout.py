@@ -303,12 +303,12 @@ def test_env_fused_epilogue_matches_default(monkeypatch):
     kw = dict(NUM_ENVS=64, MAX_EPOCH_STEPS=12, USE_GRAPHS=False, SEED=41)
     torch.manual_seed(0)
     a = make_engine(**kw)
-    monkeypatch.setenv("DPPO_ENV_FUSED", "1")
     torch.manual_seed(0)
     b = make_engine(**kw)
-    ba, _ = a.rollout_once()
+    ba, _ = a.rollout_once()           # default split env path
+    monkeypatch.setenv("DPPO_ENV_FUSED", "1")
+    bb, _ = b.rollout_once()           # fused epilogue path
     monkeypatch.delenv("DPPO_ENV_FUSED")
-    bb, _ = b.rollout_once()
     assert torch.equal(ba.states, bb.states)
     assert torch.equal(ba.actions, bb.actions)
     sa, sb = v3_views(a), v3_views(b)
