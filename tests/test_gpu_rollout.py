@@ -312,5 +312,10 @@ def test_env_fused_epilogue_matches_default(monkeypatch):
     assert torch.equal(ba.states, bb.states)
     assert torch.equal(ba.actions, bb.actions)
     sa, sb = v3_views(a), v3_views(b)
-    assert torch.equal(sa[4], sb[4])  # rewards
+    # Trajectories (states/actions/dones) are bitwise identical — the RNG
+    # slots and transition math match exactly.  Rewards alone are compared
+    # with a tolerance: the fused path accumulates the per-env reward sum
+    # as per-panel partials combined by env_finish2, a reassociation of
+    # env_finish's single-kernel sum (last-ulp differences only).
+    torch.testing.assert_close(sa[4], sb[4], rtol=1e-6, atol=1e-6)
     assert torch.equal(sa[5], sb[5])  # dones
