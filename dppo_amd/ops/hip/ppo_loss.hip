@@ -419,6 +419,127 @@ __global__ void ppo_gh_kernel(
   }
 }
 
+// LDS-tiled row-per-lane variant.  The wave-per-row kernel above keeps
+// only 2A+2 of 64 lanes active (36/64 on the flagship shape) and loads
+// pdflat twice per row — measured 0.8 TB/s.  Here a 256-thread block
+// stages a GH_TILE-row tile of every operand with full-width coalesced
+// loads into odd-stride (bank-conflict-free) LDS rows, then each of
+// GH_TILE lanes computes one row serially — no idle-lane guards and no
+// wave reductions (the serial j order matches ppo_gauss_row exactly).
+// gh rows are built in LDS, overwriting the lane's own oldflat row
+// (oldflat is consumed in the first pass only), and stored coalesced.
+constexpr int GH_TILE = 128;
+
+__launch_bounds__(256)
+__global__ void ppo_gh_tile_kernel(
+    const float* __restrict__ pdflat, const float* __restrict__ oldflat,
+    const float* __restrict__ vpred, const float* __restrict__ oldv,
+    const float* __restrict__ act, const float* __restrict__ adv,
+    const float* __restrict__ etr, float* __restrict__ gh,  // [B][ldgh]
+    const float* __restrict__ clip_dev,  // nullptr -> use `clip` arg
+    int64_t B, int A, int ldgh, float clip, float entcoeff, float vcoeff) {
+  if (clip_dev != nullptr) clip = clip_dev[0];
+  extern __shared__ float lds[];
+  const int P = 2 * A;
+  const int sp = P + 1;     // pdflat row stride (odd: P even)
+  const int sg = ldgh | 1;  // oldflat/gh shared row stride (>= P, odd)
+  const int sa = A | 1;     // act row stride (odd)
+  float* l_pd = lds;
+  float* l_og = l_pd + GH_TILE * sp;  // oldflat, later this tile's gh
+  float* l_ac = l_og + GH_TILE * sg;
+  float* l_sc = l_ac + GH_TILE * sa;  // [4][GH_TILE]: vpred|oldv|adv|etr
+  const int tid = threadIdx.x;
+  const int64_t tb = (int64_t)blockIdx.x * GH_TILE;
+  const int rows = (int)min((int64_t)GH_TILE, B - tb);
+
+  // stage: contiguous global streams -> padded LDS rows (row/col kept by
+  // increment — one div/mod per stream, not per element)
+  {
+    const int n = rows * P;
+    const int r0 = tid / P, c0 = tid % P;
+    const int dr = 256 / P, dc = 256 % P;
+    for (int g = tid, r = r0, c = c0; g < n; g += 256) {
+      l_pd[r * sp + c] = pdflat[tb * P + g];
+      r += dr; c += dc;
+      if (c >= P) { c -= P; r += 1; }
+    }
+    for (int g = tid, r = r0, c = c0; g < n; g += 256) {
+      l_og[r * sg + c] = oldflat[tb * P + g];
+      r += dr; c += dc;
+      if (c >= P) { c -= P; r += 1; }
+    }
+    const int na = rows * A;
+    const int dra = 256 / A, dca = 256 % A;
+    for (int g = tid, r = tid / A, c = tid % A; g < na; g += 256) {
+      l_ac[r * sa + c] = act[tb * A + g];
+      r += dra; c += dca;
+      if (c >= A) { c -= A; r += 1; }
+    }
+    for (int idx = tid; idx < 4 * GH_TILE; idx += 256) {
+      const int s = idx / GH_TILE, t = idx % GH_TILE;
+      if (t < rows) {
+        const float* src = s == 0 ? vpred : s == 1 ? oldv : s == 2 ? adv
+                                                                   : etr;
+        l_sc[s * GH_TILE + t] = src[tb + t];
+      }
+    }
+  }
+  __syncthreads();
+
+  // compute: one row per lane (rows <= 128 of 256 threads — the kernel
+  // is memory-bound, the idle compute lanes cost nothing)
+  if (tid < rows) {
+    const float* pd = l_pd + tid * sp;
+    const float* og = l_og + tid * sg;
+    const float* ac = l_ac + tid * sa;
+    float lp = 0.f, lo = 0.f, ent = 0.f;
+    for (int j = 0; j < A; ++j) {
+      const float aj = ac[j];
+      const float lsp = pd[A + j];
+      const float zp = (aj - pd[j]) * __expf(-lsp);
+      lp += -0.5f * zp * zp - lsp;
+      const float lso = og[A + j];
+      const float zo = (aj - og[j]) * __expf(-lso);
+      lo += -0.5f * zo * zo - lso;
+      ent += lsp;
+    }
+    const float cc = 0.5f * PPO_LOG_2PI * A;
+    GaussRow row;
+    row.logp_pi = lp - cc;
+    row.logp_old = lo - cc;
+    row.ent = ent + 0.5f * (PPO_LOG_2PI + 1.f) * A;
+    const PPORowGrads g =
+        ppo_row_grads(row, l_sc[tid], l_sc[GH_TILE + tid],
+                      l_sc[2 * GH_TILE + tid], l_sc[3 * GH_TILE + tid], B,
+                      clip, entcoeff, vcoeff, 1.f);
+    // overwrite this lane's own oldflat row (read only in the loop
+    // above, and only by this lane); z/inv_s recomputed from pd/ac so
+    // no runtime-indexed register arrays are kept (scratch-spill rule)
+    float* ghr = l_og + tid * sg;
+    for (int j = 0; j < A; ++j) {
+      const float lsp = pd[A + j];
+      const float inv_s = __expf(-lsp);
+      const float z = (ac[j] - pd[j]) * inv_s;
+      ghr[j] = g.g_logp * z * inv_s;
+      ghr[A + j] = g.g_logp * (z * z - 1.f) + g.g_ent;
+    }
+    ghr[P] = g.g_v;
+    for (int j = P + 1; j < ldgh; ++j) ghr[j] = 0.f;
+  }
+  __syncthreads();
+
+  // store: padded LDS gh rows -> contiguous [B][ldgh]
+  {
+    const int n = rows * ldgh;
+    const int dr = 256 / ldgh, dc = 256 % ldgh;
+    for (int g = tid, r = tid / ldgh, c = tid % ldgh; g < n; g += 256) {
+      gh[tb * ldgh + g] = l_og[r * sg + c];
+      r += dr; c += dc;
+      if (c >= ldgh) { c -= ldgh; r += 1; }
+    }
+  }
+}
+
 }  // namespace
 
 torch::Tensor ppo_loss_gauss_gh(torch::Tensor pdflat, torch::Tensor oldflat,
@@ -440,6 +561,23 @@ torch::Tensor ppo_loss_gauss_gh(torch::Tensor pdflat, torch::Tensor oldflat,
   hipStream_t stream = c10::hip::getCurrentHIPStream().stream();
   const float* cd =
       (clip_dev.numel() > 0) ? clip_dev.data_ptr<float>() : nullptr;
+  static const int gh_tile_env = []() {
+    const char* e = getenv("DPPO_GH_TILE");
+    return e ? atoi(e) : 1;
+  }();
+  const int lds_bytes =
+      GH_TILE * (2 * A + 1 + (ldgh | 1) + (A | 1) + 4) * (int)sizeof(float);
+  if (gh_tile_env && lds_bytes <= 65536) {
+    const int64_t grid = (B + GH_TILE - 1) / GH_TILE;
+    hipLaunchKernelGGL(ppo_gh_tile_kernel, dim3((unsigned)grid), dim3(256),
+                       lds_bytes, stream, pdflat.data_ptr<float>(),
+                       oldflat.data_ptr<float>(), vpred.data_ptr<float>(),
+                       oldv.data_ptr<float>(), act.data_ptr<float>(),
+                       adv.data_ptr<float>(), etr.data_ptr<float>(),
+                       gh.data_ptr<float>(), cd, B, A, ldgh, (float)clip,
+                       (float)entcoeff, (float)vcoeff);
+    return gh;
+  }
   hipLaunchKernelGGL(ppo_gh_kernel, dim3(2048), dim3(256), 0, stream,
                      pdflat.data_ptr<float>(), oldflat.data_ptr<float>(),
                      vpred.data_ptr<float>(), oldv.data_ptr<float>(),
