@@ -452,36 +452,50 @@ __global__ void ppo_gh_tile_kernel(
   const int64_t tb = (int64_t)blockIdx.x * GH_TILE;
   const int rows = (int)min((int64_t)GH_TILE, B - tb);
 
-  // stage: contiguous global streams -> padded LDS rows (row/col kept by
-  // increment — one div/mod per stream, not per element)
-  {
-    const int n = rows * P;
-    const int r0 = tid / P, c0 = tid % P;
-    const int dr = 256 / P, dc = 256 % P;
-    for (int g = tid, r = r0, c = c0; g < n; g += 256) {
-      l_pd[r * sp + c] = pdflat[tb * P + g];
-      r += dr; c += dc;
-      if (c >= P) { c -= P; r += 1; }
-    }
-    for (int g = tid, r = r0, c = c0; g < n; g += 256) {
-      l_og[r * sg + c] = oldflat[tb * P + g];
-      r += dr; c += dc;
-      if (c >= P) { c -= P; r += 1; }
-    }
-    const int na = rows * A;
-    const int dra = 256 / A, dca = 256 % A;
-    for (int g = tid, r = tid / A, c = tid % A; g < na; g += 256) {
-      l_ac[r * sa + c] = act[tb * A + g];
-      r += dra; c += dca;
-      if (c >= A) { c -= A; r += 1; }
-    }
-    for (int idx = tid; idx < 4 * GH_TILE; idx += 256) {
-      const int s = idx / GH_TILE, t = idx % GH_TILE;
-      if (t < rows) {
-        const float* src = s == 0 ? vpred : s == 1 ? oldv : s == 2 ? adv
-                                                                   : etr;
-        l_sc[s * GH_TILE + t] = src[tb + t];
+  // stage: contiguous global streams -> padded LDS rows.  Global side is
+  // float4 (the streams are row-major contiguous; tb*width is a multiple
+  // of 4 because GH_TILE is), LDS side is 4 scalar writes with an
+  // at-most-one row wrap per group (width >= 8 > 4).  row/col kept by
+  // increment — one div/mod per stream, not per element.
+  auto stage = [&](const float* __restrict__ src, float* __restrict__ dst,
+                   int width, int stride) {
+    const int n = rows * width;
+    if (width >= 8) {
+      const float4* s4 = reinterpret_cast<const float4*>(src);
+      const int n4 = n >> 2;
+      const int dr = 1024 / width, dc = 1024 % width;
+      int r = (tid * 4) / width, c = (tid * 4) % width;
+      for (int q = tid; q < n4; q += 256) {
+        const float4 v = s4[q];
+        int rr = r, cc = c;
+        const float vv[4] = {v.x, v.y, v.z, v.w};
+        #pragma unroll
+        for (int k = 0; k < 4; ++k) {
+          dst[rr * stride + cc] = vv[k];
+          if (++cc == width) { cc = 0; ++rr; }
+        }
+        r += dr; c += dc;
+        if (c >= width) { c -= width; ++r; }
       }
+      for (int g = (n4 << 2) + tid; g < n; g += 256)
+        dst[(g / width) * stride + g % width] = src[g];
+    } else {
+      const int dr = 256 / width, dc = 256 % width;
+      for (int g = tid, r = tid / width, c = tid % width; g < n; g += 256) {
+        dst[r * stride + c] = src[g];
+        r += dr; c += dc;
+        if (c >= width) { c -= width; r += 1; }
+      }
+    }
+  };
+  stage(pdflat + tb * P, l_pd, P, sp);
+  stage(oldflat + tb * P, l_og, P, sg);
+  stage(act + tb * A, l_ac, A, sa);
+  for (int idx = tid; idx < 4 * GH_TILE; idx += 256) {
+    const int s = idx / GH_TILE, t = idx % GH_TILE;
+    if (t < rows) {
+      const float* src = s == 0 ? vpred : s == 1 ? oldv : s == 2 ? adv : etr;
+      l_sc[s * GH_TILE + t] = src[tb + t];
     }
   }
   __syncthreads();
@@ -528,14 +542,25 @@ __global__ void ppo_gh_tile_kernel(
   }
   __syncthreads();
 
-  // store: padded LDS gh rows -> contiguous [B][ldgh]
+  // store: padded LDS gh rows -> contiguous [B][ldgh], float4 on the
+  // global side (ldgh is a multiple of 4, so n is too; wrap handling
+  // mirrors stage() — ldgh >= 4, so at most one wrap per group)
   {
-    const int n = rows * ldgh;
-    const int dr = 256 / ldgh, dc = 256 % ldgh;
-    for (int g = tid, r = tid / ldgh, c = tid % ldgh; g < n; g += 256) {
-      gh[tb * ldgh + g] = l_og[r * sg + c];
+    const int n4 = (rows * ldgh) >> 2;
+    float4* d4 = reinterpret_cast<float4*>(gh + tb * ldgh);
+    const int dr = 1024 / ldgh, dc = 1024 % ldgh;
+    int r = (tid * 4) / ldgh, c = (tid * 4) % ldgh;
+    for (int q = tid; q < n4; q += 256) {
+      float vv[4];
+      int rr = r, cc = c;
+      #pragma unroll
+      for (int k = 0; k < 4; ++k) {
+        vv[k] = l_og[rr * sg + cc];
+        if (++cc == ldgh) { cc = 0; ++rr; }
+      }
+      d4[q] = make_float4(vv[0], vv[1], vv[2], vv[3]);
       r += dr; c += dc;
-      if (c >= ldgh) { c -= ldgh; r += 1; }
+      if (c >= ldgh) { c -= ldgh; ++r; }
     }
   }
 }
