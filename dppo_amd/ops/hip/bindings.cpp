@@ -86,16 +86,17 @@ void rollout_sample(torch::Tensor pdflat, torch::Tensor actions,
                     torch::Tensor eps_dev, int64_t step, int64_t va_off,
                     double act_low, double act_high);
 
-void gemm_env_step(torch::Tensor xva, torch::Tensor M, torch::Tensor x,
-                   torch::Tensor envd, torch::Tensor horizons,
-                   torch::Tensor t, torch::Tensor epr, torch::Tensor snext,
+void gemm_env_step(torch::Tensor xva, torch::Tensor M, torch::Tensor xin,
+                   torch::Tensor xout, torch::Tensor envd,
+                   torch::Tensor horizons, torch::Tensor t,
+                   torch::Tensor epr,
                    torch::Tensor rewards, torch::Tensor dones,
                    torch::Tensor rsum, torch::Tensor seed_dev, double sigma,
                    int64_t step);
 
-void rollout_env_step(torch::Tensor x, torch::Tensor G, torch::Tensor envd,
-                      torch::Tensor horizons, torch::Tensor t,
-                      torch::Tensor epr, torch::Tensor states_next,
+void rollout_env_step(torch::Tensor xin, torch::Tensor xout, torch::Tensor G,
+                      torch::Tensor envd, torch::Tensor horizons,
+                      torch::Tensor t, torch::Tensor epr,
                       torch::Tensor rewards, torch::Tensor dones,
                       torch::Tensor seed_dev, double sigma, int64_t step);
 

@@ -82,14 +82,16 @@ struct FwdArgs {
   // accumulating per-env pred^2 partials for the reward.  Kills the
   // 2x[E][D] G round trip and most of env_finish_kernel (round-2 lever
   // #3, profiles/r01_final_v3_69M.txt).  RNG slots == env_finish's.
-  float* env_x;             // [E][D] in/out
+  const float* env_xin;     // [E][D] state at `step`
+  float* env_x;             // [E][D] out: state at `step`+1 (the v3 loop
+                            // passes the states-blob slot directly; may
+                            // alias env_xin)
   const float* envd;        // [D]
   const int* horizons;      // [E]
   const int* tcount;        // [E] (read; env_finish2 increments)
   const long long* seed_dev;
   float* env_rsum;          // [npanels][E] per-panel pred^2 partials
                             // (plain stores — deterministic, no zeroing)
-  float* snext;             // [E][D] states[st+1] or null
   float sigma;
   int step;
 };
@@ -485,7 +487,10 @@ __global__ void gemm_fwd_pipe_kernel(FwdArgs a) {
 // address (guide rule 21) — same 64 B cacheline, zero coalescing cost.
 // ---------------------------------------------------------------------------
 
-__launch_bounds__(FWD_WAVES * 64, 4)
+template <int RING>  // ring depth: 3 (36 KB, 4 blocks/CU) or 4 (48 KB,
+                     // 3 blocks/CU — two extra stages in flight cover
+                     // loaded-chip HBM latency the 3-deep ring stalls on)
+__launch_bounds__(FWD_WAVES * 64, RING == 3 ? 4 : 3)
 __global__ void gemm_fwd_glds_kernel(FwdArgs a) {
   constexpr int NT = 2;
   constexpr int PBK = 16;
@@ -493,7 +498,7 @@ __global__ void gemm_fwd_glds_kernel(FwdArgs a) {
   constexpr int XB = FWD_M * PBK * 4;      // 8192 B
   constexpr int WB = PBK * NW * 4;         // 4096 B
   constexpr int SLOT = XB + WB;
-  __shared__ __attribute__((aligned(16))) char smem[3 * SLOT];
+  __shared__ __attribute__((aligned(16))) char smem[RING * SLOT];
   const int tid = threadIdx.x;
   const int lane = tid & (WAVE - 1);
   const int wave = tid / WAVE;
@@ -532,7 +537,7 @@ __global__ void gemm_fwd_glds_kernel(FwdArgs a) {
 
   auto issue = [&](int s) {
     const int kb = s * PBK;
-    char* slot = smem + (s % 3) * SLOT;
+    char* slot = smem + (s % RING) * SLOT;
     #pragma unroll
     for (int p = 0; p < 2; ++p) {
       int col = kb + xcol[p];
@@ -550,8 +555,9 @@ __global__ void gemm_fwd_glds_kernel(FwdArgs a) {
   };
 
   const int S = (a.K + PBK - 1) / PBK;
-  issue(0);
-  if (S > 1) issue(1);
+  #pragma unroll
+  for (int p = 0; p < RING - 1; ++p)
+    if (p < S) issue(p);
 
   f32x16 acc[NT];
   #pragma unroll
@@ -560,13 +566,17 @@ __global__ void gemm_fwd_glds_kernel(FwdArgs a) {
     for (int r = 0; r < 16; ++r) acc[t][r] = 0.f;
 
   for (int s = 0; s < S; ++s) {
-    // stage s landed when only stage s+1's 3 glds remain outstanding
-    if (s + 1 < S)
+    // stage s landed when only the later in-flight stages' glds (3 each)
+    // remain outstanding: rem = how many stages beyond s are in flight
+    const int rem = min(S - 1 - s, RING - 2);
+    if (rem >= RING - 2)
+      asm volatile("s_waitcnt vmcnt(%0)" ::"i"(3 * (RING - 2)) : "memory");
+    else if (RING == 4 && rem == 1)
       asm volatile("s_waitcnt vmcnt(3)" ::: "memory");
     else
       asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
     __builtin_amdgcn_s_barrier();
-    char* slot = smem + (s % 3) * SLOT;
+    char* slot = smem + (s % RING) * SLOT;
     float* xs = (float*)slot;
     float* ws = (float*)(slot + XB);
     const int kb = s * PBK;
@@ -577,7 +587,8 @@ __global__ void gemm_fwd_glds_kernel(FwdArgs a) {
         ws[(a.K - kb) * NW + idx] = 0.f;
       __builtin_amdgcn_s_barrier();
     }
-    if (s + 2 < S) issue(s + 2);  // buf (s+2)%3 == (s-1)%3, freed above
+    // buf (s+RING-1)%RING == (s-1)%RING, freed by the barrier above
+    if (s + RING - 1 < S) issue(s + RING - 1);
     #pragma unroll 4
     for (int k2 = 0; k2 < PBK; k2 += 2) {
       const int k = k2 + k_l;
@@ -610,7 +621,7 @@ __global__ void gemm_fwd_glds_kernel(FwdArgs a) {
         const int col = n0 + t * M_WAVE + i_l;
         if (rok && col < a.N) {
           const float dv = a.envd[col];
-          const float xv = a.env_x[row * a.N + col];
+          const float xv = a.env_xin[row * a.N + col];
           float nz = 0.f;
           if (a.sigma != 0.f) {
             const float2 pr =
@@ -623,7 +634,6 @@ __global__ void gemm_fwd_glds_kernel(FwdArgs a) {
               done ? 0.1f * g_rng_normal(seed, (int)row, a.step, 5000 + col)
                    : pred;
           a.env_x[row * a.N + col] = out;
-          if (a.snext != nullptr) a.snext[row * a.N + col] = out;
         }
       }
       #pragma unroll
@@ -1139,9 +1149,18 @@ void gemm_fwd(torch::Tensor X, torch::Tensor Wt, torch::Tensor bias,
   const bool glds_ok = glds_env && !a.ablate && wt_layout == 0 &&
                        K % 4 == 0 && K >= 4 && N % 4 == 0 && N >= 4 &&
                        a.activation != 9999;
+  static const bool ring4 = []() {
+    const char* e = getenv("DPPO_FWD_RING");
+    return e && atoi(e) == 4;  // measured slower at the flagship shapes
+                               // (occupancy 3 vs 4 blocks/CU); opt-in
+  }();
   #define DISPATCH_FWD(NTV, NPANEL)                                          \
-    if (glds_ok && NTV == 2)                                                 \
-      hipLaunchKernelGGL(gemm_fwd_glds_kernel,                               \
+    if (glds_ok && NTV == 2 && ring4)                                        \
+      hipLaunchKernelGGL(gemm_fwd_glds_kernel<4>,                            \
+                         dim3((unsigned)grid_pipe, NPANEL),                  \
+                         dim3(FWD_WAVES * 64), 0, stream, a);                \
+    else if (glds_ok && NTV == 2)                                            \
+      hipLaunchKernelGGL(gemm_fwd_glds_kernel<3>,                            \
                          dim3((unsigned)grid_pipe, NPANEL),                  \
                          dim3(FWD_WAVES * 64), 0, stream, a);                \
     else if (pipe && pbk == 32)                                              \
@@ -1172,17 +1191,20 @@ void gemm_fwd(torch::Tensor X, torch::Tensor Wt, torch::Tensor bias,
   #undef DISPATCH_FWD
 }
 
-void gemm_env_step(torch::Tensor xva, torch::Tensor M, torch::Tensor x,
-                   torch::Tensor envd, torch::Tensor horizons,
-                   torch::Tensor t, torch::Tensor epr, torch::Tensor snext,
+void gemm_env_step(torch::Tensor xva, torch::Tensor M, torch::Tensor xin,
+                   torch::Tensor xout, torch::Tensor envd,
+                   torch::Tensor horizons, torch::Tensor t,
+                   torch::Tensor epr,
                    torch::Tensor rewards, torch::Tensor dones,
                    torch::Tensor rsum, torch::Tensor seed_dev, double sigma,
                    int64_t step) {
   const int64_t E = xva.size(0);
   const int K = static_cast<int>(xva.size(1));
   const int D = static_cast<int>(M.size(1));
-  TORCH_CHECK(xva.is_contiguous() && M.is_contiguous() && x.is_contiguous());
-  TORCH_CHECK(M.size(0) == K && x.size(0) == E && x.size(1) == D);
+  TORCH_CHECK(xva.is_contiguous() && M.is_contiguous() &&
+              xin.is_contiguous() && xout.is_contiguous());
+  TORCH_CHECK(M.size(0) == K && xin.size(0) == E && xin.size(1) == D);
+  TORCH_CHECK(xout.numel() == E * D);
   TORCH_CHECK(K % 4 == 0 && D % 4 == 0, "fused env step needs 16B rows");
   TORCH_CHECK(horizons.dtype() == torch::kInt32 && t.dtype() == torch::kInt32);
   const int panels_chk = (D + 63) / 64;
@@ -1195,19 +1217,19 @@ void gemm_env_step(torch::Tensor xva, torch::Tensor M, torch::Tensor x,
   a.N = D;
   a.activation = 5;
   a.ldc = D;
-  a.env_x = x.data_ptr<float>();
+  a.env_xin = xin.data_ptr<float>();
+  a.env_x = xout.data_ptr<float>();
   a.envd = envd.data_ptr<float>();
   a.horizons = horizons.data_ptr<int>();
   a.tcount = t.data_ptr<int>();
   a.seed_dev = reinterpret_cast<const long long*>(seed_dev.data_ptr<int64_t>());
   a.env_rsum = rsum.data_ptr<float>();
-  a.snext = snext.numel() ? snext.data_ptr<float>() : nullptr;
   a.sigma = (float)sigma;
   a.step = (int)step;
   hipStream_t stream = c10::hip::getCurrentHIPStream().stream();
   const int64_t tiles = (E + FWD_M - 1) / FWD_M;
   const int panels = (D + 63) / 64;
-  hipLaunchKernelGGL(gemm_fwd_glds_kernel,
+  hipLaunchKernelGGL(gemm_fwd_glds_kernel<3>,
                      dim3((unsigned)tiles, (unsigned)panels),
                      dim3(FWD_WAVES * 64), 0, stream, a);
   hipLaunchKernelGGL(env_finish2_kernel, dim3(elementwise_grid(E, 256)),
@@ -1292,8 +1314,10 @@ void dw_mfma(torch::Tensor delta, torch::Tensor acts, torch::Tensor grad_buf,
   if (dwglds_env && !a.ablate && out_dim <= 64 &&
       (glds_wide || glds_narrow) && out_dim % 4 == 0 && in_dim % 4 == 0 &&
       B >= 4096) {
+    const char* sc_env = getenv("DPPO_DW_SPLITS");  // re-read: sweepable
+    const int64_t split_cap = sc_env ? (int64_t)atoll(sc_env) : (int64_t)1024;
     a.splits = static_cast<int>(
-        std::min<int64_t>(1024, std::max<int64_t>(256, B / 4096)));
+        std::min<int64_t>(split_cap, std::max<int64_t>(256, B / 4096)));
     a.splits = static_cast<int>(
         std::min<int64_t>(a.splits, std::max<int64_t>(1, B / DWG_K)));
     auto& slab2 = dw_scratch(0, (int64_t)a.splits * out_dim * in_dim,

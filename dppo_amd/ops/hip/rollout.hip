@@ -645,13 +645,16 @@ __global__ void sample_kernel(const float* __restrict__ pdflat,  // [E][2A]
 // wave per env: x' = tanh(x*d + G + sigma*noise), reward, done/reset and
 // episode bookkeeping.  G = [x@V | act] @ [U; B] comes from the GEMM.
 __global__ void env_finish_kernel(
-    float* __restrict__ x,             // [E][D] in/out
+    const float* __restrict__ xin,     // [E][D] state at `step`
+    float* __restrict__ xout,          // [E][D] state at `step`+1 (the v3
+                                       // loop passes the states-blob slot
+                                       // directly — one stream, no env.x
+                                       // mirror write; xout may alias xin)
     const float* __restrict__ G,       // [E][D]
     const float* __restrict__ envd,    // [D] diagonal
     const int* __restrict__ horizons,  // [E]
     int* __restrict__ t,               // [E]
     float* __restrict__ epr,           // [E]
-    float* __restrict__ states_next,   // [E][D] out_states[step+1] or null
     float* __restrict__ rewards,       // [E] out_rewards[step]
     float* __restrict__ dones,         // [E] out_dones[step]
     const long* __restrict__ seed_dev, float sigma, int step, int64_t E,
@@ -674,7 +677,7 @@ __global__ void env_finish_kernel(
   int it = 0;
   for (int c = lane; c < nc4; c += WAVE, ++it) {
     const int d = 4 * c;
-    const float4 xv = *reinterpret_cast<const float4*>(&x[e * D + d]);
+    const float4 xv = *reinterpret_cast<const float4*>(&xin[e * D + d]);
     const float4 gv = *reinterpret_cast<const float4*>(&G[e * D + d]);
     const float4 dv = *reinterpret_cast<const float4*>(&envd[d]);
     float nz[4] = {0.f, 0.f, 0.f, 0.f};
@@ -718,9 +721,7 @@ __global__ void env_finish_kernel(
       v.z = 0.1f * rng_normal(seed, (int)e, step, 5000 + d + 2);
       v.w = 0.1f * rng_normal(seed, (int)e, step, 5000 + d + 3);
     }
-    *reinterpret_cast<float4*>(&x[e * D + d]) = v;
-    if (states_next != nullptr)
-      *reinterpret_cast<float4*>(&states_next[e * D + d]) = v;
+    *reinterpret_cast<float4*>(&xout[e * D + d]) = v;
   }
 }
 
@@ -926,14 +927,15 @@ void rollout_sample(torch::Tensor pdflat, torch::Tensor actions,
                      (int)va_off, (float)act_low, (float)act_high);
 }
 
-void rollout_env_step(torch::Tensor x, torch::Tensor G, torch::Tensor envd,
-                      torch::Tensor horizons, torch::Tensor t,
-                      torch::Tensor epr, torch::Tensor states_next,
+void rollout_env_step(torch::Tensor xin, torch::Tensor xout, torch::Tensor G,
+                      torch::Tensor envd, torch::Tensor horizons,
+                      torch::Tensor t, torch::Tensor epr,
                       torch::Tensor rewards, torch::Tensor dones,
                       torch::Tensor seed_dev, double sigma, int64_t step) {
-  const int64_t E = x.size(0);
-  const int D = static_cast<int>(x.size(1));
-  TORCH_CHECK(x.is_cuda() && x.is_contiguous() && G.numel() >= E * D);
+  const int64_t E = xin.size(0);
+  const int D = static_cast<int>(xin.size(1));
+  TORCH_CHECK(xin.is_cuda() && xin.is_contiguous() && G.numel() >= E * D);
+  TORCH_CHECK(xout.is_contiguous() && xout.numel() == E * D);
   TORCH_CHECK(D <= MAX_D && D % 4 == 0, "env_finish_kernel is float4-wide");
   TORCH_CHECK(horizons.dtype() == torch::kInt32 && t.dtype() == torch::kInt32);
   hipStream_t stream = c10::hip::getCurrentHIPStream().stream();
@@ -941,12 +943,10 @@ void rollout_env_step(torch::Tensor x, torch::Tensor G, torch::Tensor envd,
   const int64_t blocks = (waves * WAVE + 255) / 256;
   hipLaunchKernelGGL(env_finish_kernel,
                      dim3((unsigned)std::min<int64_t>(blocks, 1 << 26)),
-                     dim3(256), 0, stream, x.data_ptr<float>(),
-                     G.data_ptr<float>(), envd.data_ptr<float>(),
-                     horizons.data_ptr<int>(), t.data_ptr<int>(),
-                     epr.data_ptr<float>(),
-                     states_next.numel() ? states_next.data_ptr<float>()
-                                         : nullptr,
+                     dim3(256), 0, stream, xin.data_ptr<float>(),
+                     xout.data_ptr<float>(), G.data_ptr<float>(),
+                     envd.data_ptr<float>(), horizons.data_ptr<int>(),
+                     t.data_ptr<int>(), epr.data_ptr<float>(),
                      rewards.data_ptr<float>(), dones.data_ptr<float>(),
                      seed_dev.data_ptr<int64_t>(), (float)sigma, (int)step, E,
                      D);

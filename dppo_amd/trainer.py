@@ -558,7 +558,12 @@ class DPPOEngine:
             ext.rollout_sample(pdflats[st], actions[st], v3["xva"],
                                v3["seed_dev"], v3["eps_dev"], st,
                                v3["va_off"], low, high)
-            nxt = states[st + 1] if st + 1 < T else v3["empty"]
+            # the env transition reads states[st] and writes states[st+1]
+            # directly (policy input already comes from the blob) — env.x
+            # only receives the FINAL state, for the bootstrap forward and
+            # the next round's states[0]: one [E][D] stream per step, no
+            # mirror write
+            xout = states[st + 1] if st + 1 < T else env.x
             if self._env_fused():
                 # G = [XV|act] @ [U;B] with the env transition fused into
                 # the GEMM epilogue (identical math and RNG slots).  Saves
@@ -566,15 +571,15 @@ class DPPOEngine:
                 # (the separate env_finish shares one hash per dim pair):
                 # measured net -4ms/round at the flagship config, so the
                 # split path stays the default (DPPO_ENV_FUSED=1 opts in).
-                ext.gemm_env_step(v3["xva"], v3["M"], env.x, env.d,
-                                  env.horizons_i32, env.t, self.epr, nxt,
+                ext.gemm_env_step(v3["xva"], v3["M"], xin, xout, env.d,
+                                  env.horizons_i32, env.t, self.epr,
                                   rewards[st], dones[st], v3["rsum"],
                                   v3["seed_dev"], float(env.NOISE), st)
             else:
                 ext.gemm_fwd(v3["xva"], v3["M"], v3["bz_D"], 2, 0, v3["G"],
                              v3["G"], v3["G"], 0, 0, 0)
-                ext.rollout_env_step(env.x, v3["G"], env.d,
-                                     env.horizons_i32, env.t, self.epr, nxt,
+                ext.rollout_env_step(xin, xout, v3["G"], env.d,
+                                     env.horizons_i32, env.t, self.epr,
                                      rewards[st], dones[st], v3["seed_dev"],
                                      float(env.NOISE), st)
         # bootstrap value V(x_T)
