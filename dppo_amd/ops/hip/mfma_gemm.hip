@@ -881,7 +881,7 @@ constexpr int DWG_K = 16;        // k rows per stage
 // of NTW 32-col tiles (threads = MW*NWV*64).  <2,4,3> covers the wide
 // dW1 tile [64][384]; <2,2,1> covers the narrow [64][64] shapes (dW2,
 // padded heads dW) without idling 3/4 of the block.
-template <int MW, int NWV, int NTW>
+template <int MW, int NWV, int NTW, int RING = 3>
 __launch_bounds__(MW * NWV * 64, 2)
 __global__ void dw_glds_kernel(DwArgs a) {
   constexpr int THREADS = MW * NWV * 64;
@@ -892,8 +892,11 @@ __global__ void dw_glds_kernel(DwArgs a) {
   constexpr int DWG_SLOT = DWG_DB + DWG_AB;
   constexpr int DWG_PIECES = DWG_SLOT / 16;
   constexpr int DWG_PPT = (DWG_PIECES + THREADS - 1) / THREADS;
-  __shared__ __attribute__((aligned(16))) char smem[3 * DWG_SLOT + 4096];
-  char* const pad = smem + 3 * DWG_SLOT;  // dummy glds target (uniformity)
+  // RING: the wide <2,4,3> tile is 28 KB/stage -> 1 block/CU at ANY ring
+  // depth, so a 5-deep ring (4 stages = 112 KB in flight) is free and
+  // covers loaded-chip HBM latency the 3-deep ring stalled on
+  __shared__ __attribute__((aligned(16))) char smem[RING * DWG_SLOT + 4096];
+  char* const pad = smem + RING * DWG_SLOT;  // dummy glds target (uniformity)
   const int tid = threadIdx.x;
   const int lane = tid & (WAVE - 1);
   const int wave = tid >> 6;
@@ -902,6 +905,10 @@ __global__ void dw_glds_kernel(DwArgs a) {
   const int wm = wave / NWV;              // m-tile of 32
   const int wn = wave % NWV;              // n-group of NTW*32
   const int split = blockIdx.x;
+  // n-panel (grid.y): narrow tiles with small per-stage LDS let 2 blocks
+  // co-reside per CU so stage-barrier bubbles overlap across blocks; the
+  // (small) delta stage is duplicated per panel
+  const int nb0 = (int)blockIdx.y * DWG_N;
 
   const int64_t rows_per = (a.B + a.splits - 1) / a.splits;
   const int64_t k0 = (int64_t)split * rows_per;
@@ -929,7 +936,7 @@ __global__ void dw_glds_kernel(DwArgs a) {
     } else if (q < DWG_PIECES) {         // acts piece
       const int qq = q - DWG_DB / 16;
       const int kq = qq / (DWG_N / 4);
-      int c4 = (qq % (DWG_N / 4)) * 4;
+      int c4 = nb0 + (qq % (DWG_N / 4)) * 4;
       c4 = c4 + 3 < a.in_dim ? c4 : (a.in_dim > 4 ? a.in_dim - 4 : 0);
       base[p] = a.acts + c4;
       krow_p[p] = kq;
@@ -944,7 +951,7 @@ __global__ void dw_glds_kernel(DwArgs a) {
   }
 
   auto issue = [&](int s) {
-    char* slot = smem + (s % 3) * DWG_SLOT;
+    char* slot = smem + (s % RING) * DWG_SLOT;
     #pragma unroll
     for (int p = 0; p < DWG_PPT; ++p) {
       int64_t row = k0 + (int64_t)s * DWG_K + krow_p[p];
@@ -958,8 +965,9 @@ __global__ void dw_glds_kernel(DwArgs a) {
     }
   };
 
-  issue(0);
-  if (S > 1) issue(1);
+  #pragma unroll
+  for (int p = 0; p < RING - 1; ++p)
+    if (p < S) issue(p);
 
   f32x16 acc[NTW];
   #pragma unroll
@@ -970,12 +978,19 @@ __global__ void dw_glds_kernel(DwArgs a) {
   const int mcol = wm * M_WAVE + i_l;
 
   for (int s = 0; s < S; ++s) {
-    if (s + 1 < S)
+    // stage s landed when only the rem in-flight later stages' loads
+    // (DWG_PPT each) remain outstanding
+    const int rem = min(S - 1 - s, RING - 2);
+    if (RING >= 5 && rem >= 3)
+      asm volatile("s_waitcnt vmcnt(%0)" :: "i"(3 * DWG_PPT) : "memory");
+    else if (RING >= 4 && rem == 2)
+      asm volatile("s_waitcnt vmcnt(%0)" :: "i"(2 * DWG_PPT) : "memory");
+    else if (rem == 1)
       asm volatile("s_waitcnt vmcnt(%0)" :: "i"(DWG_PPT) : "memory");
     else
       asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
     __builtin_amdgcn_s_barrier();
-    char* slot = smem + (s % 3) * DWG_SLOT;
+    char* slot = smem + (s % RING) * DWG_SLOT;
     float* ds = (float*)slot;                    // [16][DWG_M] delta
     float* as = (float*)(slot + DWG_DB);         // [16][DWG_N] acts
     const int64_t kb = k0 + (int64_t)s * DWG_K;
@@ -986,7 +1001,8 @@ __global__ void dw_glds_kernel(DwArgs a) {
         ds[kval * DWG_M + idx] = 0.f;
       __builtin_amdgcn_s_barrier();
     }
-    if (s + 2 < S) issue(s + 2);  // buf (s+2)%3 == (s-1)%3, freed above
+    // buf (s+RING-1)%RING == (s-1)%RING, freed by the barrier above
+    if (s + RING - 1 < S) issue(s + RING - 1);
     #pragma unroll 4
     for (int k2 = 0; k2 < DWG_K; k2 += 2) {
       const float av = ds[(k2 + k_l) * DWG_M + mcol];
@@ -1004,7 +1020,7 @@ __global__ void dw_glds_kernel(DwArgs a) {
   float* slab = a.slab + (int64_t)split * a.out_dim * a.in_dim;
   #pragma unroll
   for (int t = 0; t < NTW; ++t) {
-    const int col = wn * (NTW * M_WAVE) + t * M_WAVE + i_l;
+    const int col = nb0 + wn * (NTW * M_WAVE) + t * M_WAVE + i_l;
     if (col < a.in_dim) {
       #pragma unroll
       for (int r = 0; r < 16; ++r) {
@@ -1014,7 +1030,7 @@ __global__ void dw_glds_kernel(DwArgs a) {
       }
     }
   }
-  if (wn == 0) {
+  if (wn == 0 && blockIdx.y == 0) {
     const float other = __shfl(dbacc, lane ^ 32, WAVE);
     if (lane < 32 && mcol < a.out_dim)
       a.db_slab[(int64_t)split * a.out_dim + mcol] = dbacc + other;
@@ -1327,8 +1343,13 @@ void dw_mfma(torch::Tensor delta, torch::Tensor acts, torch::Tensor grad_buf,
     a.slab = slab2.data_ptr<float>();
     a.db_slab = db_slab2.data_ptr<float>();
     if (glds_wide)
-      hipLaunchKernelGGL((dw_glds_kernel<2, 4, 3>), dim3((unsigned)a.splits),
-                         dim3(512), 0, stream, a);
+      // single 384-wide tile, 5-deep ring (measured: 2-panel <2,2,3,4>
+      // grids with 2 blocks/CU ran 6% SLOWER — the duplicated delta
+      // stage and shorter MFMA chains cost more than the cross-block
+      // barrier overlap bought; both are ~60% of the f32-MFMA floor)
+      hipLaunchKernelGGL((dw_glds_kernel<2, 4, 3, 5>),
+                         dim3((unsigned)a.splits, 1), dim3(512), 0,
+                         stream, a);
     else
       hipLaunchKernelGGL((dw_glds_kernel<2, 2, 1>), dim3((unsigned)a.splits),
                          dim3(256), 0, stream, a);
