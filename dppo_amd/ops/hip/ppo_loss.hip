@@ -25,6 +25,44 @@
 
 namespace {
 
+// Stage a [rows][width] contiguous global stream into padded LDS rows.
+// Global side is float4 where width >= 8 (the caller guarantees the
+// base is 16-B aligned: tile_base*width is a multiple of 4), LDS side
+// is 4 scalar writes with an at-most-one row wrap per group.  row/col
+// kept by increment — one div/mod per stream, not per element.
+DEV_INLINE void stage_tile(const float* __restrict__ src,
+                           float* __restrict__ dst, int width, int stride,
+                           int rows, int tid) {
+  const int n = rows * width;
+  if (width >= 8) {
+    const float4* s4 = reinterpret_cast<const float4*>(src);
+    const int n4 = n >> 2;
+    const int dr = 1024 / width, dc = 1024 % width;
+    int r = (tid * 4) / width, c = (tid * 4) % width;
+    for (int q = tid; q < n4; q += 256) {
+      const float4 v = s4[q];
+      int rr = r, cc = c;
+      const float vv[4] = {v.x, v.y, v.z, v.w};
+      #pragma unroll
+      for (int k = 0; k < 4; ++k) {
+        dst[rr * stride + cc] = vv[k];
+        if (++cc == width) { cc = 0; ++rr; }
+      }
+      r += dr; c += dc;
+      if (c >= width) { c -= width; ++r; }
+    }
+    for (int g = (n4 << 2) + tid; g < n; g += 256)
+      dst[(g / width) * stride + g % width] = src[g];
+  } else {
+    const int dr = 256 / width, dc = 256 % width;
+    for (int g = tid, r = tid / width, c = tid % width; g < n; g += 256) {
+      dst[r * stride + c] = src[g];
+      r += dr; c += dc;
+      if (c >= width) { c -= width; r += 1; }
+    }
+  }
+}
+
 DEV_INLINE GaussRow gauss_row(const float* __restrict__ pdpi,
                               const float* __restrict__ pdold,
                               const float* __restrict__ act,
@@ -181,6 +219,108 @@ __global__ void ppo_gauss_fwd_wide_kernel(
   }
 }
 
+// LDS-tiled row-per-lane forward (same staging pattern as
+// ppo_gh_tile_kernel below: the wave-per-row kernel above keeps only A
+// of 64 lanes loading and measured 0.88 TB/s).  A 256-thread block
+// stages a 128-row tile coalesced, each of 128 lanes computes its row's
+// surrogate/entropy/value terms serially, then one block reduction
+// feeds the three double accumulators (one atomicAdd triple per block).
+__launch_bounds__(256)
+__global__ void ppo_gauss_fwd_tile_kernel(
+    const float* __restrict__ pdpi, const float* __restrict__ pdold,
+    const float* __restrict__ vpred, const float* __restrict__ oldv,
+    const float* __restrict__ act, const float* __restrict__ adv,
+    const float* __restrict__ etr, double* __restrict__ acc,
+    int64_t B, int A, float clip) {
+  constexpr int TILE = 128;
+  extern __shared__ float lds[];
+  const int P = 2 * A;
+  const int sp = P + 1;
+  const int sa = A | 1;
+  float* l_pd = lds;
+  float* l_og = l_pd + TILE * sp;
+  float* l_ac = l_og + TILE * sp;
+  float* l_sc = l_ac + TILE * sa;  // [4][TILE]: vpred|oldv|adv|etr
+  const int tid = threadIdx.x;
+  const int64_t ntiles = (B + TILE - 1) / TILE;
+
+  // grid-stride over tiles, per-thread accumulators, ONE atomic triple
+  // per block at the end: one atomic per TILE serialized 3*(B/128)
+  // same-address f64 atomics at L2 and dominated the kernel
+  float pol = 0.f, ent = 0.f, val = 0.f;
+  for (int64_t tile = blockIdx.x; tile < ntiles; tile += gridDim.x) {
+  const int64_t tb = tile * TILE;
+  const int rows = (int)min((int64_t)TILE, B - tb);
+
+  stage_tile(pdpi + tb * P, l_pd, P, sp, rows, tid);
+  stage_tile(pdold + tb * P, l_og, P, sp, rows, tid);
+  stage_tile(act + tb * A, l_ac, A, sa, rows, tid);
+  for (int idx = tid; idx < 4 * TILE; idx += 256) {
+    const int s = idx / TILE, t = idx % TILE;
+    if (t < rows) {
+      const float* src = s == 0 ? vpred : s == 1 ? oldv : s == 2 ? adv : etr;
+      l_sc[s * TILE + t] = src[tb + t];
+    }
+  }
+  __syncthreads();
+
+  if (tid < rows) {
+    const float* pd = l_pd + tid * sp;
+    const float* og = l_og + tid * sp;
+    const float* ac = l_ac + tid * sa;
+    float lp = 0.f, lo_ = 0.f, es = 0.f;
+    for (int j = 0; j < A; ++j) {
+      const float aj = ac[j];
+      const float ls = pd[A + j];
+      const float zp = (aj - pd[j]) * __expf(-ls);
+      lp += -0.5f * zp * zp - ls;
+      const float lso = og[A + j];
+      const float zo = (aj - og[j]) * __expf(-lso);
+      lo_ += -0.5f * zo * zo - lso;
+      es += ls;
+    }
+    const float ratio = __expf(lp - lo_);  // the logp constants cancel
+    const float ab = l_sc[2 * TILE + tid];
+    const float surr1 = ratio * ab;
+    const float rc = fminf(fmaxf(ratio, 1.f - clip), 1.f + clip);
+    pol += fminf(surr1, rc * ab);
+    ent += es + 0.5f * (PPO_LOG_2PI + 1.f) * A;
+    const float vp = l_sc[tid], ov = l_sc[TILE + tid];
+    const float et = l_sc[3 * TILE + tid];
+    const float d1 = vp - et;
+    const float dc = fminf(fmaxf(vp - ov, -clip), clip);
+    const float d2 = ov + dc - et;
+    val += fmaxf(d1 * d1, d2 * d2);
+  }
+  __syncthreads();  // lanes done reading before the next tile restages
+  }
+  // block reduce (wave sums -> LDS -> wave 0) then one atomic triple
+  __syncthreads();  // staging regions reused for the 4x3 wave partials
+  const int lane = tid & (WAVE - 1);
+  const int wave = tid / WAVE;
+  pol = wave_reduce_sum(pol);
+  ent = wave_reduce_sum(ent);
+  val = wave_reduce_sum(val);
+  if (lane == 0) {
+    lds[wave * 3 + 0] = pol;
+    lds[wave * 3 + 1] = ent;
+    lds[wave * 3 + 2] = val;
+  }
+  __syncthreads();
+  if (tid == 0) {
+    double p = 0.0, e = 0.0, v = 0.0;
+    #pragma unroll
+    for (int w = 0; w < 4; ++w) {
+      p += lds[w * 3 + 0];
+      e += lds[w * 3 + 1];
+      v += lds[w * 3 + 2];
+    }
+    atomicAdd(&acc[0], p);
+    atomicAdd(&acc[1], e);
+    atomicAdd(&acc[2], v);
+  }
+}
+
 __global__ void ppo_gauss_finalize_kernel(const double* __restrict__ acc,
                                           float* __restrict__ losses,  // [4]
                                           int64_t B, float entcoeff,
@@ -280,6 +420,26 @@ torch::Tensor ppo_loss_gauss_fwd(torch::Tensor pdpi, torch::Tensor pdold,
   auto losses = torch::empty({4}, pdpi.options());
   hipStream_t stream = c10::hip::getCurrentHIPStream().stream();
   const int block = 256;
+  static const int fwd_tile_env = []() {
+    const char* e = getenv("DPPO_GH_TILE");  // same knob as the gh kernel
+    return e ? atoi(e) : 1;
+  }();
+  const int fwd_lds =
+      128 * (2 * (2 * A + 1) + (A | 1) + 4) * (int)sizeof(float);
+  if (fwd_tile_env && A <= WAVE && fwd_lds <= 65536) {
+    const int64_t grid = std::min<int64_t>(1024, (B + 127) / 128);
+    hipLaunchKernelGGL(ppo_gauss_fwd_tile_kernel, dim3((unsigned)grid),
+                       dim3(block), fwd_lds, stream, pdpi.data_ptr<float>(),
+                       pdold.data_ptr<float>(), vpred.data_ptr<float>(),
+                       oldv.data_ptr<float>(), act.data_ptr<float>(),
+                       adv.data_ptr<float>(), etr.data_ptr<float>(),
+                       acc.data_ptr<double>(), B, A, (float)clip);
+    hipLaunchKernelGGL(ppo_gauss_finalize_kernel, dim3(1), dim3(1), 0,
+                       stream, acc.data_ptr<double>(),
+                       losses.data_ptr<float>(), B, (float)entcoeff,
+                       (float)vcoeff);
+    return losses;
+  }
   auto* fwd = (A <= WAVE) ? &ppo_gauss_fwd_kernel : &ppo_gauss_fwd_wide_kernel;
   hipLaunchKernelGGL(fwd, dim3(2048),
                      dim3(block), 0, stream, pdpi.data_ptr<float>(),
@@ -430,6 +590,7 @@ __global__ void ppo_gh_kernel(
 // (oldflat is consumed in the first pass only), and stored coalesced.
 constexpr int GH_TILE = 128;
 
+
 __launch_bounds__(256)
 __global__ void ppo_gh_tile_kernel(
     const float* __restrict__ pdflat, const float* __restrict__ oldflat,
@@ -452,45 +613,9 @@ __global__ void ppo_gh_tile_kernel(
   const int64_t tb = (int64_t)blockIdx.x * GH_TILE;
   const int rows = (int)min((int64_t)GH_TILE, B - tb);
 
-  // stage: contiguous global streams -> padded LDS rows.  Global side is
-  // float4 (the streams are row-major contiguous; tb*width is a multiple
-  // of 4 because GH_TILE is), LDS side is 4 scalar writes with an
-  // at-most-one row wrap per group (width >= 8 > 4).  row/col kept by
-  // increment — one div/mod per stream, not per element.
-  auto stage = [&](const float* __restrict__ src, float* __restrict__ dst,
-                   int width, int stride) {
-    const int n = rows * width;
-    if (width >= 8) {
-      const float4* s4 = reinterpret_cast<const float4*>(src);
-      const int n4 = n >> 2;
-      const int dr = 1024 / width, dc = 1024 % width;
-      int r = (tid * 4) / width, c = (tid * 4) % width;
-      for (int q = tid; q < n4; q += 256) {
-        const float4 v = s4[q];
-        int rr = r, cc = c;
-        const float vv[4] = {v.x, v.y, v.z, v.w};
-        #pragma unroll
-        for (int k = 0; k < 4; ++k) {
-          dst[rr * stride + cc] = vv[k];
-          if (++cc == width) { cc = 0; ++rr; }
-        }
-        r += dr; c += dc;
-        if (c >= width) { c -= width; ++r; }
-      }
-      for (int g = (n4 << 2) + tid; g < n; g += 256)
-        dst[(g / width) * stride + g % width] = src[g];
-    } else {
-      const int dr = 256 / width, dc = 256 % width;
-      for (int g = tid, r = tid / width, c = tid % width; g < n; g += 256) {
-        dst[r * stride + c] = src[g];
-        r += dr; c += dc;
-        if (c >= width) { c -= width; r += 1; }
-      }
-    }
-  };
-  stage(pdflat + tb * P, l_pd, P, sp);
-  stage(oldflat + tb * P, l_og, P, sg);
-  stage(act + tb * A, l_ac, A, sa);
+  stage_tile(pdflat + tb * P, l_pd, P, sp, rows, tid);
+  stage_tile(oldflat + tb * P, l_og, P, sg, rows, tid);
+  stage_tile(act + tb * A, l_ac, A, sa, rows, tid);
   for (int idx = tid; idx < 4 * GH_TILE; idx += 256) {
     const int s = idx / GH_TILE, t = idx % GH_TILE;
     if (t < rows) {
