@@ -535,8 +535,20 @@ __global__ void gemm_fwd_glds_kernel(FwdArgs a) {
     wc4 = c;
   }
 
+  // per-block K-phase rotation: co-resident blocks otherwise hit their
+  // stage barriers in lockstep (PMC: MFMA pipe 51% busy, 34% of wave
+  // time parked on vmcnt/barrier) — rotating the stage order by
+  // blockIdx desynchronizes the convoy at unchanged traffic.  fp32
+  // accumulation order changes per block (still deterministic per
+  // shape; numerics tests compare against eager to tolerance).
+  const int S_ = (a.K + PBK - 1) / PBK;
+  const int phase = (int)(blockIdx.x % (unsigned)S_);
+  auto smap = [&](int s) {
+    const int t = s + phase;
+    return t >= S_ ? t - S_ : t;
+  };
   auto issue = [&](int s) {
-    const int kb = s * PBK;
+    const int kb = smap(s) * PBK;
     char* slot = smem + (s % RING) * SLOT;
     #pragma unroll
     for (int p = 0; p < 2; ++p) {
@@ -579,7 +591,7 @@ __global__ void gemm_fwd_glds_kernel(FwdArgs a) {
     char* slot = smem + (s % RING) * SLOT;
     float* xs = (float*)slot;
     float* ws = (float*)(slot + XB);
-    const int kb = s * PBK;
+    const int kb = smap(s) * PBK;
     if (kb + PBK > a.K) {
       // zero the W rows past K so the clamped/garbage X columns of the
       // tail stage contribute nothing
