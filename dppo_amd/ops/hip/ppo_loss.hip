@@ -420,10 +420,8 @@ torch::Tensor ppo_loss_gauss_fwd(torch::Tensor pdpi, torch::Tensor pdold,
   auto losses = torch::empty({4}, pdpi.options());
   hipStream_t stream = c10::hip::getCurrentHIPStream().stream();
   const int block = 256;
-  static const int fwd_tile_env = []() {
-    const char* e = getenv("DPPO_GH_TILE");  // same knob as the gh kernel
-    return e ? atoi(e) : 1;
-  }();
+  const char* fte = getenv("DPPO_GH_TILE");  // same knob as the gh kernel
+  const int fwd_tile_env = fte ? atoi(fte) : 1;
   const int fwd_lds =
       128 * (2 * (2 * A + 1) + (A | 1) + 4) * (int)sizeof(float);
   if (fwd_tile_env && A <= WAVE && fwd_lds <= 65536) {
@@ -711,10 +709,8 @@ torch::Tensor ppo_loss_gauss_gh(torch::Tensor pdflat, torch::Tensor oldflat,
   hipStream_t stream = c10::hip::getCurrentHIPStream().stream();
   const float* cd =
       (clip_dev.numel() > 0) ? clip_dev.data_ptr<float>() : nullptr;
-  static const int gh_tile_env = []() {
-    const char* e = getenv("DPPO_GH_TILE");
-    return e ? atoi(e) : 1;
-  }();
+  const char* ghe = getenv("DPPO_GH_TILE");  // re-read: testable per call
+  const int gh_tile_env = ghe ? atoi(ghe) : 1;
   const int lds_bytes =
       GH_TILE * (2 * A + 1 + (ldgh | 1) + (A | 1) + 4) * (int)sizeof(float);
   if (gh_tile_env && lds_bytes <= 65536) {

@@ -260,3 +260,31 @@ def test_adam_dev_matches_torch(ext):
                           0.9, 0.999, 1e-8)
     assert int(step_dev.item()) == 3
     torch.testing.assert_close(p2, p1.detach(), atol=2e-6, rtol=2e-5)
+
+
+def test_ppo_loss_fwd_tile_matches_legacy(ext, monkeypatch):
+    """The LDS-tiled forward and the legacy wave-per-row kernel agree
+    (same math, different reduction order; DPPO_GH_TILE picks the path
+    per call)."""
+    _, pdflat, oldflat, v, oldv, a, adv, etr = _loss_case(B=8192 + 37, seed=7)
+    monkeypatch.setenv("DPPO_GH_TILE", "1")
+    lt = ext.ppo_loss_gauss_fwd(pdflat, oldflat, v, oldv, a, adv, etr,
+                                0.2, 0.01, 0.5)
+    monkeypatch.setenv("DPPO_GH_TILE", "0")
+    ll = ext.ppo_loss_gauss_fwd(pdflat, oldflat, v, oldv, a, adv, etr,
+                                0.2, 0.01, 0.5)
+    torch.testing.assert_close(lt, ll, atol=2e-6, rtol=1e-5)
+
+
+def test_gh_tile_matches_legacy(ext, monkeypatch):
+    """ppo_gh_tile_kernel vs the wave-per-row gh kernel: per-row grads
+    are bitwise-comparable up to fp reassociation of the A-dim sums."""
+    _, pdflat, oldflat, v, oldv, a, adv, etr = _loss_case(B=4096 + 13, seed=9)
+    cde = torch.empty(0, device="cuda")
+    monkeypatch.setenv("DPPO_GH_TILE", "1")
+    gt = ext.ppo_loss_gauss_gh(pdflat, oldflat, v, oldv, a, adv, etr,
+                               0.2, 0.01, 0.5, cde)
+    monkeypatch.setenv("DPPO_GH_TILE", "0")
+    gl = ext.ppo_loss_gauss_gh(pdflat, oldflat, v, oldv, a, adv, etr,
+                               0.2, 0.01, 0.5, cde)
+    torch.testing.assert_close(gt, gl, atol=1e-6, rtol=1e-5)
