@@ -196,6 +196,8 @@ class DPPOEngine:
         from .ops import use_hip
 
         c = self.cfg
+        if os.environ.get("DPPO_WIDE") == "0":  # A/B: autocast/rocBLAS path
+            return False
         if self._act_kind != "box" or c.DTYPE != "bfloat16":
             return False
         if c.ACTIVATION != "tanh" or c.MINIBATCH_SIZE != 0:
