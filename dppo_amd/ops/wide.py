@@ -99,15 +99,24 @@ class WideBF16Path:
             self._fwd_bufs[key] = bufs
         return bufs
 
+    @staticmethod
+    def _dualw() -> bool:
+        """Transposed dual-writes from the GEMM epilogues (default on;
+        DPPO_WIDE_DUALW=0 restores the separate transpose passes)."""
+        return os.environ.get("DPPO_WIDE_DUALW", "1") != "0"
+
     @torch.no_grad()
-    def forward(self, obs: torch.Tensor, out=None):
+    def forward(self, obs: torch.Tensor, out=None, ct=None):
         """(v, pdflat) in f32, every GEMM on the hand bf16 kernels.
 
         `out = (h_views, pd_view, v_view)` redirects the layer outputs
         into caller-owned buffers — the graphed rollout records its
         activations straight into the update pipeline's buffers so the
         first update step can skip its forward entirely (the parameters
-        are unchanged between rollout and step 1)."""
+        are unchanged between rollout and step 1).  `ct = (hT_views,
+        ldt)` additionally dual-writes the TRANSPOSED activations
+        (column-offset views of the update pipeline's hT buffers), so
+        step 1 skips its hT transposes too."""
         if self._weights_dirty:
             self.refresh_weights()
         M = obs.shape[0]
@@ -116,13 +125,21 @@ class WideBF16Path:
             hs, pd_o, v_o = s["h"], s["pd"], s["v"]
         else:
             hs, pd_o, v_o = out
+        if out is not None:  # recording pass: remember whether hT rode along
+            self._hT_from_rollout = ct is not None
         s["x"].copy_(obs)
         x = s["x"]
         pi = self.eng.pi
         for l in range(len(self.H)):
-            self.ext.bf16_mm256(x, self.W_bf[l], hs[l], 1,
-                                pi.hidden[l].bias.detach(), self._b,
-                                self._f, 0, self._b, 0, self._f, 0)
+            if ct is not None:
+                self.ext.bf16_mm256(x, self.W_bf[l], hs[l], 1,
+                                    pi.hidden[l].bias.detach(), self._b,
+                                    self._f, 0, ct[0][l], ct[1],
+                                    self._f, 0)
+            else:
+                self.ext.bf16_mm256(x, self.W_bf[l], hs[l], 1,
+                                    pi.hidden[l].bias.detach(), self._b,
+                                    self._f, 0, self._b, 0, self._f, 0)
             x = hs[l]
         self.ext.bf16_mm_small(x, self.whcat, pd_o, v_o, self._b,
                                self._f, 0, 0, 0, 4, M, self.P + 1, self.P,
@@ -321,12 +338,14 @@ class WideBF16Path:
             x = u["x"]
             ev_hT = [None] * nH
             if step_i == 0 and skip_first:
-                for l in range(nH):
-                    fork()
-                    with side_ctx():
-                        ext.bf16_transpose(u["h"][l], u["hT"][l], self._f,
-                                           0, B, self.H[l], self.H[l], B)
-                    ev_hT[l] = join()
+                if not (dualw and getattr(self, "_hT_from_rollout", False)):
+                    for l in range(nH):
+                        fork()
+                        with side_ctx():
+                            ext.bf16_transpose(u["h"][l], u["hT"][l],
+                                               self._f, 0, B, self.H[l],
+                                               self.H[l], B)
+                        ev_hT[l] = join()
             else:
                 for l in range(nH):
                     if dualw:

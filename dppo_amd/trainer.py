@@ -741,7 +741,13 @@ class DPPOEngine:
         for t in range(T):
             outs = ([u["h"][l].narrow(0, t * E, E) for l in range(len(wide.H))],
                     u["pd"].narrow(0, t * E, E), u["v"].narrow(0, t * E, E))
-            v, pdflat = wide.forward(obs, out=outs)
+            # dual-write the transposed activations into the update
+            # pipeline's hT buffers (column-offset views; the GEMM writes
+            # CT[col][t*E + rb] through the full-B row stride), so update
+            # step 1 skips its hT transposes as well as its forward
+            ct = (([u["hT"][l][:, t * E:] for l in range(len(wide.H))],
+                   T * E) if wide._dualw() else None)
+            v, pdflat = wide.forward(obs, out=outs, ct=ct)
             pd = self.pi.pdtype.pdfromflat(pdflat)
             a = pd.sample()
             explore = torch.rand(E, device=self.device) < eps
