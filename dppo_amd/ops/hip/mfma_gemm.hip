@@ -487,10 +487,11 @@ __global__ void gemm_fwd_pipe_kernel(FwdArgs a) {
 // address (guide rule 21) — same 64 B cacheline, zero coalescing cost.
 // ---------------------------------------------------------------------------
 
-template <int RING>  // ring depth: 3 (36 KB, 4 blocks/CU) or 4 (48 KB,
-                     // 3 blocks/CU — two extra stages in flight cover
-                     // loaded-chip HBM latency the 3-deep ring stalls on)
-__launch_bounds__(FWD_WAVES * 64, RING == 3 ? 4 : 3)
+template <int RING>  // ring depth: 2 (24 KB, 6 blocks/CU — overlap via
+                     // MORE co-resident blocks), 3 (36 KB, 4 blocks/CU,
+                     // default), 4 (48 KB, 3 blocks/CU — deeper
+                     // in-flight staging; measured slower)
+__launch_bounds__(FWD_WAVES * 64, RING == 2 ? 6 : RING == 3 ? 4 : 3)
 __global__ void gemm_fwd_glds_kernel(FwdArgs a) {
   constexpr int NT = 2;
   constexpr int PBK = 16;
@@ -1177,13 +1178,21 @@ void gemm_fwd(torch::Tensor X, torch::Tensor Wt, torch::Tensor bias,
   const bool glds_ok = glds_env && !a.ablate && wt_layout == 0 &&
                        K % 4 == 0 && K >= 4 && N % 4 == 0 && N >= 4 &&
                        a.activation != 9999;
-  static const bool ring4 = []() {
+  static const int ring_env = []() {
     const char* e = getenv("DPPO_FWD_RING");
-    return e && atoi(e) == 4;  // measured slower at the flagship shapes
-                               // (occupancy 3 vs 4 blocks/CU); opt-in
+    return e ? atoi(e) : 0;  // 0 = by shape (below); 2/3/4 force
   }();
+  // by shape: short-K GEMMs (few stages) run the 2-deep ring — 24 KB
+  // LDS lets 6 blocks/CU co-reside and prologue/drain is a large
+  // fraction of S (L2t K=64: 179 -> 157 us at B=1M); long-K streams
+  // keep the 3-deep ring (L1t K=376: wash; ring 4 measured slower)
+  const int ring_sel = ring_env ? ring_env : (a.K <= 128 ? 2 : 3);
   #define DISPATCH_FWD(NTV, NPANEL)                                          \
-    if (glds_ok && NTV == 2 && ring4)                                        \
+    if (glds_ok && NTV == 2 && ring_sel == 2)                                \
+      hipLaunchKernelGGL(gemm_fwd_glds_kernel<2>,                            \
+                         dim3((unsigned)grid_pipe, NPANEL),                  \
+                         dim3(FWD_WAVES * 64), 0, stream, a);                \
+    else if (glds_ok && NTV == 2 && ring_sel == 4)                           \
       hipLaunchKernelGGL(gemm_fwd_glds_kernel<4>,                            \
                          dim3((unsigned)grid_pipe, NPANEL),                  \
                          dim3(FWD_WAVES * 64), 0, stream, a);                \
