@@ -251,7 +251,13 @@ __global__ void gemm_fwd_kernel(FwdArgs a) {
               if (a.activation == 0) x = fmaxf(x, 0.f);
               else if (a.activation == 1) x = fast_tanhf(x);
             }
-            if (a.heads) {
+            if (a.heads == 2) {
+              // padded heads: N includes pad columns so N%4==0 shapes
+              // can take the glds kernel; v sits at N-2, cols >= N-1
+              // are the zero-weight pad (discarded)
+              if (col == a.N - 2) a.v[row] = x;
+              else if (col < a.N - 2) a.C[row * a.ldc + col] = x;
+            } else if (a.heads) {
               if (col == a.N - 1) a.v[row] = x;
               else a.C[row * a.ldc + col] = x;
             } else {
@@ -453,7 +459,13 @@ __global__ void gemm_fwd_pipe_kernel(FwdArgs a) {
               if (a.activation == 0) x = fmaxf(x, 0.f);
               else if (a.activation == 1) x = fast_tanhf(x);
             }
-            if (a.heads) {
+            if (a.heads == 2) {
+              // padded heads: N includes pad columns so N%4==0 shapes
+              // can take the glds kernel; v sits at N-2, cols >= N-1
+              // are the zero-weight pad (discarded)
+              if (col == a.N - 2) a.v[row] = x;
+              else if (col < a.N - 2) a.C[row * a.ldc + col] = x;
+            } else if (a.heads) {
               if (col == a.N - 1) a.v[row] = x;
               else a.C[row * a.ldc + col] = x;
             } else {
@@ -679,7 +691,10 @@ __global__ void gemm_fwd_glds_kernel(FwdArgs a) {
             if (a.activation == 0) x = fmaxf(x, 0.f);
             else if (a.activation == 1) x = fast_tanhf(x);
           }
-          if (a.heads) {
+          if (a.heads == 2) {
+            if (col == a.N - 2) a.v[row] = x;
+            else if (col < a.N - 2) a.C[row * a.ldc + col] = x;
+          } else if (a.heads) {
             if (col == a.N - 1) a.v[row] = x;
             else a.C[row * a.ldc + col] = x;
           } else {
@@ -1116,13 +1131,13 @@ void gemm_fwd(torch::Tensor X, torch::Tensor Wt, torch::Tensor bias,
   TORCH_CHECK(C.is_contiguous());
   TORCH_CHECK((wt_layout ? Wt.size(1) : Wt.size(0)) == K);
   TORCH_CHECK(activation >= 3 || bias.numel() == N);
-  const int ldc_eff =
-      static_cast<int>(ldc > 0 ? ldc : (heads ? N - 1 : N));
-  TORCH_CHECK(ldc_eff >= (heads ? N - 1 : N));
+  const int ncols = static_cast<int>(heads == 2 ? N - 2 : heads ? N - 1 : N);
+  const int ldc_eff = static_cast<int>(ldc > 0 ? ldc : ncols);
+  TORCH_CHECK(ldc_eff >= ncols);
   if (heads) {
-    TORCH_CHECK(C.numel() >= (B - 1) * ldc_eff + N - 1 && v.numel() == B);
+    TORCH_CHECK(C.numel() >= (B - 1) * ldc_eff + ncols && v.numel() == B);
   } else {
-    TORCH_CHECK(C.numel() >= (B - 1) * ldc_eff + N);
+    TORCH_CHECK(C.numel() >= (B - 1) * ldc_eff + ncols);
   }
 
   FwdArgs a{};

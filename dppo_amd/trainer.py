@@ -489,6 +489,14 @@ class DPPOEngine:
             bz_D=torch.zeros(D, device=dev),
             wh=torch.empty(P + 1, H[-1], device=dev),
             bh=torch.empty(P + 1, device=dev),
+            # padded transposed heads weights: N = P+2 (%4 == 0 when
+            # P%4 == 2) routes the heads GEMM through the glds kernel
+            # (wt_layout 0, heads mode 2: v at col N-2, col N-1 is the
+            # zero-weight pad) instead of the pipe kernel
+            whT_pad=(torch.zeros(H[-1], P + 2, device=dev)
+                     if (P + 2) % 4 == 0
+                     and os.environ.get("DPPO_HEADS_PAD") != "0" else None),
+            bh_pad=torch.zeros(P + 2, device=dev),
             pd_scratch=torch.empty(E, P, device=dev),
             seed_dev=torch.zeros(1, dtype=torch.int64, device=dev),
             eps_dev=torch.zeros(1, dtype=torch.float32, device=dev),
@@ -530,6 +538,10 @@ class DPPOEngine:
         wh[P:].copy_(self.pi.vf.weight.detach())
         bh[:P].copy_(self.pi.pi.bias.detach())
         bh[P:].copy_(self.pi.vf.bias.detach())
+        whT_pad = v3.get("whT_pad")
+        if whT_pad is not None:
+            whT_pad[:, :P + 1].copy_(wh.t())
+            v3["bh_pad"][:P + 1].copy_(bh)
         wts = self._wt_bufs()
         for l, lay in enumerate(self.pi.hidden):
             wts[l].copy_(lay.weight.detach().t())
@@ -552,8 +564,12 @@ class DPPOEngine:
                              self.pi.hidden[l].bias.detach(), act_code, 0,
                              hl, hl, hl, 0, 0, 0)
                 h = hl
-            ext.gemm_fwd(h, wh, bh, 2, 1, pdflats[st], values[st],
-                         pdflats[st], 1, 0, 0)
+            if whT_pad is not None:
+                ext.gemm_fwd(h, whT_pad, v3["bh_pad"], 2, 2, pdflats[st],
+                             values[st], pdflats[st], 0, 0, 0)
+            else:
+                ext.gemm_fwd(h, wh, bh, 2, 1, pdflats[st], values[st],
+                             pdflats[st], 1, 0, 0)
             # XV = x @ V into the concat buffer's first rank columns
             ext.gemm_fwd(xin, v3["V"], v3["bz_r"], 2, 0, v3["xva"],
                          v3["xva"], v3["xva"], 0, 0, v3["kw"])
@@ -592,8 +608,12 @@ class DPPOEngine:
                          self.pi.hidden[l].bias.detach(), act_code, 0,
                          hl, hl, hl, 0, 0, 0)
             h = hl
-        ext.gemm_fwd(h, wh, bh, 2, 1, v3["pd_scratch"], boot_v,
-                     v3["pd_scratch"], 1, 0, 0)
+        if whT_pad is not None:
+            ext.gemm_fwd(h, whT_pad, v3["bh_pad"], 2, 2, v3["pd_scratch"],
+                         boot_v, v3["pd_scratch"], 0, 0, 0)
+        else:
+            ext.gemm_fwd(h, wh, bh, 2, 1, v3["pd_scratch"], boot_v,
+                         v3["pd_scratch"], 1, 0, 0)
         return ext.rollout_moments(rewards, dones, v3["epr_before"], T, E)
 
     @torch.no_grad()
@@ -1478,7 +1498,17 @@ class DPPOEngine:
         P = 2 * self.act_space.shape[0]
         pdflat = torch.empty(B, P, device=states.device, dtype=states.dtype)
         v = torch.empty(B, device=states.device, dtype=states.dtype)
-        ext.gemm_fwd(x, Wh_cat, bh, 2, 1, pdflat, v, pdflat, 1, 0, 0)
+        if (P + 2) % 4 == 0 and os.environ.get("DPPO_HEADS_PAD") != "0":
+            # padded transposed heads (N=P+2, %4==0): glds kernel with
+            # heads mode 2 instead of the layout-1 pipe kernel
+            WhT_pad = torch.zeros(Wh_cat.shape[1], P + 2,
+                                  device=states.device)
+            WhT_pad[:, :P + 1] = Wh_cat.t()
+            bh2 = torch.zeros(P + 2, device=states.device)
+            bh2[:P + 1] = bh
+            ext.gemm_fwd(x, WhT_pad, bh2, 2, 2, pdflat, v, pdflat, 0, 0, 0)
+        else:
+            ext.gemm_fwd(x, Wh_cat, bh, 2, 1, pdflat, v, pdflat, 1, 0, 0)
         self._Wh_pad = Wh_pad
         return acts, a_views, v, pdflat
 
