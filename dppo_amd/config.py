@@ -30,7 +30,7 @@ import json
 import multiprocessing
 import os
 from dataclasses import dataclass, field
-from typing import Any, Dict, Optional, Sequence, Tuple
+from typing import Any, Dict, Tuple
 
 
 def _default_num_workers() -> int:

@@ -25,7 +25,6 @@ from __future__ import annotations
 import math
 from typing import List
 
-import numpy as np
 import torch
 
 from . import spaces

@@ -19,7 +19,7 @@ gradients ride along with the delta transposes).
 from __future__ import annotations
 
 import os
-from typing import List, Optional
+from typing import List
 
 import torch
 
