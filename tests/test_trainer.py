@@ -213,3 +213,45 @@ def test_eval_losses_equal_recompute():
                 for k, v in eng._losses(batch, 1.0, recorded_pi=False).items()}
     for k in fast:
         assert abs(fast[k] - slow[k]) < 1e-5, (k, fast[k], slow[k])
+
+
+@pytest.mark.parametrize("hidden,act", [
+    ((16, 16), "relu"),
+    ((16, 16, 16), "tanh"),   # 3 hidden layers (engine/kernel-gate max)
+    ((24,), "relu"),
+])
+def test_config_matrix_rounds(hidden, act):
+    """Hidden-depth x activation matrix: every combination steps through
+    a full round (rollout + GAE + update) and stays finite."""
+    eng = DPPOEngine(small_cfg(HIDDEN_SIZES=hidden, ACTIVATION=act),
+                     comm=Comm(device="cpu"))
+    before = eng.flat_pi.flat_param.clone()
+    eng.train_round()
+    assert torch.isfinite(eng.flat_pi.flat_param).all()
+    assert not torch.equal(before, eng.flat_pi.flat_param)
+
+
+def test_batch_curation_single_rank_is_identity():
+    """BATCH_CURATION=True with world_size 1 must degenerate to the
+    rank-local default (the sort over one batch is the identity)."""
+    torch.manual_seed(0)
+    a = DPPOEngine(small_cfg(SEED=11, BATCH_CURATION=True),
+                   comm=Comm(device="cpu"))
+    torch.manual_seed(0)
+    b = DPPOEngine(small_cfg(SEED=11, BATCH_CURATION=False),
+                   comm=Comm(device="cpu"))
+    a.train_round()
+    b.train_round()
+    assert torch.equal(a.flat_pi.flat_param, b.flat_pi.flat_param)
+
+
+def test_minibatch_with_curation_and_multi_hidden():
+    """Interaction case: minibatched epochs + curation flag + 2 hidden
+    layers (the config-4 scheme with every optional feature on)."""
+    eng = DPPOEngine(small_cfg(MINIBATCH_SIZE=32, NUM_ENVS=8,
+                               MAX_EPOCH_STEPS=16, HIDDEN_SIZES=(16, 16),
+                               BATCH_CURATION=True),
+                     comm=Comm(device="cpu"))
+    eng.train_round()
+    eng.train_round()
+    assert torch.isfinite(eng.flat_pi.flat_param).all()
