@@ -240,7 +240,11 @@ def test_batch_curation_single_rank_is_identity():
     torch.manual_seed(0)
     b = DPPOEngine(small_cfg(SEED=11, BATCH_CURATION=False),
                    comm=Comm(device="cpu"))
+    # the eager rollout samples from the GLOBAL torch generator: pin it
+    # before each round so the two engines see identical draws
+    torch.manual_seed(123)
     a.train_round()
+    torch.manual_seed(123)
     b.train_round()
     assert torch.equal(a.flat_pi.flat_param, b.flat_pi.flat_param)
 
