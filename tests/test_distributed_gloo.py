@@ -352,3 +352,53 @@ def test_drift_guard_broadcast_heals_divergence(tmp_path):
     for r in range(WORLD):
         with open(tmp_path / f"dg{r}.json") as f:
             assert json.load(f)["identical"], "drift guard did not heal"
+
+
+def _curation4_worker(rank, world, port, out_dir):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    os.environ["RANK"] = str(rank)
+    os.environ["WORLD_SIZE"] = str(world)
+    from dppo_amd.parallel.comm import Comm
+    from dppo_amd.trainer import DPPOEngine
+
+    cfg = DPPOConfig(
+        GAME="Pendulum-v1", NUM_ENVS=2, MAX_EPOCH_STEPS=8, EPOCH_MAX=8,
+        STOP_EPOCH=8, LEARNING_RATE=1e-3, NUM_WORKERS=world,
+        LOG_FILE_PATH=os.path.join(out_dir, "logs4"), DEVICE="cpu",
+        BROADCAST_INTERVAL=0, BATCH_CURATION=True,
+    )
+    comm = Comm(backend="gloo", device="cpu")
+    eng = DPPOEngine(cfg, comm=comm)
+    batch = eng.collect()
+    states_all = comm.all_gather_rows(batch.states.reshape(-1).clone())
+
+    results = {}
+    # scripted 4-rank scenario: ranks 1 and 3 valid with EQUAL keys
+    # (stable sort keeps rank 1 first), ranks 0 and 2 invalid.
+    # Reference drain-sort-assign (Chief.py:33-53): sorted valid order is
+    # [1, 3]; towers 0..3 get sources [1, 3, 1, 3] (order[i % V] recycle).
+    g = torch.zeros(world, 11)
+    g[1, 10] = 1.0
+    g[3, 10] = 1.0
+    g[:, 2] = torch.tensor([9.0, 4.0, 9.0, 4.0])  # invalid 9s never win
+    eng._curate_batches(batch, g)
+    expect_src = [1, 3, 1, 3][rank]
+    results["assign_ok"] = bool(torch.equal(
+        batch.states.reshape(-1), states_all[expect_src]))
+    with open(os.path.join(out_dir, f"cur4_{rank}.json"), "w") as f:
+        json.dump(results, f)
+    comm.shutdown()
+
+
+@pytest.mark.timeout(300)
+def test_batch_curation_four_rank_ties_and_recycle(tmp_path):
+    """4 ranks, duplicate sort keys, two invalid ranks: the stable sort
+    tie-break and the order[i % V] best-batch recycle both match the
+    reference's drain-sort-assign semantics (Chief.py:33-53)."""
+    mp.spawn(_curation4_worker, args=(4, 29833, str(tmp_path)),
+             nprocs=4, join=True)
+    for r in range(4):
+        with open(tmp_path / f"cur4_{r}.json") as f:
+            d = json.load(f)
+        assert d["assign_ok"], f"rank {r} trained on the wrong batch"
