@@ -3,10 +3,12 @@ shapes, A/B'd against torch/rocBLAS bf16 matmul on the same random data
 (guide §5.4 rule 25: random operands, within-probe interleave)."""
 
 import argparse
+import os
 import sys
 import time
 
-sys.path.insert(0, "/root/repo")
+sys.path.insert(0, os.path.dirname(os.path.dirname(
+    os.path.abspath(__file__))))
 
 import torch
 

@@ -1,6 +1,8 @@
 """Minimal mm256 launcher for rocprofv3 PMC runs."""
+import os
 import sys
-sys.path.insert(0, "/root/repo")
+sys.path.insert(0, os.path.dirname(os.path.dirname(
+    os.path.abspath(__file__))))
 import torch
 from dppo_amd.ops import require_hip_ext
 

@@ -9,7 +9,8 @@ import os
 import sys
 import time
 
-sys.path.insert(0, "/root/repo")
+sys.path.insert(0, os.path.dirname(os.path.dirname(
+    os.path.abspath(__file__))))
 import torch
 
 from dppo_amd.ops import require_hip_ext

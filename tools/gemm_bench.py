@@ -1,7 +1,9 @@
 """Microbench: the update-path GEMM shapes in isolation (B=1M rows)."""
+import os
 import sys, time
 import torch
-sys.path.insert(0, "/root/repo")
+sys.path.insert(0, os.path.dirname(os.path.dirname(
+    os.path.abspath(__file__))))
 from dppo_amd.ops import require_hip_ext
 
 ext = require_hip_ext()

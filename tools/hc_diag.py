@@ -1,5 +1,7 @@
+import os
 import sys, time, torch
-sys.path.insert(0, "/root/repo")
+sys.path.insert(0, os.path.dirname(os.path.dirname(
+    os.path.abspath(__file__))))
 from dppo_amd.config import DPPOConfig
 from dppo_amd.parallel.comm import Comm
 from dppo_amd.trainer import DPPOEngine

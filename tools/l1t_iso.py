@@ -1,6 +1,8 @@
 """Isolated L1t fwd GEMM (B=1M, 376->64) for PMC attribution."""
+import os
 import sys, torch
-sys.path.insert(0, "/root/repo")
+sys.path.insert(0, os.path.dirname(os.path.dirname(
+    os.path.abspath(__file__))))
 from dppo_amd.ops import require_hip_ext
 ext = require_hip_ext()
 B, D, H = 1048576, 376, 64

@@ -2,12 +2,14 @@
 
 Usage: python tools/preset_diag.py <preset> [NUM_ENVS] [rounds]
 """
+import os
 import sys
 import time
 
 import torch
 
-sys.path.insert(0, "/root/repo")
+sys.path.insert(0, os.path.dirname(os.path.dirname(
+    os.path.abspath(__file__))))
 from bench import PRESETS  # noqa: E402
 from dppo_amd.config import DPPOConfig  # noqa: E402
 from dppo_amd.parallel.comm import Comm  # noqa: E402

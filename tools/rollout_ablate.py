@@ -1,7 +1,9 @@
 """Phase ablation of the rollout kernel (guide §5.4: ablate before
 optimizing).  Runtime mask: 1=trunk 2=heads 4=sampling 8=env 16=writes."""
+import os
 import sys, time, torch
-sys.path.insert(0, "/root/repo")
+sys.path.insert(0, os.path.dirname(os.path.dirname(
+    os.path.abspath(__file__))))
 from dppo_amd.config import DPPOConfig
 from dppo_amd.parallel.comm import Comm
 from dppo_amd.trainer import DPPOEngine

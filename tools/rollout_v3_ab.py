@@ -1,6 +1,7 @@
 """A/B: fused whole-rollout kernel vs v3 per-step GEMM rollout."""
 import os, sys, time, torch
-sys.path.insert(0, "/root/repo")
+sys.path.insert(0, os.path.dirname(os.path.dirname(
+    os.path.abspath(__file__))))
 from dppo_amd.config import DPPOConfig
 from dppo_amd.parallel.comm import Comm
 from dppo_amd.trainer import DPPOEngine

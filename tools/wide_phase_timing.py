@@ -1,5 +1,7 @@
+import os
 import sys, json, os
-sys.path.insert(0, "/root/repo")
+sys.path.insert(0, os.path.dirname(os.path.dirname(
+    os.path.abspath(__file__))))
 os.environ["DPPO_TIMER_SYNC"] = "1"
 import torch
 from dppo_amd.config import DPPOConfig
