@@ -116,11 +116,16 @@ DEV_INLINE short8 frag_read(const char* img, int row, int colb) {
   return *(const short8*)(img + byte);
 }
 
-template <int NPH, bool TWO_BARRIERS, bool STATIC_PRIO = false>
+template <int NPH, bool TWO_BARRIERS, bool STATIC_PRIO = false,
+          int NIMG = 2>
 __launch_bounds__(512, 2)
 __global__ void bf16_mm256_kernel(MM256Args a) {
   constexpr int FMPP = 8 / NPH;  // fm-blocks computed per phase
-  __shared__ __attribute__((aligned(16))) char smem[LDS_BYTES];
+  // NIMG == 3: a third A image (5 x 32 KB = the full 160 KB CU; still
+  // 1 block/CU so occupancy is unchanged) lets ph0 stage A TWO tiles
+  // ahead — the tile-seam wait then only covers the t+2 stages
+  // (vmcnt(8)), never tile t+1's A, which landed a full tile earlier.
+  __shared__ __attribute__((aligned(16))) char smem[(NIMG + 2) * IMG];
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wave = tid >> 6;
@@ -138,8 +143,13 @@ __global__ void bf16_mm256_kernel(MM256Args a) {
 
   char* const imgA0 = smem;
   char* const imgA1 = smem + IMG;
-  char* const imgB0 = smem + 2 * IMG;
-  char* const imgB1 = smem + 3 * IMG;
+  char* const imgA2 = smem + 2 * IMG;  // NIMG == 3 only
+  char* const imgB0 = smem + NIMG * IMG;
+  char* const imgB1 = smem + (NIMG + 1) * IMG;
+  auto imgA = [&](int t) {
+    return NIMG == 2 ? ((t & 1) ? imgA1 : imgA0)
+                     : (t % 3 == 0 ? imgA0 : t % 3 == 1 ? imgA1 : imgA2);
+  };
   const long rowA = (long)bm * 256;
   const long rowB = (long)bn * 256;
   const int NT = (int)(a.K / BK);
@@ -173,18 +183,24 @@ __global__ void bf16_mm256_kernel(MM256Args a) {
   stage_piece(sB10, 0, imgB0, 1, 0, wave);
   stage_piece(sB11, 0, imgB0, 1, 1, wave);
   if (NT > 1) {
+    if (NIMG == 3) {  // A(1) rides the prologue; ph0 then stages t+2
+      stage_piece(sA00, BK, imgA1, 0, 0, wave);
+      stage_piece(sA01, BK, imgA1, 0, 1, wave);
+      stage_piece(sA10, BK, imgA1, 1, 0, wave);
+      stage_piece(sA11, BK, imgA1, 1, 1, wave);
+    }
     stage_piece(sB00, BK, imgB1, 0, 0, wave);
     stage_piece(sB01, BK, imgB1, 0, 1, wave);
     stage_piece(sB10, BK, imgB1, 1, 0, wave);
     stage_piece(sB11, BK, imgB1, 1, 1, wave);
-    asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+    asm volatile("s_waitcnt vmcnt(%0)" ::"i"(4 * (NIMG - 1)) : "memory");
   } else {
     asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
   }
   __builtin_amdgcn_s_barrier();
 
   for (int t = 0; t < NT; ++t) {
-    const char* iA = (t & 1) ? imgA1 : imgA0;
+    const char* iA = imgA(t);
     const char* iB = (t & 1) ? imgB1 : imgB0;
     // B fragments: one read pass, live for the whole tile
     short8 bfr[4][2];
@@ -214,12 +230,13 @@ __global__ void bf16_mm256_kernel(MM256Args a) {
       //   free after the first phase's B-fragment reads, sealed by the
       //   phase-end barrier).
       if (ph == 0) {
-        if (t + 1 < NT) {
-          char* img = ((t + 1) & 1) ? imgA1 : imgA0;
-          stage_piece(sA00, (long)(t + 1) * BK, img, 0, 0, wave);
-          stage_piece(sA01, (long)(t + 1) * BK, img, 0, 1, wave);
-          stage_piece(sA10, (long)(t + 1) * BK, img, 1, 0, wave);
-          stage_piece(sA11, (long)(t + 1) * BK, img, 1, 1, wave);
+        const int ta = t + NIMG - 1;  // NIMG=2: t+1; NIMG=3: t+2
+        if (ta < NT) {
+          char* img = imgA(ta);
+          stage_piece(sA00, (long)ta * BK, img, 0, 0, wave);
+          stage_piece(sA01, (long)ta * BK, img, 0, 1, wave);
+          stage_piece(sA10, (long)ta * BK, img, 1, 0, wave);
+          stage_piece(sA11, (long)ta * BK, img, 1, 1, wave);
         }
       } else if (ph == 1) {
         if (t + 2 < NT) {
@@ -262,7 +279,8 @@ __global__ void bf16_mm256_kernel(MM256Args a) {
         // anyone reads them after this barrier.  Steady state leaves the
         // two B halves of t+2 in flight (4 glds); at the tail drain all.
         if (t + 2 < NT)
-          asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+          asm volatile("s_waitcnt vmcnt(%0)" ::"i"(4 * (NIMG - 1))
+                       : "memory");
         else
           asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
       }
@@ -866,7 +884,15 @@ void bf16_mm256(torch::Tensor A, torch::Tensor B, torch::Tensor C,
       const char* e = getenv("DPPO_MM256_SPRIO");
       return e != nullptr && e[0] == '1';
     }();
-    if (sprio)
+    static const int nimg = []() {
+      const char* e = getenv("DPPO_MM256_IMG");
+      return e ? atoi(e) : 2;  // 3 = third A image (UNVALIDATED on
+                               // hardware this round — ROADMAP #1a)
+    }();
+    if (nimg == 3)
+      hipLaunchKernelGGL((bf16_mm256_kernel<2, false, false, 3>),
+                         dim3((unsigned)grid), dim3(512), 0, stream, a);
+    else if (sprio)
       hipLaunchKernelGGL((bf16_mm256_kernel<2, false, true>),
                          dim3((unsigned)grid), dim3(512), 0, stream, a);
     else
