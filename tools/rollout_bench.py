@@ -1,3 +1,4 @@
+import os
 import torch, time, sys
 sys.path.insert(0, os.path.dirname(os.path.dirname(
     os.path.abspath(__file__))))

@@ -1,5 +1,6 @@
 """Learning-curve run: logs per-round stats JSONL (evidence that the
 optimized path LEARNS, not just runs fast)."""
+import os
 import json, sys, time, torch
 sys.path.insert(0, os.path.dirname(os.path.dirname(
     os.path.abspath(__file__))))
