@@ -81,3 +81,9 @@ def load_state(path: str, engine: DPPOEngine) -> None:
     # invariant explicit
     engine.comm.broadcast_(engine.flat_pi.flat_param, src=0)
     engine.comm.broadcast_(engine.flat_old.flat_param, src=0)
+    # the wide-config path caches bf16 weight copies keyed on a dirty
+    # flag that updates set — a restore changes params outside that
+    # flow, so invalidate explicitly (stale-weight rollout otherwise)
+    wide = getattr(engine, "_wide_path", None)
+    if wide is not None:
+        wide.mark_dirty()
